@@ -1,0 +1,110 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: PyTorchJob Llama-3-8B DDP bf16 training.
+
+Measures BASELINE.json's headline metric (tokens/sec at 1/2/4/8 workers) on
+synthetic data with random-init weights. Run directly for N=1 or under
+`python -m torch.distributed.run --nnodes=1 --nproc-per-node N` for N>1
+(reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env).
+
+Prints ONE JSON line from rank 0 with the whole-job aggregate.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from kubeflow_amd.models import build_model  # noqa: E402
+from kubeflow_amd.parallel import dist as kdist  # noqa: E402
+from kubeflow_amd.runtime import Trainer, TrainConfig  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", default=os.environ.get("KF_BENCH_MODEL", "llama3-8b"))
+    ap.add_argument("--seq-len", type=int,
+                    default=int(os.environ.get("KF_BENCH_SEQ", "4096")))
+    ap.add_argument("--micro-batch", type=int,
+                    default=int(os.environ.get("KF_BENCH_MB", "2")))
+    args = ap.parse_args()
+
+    rank, world, device = kdist.init_distributed()
+    if world != args.gpus and rank == 0:
+        print(f"# note: WORLD_SIZE={world} != --gpus {args.gpus}; using {world}",
+              file=sys.stderr)
+    n = world
+
+    torch.manual_seed(1234 + rank)
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    model = build_model(args.model, device=device, dtype=dtype)
+    cfg = model.cfg
+    trainer = Trainer(model, TrainConfig(warmup_steps=2, lr=3e-4))
+
+    B, S = args.micro_batch, args.seq_len
+    tokens = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+    targets = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+
+    t_start = time.time()
+    for _ in range(args.warmup):
+        loss = trainer.step(tokens, targets)
+    kdist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        loss = trainer.step(tokens, targets)
+    kdist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t1 = time.time()
+
+    elapsed = t1 - t0
+    # MAX over ranks (all ranks hit the same barriers; reduce to be exact)
+    et = torch.tensor([elapsed], dtype=torch.float64)
+    if n > 1:
+        import torch.distributed as dist
+        if dist.get_backend() == "nccl":
+            et = et.to(device)
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+    elapsed = float(et.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    tokens_per_step = B * S * n
+    value = tokens_per_step * args.steps / elapsed
+
+    if rank == 0:
+        out = {
+            "metric": "pytorchjob_train_tokens_per_s",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": B * n,
+                "seq_len": S,
+                "parallelism": f"dp{n}",
+                "loss": round(float(loss.item()), 4),
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

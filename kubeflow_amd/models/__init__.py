@@ -1,0 +1,58 @@
+"""Model registry for PyTorchJob / InferenceService / Katib specs.
+
+Jobs name models the way Kubeflow CRs name images: a string key in the spec.
+"""
+from __future__ import annotations
+
+from typing import Callable, Dict
+
+from .llama import (LlamaConfig, LlamaModel, llama3_8b, llama3_1b, llama_tiny)
+from .bert import (BertConfig, BertClassifier, bert_base, bert_base_hd128,
+                   bert_tiny)
+from .mlp import MnistMLP
+
+__all__ = [
+    "LlamaConfig", "LlamaModel", "llama3_8b", "llama3_1b", "llama_tiny",
+    "BertConfig", "BertClassifier", "bert_base", "bert_base_hd128",
+    "bert_tiny", "MnistMLP", "build_model", "MODEL_REGISTRY",
+]
+
+
+def _llama(cfg_fn):
+    def build(device=None, dtype=None, **kw):
+        import torch
+        return LlamaModel(cfg_fn(), device=device,
+                          dtype=dtype or torch.bfloat16)
+    return build
+
+
+def _bert(cfg_fn):
+    def build(device=None, dtype=None, **kw):
+        import torch
+        return BertClassifier(cfg_fn(), device=device,
+                              dtype=dtype or torch.bfloat16)
+    return build
+
+
+def _mlp(device=None, dtype=None, **kw):
+    m = MnistMLP()
+    if device is not None:
+        m = m.to(device)
+    return m
+
+
+MODEL_REGISTRY: Dict[str, Callable] = {
+    "llama3-8b": _llama(llama3_8b),
+    "llama3-1b": _llama(llama3_1b),
+    "llama-tiny": _llama(llama_tiny),
+    "bert-base": _bert(bert_base),
+    "bert-base-hd128": _bert(bert_base_hd128),
+    "bert-tiny": _bert(bert_tiny),
+    "mnist-mlp": _mlp,
+}
+
+
+def build_model(name: str, device=None, dtype=None, **kw):
+    if name not in MODEL_REGISTRY:
+        raise KeyError(f"unknown model {name!r}; known: {sorted(MODEL_REGISTRY)}")
+    return MODEL_REGISTRY[name](device=device, dtype=dtype, **kw)

@@ -1,0 +1,167 @@
+"""Llama-3-family decoder for the PyTorchJob flagship workload.
+
+MI355X-first design decisions:
+  * bshd tensor layout end-to-end (no transposes around attention — the HIP
+    flash kernel consumes [B,S,H,D] directly).
+  * fused QKV and fused gate+up projections: plain library GEMMs go through
+    torch.nn.functional.linear (hipBLASLt); everything between GEMMs is a
+    hand-written CDNA4 kernel (kubeflow_amd.ops): RMSNorm, RoPE, flash
+    attention, fused softmax-CE over the 128256 vocab.
+  * bf16 parameters (fp32 master copies live in the optimizer, not here).
+
+Reference-behavior anchor: the model that BASELINE.json's PyTorchJob metric
+is quoted on (Llama-3-8B: hidden 4096, 32 layers, 32 q / 8 kv heads,
+ffn 14336, vocab 128256).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from kubeflow_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama3-8b"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    ffn_dim: int = 14336
+    rope_theta: float = 500000.0
+    max_seq_len: int = 8192
+    norm_eps: float = 1e-5
+    init_std: float = 0.02
+
+    @property
+    def n_params(self) -> int:
+        h, v = self.hidden_size, self.vocab_size
+        qkv = h * (self.n_heads + 2 * self.n_kv_heads) * self.head_dim
+        o = self.n_heads * self.head_dim * h
+        mlp = 3 * h * self.ffn_dim
+        return 2 * v * h + self.n_layers * (qkv + o + mlp + 2 * h) + h
+
+
+def llama3_8b() -> LlamaConfig:
+    return LlamaConfig()
+
+
+def llama3_1b() -> LlamaConfig:
+    """Scaled-down config for quick GPU smoke/bench runs."""
+    return LlamaConfig(name="llama3-1b", hidden_size=2048, n_layers=16,
+                       n_heads=16, n_kv_heads=8, ffn_dim=8192)
+
+
+def llama_tiny(vocab: int = 512) -> LlamaConfig:
+    """Tiny config for tests. head_dim stays 128 (HIP kernel contract)."""
+    return LlamaConfig(name="llama-tiny", vocab_size=vocab, hidden_size=256,
+                       n_layers=2, n_heads=2, n_kv_heads=1, ffn_dim=512,
+                       max_seq_len=512)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.rms_norm(x, self.weight, self.eps)
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        h, d = cfg.hidden_size, cfg.head_dim
+        self.cfg = cfg
+        self.attn_norm = RMSNorm(h, cfg.norm_eps)
+        self.wqkv = nn.Linear(h, (cfg.n_heads + 2 * cfg.n_kv_heads) * d,
+                              bias=False)
+        self.wo = nn.Linear(cfg.n_heads * d, h, bias=False)
+        self.mlp_norm = RMSNorm(h, cfg.norm_eps)
+        self.w13 = nn.Linear(h, 2 * cfg.ffn_dim, bias=False)  # gate ++ up
+        self.w2 = nn.Linear(cfg.ffn_dim, h, bias=False)
+
+    def forward(self, x, cos, sin, pos_offset: int = 0, kv_cache=None):
+        cfg = self.cfg
+        B, S, _ = x.shape
+        qkv = F.linear(self.attn_norm(x), self.wqkv.weight)
+        q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
+                             cfg.n_kv_heads * cfg.head_dim,
+                             cfg.n_kv_heads * cfg.head_dim], dim=-1)
+        q = q.view(B, S, cfg.n_heads, cfg.head_dim)
+        k = k.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        v = v.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        q, k = ops.rope(q, k, cos, sin, pos_offset)
+        if kv_cache is not None:
+            k, v = kv_cache.update(k, v, pos_offset)
+            o = ops.flash_attention(q, k, v, causal=(S > 1))
+        else:
+            o = ops.flash_attention(q, k, v, causal=True)
+        o = self.wo(o.reshape(B, S, cfg.n_heads * cfg.head_dim))
+        x = x + o
+        g, u = F.linear(self.mlp_norm(x), self.w13.weight).split(
+            [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
+        return x + self.w2(F.silu(g) * u)
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig, device=None, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        factory = dict(device=device, dtype=dtype)
+        with torch.device(device if device is not None else "cpu"):
+            self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+            self.layers = nn.ModuleList(
+                [LlamaBlock(cfg) for _ in range(cfg.n_layers)])
+            self.final_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
+            self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
+                                     bias=False)
+        self.to(**{k: v for k, v in factory.items() if v is not None})
+        cos, sin = ops.rope_cos_sin(cfg.max_seq_len, cfg.head_dim,
+                                    cfg.rope_theta, device=device)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.init_weights()
+
+    @torch.no_grad()
+    def init_weights(self):
+        std = self.cfg.init_std
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                p.normal_(0.0, std)
+                # scaled init for residual-out projections (GPT-2 style)
+                if name.endswith(("wo.weight", "w2.weight")):
+                    p.mul_(1.0 / math.sqrt(2 * self.cfg.n_layers))
+            else:
+                p.fill_(1.0)
+
+    def forward(self, tokens: torch.Tensor, targets: torch.Tensor = None,
+                pos_offset: int = 0, kv_caches=None):
+        """tokens [B,S] int64 -> loss (if targets) else logits [B,S,V]."""
+        x = self.embed(tokens)
+        cos, sin = self.rope_cos, self.rope_sin
+        for i, layer in enumerate(self.layers):
+            cache = kv_caches[i] if kv_caches is not None else None
+            x = layer(x, cos, sin, pos_offset, cache)
+        x = self.final_norm(x)
+        logits = F.linear(x, self.lm_head.weight)
+        if targets is None:
+            return logits
+        T = logits.shape[0] * logits.shape[1]
+        return ops.cross_entropy(logits.view(T, -1), targets.view(T))
+
+    def flops_per_token(self, seq_len: int) -> float:
+        """Approximate training FLOPs/token (fwd+bwd) for MFU reporting."""
+        n = self.cfg.n_params
+        # attention: 2*2*S*D per head-pair term, causal halves it
+        attn = (3 * 2 * 2 * self.cfg.n_layers * self.cfg.n_heads *
+                self.cfg.head_dim * seq_len / 2)
+        return 6 * n + attn

@@ -1,0 +1,266 @@
+"""kubeflow_amd.ops — the hot-op library (CDNA4 HIP kernels + CPU reference).
+
+Dispatch policy:
+  * CUDA (= ROCm/HIP) tensors -> hand-written gfx950 kernels in libkfops.so,
+    loaded via ctypes (`_backend`). Missing library on a GPU box raises —
+    no silent eager fallback (KF_NATIVE_KERNELS=0 is the explicit escape
+    hatch for bisection).
+  * CPU tensors -> the pure-torch reference implementations (differentiable),
+    so the full stack runs in CPU-only CI.
+
+All ops are exposed as autograd-capable functions:
+  rms_norm, rope, flash_attention, cross_entropy, fused_adamw (no autograd).
+"""
+from __future__ import annotations
+
+import ctypes
+from typing import Optional
+
+import torch
+
+from . import reference
+from . import _backend
+
+__all__ = [
+    "rms_norm", "rope", "flash_attention", "cross_entropy", "fused_adamw",
+    "rope_cos_sin", "native_available",
+]
+
+rope_cos_sin = reference.rope_cos_sin
+
+
+def native_available() -> bool:
+    return _backend.try_load() is not None
+
+
+def _stream() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _p(t: Optional[torch.Tensor]) -> ctypes.c_void_p:
+    return ctypes.c_void_p(0 if t is None else t.data_ptr())
+
+
+def _fp(t: Optional[torch.Tensor]):
+    return ctypes.cast(0 if t is None else t.data_ptr(),
+                       ctypes.POINTER(ctypes.c_float))
+
+
+def _ip(t: torch.Tensor):
+    return ctypes.cast(t.data_ptr(), ctypes.POINTER(ctypes.c_int64))
+
+
+def _use_native(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if _backend.native_enabled():
+        _backend.require()
+        return True
+    return False
+
+
+# --------------------------------------------------------------- RMSNorm --
+
+class _RMSNormHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        lib = _backend.require()
+        shape = x.shape
+        x2 = x.contiguous().view(-1, shape[-1])
+        rows, cols = x2.shape
+        y = torch.empty_like(x2)
+        rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+        _backend.check(
+            lib.kf_rmsnorm_fwd(_p(y), _fp(rstd), _p(x2), _p(weight), rows,
+                               cols, float(eps), _stream()), "rmsnorm_fwd")
+        ctx.save_for_backward(x2, weight, rstd)
+        ctx.shape = shape
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = _backend.require()
+        x2, weight, rstd = ctx.saved_tensors
+        rows, cols = x2.shape
+        dy2 = dy.contiguous().view(rows, cols)
+        dx = torch.empty_like(x2)
+        dw = torch.empty_like(weight)
+        nparts = lib.kf_rmsnorm_bwd_nparts(rows)
+        dw_part = torch.empty(nparts * cols, dtype=torch.float32,
+                              device=x2.device)
+        _backend.check(
+            lib.kf_rmsnorm_bwd(_p(dx), _p(dw), _fp(dw_part), _p(dy2), _p(x2),
+                               _p(weight), _fp(rstd), rows, cols, _stream()),
+            "rmsnorm_bwd")
+        return dx.view(ctx.shape), dw, None
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5):
+    if _use_native(x):
+        return _RMSNormHip.apply(x, weight, eps)
+    return reference.rms_norm(x, weight, eps)
+
+
+# ------------------------------------------------------------------ RoPE --
+
+class _RopeHip(torch.autograd.Function):
+    """Joint rotary embedding on (q, k); in-place rotation on fresh clones."""
+
+    @staticmethod
+    def forward(ctx, q, k, cos, sin, pos_offset):
+        lib = _backend.require()
+        B, S, Hq, D = q.shape
+        Hkv = k.shape[2]
+        q = q.contiguous().clone(memory_format=torch.contiguous_format)
+        k = k.contiguous().clone(memory_format=torch.contiguous_format)
+        _backend.check(
+            lib.kf_rope(_p(q), _p(k), _fp(cos), _fp(sin), B, S, Hq, Hkv, D,
+                        pos_offset, 0, _stream()), "rope_fwd")
+        ctx.save_for_backward(cos, sin)
+        ctx.dims = (B, S, Hq, Hkv, D, pos_offset)
+        return q, k
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        lib = _backend.require()
+        cos, sin = ctx.saved_tensors
+        B, S, Hq, Hkv, D, pos_offset = ctx.dims
+        dq = dq.contiguous().clone(memory_format=torch.contiguous_format)
+        dk = dk.contiguous().clone(memory_format=torch.contiguous_format)
+        _backend.check(
+            lib.kf_rope(_p(dq), _p(dk), _fp(cos), _fp(sin), B, S, Hq, Hkv, D,
+                        pos_offset, 1, _stream()), "rope_bwd")
+        return dq, dk, None, None, None
+
+
+def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
+         sin: torch.Tensor, pos_offset: int = 0):
+    """q [B,S,Hq,D], k [B,S,Hkv,D]; cos/sin fp32 [>=S+off, D/2]."""
+    if _use_native(q):
+        return _RopeHip.apply(q, k, cos, sin, pos_offset)
+    return (reference.rope_apply(q, cos, sin, pos_offset),
+            reference.rope_apply(k, cos, sin, pos_offset))
+
+
+# ------------------------------------------------------- Flash attention --
+
+class _FlashAttnHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        lib = _backend.require()
+        B, S, Hq, D = q.shape
+        Hkv = k.shape[2]
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o = torch.empty_like(q)
+        lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
+        _backend.check(
+            lib.kf_attn_fwd(_p(o), _fp(lse), _p(q), _p(k), _p(v), B, S, Hq,
+                            Hkv, D, float(scale), int(causal), _stream()),
+            "attn_fwd")
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.meta = (causal, scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        lib = _backend.require()
+        q, k, v, o, lse = ctx.saved_tensors
+        causal, scale = ctx.meta
+        B, S, Hq, D = q.shape
+        Hkv = k.shape[2]
+        dout = dout.contiguous()
+        dq = torch.empty_like(q)
+        dk = torch.empty_like(k)
+        dv = torch.empty_like(v)
+        delta = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
+        _backend.check(
+            lib.kf_attn_bwd(_p(dq), _p(dk), _p(dv), _p(dout), _p(q), _p(k),
+                            _p(v), _p(o), _fp(lse), _fp(delta), B, S, Hq, Hkv,
+                            D, float(scale), int(causal), _stream()),
+            "attn_bwd")
+        return dq, dk, dv, None, None
+
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    causal: bool = True, scale: Optional[float] = None):
+    """bshd layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] -> o [B,S,Hq,D]."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if _use_native(q):
+        return _FlashAttnHip.apply(q, k, v, causal, scale)
+    # reference path works in bhsd
+    qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))
+    o = reference.sdpa(qt, kt, vt, causal=causal, scale=scale)
+    return o.transpose(1, 2)
+
+
+# --------------------------------------------------------- Cross entropy --
+
+class _CrossEntropyHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        lib = _backend.require()
+        T, V = logits.shape
+        logits = logits.contiguous()
+        loss_sum = torch.zeros(1, dtype=torch.float32, device=logits.device)
+        lse = torch.empty(T, dtype=torch.float32, device=logits.device)
+        _backend.check(
+            lib.kf_ce_fwd(_fp(loss_sum), _fp(lse), _p(logits), _ip(targets),
+                          T, V, ignore_index, _stream()), "ce_fwd")
+        valid = (targets != ignore_index).sum()
+        inv_valid = torch.where(valid > 0, 1.0 / valid.float(),
+                                torch.zeros((), device=logits.device))
+        ctx.save_for_backward(logits, lse, targets, inv_valid)
+        ctx.ignore_index = ignore_index
+        return (loss_sum * inv_valid).squeeze(0)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        lib = _backend.require()
+        logits, lse, targets, inv_valid = ctx.saved_tensors
+        T, V = logits.shape
+        scale = (grad_out.float() * inv_valid).contiguous()
+        dlogits = torch.empty_like(logits)
+        _backend.check(
+            lib.kf_ce_bwd(_p(dlogits), _p(logits), _fp(lse), _ip(targets),
+                          _fp(scale), T, V, ctx.ignore_index, _stream()),
+            "ce_bwd")
+        return dlogits, None, None
+
+
+def cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
+                  ignore_index: int = -100):
+    """Mean CE over non-ignored rows. logits [T,V] bf16, targets [T] int64."""
+    if _use_native(logits):
+        return _CrossEntropyHip.apply(logits, targets, ignore_index)
+    return reference.softmax_cross_entropy(logits, targets, ignore_index)
+
+
+# ----------------------------------------------------------- Fused AdamW --
+
+def fused_adamw(p16: torch.Tensor, p32: torch.Tensor, grad: torch.Tensor,
+                m: torch.Tensor, v: torch.Tensor,
+                wd_mask: Optional[torch.Tensor], lr: float, beta1: float,
+                beta2: float, eps: float, weight_decay: float, step: int):
+    """One fused update over a flat bf16 parameter shard + fp32 state.
+
+    p16/grad bf16 [N]; p32/m/v fp32 [N]; wd_mask fp32 {0,1} [N] or None.
+    """
+    if p16.is_cuda and _backend.native_enabled():
+        lib = _backend.require()
+        _backend.check(
+            lib.kf_adamw(_p(p16), _fp(p32), _p(grad), _fp(m), _fp(v),
+                         _fp(wd_mask), p16.numel(), lr, beta1, beta2, eps,
+                         weight_decay, step, _stream()), "adamw")
+        return
+    # reference path (CPU tests)
+    g32 = grad.float()
+    m.mul_(beta1).add_(g32, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(g32, g32, value=1 - beta2)
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    denom = (v / bc2).sqrt().add_(eps)
+    decay = weight_decay if wd_mask is None else wd_mask * weight_decay
+    p32.mul_(1 - lr * decay)
+    p32.addcdiv_(m / bc1, denom, value=-lr)
+    p16.copy_(p32.to(torch.bfloat16))

@@ -1,0 +1,91 @@
+"""ctypes loader for the in-tree CDNA4 kernel library.
+
+Policy (per the project's "no silent eager fallback" rule):
+  * On a machine WITH a GPU, the HIP library is required — a missing or
+    unloadable libkfops.so raises at first use, unless the user explicitly
+    sets KF_NATIVE_KERNELS=0 (bisection escape hatch).
+  * On CPU-only machines (CI container), ops fall back to the pure-torch
+    reference implementations so unit tests run without a GPU.
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from pathlib import Path
+
+_LIB_PATH = Path(__file__).resolve().parent / "libkfops.so"
+_lib = None
+_load_error: str | None = None
+
+
+def native_enabled() -> bool:
+    return os.environ.get("KF_NATIVE_KERNELS", "1") != "0"
+
+
+def try_load():
+    """Load libkfops.so if present; returns the CDLL or None."""
+    global _lib, _load_error
+    if _lib is not None:
+        return _lib
+    if not native_enabled():
+        _load_error = "disabled via KF_NATIVE_KERNELS=0"
+        return None
+    if not _LIB_PATH.exists():
+        _load_error = f"{_LIB_PATH} not built (run kubeflow_amd.ops.build_ext)"
+        return None
+    try:
+        _lib = ctypes.CDLL(str(_LIB_PATH))
+    except OSError as e:  # pragma: no cover
+        _load_error = str(e)
+        return None
+    _configure(_lib)
+    return _lib
+
+
+def require():
+    """Return the library, raising loudly if it should be there but is not."""
+    lib = try_load()
+    if lib is None:
+        raise RuntimeError(
+            "kubeflow_amd native kernels unavailable on a GPU machine: "
+            f"{_load_error}. Build with `python -m kubeflow_amd.ops.build_ext` "
+            "or set KF_NATIVE_KERNELS=0 to explicitly allow the (slow) "
+            "reference path.")
+    return lib
+
+
+def _configure(lib: ctypes.CDLL) -> None:
+    c = ctypes
+    P, F, I64, I32, FP = c.c_void_p, c.c_float, c.c_int64, c.c_int, c.POINTER(c.c_float)
+    PI64 = c.POINTER(c.c_int64)
+    lib.kf_rmsnorm_fwd.restype = I32
+    lib.kf_rmsnorm_fwd.argtypes = [P, FP, P, P, I64, I64, F, P]
+    lib.kf_rmsnorm_bwd.restype = I32
+    lib.kf_rmsnorm_bwd.argtypes = [P, P, FP, P, P, P, FP, I64, I64, P]
+    lib.kf_rmsnorm_bwd_nparts.restype = I64
+    lib.kf_rmsnorm_bwd_nparts.argtypes = [I64]
+    lib.kf_rope.restype = I32
+    lib.kf_rope.argtypes = [P, P, FP, FP, I64, I64, I64, I64, I64, I64, I32, P]
+    lib.kf_adamw.restype = I32
+    lib.kf_adamw.argtypes = [P, FP, P, FP, FP, FP, I64, F, F, F, F, F, I64, P]
+    lib.kf_ce_fwd.restype = I32
+    lib.kf_ce_fwd.argtypes = [FP, FP, P, PI64, I64, I64, I64, P]
+    lib.kf_ce_bwd.restype = I32
+    lib.kf_ce_bwd.argtypes = [P, P, FP, PI64, FP, I64, I64, I64, P]
+    lib.kf_attn_fwd.restype = I32
+    lib.kf_attn_fwd.argtypes = [P, FP, P, P, P, I64, I64, I64, I64, I64, F, I32, P]
+    if hasattr(lib, "kf_attn_bwd"):
+        lib.kf_attn_bwd.restype = I32
+        lib.kf_attn_bwd.argtypes = [P, P, P, P, P, P, P, P, FP, FP, I64, I64,
+                                    I64, I64, I64, F, I32, P]
+    if hasattr(lib, "kf_layernorm_fwd"):
+        lib.kf_layernorm_fwd.restype = I32
+        lib.kf_layernorm_fwd.argtypes = [P, FP, FP, P, P, P, I64, I64, F, P]
+    if hasattr(lib, "kf_layernorm_bwd"):
+        lib.kf_layernorm_bwd.restype = I32
+        lib.kf_layernorm_bwd.argtypes = [P, P, P, FP, P, P, P, FP, FP, I64, I64, P]
+
+
+def check(err: int, name: str) -> None:
+    if err != 0:
+        raise RuntimeError(f"HIP kernel {name} failed with hipError_t={err}")

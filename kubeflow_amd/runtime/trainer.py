@@ -1,0 +1,117 @@
+"""Training worker runtime — the loop that runs inside a PyTorchJob rank.
+
+This is the half of Kubeflow that lives in the sibling training-operator's
+worker pods; here it is a first-class component:
+FlatParamSpace + BucketedDDP (RCCL/xGMI) + fused-AdamW (single HIP kernel
+over the whole flat model) + cosine LR schedule + checkpoint save/load with
+a PyTorchJob-compatible directory layout (checkpoint.py).
+"""
+from __future__ import annotations
+
+import math
+import os
+import time
+from dataclasses import dataclass, field
+from typing import Callable, Optional
+
+import torch
+
+from kubeflow_amd import ops
+from kubeflow_amd.parallel import FlatParamSpace, BucketedDDP
+from kubeflow_amd.parallel import dist as kdist
+
+
+@dataclass
+class TrainConfig:
+    lr: float = 3e-4
+    beta1: float = 0.9
+    beta2: float = 0.95
+    eps: float = 1e-8
+    weight_decay: float = 0.1
+    warmup_steps: int = 10
+    lr_decay_steps: int = 10000
+    min_lr_ratio: float = 0.1
+    grad_accum: int = 1
+    bucket_mb: Optional[float] = None
+    use_wd_mask: bool = True
+
+
+class Trainer:
+    """Owns the flat parameter space, DDP overlap and the fused optimizer."""
+
+    def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig()):
+        self.model = model
+        self.cfg = cfg
+        self.flat = FlatParamSpace(model)
+        self.ddp = BucketedDDP(self.flat, bucket_mb=cfg.bucket_mb)
+        dev = self.flat.device
+        self.p32 = self.flat.data.float()
+        self.m = torch.zeros_like(self.p32)
+        self.v = torch.zeros_like(self.p32)
+        self.wd_mask = self.flat.build_wd_mask() if cfg.use_wd_mask else None
+        self.step_num = 0
+
+    # ------------------------------------------------------------ schedule
+    def lr_at(self, step: int) -> float:
+        cfg = self.cfg
+        if step < cfg.warmup_steps:
+            return cfg.lr * (step + 1) / max(1, cfg.warmup_steps)
+        t = min(1.0, (step - cfg.warmup_steps) /
+                max(1, cfg.lr_decay_steps - cfg.warmup_steps))
+        floor = cfg.lr * cfg.min_lr_ratio
+        return floor + 0.5 * (cfg.lr - floor) * (1 + math.cos(math.pi * t))
+
+    # ---------------------------------------------------------------- step
+    def step(self, *batch) -> torch.Tensor:
+        """One optimizer step over `grad_accum` micro-batches.
+
+        `batch` is either (inputs, targets) tensors or a callable returning
+        them per micro-step. Returns the (device) loss of the last micro.
+        """
+        cfg = self.cfg
+        self.flat.zero_grad()
+        loss = None
+        for micro in range(cfg.grad_accum):
+            if len(batch) == 1 and callable(batch[0]):
+                inputs, targets = batch[0](micro)
+            else:
+                inputs, targets = batch
+            final = micro == cfg.grad_accum - 1
+            if final:
+                self.ddp.prepare_step()
+            loss = self.model(inputs, targets)
+            if cfg.grad_accum > 1:
+                (loss / cfg.grad_accum).backward()
+            else:
+                loss.backward()
+        self.ddp.finalize()
+        self.step_num += 1
+        lr = self.lr_at(self.step_num - 1)
+        ops.fused_adamw(self.flat.data, self.p32, self.flat.grad, self.m,
+                        self.v, self.wd_mask, lr, cfg.beta1, cfg.beta2,
+                        cfg.eps, cfg.weight_decay, self.step_num)
+        return loss.detach()
+
+    # ---------------------------------------------------------- checkpoint
+    def state_dict(self) -> dict:
+        return {
+            "step": self.step_num,
+            "flat_data": self.flat.data,
+            "p32": self.p32,
+            "m": self.m,
+            "v": self.v,
+            "param_names": self.flat.names,
+            "rng": torch.get_rng_state(),
+            "cuda_rng": (torch.cuda.get_rng_state()
+                         if torch.cuda.is_available() else None),
+        }
+
+    def load_state_dict(self, sd: dict):
+        self.step_num = sd["step"]
+        self.flat.data.copy_(sd["flat_data"])
+        self.p32.copy_(sd["p32"])
+        self.m.copy_(sd["m"])
+        self.v.copy_(sd["v"])
+        torch.set_rng_state(sd["rng"].cpu() if hasattr(sd["rng"], "cpu") else sd["rng"])
+        if sd.get("cuda_rng") is not None and torch.cuda.is_available():
+            torch.cuda.set_rng_state(sd["cuda_rng"])

@@ -1,0 +1,96 @@
+"""CPU tests: flat parameter space, trainer loop, checkpoint round-trip."""
+import torch
+import pytest
+
+from kubeflow_amd.models import MnistMLP, build_model
+from kubeflow_amd.parallel import FlatParamSpace
+from kubeflow_amd.runtime import Trainer, TrainConfig
+
+
+def test_flat_param_space_views():
+    torch.manual_seed(0)
+    m = MnistMLP(in_dim=16, hidden=8, n_classes=4)
+    orig = {n: p.detach().clone() for n, p in m.named_parameters()}
+    flat = FlatParamSpace(m, dtype=torch.float32)
+    for n, p in m.named_parameters():
+        assert torch.allclose(p.data, orig[n])
+        assert p.data.data_ptr() >= flat.data.data_ptr()
+        assert p.grad is not None
+    # writing flat propagates to params
+    flat.data.fill_(0.5)
+    for _, p in m.named_parameters():
+        assert (p.data == 0.5).all()
+
+
+def test_wd_mask_matrices_only():
+    m = MnistMLP(in_dim=16, hidden=8, n_classes=4)
+    flat = FlatParamSpace(m, dtype=torch.float32)
+    mask = flat.build_wd_mask()
+    for (off, n), p in zip(flat.slices, flat.params):
+        expect = 1.0 if p.dim() >= 2 else 0.0
+        assert (mask[off:off + n] == expect).all()
+
+
+def test_trainer_loss_decreases_mlp():
+    torch.manual_seed(0)
+    m = MnistMLP(in_dim=32, hidden=64, n_classes=4)
+    tr = Trainer(m, TrainConfig(lr=1e-2, warmup_steps=2, weight_decay=0.0))
+    x = torch.randn(64, 32)
+    y = torch.randint(0, 4, (64,))
+    losses = [float(tr.step(x, y)) for _ in range(30)]
+    assert losses[-1] < losses[0] * 0.5, losses[::10]
+
+
+def test_trainer_llama_tiny_cpu():
+    torch.manual_seed(0)
+    model = build_model("llama-tiny", dtype=torch.float32)
+    tr = Trainer(model, TrainConfig(lr=1e-3, warmup_steps=2))
+    tokens = torch.randint(0, model.cfg.vocab_size, (2, 64))
+    targets = torch.randint(0, model.cfg.vocab_size, (2, 64))
+    losses = [float(tr.step(tokens, targets)) for _ in range(8)]
+    assert all(l == l for l in losses)  # no NaN
+    assert losses[-1] < losses[0]
+
+
+def test_checkpoint_roundtrip_determinism():
+    torch.manual_seed(0)
+    m1 = MnistMLP(in_dim=16, hidden=16, n_classes=4)
+    tr1 = Trainer(m1, TrainConfig(lr=1e-2))
+    x = torch.randn(16, 16)
+    y = torch.randint(0, 4, (16,))
+    for _ in range(3):
+        tr1.step(x, y)
+    sd = {k: (v.clone() if torch.is_tensor(v) else v)
+          for k, v in tr1.state_dict().items()}
+
+    # continue 2 more steps, remember losses
+    ref = [float(tr1.step(x, y)) for _ in range(2)]
+
+    torch.manual_seed(123)  # different RNG state; checkpoint must restore all
+    m2 = MnistMLP(in_dim=16, hidden=16, n_classes=4)
+    tr2 = Trainer(m2, TrainConfig(lr=1e-2))
+    tr2.load_state_dict(sd)
+    got = [float(tr2.step(x, y)) for _ in range(2)]
+    assert got == pytest.approx(ref, rel=1e-6)
+
+
+def test_grad_accum_equivalence():
+    torch.manual_seed(0)
+    x = torch.randn(8, 32)
+    y = torch.randint(0, 4, (8,))
+
+    torch.manual_seed(1)
+    m1 = MnistMLP(in_dim=32, hidden=16, n_classes=4)
+    tr1 = Trainer(m1, TrainConfig(lr=1e-2, grad_accum=1, weight_decay=0.0))
+    tr1.step(x, y)
+
+    torch.manual_seed(1)
+    m2 = MnistMLP(in_dim=32, hidden=16, n_classes=4)
+    tr2 = Trainer(m2, TrainConfig(lr=1e-2, grad_accum=2, weight_decay=0.0))
+    half = [(x[:4], y[:4]), (x[4:], y[4:])]
+    tr2.step(lambda i: half[i])
+
+    # same data split across two micros averages to ~the same update
+    p1 = torch.cat([p.flatten() for p in m1.parameters()])
+    p2 = torch.cat([p.flatten() for p in m2.parameters()])
+    assert torch.allclose(p1, p2, atol=1e-4)

@@ -1,0 +1,200 @@
+"""GPU numerics tests: every CDNA4 HIP kernel vs the fp32 torch reference.
+
+Per the project test strategy (SURVEY.md §4 rebuild mapping): "numerics
+tests for a HIP kernel compare it against a plain PyTorch fp32 reference of
+the same op". Tolerances account for bf16 storage (rel ~1e-2).
+"""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from kubeflow_amd import ops
+from kubeflow_amd.ops import reference as R
+
+
+def _relerr(a, b):
+    a, b = a.float(), b.float()
+    return ((a - b).norm() / (b.norm() + 1e-12)).item()
+
+
+@pytest.fixture(autouse=True)
+def _require_native():
+    from kubeflow_amd.ops import _backend
+    _backend.require()  # fail loudly if .so missing on the GPU box
+
+
+# ---------------------------------------------------------------- RMSNorm
+
+@pytest.mark.parametrize("rows,cols", [(128, 4096), (1024, 768), (8, 256)])
+def test_rmsnorm_fwd(rows, cols, gpu_device):
+    torch.manual_seed(0)
+    x = torch.randn(rows, cols, device=gpu_device, dtype=torch.bfloat16)
+    w = torch.randn(cols, device=gpu_device, dtype=torch.bfloat16)
+    y = ops.rms_norm(x, w)
+    ref = R.rms_norm(x.float().cpu(), w.float().cpu()).to(torch.bfloat16)
+    assert _relerr(y.cpu(), ref) < 2e-2
+
+
+def test_rmsnorm_bwd(gpu_device):
+    torch.manual_seed(0)
+    rows, cols = 512, 1024
+    x = torch.randn(rows, cols, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(cols, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.rms_norm(x, w)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    yr = R.rms_norm(xr, wr)
+    yr.backward(dy.float().cpu())
+
+    assert _relerr(x.grad.cpu(), xr.grad) < 3e-2
+    assert _relerr(w.grad.cpu(), wr.grad) < 3e-2
+
+
+# ------------------------------------------------------------------- RoPE
+
+def test_rope_fwd_bwd(gpu_device):
+    torch.manual_seed(0)
+    B, S, Hq, Hkv, D = 2, 128, 4, 2, 128
+    q = torch.randn(B, S, Hq, D, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    cos, sin = ops.rope_cos_sin(S, D, device=gpu_device)
+    q2, k2 = ops.rope(q, k, cos, sin)
+
+    qr = q.detach().float().cpu().requires_grad_(True)
+    kr = k.detach().float().cpu().requires_grad_(True)
+    ccpu, scpu = ops.rope_cos_sin(S, D)
+    q2r = R.rope_apply(qr, ccpu, scpu)
+    k2r = R.rope_apply(kr, ccpu, scpu)
+    assert _relerr(q2.cpu(), q2r) < 2e-2
+    assert _relerr(k2.cpu(), k2r) < 2e-2
+
+    dq2 = torch.randn_like(q2)
+    dk2 = torch.randn_like(k2)
+    qg = torch.autograd.grad(
+        (q2r * dq2.float().cpu()).sum() + (k2r * dk2.float().cpu()).sum(),
+        [qr, kr])
+    ((q2 * dq2).sum() + (k2 * dk2).sum()).backward()
+    assert _relerr(q.grad.cpu(), qg[0]) < 2e-2
+    assert _relerr(k.grad.cpu(), qg[1]) < 2e-2
+
+
+# -------------------------------------------------------- Flash attention
+
+@pytest.mark.parametrize("B,S,Hq,Hkv,causal", [
+    (2, 128, 4, 1, True),
+    (1, 256, 8, 2, True),
+    (2, 128, 4, 4, False),
+    (1, 512, 32, 8, True),   # llama-3 8B head config
+])
+def test_attention_fwd(B, S, Hq, Hkv, causal, gpu_device):
+    torch.manual_seed(0)
+    D = 128
+    q = torch.randn(B, S, Hq, D, device=gpu_device, dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device=gpu_device, dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device=gpu_device, dtype=torch.bfloat16)
+    o = ops.flash_attention(q, k, v, causal=causal)
+    ref = R.sdpa(q.float().cpu().transpose(1, 2),
+                 k.float().cpu().transpose(1, 2),
+                 v.float().cpu().transpose(1, 2),
+                 causal=causal).transpose(1, 2)
+    assert _relerr(o.cpu(), ref) < 2e-2
+
+
+def test_attention_bwd(gpu_device):
+    torch.manual_seed(0)
+    B, S, Hq, Hkv, D = 1, 256, 4, 2, 128
+    q = torch.randn(B, S, Hq, D, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, S, Hkv, D, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = ops.flash_attention(q, k, v, causal=True)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    qr = q.detach().float().cpu().transpose(1, 2).requires_grad_(True)
+    kr = k.detach().float().cpu().transpose(1, 2).requires_grad_(True)
+    vr = v.detach().float().cpu().transpose(1, 2).requires_grad_(True)
+    orf = R.sdpa(qr, kr, vr, causal=True, scale=D ** -0.5)
+    orf.backward(do.float().cpu().transpose(1, 2))
+
+    assert _relerr(q.grad.cpu(), qr.grad.transpose(1, 2)) < 4e-2
+    assert _relerr(k.grad.cpu(), kr.grad.transpose(1, 2)) < 4e-2
+    assert _relerr(v.grad.cpu(), vr.grad.transpose(1, 2)) < 4e-2
+
+
+# ---------------------------------------------------------- Cross entropy
+
+@pytest.mark.parametrize("T,V", [(64, 1000), (256, 128256)])
+def test_cross_entropy(T, V, gpu_device):
+    torch.manual_seed(0)
+    logits = torch.randn(T, V, device=gpu_device, dtype=torch.bfloat16,
+                         requires_grad=True)
+    targets = torch.randint(0, V, (T,), device=gpu_device)
+    targets[::7] = -100
+    loss = ops.cross_entropy(logits, targets)
+    loss.backward()
+
+    lr = logits.detach().float().cpu().requires_grad_(True)
+    ref = R.softmax_cross_entropy(lr, targets.cpu())
+    ref.backward()
+    assert abs(loss.item() - ref.item()) / abs(ref.item()) < 1e-2
+    assert _relerr(logits.grad.cpu(), lr.grad) < 2e-2
+
+
+# ------------------------------------------------------------ Fused AdamW
+
+def test_fused_adamw_matches_reference(gpu_device):
+    torch.manual_seed(0)
+    n = 4096 + 64
+    p32 = torch.randn(n, device=gpu_device)
+    p16 = p32.to(torch.bfloat16)
+    m = torch.zeros_like(p32)
+    v = torch.zeros_like(p32)
+    mask = (torch.rand(n, device=gpu_device) > 0.5).float()
+
+    p32r = p32.cpu().clone()
+    mr = torch.zeros(n)
+    vr = torch.zeros(n)
+
+    for step in range(1, 4):
+        g16 = torch.randn(n, device=gpu_device).to(torch.bfloat16)
+        ops.fused_adamw(p16, p32, g16, m, v, mask, 1e-2, 0.9, 0.95, 1e-8,
+                        0.1, step)
+        # reference with same masked decay
+        g32 = g16.float().cpu()
+        mr.mul_(0.9).add_(g32, alpha=0.1)
+        vr.mul_(0.95).addcmul_(g32, g32, value=0.05)
+        bc1 = 1 - 0.9 ** step
+        bc2 = 1 - 0.95 ** step
+        denom = (vr / bc2).sqrt().add_(1e-8)
+        p32r.mul_(1 - 1e-2 * 0.1 * mask.cpu())
+        p32r.addcdiv_(mr / bc1, denom, value=-1e-2)
+    assert _relerr(p32.cpu(), p32r) < 1e-4
+    assert _relerr(p16.float().cpu(), p32r) < 1e-2
+
+
+# ------------------------------------------------------- End-to-end train
+
+def test_llama_tiny_train_step(gpu_device):
+    from kubeflow_amd.models import build_model
+    from kubeflow_amd.runtime import Trainer, TrainConfig
+    torch.manual_seed(0)
+    model = build_model("llama-tiny", device=gpu_device)
+    tr = Trainer(model, TrainConfig(lr=1e-3, warmup_steps=2))
+    tokens = torch.randint(0, model.cfg.vocab_size, (2, 128), device=gpu_device)
+    targets = torch.randint(0, model.cfg.vocab_size, (2, 128), device=gpu_device)
+    losses = [float(tr.step(tokens, targets)) for _ in range(10)]
+    assert all(l == l for l in losses), losses
+    assert losses[-1] < losses[0], losses
