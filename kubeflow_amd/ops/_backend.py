@@ -30,9 +30,15 @@ def try_load():
     if not native_enabled():
         _load_error = "disabled via KF_NATIVE_KERNELS=0"
         return None
-    if not _LIB_PATH.exists():
-        _load_error = f"{_LIB_PATH} not built (run kubeflow_amd.ops.build_ext)"
-        return None
+    from . import build_ext
+    if build_ext.needs_build():
+        # stale or missing .so — rebuild (sources are authoritative; a stale
+        # library with a changed C ABI segfaults at call time)
+        try:
+            build_ext.build(verbose=True)
+        except Exception as e:
+            _load_error = f"build failed: {e}"
+            return None
     try:
         _lib = ctypes.CDLL(str(_LIB_PATH))
     except OSError as e:  # pragma: no cover
