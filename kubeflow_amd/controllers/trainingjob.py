@@ -1,0 +1,207 @@
+"""PyTorchJob / TFJob controller — gang-scheduled distributed training.
+
+Replaces the sibling-repo training-operator the reference integrates with
+(SURVEY.md §2.12): same CRD shape —
+
+    spec.pytorchReplicaSpecs.{Master,Worker}.replicas     (tfReplicaSpecs for
+    spec....restartPolicy: Never|OnFailure                 TFJob)
+    spec....template: the worker spec {model, steps, micro_batch, seq_len,
+                      lr, gpus_per_replica, ...}
+    status.conditions[]: Created/Running/Succeeded/Failed/Restarting
+    status.replicaStatuses.{Master,Worker}: {active, succeeded, failed}
+
+— but reconciled onto a gang of one-process-per-GPU workers over RCCL/xGMI
+instead of pods. Worker heartbeats (rank-*/status.json) feed job status the
+way pod state + events feed notebook status in the reference
+(apps/common/status.py:10-99).
+"""
+from __future__ import annotations
+
+import json
+import os
+import shutil
+import time
+from typing import Dict, Optional
+
+from kubeflow_amd.api import ObjectStore, set_condition, get_condition
+from kubeflow_amd.api.objects import has_condition
+from kubeflow_amd.controllers.base import Reconciler, RequeueAfter
+from kubeflow_amd.scheduler import (GangScheduler, InsufficientResources,
+                                    ProcessGang, launch_gang)
+
+TERMINAL = ("Succeeded", "Failed")
+
+
+class TrainingJobReconciler(Reconciler):
+    kind = "PyTorchJob"
+    replica_field = "pytorchReplicaSpecs"
+
+    def __init__(self, store: ObjectStore, scheduler: GangScheduler,
+                 jobs_dir: str, poll_period: float = 1.0):
+        super().__init__(store)
+        self.scheduler = scheduler
+        self.jobs_dir = jobs_dir
+        self.poll_period = poll_period
+        self.gangs: Dict[str, ProcessGang] = {}
+        self.restarts: Dict[str, int] = {}
+
+    # ------------------------------------------------------------- helpers
+    def _replicas(self, job) -> int:
+        specs = job["spec"].get(self.replica_field, {})
+        return sum(int(r.get("replicas", 1)) for r in specs.values())
+
+    def _template(self, job) -> dict:
+        specs = job["spec"].get(self.replica_field, {})
+        for role in ("Master", "Chief", "Worker"):
+            if role in specs and "template" in specs[role]:
+                return dict(specs[role]["template"])
+        return {}
+
+    def _restart_policy(self, job) -> str:
+        specs = job["spec"].get(self.replica_field, {})
+        for r in specs.values():
+            if "restartPolicy" in r:
+                return r["restartPolicy"]
+        return "Never"
+
+    def _workdir(self, job) -> str:
+        m = job["metadata"]
+        return os.path.join(self.jobs_dir, m.get("namespace") or "default",
+                            f"{m['name']}-{m['uid'][:8]}")
+
+    # ----------------------------------------------------------- reconcile
+    def reconcile(self, namespace: Optional[str], name: str) -> None:
+        job = self.store.get(self.kind, name, namespace)
+        uid = job["metadata"]["uid"]
+
+        if any(has_condition(job, t) for t in TERMINAL):
+            self._cleanup(uid)
+            return
+
+        gang = self.gangs.get(uid)
+        if gang is None:
+            self._start(job)
+            raise RequeueAfter(self.poll_period)
+
+        self._sync_status(job, gang)
+        raise RequeueAfter(self.poll_period)
+
+    def _start(self, job):
+        uid = job["metadata"]["uid"]
+        n = self._replicas(job)
+        template = self._template(job)
+        gpus_per = int(template.get("gpus_per_replica", 1))
+        import torch
+        want_gpu = gpus_per > 0 and (
+            torch.cuda.is_available() or os.environ.get("KF_FAKE_GPUS"))
+        try:
+            if want_gpu:
+                alloc = self.scheduler.allocate(uid, n * gpus_per)
+                gpu_indices = alloc.gpu_indices
+            else:
+                self.scheduler.allocate(uid, 0)
+                gpu_indices = []
+        except InsufficientResources as e:
+            if not has_condition(job, "Created"):
+                set_condition(job, "Created", "True", "JobCreated",
+                              "waiting for GPUs")
+                self.store.update(job, check_version=False)
+            self.store.record_event(job, "InsufficientResources", str(e),
+                                    "Warning")
+            raise RequeueAfter(2.0)
+
+        workdir = self._workdir(job)
+        spec = dict(template)
+        spec.setdefault("model", "mnist-mlp")
+        spec["world_size"] = n
+        poddefaults = self.store.list("PodDefault",
+                                      job["metadata"].get("namespace"))
+        numa = {g.index: g.numa_node for g in self.scheduler.inv.gpus}
+        gang = launch_gang(uid, workdir, spec, gpu_indices,
+                           poddefaults=poddefaults,
+                           labels=job["metadata"].get("labels", {}),
+                           numa_nodes=numa)
+        self.gangs[uid] = gang
+        set_condition(job, "Created", "True", "JobCreated", "gang launched")
+        set_condition(job, "Running", "True", "JobRunning",
+                      f"{n} replicas on GPUs {gpu_indices or 'cpu'}")
+        job["status"]["startTime"] = job["status"].get("startTime") or time.time()
+        self.store.update(job, check_version=False)
+        self.store.record_event(job, "SuccessfulCreate",
+                                f"created gang of {n} workers")
+
+    def _read_rank_status(self, gang: ProcessGang) -> Dict[int, dict]:
+        out = {}
+        for r in gang.ranks:
+            path = os.path.join(gang.workdir, f"rank-{r.rank}", "status.json")
+            try:
+                with open(path) as f:
+                    out[r.rank] = json.load(f)
+            except (OSError, json.JSONDecodeError):
+                out[r.rank] = {"state": "initializing"}
+        return out
+
+    def _sync_status(self, job, gang: ProcessGang):
+        uid = job["metadata"]["uid"]
+        state = gang.poll()
+        ranks = self._read_rank_status(gang)
+        active = sum(1 for s in ranks.values() if s.get("state") == "running")
+        succeeded = sum(1 for s in ranks.values()
+                        if s.get("state") == "succeeded")
+        failed = sum(1 for s in ranks.values() if s.get("state") == "failed")
+        job["status"]["replicaStatuses"] = {
+            "Worker": {"active": active, "succeeded": succeeded,
+                       "failed": failed}}
+        r0 = ranks.get(0, {})
+        if r0.get("metrics"):
+            job["status"]["trainingMetrics"] = r0["metrics"]
+            job["status"]["step"] = r0.get("step", 0)
+
+        if state == "Succeeded":
+            set_condition(job, "Running", "False", "JobFinished", "")
+            set_condition(job, "Succeeded", "True", "JobSucceeded",
+                          "all ranks exited 0")
+            job["status"]["completionTime"] = time.time()
+            self.store.record_event(job, "JobSucceeded", "training complete")
+            self.scheduler.release(uid)
+            self.gangs.pop(uid, None)
+        elif state == "Failed":
+            policy = self._restart_policy(job)
+            limit = int(job["spec"].get("backoffLimit", 3))
+            nrestart = self.restarts.get(uid, 0)
+            err = next((s.get("error") for s in ranks.values()
+                        if s.get("error")), "worker exited nonzero")
+            if policy == "OnFailure" and nrestart < limit:
+                self.restarts[uid] = nrestart + 1
+                self.scheduler.release(uid)
+                self.gangs.pop(uid, None)
+                set_condition(job, "Restarting", "True", "JobRestarting",
+                              f"restart {nrestart + 1}/{limit}: {err}")
+                self.store.record_event(job, "JobRestarting", err, "Warning")
+            else:
+                set_condition(job, "Running", "False", "JobFinished", "")
+                set_condition(job, "Failed", "True", "JobFailed", err)
+                job["status"]["completionTime"] = time.time()
+                self.store.record_event(job, "JobFailed", err, "Warning")
+                self.scheduler.release(uid)
+                self.gangs.pop(uid, None)
+        self.store.update(job, check_version=False)
+
+    def _cleanup(self, uid: str):
+        gang = self.gangs.pop(uid, None)
+        if gang is not None:
+            gang.terminate_and_wait()
+        self.scheduler.release(uid)
+
+    def shutdown(self):
+        for gang in list(self.gangs.values()):
+            gang.terminate_and_wait()
+        self.gangs.clear()
+
+
+class TFJobReconciler(TrainingJobReconciler):
+    """TFJob parity: same reconcile machinery, tfReplicaSpecs field.
+    (The reference treats TFJob/PyTorchJob as parallel integrations —
+    testing/kf_is_ready_test.py:99-113 expects both operators.)"""
+    kind = "TFJob"
+    replica_field = "tfReplicaSpecs"

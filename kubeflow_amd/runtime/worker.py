@@ -1,0 +1,140 @@
+"""Training worker process entry — what runs inside a PyTorchJob rank.
+
+Launched by the gang launcher as
+    python -m kubeflow_amd.runtime.worker --spec <workdir>/spec.json
+with RANK/WORLD_SIZE/LOCAL_RANK/MASTER_* set. Heartbeats go to
+<workdir>/rank-<r>/status.json (atomic rename); the PyTorchJob controller
+derives CR status.conditions from these the way the reference derives
+notebook status from pod state + events (apps/common/status.py:10-99).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import signal
+import sys
+import time
+
+import torch
+
+from kubeflow_amd.models import build_model
+from kubeflow_amd.parallel import dist as kdist
+from kubeflow_amd.runtime import Trainer, TrainConfig
+from kubeflow_amd.runtime import checkpoint as ckpt
+
+
+def write_status(rank_dir: str, state: str, step: int = 0, loss=None,
+                 metrics=None, error: str = ""):
+    payload = {"state": state, "step": step,
+               "loss": None if loss is None else float(loss),
+               "metrics": metrics or {}, "error": error, "ts": time.time()}
+    tmp = os.path.join(rank_dir, ".status.tmp")
+    with open(tmp, "w") as f:
+        json.dump(payload, f)
+    os.replace(tmp, os.path.join(rank_dir, "status.json"))
+
+
+def synthetic_batch(spec: dict, cfg, device, rank: int, step: int):
+    """Deterministic-per-(rank,step) synthetic data of the model's shape."""
+    seed = (int(spec.get("seed", 0)) * 1000003 + rank * 9176 + step) % (2**31)
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    kind = spec.get("task") or (
+        "mlp" if cfg is None else
+        ("classify" if hasattr(cfg, "n_classes") else "lm"))
+    if kind == "lm":
+        B = int(spec.get("micro_batch", 2))
+        S = int(spec.get("seq_len", 4096))
+        V = cfg.vocab_size
+        toks = torch.randint(0, V, (B, S + 1), generator=g)
+        return (toks[:, :-1].to(device), toks[:, 1:].contiguous().to(device))
+    if kind == "classify":
+        B = int(spec.get("micro_batch", 32))
+        S = int(spec.get("seq_len", 128))
+        V = cfg.vocab_size
+        x = torch.randint(0, V, (B, S), generator=g)
+        y = torch.randint(0, cfg.n_classes, (B,), generator=g)
+        return x.to(device), y.to(device)
+    # mnist-style dense input
+    B = int(spec.get("micro_batch", 64))
+    x = torch.randn(B, 784, generator=g)
+    y = torch.randint(0, 10, (B,), generator=g)
+    return x.to(device), y.to(device)
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--spec", required=True)
+    args = ap.parse_args(argv)
+    with open(args.spec) as f:
+        spec = json.load(f)
+
+    rank = kdist.env_rank()
+    workdir = os.environ.get("KF_JOB_WORKDIR",
+                             os.path.dirname(os.path.abspath(args.spec)))
+    rank_dir = os.path.join(workdir, f"rank-{rank}")
+    os.makedirs(rank_dir, exist_ok=True)
+    write_status(rank_dir, "initializing")
+
+    stop = {"flag": False}
+    signal.signal(signal.SIGTERM, lambda *_: stop.update(flag=True))
+
+    try:
+        rank, world, device = kdist.init_distributed()
+        torch.manual_seed(int(spec.get("seed", 0)) + rank)
+        dtype = (torch.bfloat16 if device.type == "cuda"
+                 and spec.get("dtype", "bf16") == "bf16" else torch.float32)
+        model = build_model(spec["model"], device=device, dtype=dtype)
+        cfg = getattr(model, "cfg", None)
+        tcfg = TrainConfig(
+            lr=float(spec.get("lr", 3e-4)),
+            weight_decay=float(spec.get("weight_decay", 0.1)),
+            warmup_steps=int(spec.get("warmup_steps", 10)),
+            lr_decay_steps=int(spec.get("steps", 100)),
+            grad_accum=int(spec.get("grad_accum", 1)),
+        )
+        trainer = Trainer(model, tcfg)
+
+        ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir, "checkpoints")
+        save_every = int(spec.get("save_every", 0))
+        start_step = 0
+        if spec.get("resume", True) and ckpt.latest_dir(ckpt_dir):
+            start_step = ckpt.load(trainer, ckpt_dir, rank)
+
+        steps = int(spec.get("steps", 100))
+        status_every = int(spec.get("status_every", 5))
+        write_status(rank_dir, "running", start_step)
+        loss = None
+        ema = None
+        for step in range(start_step, steps):
+            if stop["flag"]:
+                write_status(rank_dir, "failed", step, loss,
+                             error="terminated")
+                return 143
+            x, y = synthetic_batch(spec, cfg, device, rank, step)
+            loss = trainer.step(x, y)
+            if (step + 1) % status_every == 0 or step + 1 == steps:
+                lval = float(loss)
+                ema = lval if ema is None else 0.9 * ema + 0.1 * lval
+                write_status(rank_dir, "running", step + 1, lval,
+                             metrics={"loss": lval, "loss_ema": ema,
+                                      "lr": trainer.lr_at(step)})
+            if save_every and (step + 1) % save_every == 0:
+                kdist.barrier()
+                ckpt.save(trainer, ckpt_dir, spec["model"], rank, world)
+        if spec.get("save_final", True):
+            kdist.barrier()
+            ckpt.save(trainer, ckpt_dir, spec["model"], rank, world)
+        write_status(rank_dir, "succeeded", steps, loss,
+                     metrics={"loss": None if loss is None else float(loss),
+                              "loss_ema": ema})
+        return 0
+    except Exception as e:  # surface the error to the controller
+        import traceback
+        traceback.print_exc()
+        write_status(rank_dir, "failed", error=f"{type(e).__name__}: {e}")
+        return 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,183 @@
+"""Process gang launcher — one worker process per MI355X.
+
+Replaces kubelet + CRI + the admission webhook's pod mutation
+("in the rebuild this becomes a synchronous defaults
+injection pass inside the launcher"):
+
+  * per-rank env: RANK/WORLD_SIZE/LOCAL_RANK/MASTER_ADDR/MASTER_PORT +
+    HIP_VISIBLE_DEVICES pinning one GPU per process (the extension seam for
+    future TP/PP strategies);
+  * PodDefault injection: env/annotations from matching PodDefault objects
+    are merged into the worker env before spawn (same selector + conflict
+    semantics as admission-webhook/main.go:69-94,369-441);
+  * gang semantics: all ranks spawn together; if any rank dies, the whole
+    gang is killed (kill-on-peer-death); children get PR_SET_PDEATHSIG so
+    a dead controller never leaks GPU processes;
+  * NUMA affinity: worker is bound to its GPU's NUMA node via numactl when
+    available (falls back to no pinning).
+"""
+from __future__ import annotations
+
+import ctypes
+import json
+import os
+import shutil
+import signal
+import socket
+import subprocess
+import sys
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+PR_SET_PDEATHSIG = 1
+
+
+def _preexec():
+    # die with the parent (controller) — no leaked GPU processes
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    libc.prctl(PR_SET_PDEATHSIG, signal.SIGKILL)
+    os.setpgrp()
+
+
+def free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def merge_poddefaults(env: Dict[str, str], labels: Dict[str, str],
+                      poddefaults: List[dict]) -> Dict[str, str]:
+    """Merge env from PodDefaults whose selector matches `labels`.
+    Conflicting values across defaults raise, mirroring
+    safeToApplyPodDefaultsOnPod (admission-webhook/main.go:98-132)."""
+    out = dict(env)
+    origin: Dict[str, str] = {}
+    for pd in poddefaults:
+        sel = pd.get("spec", {}).get("selector", {}).get("matchLabels", {})
+        if not all(labels.get(k) == v for k, v in sel.items()):
+            continue
+        name = pd["metadata"]["name"]
+        for e in pd.get("spec", {}).get("env", []):
+            k, v = e["name"], str(e.get("value", ""))
+            if k in origin and out.get(k) != v:
+                raise ValueError(
+                    f"PodDefault conflict on env {k}: {origin[k]} vs {name}")
+            if k not in env:  # explicit job env wins over defaults
+                out[k] = v
+                origin[k] = name
+    return out
+
+
+@dataclass
+class RankProc:
+    rank: int
+    proc: subprocess.Popen
+    log_path: str
+
+
+class ProcessGang:
+    """A running gang of rank processes for one job."""
+
+    def __init__(self, job_uid: str, workdir: str):
+        self.job_uid = job_uid
+        self.workdir = workdir
+        self.ranks: List[RankProc] = []
+        self._killed = False
+        self._lock = threading.Lock()
+        self.started_at = time.time()
+
+    def poll(self) -> Optional[str]:
+        """None while running; 'Succeeded' when all ranks exited 0;
+        'Failed' otherwise. Enforces kill-on-peer-death."""
+        with self._lock:
+            codes = [r.proc.poll() for r in self.ranks]
+            if any(c not in (None, 0) for c in codes):
+                self._kill_locked()
+                return "Failed"
+            if all(c == 0 for c in codes):
+                return "Succeeded"
+            return None
+
+    def kill(self):
+        with self._lock:
+            self._kill_locked()
+
+    def _kill_locked(self):
+        if self._killed:
+            return
+        self._killed = True
+        for r in self.ranks:
+            if r.proc.poll() is None:
+                try:
+                    os.killpg(r.proc.pid, signal.SIGTERM)
+                except ProcessLookupError:
+                    pass
+        deadline = time.time() + 5
+        for r in self.ranks:
+            try:
+                r.proc.wait(timeout=max(0.1, deadline - time.time()))
+            except subprocess.TimeoutExpired:
+                try:
+                    os.killpg(r.proc.pid, signal.SIGKILL)
+                except ProcessLookupError:
+                    pass
+
+    def terminate_and_wait(self, timeout: float = 10.0):
+        self.kill()
+        for r in self.ranks:
+            try:
+                r.proc.wait(timeout=timeout)
+            except subprocess.TimeoutExpired:
+                pass
+
+
+def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
+                poddefaults: Optional[List[dict]] = None,
+                labels: Optional[Dict[str, str]] = None,
+                numa_nodes: Optional[Dict[int, int]] = None,
+                entry_module: str = "kubeflow_amd.runtime.worker") -> ProcessGang:
+    """Spawn one process per rank. CPU jobs pass gpu_indices=[] and
+    spec['world_size'] ranks run on CPU (gloo)."""
+    os.makedirs(workdir, exist_ok=True)
+    world = max(1, len(gpu_indices) or int(spec.get("world_size", 1)))
+    port = free_port()
+    spec_path = os.path.join(workdir, "spec.json")
+    with open(spec_path, "w") as f:
+        json.dump(spec, f, indent=2)
+
+    gang = ProcessGang(job_uid, workdir)
+    repo_root = os.path.dirname(os.path.dirname(
+        os.path.dirname(os.path.abspath(__file__))))
+    for rank in range(world):
+        env = dict(os.environ)
+        env.update({
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+            "KF_JOB_UID": job_uid,
+            "KF_JOB_WORKDIR": workdir,
+            "PYTHONPATH": repo_root + os.pathsep + env.get("PYTHONPATH", ""),
+        })
+        env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+        if gpu_indices:
+            env["HIP_VISIBLE_DEVICES"] = str(gpu_indices[rank])
+            env["LOCAL_RANK"] = "0"  # each proc sees exactly one GPU
+        env = merge_poddefaults(env, labels or {}, poddefaults or [])
+
+        cmd = [sys.executable, "-m", entry_module, "--spec", spec_path]
+        if gpu_indices and numa_nodes and shutil.which("numactl"):
+            node = numa_nodes.get(gpu_indices[rank], 0)
+            cmd = ["numactl", f"--cpunodebind={node}",
+                   f"--preferred={node}"] + cmd
+        rank_dir = os.path.join(workdir, f"rank-{rank}")
+        os.makedirs(rank_dir, exist_ok=True)
+        log_path = os.path.join(rank_dir, "worker.log")
+        logf = open(log_path, "w")
+        proc = subprocess.Popen(cmd, env=env, stdout=logf, stderr=logf,
+                                cwd=rank_dir, preexec_fn=_preexec)
+        gang.ranks.append(RankProc(rank, proc, log_path))
+    return gang
