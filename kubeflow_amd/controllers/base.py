@@ -46,6 +46,10 @@ class Reconciler:
     def reconcile(self, namespace: Optional[str], name: str) -> None:
         raise NotImplementedError
 
+    def on_deleted(self, namespace: Optional[str], name: str) -> None:
+        """Called when the primary object no longer exists (cleanup hook —
+        the finalizer analog of profile_controller.go:277-312)."""
+
     def map_event(self, ev: Event) -> Optional[Key]:
         """Map a watched (non-primary) object event to a primary key — the
         equivalent of the reference's owner-reference/EnqueueRequestsFrom
@@ -104,7 +108,10 @@ class _Worker(threading.Thread):
         except ConflictError:
             self.enqueue(key, 0.05)
         except NotFoundError:
-            pass  # deleted while queued — converged by definition
+            try:
+                self.rec.on_deleted(*key)
+            except Exception:
+                traceback.print_exc()
         except Exception:
             self.error_count += 1
             traceback.print_exc()

@@ -44,6 +44,7 @@ class TrainingJobReconciler(Reconciler):
         self.poll_period = poll_period
         self.gangs: Dict[str, ProcessGang] = {}
         self.restarts: Dict[str, int] = {}
+        self.key_uid: Dict[tuple, str] = {}
 
     # ------------------------------------------------------------- helpers
     def _replicas(self, job) -> int:
@@ -73,6 +74,7 @@ class TrainingJobReconciler(Reconciler):
     def reconcile(self, namespace: Optional[str], name: str) -> None:
         job = self.store.get(self.kind, name, namespace)
         uid = job["metadata"]["uid"]
+        self.key_uid[(namespace, name)] = uid
 
         if any(has_condition(job, t) for t in TERMINAL):
             self._cleanup(uid)
@@ -192,6 +194,11 @@ class TrainingJobReconciler(Reconciler):
         if gang is not None:
             gang.terminate_and_wait()
         self.scheduler.release(uid)
+
+    def on_deleted(self, namespace, name):
+        uid = self.key_uid.pop((namespace, name), None)
+        if uid:
+            self._cleanup(uid)
 
     def shutdown(self):
         for gang in list(self.gangs.values()):

@@ -48,6 +48,19 @@ def bert_tiny() -> BertConfig:
                       max_seq_len=128)
 
 
+class KfLayerNorm(nn.Module):
+    """LayerNorm over the HIP kernel (CPU falls back to reference)."""
+
+    def __init__(self, dim: int, eps: float = 1e-12):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.layer_norm(x, self.weight, self.bias, self.eps)
+
+
 class BertLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
@@ -55,10 +68,10 @@ class BertLayer(nn.Module):
         self.cfg = cfg
         self.wqkv = nn.Linear(h, 3 * hd)
         self.wo = nn.Linear(hd, h)
-        self.ln1 = nn.LayerNorm(h, eps=cfg.norm_eps)
+        self.ln1 = KfLayerNorm(h, eps=cfg.norm_eps)
         self.fc1 = nn.Linear(h, cfg.ffn_dim)
         self.fc2 = nn.Linear(cfg.ffn_dim, h)
-        self.ln2 = nn.LayerNorm(h, eps=cfg.norm_eps)
+        self.ln2 = KfLayerNorm(h, eps=cfg.norm_eps)
 
     def forward(self, x):
         cfg = self.cfg
@@ -80,7 +93,7 @@ class BertClassifier(nn.Module):
         with torch.device(device if device is not None else "cpu"):
             self.tok_embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
             self.pos_embed = nn.Embedding(cfg.max_seq_len, cfg.hidden_size)
-            self.embed_ln = nn.LayerNorm(cfg.hidden_size, eps=cfg.norm_eps)
+            self.embed_ln = KfLayerNorm(cfg.hidden_size, eps=cfg.norm_eps)
             self.layers = nn.ModuleList(
                 [BertLayer(cfg) for _ in range(cfg.n_layers)])
             self.classifier = nn.Linear(cfg.hidden_size, cfg.n_classes)
@@ -96,8 +109,9 @@ class BertClassifier(nn.Module):
             else:
                 p.zero_()
         for mod in self.modules():
-            if isinstance(mod, nn.LayerNorm):
+            if isinstance(mod, (nn.LayerNorm, KfLayerNorm)):
                 mod.weight.fill_(1.0)
+                mod.bias.zero_()
 
     def forward(self, tokens, targets=None):
         B, S = tokens.shape

@@ -22,8 +22,8 @@ from . import reference
 from . import _backend
 
 __all__ = [
-    "rms_norm", "rope", "flash_attention", "cross_entropy", "fused_adamw",
-    "rope_cos_sin", "native_available",
+    "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
+    "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
 ]
 
 rope_cos_sin = reference.rope_cos_sin
@@ -48,6 +48,18 @@ def _fp(t: Optional[torch.Tensor]):
 
 def _ip(t: torch.Tensor):
     return ctypes.cast(t.data_ptr(), ctypes.POINTER(ctypes.c_int64))
+
+
+def _ip_or_null(t):
+    if t is None:
+        return ctypes.cast(0, ctypes.POINTER(ctypes.c_int64))
+    return ctypes.cast(t.data_ptr(), ctypes.POINTER(ctypes.c_int64))
+
+
+def _i32p(t):
+    if t is None:
+        return ctypes.cast(0, ctypes.POINTER(ctypes.c_int32))
+    return ctypes.cast(t.data_ptr(), ctypes.POINTER(ctypes.c_int32))
 
 
 def _use_native(t: torch.Tensor) -> bool:
@@ -107,37 +119,49 @@ class _RopeHip(torch.autograd.Function):
     """Joint rotary embedding on (q, k); in-place rotation on fresh clones."""
 
     @staticmethod
-    def forward(ctx, q, k, cos, sin, pos_offset):
+    def forward(ctx, q, k, cos, sin, pos_offset, positions):
         lib = _backend.require()
         B, S, Hq, D = q.shape
         Hkv = k.shape[2]
         q = q.contiguous().clone(memory_format=torch.contiguous_format)
         k = k.contiguous().clone(memory_format=torch.contiguous_format)
         _backend.check(
-            lib.kf_rope(_p(q), _p(k), _fp(cos), _fp(sin), B, S, Hq, Hkv, D,
+            lib.kf_rope(_p(q), _p(k), _fp(cos), _fp(sin),
+                        _ip_or_null(positions), B, S, Hq, Hkv, D,
                         pos_offset, 0, _stream()), "rope_fwd")
-        ctx.save_for_backward(cos, sin)
+        ctx.save_for_backward(cos, sin, *(
+            [positions] if positions is not None else []))
         ctx.dims = (B, S, Hq, Hkv, D, pos_offset)
         return q, k
 
     @staticmethod
     def backward(ctx, dq, dk):
         lib = _backend.require()
-        cos, sin = ctx.saved_tensors
+        cos, sin = ctx.saved_tensors[0], ctx.saved_tensors[1]
+        positions = ctx.saved_tensors[2] if len(ctx.saved_tensors) > 2 else None
         B, S, Hq, Hkv, D, pos_offset = ctx.dims
         dq = dq.contiguous().clone(memory_format=torch.contiguous_format)
         dk = dk.contiguous().clone(memory_format=torch.contiguous_format)
         _backend.check(
-            lib.kf_rope(_p(dq), _p(dk), _fp(cos), _fp(sin), B, S, Hq, Hkv, D,
+            lib.kf_rope(_p(dq), _p(dk), _fp(cos), _fp(sin),
+                        _ip_or_null(positions), B, S, Hq, Hkv, D,
                         pos_offset, 1, _stream()), "rope_bwd")
-        return dq, dk, None, None, None
+        return dq, dk, None, None, None, None
 
 
 def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
-         sin: torch.Tensor, pos_offset: int = 0):
-    """q [B,S,Hq,D], k [B,S,Hkv,D]; cos/sin fp32 [>=S+off, D/2]."""
+         sin: torch.Tensor, pos_offset: int = 0,
+         positions: Optional[torch.Tensor] = None):
+    """q [B,S,Hq,D], k [B,S,Hkv,D]; cos/sin fp32 [>=S+off, D/2].
+    positions: optional int64 [B] per-row base position (serving decode)."""
     if _use_native(q):
-        return _RopeHip.apply(q, k, cos, sin, pos_offset)
+        return _RopeHip.apply(q, k, cos, sin, pos_offset, positions)
+    if positions is not None:
+        qs = [reference.rope_apply(q[i:i+1], cos, sin, int(positions[i]))
+              for i in range(q.shape[0])]
+        ks = [reference.rope_apply(k[i:i+1], cos, sin, int(positions[i]))
+              for i in range(k.shape[0])]
+        return torch.cat(qs), torch.cat(ks)
     return (reference.rope_apply(q, cos, sin, pos_offset),
             reference.rope_apply(k, cos, sin, pos_offset))
 
@@ -192,6 +216,89 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))
     o = reference.sdpa(qt, kt, vt, causal=causal, scale=scale)
     return o.transpose(1, 2)
+
+
+# --------------------------------------------------------------- LayerNorm --
+
+class _LayerNormHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        lib = _backend.require()
+        shape = x.shape
+        x2 = x.contiguous().view(-1, shape[-1])
+        rows, cols = x2.shape
+        y = torch.empty_like(x2)
+        mu = torch.empty(rows, dtype=torch.float32, device=x.device)
+        rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+        _backend.check(
+            lib.kf_layernorm_fwd(_p(y), _fp(mu), _fp(rstd), _p(x2), _p(weight),
+                                 _p(bias), rows, cols, float(eps), _stream()),
+            "layernorm_fwd")
+        ctx.save_for_backward(x2, weight, mu, rstd)
+        ctx.shape = shape
+        return y.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = _backend.require()
+        x2, weight, mu, rstd = ctx.saved_tensors
+        rows, cols = x2.shape
+        dy2 = dy.contiguous().view(rows, cols)
+        dx = torch.empty_like(x2)
+        dw = torch.empty_like(weight)
+        db = torch.empty_like(weight)
+        nparts = lib.kf_layernorm_bwd_nparts(rows)
+        part = torch.empty(2 * nparts * cols, dtype=torch.float32,
+                           device=x2.device)
+        _backend.check(
+            lib.kf_layernorm_bwd(_p(dx), _p(dw), _p(db), _fp(part), _p(dy2),
+                                 _p(x2), _p(weight), _fp(mu), _fp(rstd), rows,
+                                 cols, _stream()), "layernorm_bwd")
+        return dx.view(ctx.shape), dw, db, None
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+               eps: float = 1e-5):
+    if _use_native(x):
+        return _LayerNormHip.apply(x, weight, bias, eps)
+    return reference.layer_norm(x, weight, bias, eps)
+
+
+# --------------------------------------------------------- Decode attention --
+
+def attention_decode(q: torch.Tensor, kcache: torch.Tensor,
+                     vcache: torch.Tensor, slots: torch.Tensor,
+                     lens: torch.Tensor, scale: Optional[float] = None):
+    """Single-token GQA attention over the KV cache (no autograd — serving).
+
+    q [N,Hq,D] bf16; kcache/vcache [SLOTS,SMAX,Hkv,D] bf16;
+    slots/lens int32 [N]. Returns o [N,Hq,D].
+    """
+    N, Hq, D = q.shape
+    Hkv = kcache.shape[2]
+    if scale is None:
+        scale = D ** -0.5
+    if _use_native(q):
+        lib = _backend.require()
+        q = q.contiguous()
+        out = torch.empty_like(q)
+        _backend.check(
+            lib.kf_attn_decode(_p(out), _p(q), _p(kcache), _p(vcache),
+                               _i32p(slots), _i32p(lens), N, kcache.shape[1],
+                               Hq, Hkv, D, float(scale), _stream()),
+            "attn_decode")
+        return out
+    # reference path: per-sequence sdpa over the cached prefix
+    outs = []
+    for i in range(N):
+        L = int(lens[i])
+        s = int(slots[i])
+        kk = kcache[s, :L].unsqueeze(0).transpose(1, 2)   # [1,Hkv,L,D]
+        vv = vcache[s, :L].unsqueeze(0).transpose(1, 2)
+        qq = q[i].view(1, 1, Hq, D).transpose(1, 2)       # [1,Hq,1,D]
+        o = reference.sdpa(qq, kk, vv, causal=False, scale=scale)
+        outs.append(o.transpose(1, 2).reshape(1, Hq, D))
+    return torch.cat(outs)
 
 
 # --------------------------------------------------------- Cross entropy --

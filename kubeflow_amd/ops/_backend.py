@@ -64,6 +64,7 @@ def _configure(lib: ctypes.CDLL) -> None:
     c = ctypes
     P, F, I64, I32, FP = c.c_void_p, c.c_float, c.c_int64, c.c_int, c.POINTER(c.c_float)
     PI64 = c.POINTER(c.c_int64)
+    PI32 = c.POINTER(c.c_int32)
     lib.kf_rmsnorm_fwd.restype = I32
     lib.kf_rmsnorm_fwd.argtypes = [P, FP, P, P, I64, I64, F, P]
     lib.kf_rmsnorm_bwd.restype = I32
@@ -71,7 +72,7 @@ def _configure(lib: ctypes.CDLL) -> None:
     lib.kf_rmsnorm_bwd_nparts.restype = I64
     lib.kf_rmsnorm_bwd_nparts.argtypes = [I64]
     lib.kf_rope.restype = I32
-    lib.kf_rope.argtypes = [P, P, FP, FP, I64, I64, I64, I64, I64, I64, I32, P]
+    lib.kf_rope.argtypes = [P, P, FP, FP, PI64, I64, I64, I64, I64, I64, I64, I32, P]
     lib.kf_adamw.restype = I32
     lib.kf_adamw.argtypes = [P, FP, P, FP, FP, FP, I64, F, F, F, F, F, I64, P]
     lib.kf_ce_fwd.restype = I32
@@ -84,12 +85,15 @@ def _configure(lib: ctypes.CDLL) -> None:
         lib.kf_attn_bwd.restype = I32
         lib.kf_attn_bwd.argtypes = [P, P, P, P, P, P, P, P, FP, FP, I64, I64,
                                     I64, I64, I64, F, I32, P]
-    if hasattr(lib, "kf_layernorm_fwd"):
-        lib.kf_layernorm_fwd.restype = I32
-        lib.kf_layernorm_fwd.argtypes = [P, FP, FP, P, P, P, I64, I64, F, P]
-    if hasattr(lib, "kf_layernorm_bwd"):
-        lib.kf_layernorm_bwd.restype = I32
-        lib.kf_layernorm_bwd.argtypes = [P, P, P, FP, P, P, P, FP, FP, I64, I64, P]
+    lib.kf_layernorm_fwd.restype = I32
+    lib.kf_layernorm_fwd.argtypes = [P, FP, FP, P, P, P, I64, I64, F, P]
+    lib.kf_layernorm_bwd.restype = I32
+    lib.kf_layernorm_bwd.argtypes = [P, P, P, FP, P, P, P, FP, FP, I64, I64, P]
+    lib.kf_layernorm_bwd_nparts.restype = I64
+    lib.kf_layernorm_bwd_nparts.argtypes = [I64]
+    lib.kf_attn_decode.restype = I32
+    lib.kf_attn_decode.argtypes = [P, P, P, P, PI32, PI32, I64, I64, I64, I64,
+                                   I64, F, P]
 
 
 def check(err: int, name: str) -> None:

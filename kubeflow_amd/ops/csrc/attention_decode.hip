@@ -1,0 +1,136 @@
+// attention_decode.hip — single-token GQA attention over the KV cache.
+//
+// The serving-side decode hot loop (SURVEY.md §2.13 "decode = GEMV-like"):
+// memory-bound streaming of the KV cache. One block per (sequence, kv-head);
+// the GQA group's q-heads (one per wave) share each staged K/V tile, so the
+// cache is read ONCE per kv-head regardless of group size — the main
+// bandwidth lever for GQA decode.
+//
+//   q      [N, Hq, 128] bf16   (one new token per active sequence)
+//   kcache [SLOTS, SMAX, Hkv, 128] bf16 (same layout for vcache)
+//   slots  [N] int32 cache slot per sequence; lens [N] tokens to attend over
+//   out    [N, Hq, 128] bf16
+//
+// Per 64-row cache tile: all waves cooperatively stage K,V into swizzled
+// LDS; each wave owns one q-head of the group: lane r computes dot(q, K[r])
+// (one cache row per lane), online-softmax in wave registers, then lanes
+// switch to owning 2 d-elements each for the PV accumulation. Waves beyond
+// the group size still participate in staging/barriers (no divergent
+// __syncthreads). Heads beyond nwaves loop in outer chunks (extra cache
+// passes — does not occur for the Llama-3 g=4 config).
+
+#include "kf_common.h"
+
+#define AD_D 128
+#define AD_TILE 64
+
+// swizzle for [64 rows][256B] LDS tiles: byte ^= ((row&7)<<4) (guide §6 G4)
+__device__ __forceinline__ int kf_swzd(int row, int byte_in_row) {
+  return row * (AD_D * 2) + (byte_in_row ^ ((row & 7) << 4));
+}
+
+__global__ __launch_bounds__(256) void kf_attn_decode_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ q,
+    const unsigned short* __restrict__ kcache,
+    const unsigned short* __restrict__ vcache,
+    const int* __restrict__ slots, const int* __restrict__ lens,
+    int64_t smax, int Hq, int Hkv, float scale) {
+  __shared__ unsigned char k_lds[AD_TILE * AD_D * 2];
+  __shared__ unsigned char v_lds[AD_TILE * AD_D * 2];
+  __shared__ float p_lds[4][AD_TILE];
+
+  const int n = blockIdx.x, hkv = blockIdx.y;
+  const int g = Hq / Hkv;
+  const int nw = blockDim.x / KF_WAVE;
+  const int w = threadIdx.x / KF_WAVE;
+  const int lane = threadIdx.x & (KF_WAVE - 1);
+  const int len = lens[n];
+  const int64_t slot = slots[n];
+  const int64_t cbase = (slot * smax * Hkv + hkv) * AD_D;
+  const int64_t cstride = (int64_t)Hkv * AD_D;
+
+  for (int hbase = 0; hbase < g; hbase += nw) {
+    const int hg = hbase + w;
+    const bool active = hg < g;
+    const int hq = hkv * g + (active ? hg : 0);
+    const unsigned short* qrow = q + ((int64_t)n * Hq + hq) * AD_D;
+
+    float qreg[AD_D / 8][8];
+#pragma unroll
+    for (int i = 0; i < AD_D / 8; ++i) {
+      kf_short8 qv = *reinterpret_cast<const kf_short8*>(qrow + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        qreg[i][j] = kf_bf16_to_f32((unsigned short)qv[j]);
+    }
+    float m_run = -INFINITY, l_run = 0.f;
+    float o0 = 0.f, o1 = 0.f;  // lane's d-elements (d = 2*lane, 2*lane+1)
+
+    for (int t0 = 0; t0 < len; t0 += AD_TILE) {
+      const int rows = min(AD_TILE, len - t0);
+      __syncthreads();
+      for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
+        const int r = vi >> 4, c8 = vi & 15;
+        kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
+            kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+        *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kv8;
+        kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
+            vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+        *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vv8;
+      }
+      __syncthreads();
+
+      float s = -INFINITY;
+      if (lane < rows) {
+        float acc = 0.f;
+#pragma unroll
+        for (int i = 0; i < AD_D / 8; ++i) {
+          kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
+              k_lds + kf_swzd(lane, i * 16));
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc += qreg[i][j] * kf_bf16_to_f32((unsigned short)kv8[j]);
+        }
+        s = acc * scale;
+      }
+      const float tile_max = kf_wave_max(s);
+      const float m_new = fmaxf(m_run, tile_max);
+      const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+      const float p = (s == -INFINITY) ? 0.f : __expf(s - m_new);
+      m_run = m_new;
+      l_run = l_run * alpha + kf_wave_sum(p);
+      o0 *= alpha;
+      o1 *= alpha;
+      p_lds[w][lane] = p;
+      __builtin_amdgcn_s_waitcnt(0);  // wave-local p_lds write->read fence
+      for (int r = 0; r < rows; ++r) {
+        const float pr = p_lds[w][r];
+        const unsigned int vv = *reinterpret_cast<const unsigned int*>(
+            v_lds + kf_swzd(r, lane * 4));
+        o0 += pr * kf_bf16_to_f32((unsigned short)(vv & 0xffff));
+        o1 += pr * kf_bf16_to_f32((unsigned short)(vv >> 16));
+      }
+    }
+    if (active) {
+      const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+      unsigned short* orow = out + ((int64_t)n * Hq + hq) * AD_D;
+      orow[lane * 2] = kf_f32_to_bf16(o0 * inv_l);
+      orow[lane * 2 + 1] = kf_f32_to_bf16(o1 * inv_l);
+    }
+  }
+}
+
+KF_EXPORT int kf_attn_decode(void* out, const void* q, const void* kcache,
+                             const void* vcache, const int* slots,
+                             const int* lens, int64_t N, int64_t smax,
+                             int64_t Hq, int64_t Hkv, int64_t D, float scale,
+                             void* stream) {
+  if (D != AD_D || Hq % Hkv) return (int)hipErrorInvalidValue;
+  dim3 grid((unsigned)N, (unsigned)Hkv);
+  hipLaunchKernelGGL(kf_attn_decode_kernel, grid, dim3(256), 0,
+                     (hipStream_t)stream, (unsigned short*)out,
+                     (const unsigned short*)q, (const unsigned short*)kcache,
+                     (const unsigned short*)vcache, slots, lens, smax,
+                     (int)Hq, (int)Hkv, scale);
+  return (int)hipGetLastError();
+}

@@ -1,0 +1,171 @@
+// layernorm.hip — fused LayerNorm forward/backward for CDNA4 (gfx950).
+//
+// BERT-path twin of rmsnorm.hip ("rmsnorm + layernorm fused
+// kernels; LN for BERT"). Same structure: one workgroup per row grid-stride,
+// short8-vectorized bf16, fp32 block reductions, deterministic dw/db via
+// per-block partials + a column-sum kernel.
+//
+//   y = (x - mu) * rsqrt(var + eps) * w + b          (saves mu, rstd)
+//   dx = rstd * (g - mean(g) - xhat * mean(g*xhat)), g = dy*w
+//   dw_j = sum_rows dy_j * xhat_j ; db_j = sum_rows dy_j
+
+#include "kf_common.h"
+
+#define LN_BLOCK 256
+#define LN_VEC 8
+
+__global__ __launch_bounds__(LN_BLOCK) void kf_layernorm_fwd_kernel(
+    unsigned short* __restrict__ y, float* __restrict__ mu_out,
+    float* __restrict__ rstd_out, const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w, const unsigned short* __restrict__ b,
+    int64_t rows, int cols, float eps) {
+  __shared__ float scratch[LN_BLOCK / KF_WAVE];
+  const int nvec = cols / LN_VEC;
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* xr = x + row * cols;
+    unsigned short* yr = y + row * cols;
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK) {
+      kf_short8 xv = *reinterpret_cast<const kf_short8*>(xr + i * LN_VEC);
+#pragma unroll
+      for (int j = 0; j < LN_VEC; ++j) {
+        float f = kf_bf16_to_f32((unsigned short)xv[j]);
+        s1 += f;
+        s2 += f * f;
+      }
+    }
+    s1 = kf_block_reduce(s1, scratch, KfSum{}, 0.f);
+    s2 = kf_block_reduce(s2, scratch, KfSum{}, 0.f);
+    const float mu = s1 / (float)cols;
+    const float var = s2 / (float)cols - mu * mu;
+    const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (threadIdx.x == 0) {
+      if (mu_out) mu_out[row] = mu;
+      if (rstd_out) rstd_out[row] = rstd;
+    }
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK) {
+      kf_short8 xv = *reinterpret_cast<const kf_short8*>(xr + i * LN_VEC);
+      kf_short8 wv = *reinterpret_cast<const kf_short8*>(w + i * LN_VEC);
+      kf_short8 bv = *reinterpret_cast<const kf_short8*>(b + i * LN_VEC);
+      kf_short8 ov;
+#pragma unroll
+      for (int j = 0; j < LN_VEC; ++j) {
+        float f = (kf_bf16_to_f32((unsigned short)xv[j]) - mu) * rstd;
+        ov[j] = (short)kf_f32_to_bf16(
+            f * kf_bf16_to_f32((unsigned short)wv[j]) +
+            kf_bf16_to_f32((unsigned short)bv[j]));
+      }
+      *reinterpret_cast<kf_short8*>(yr + i * LN_VEC) = ov;
+    }
+  }
+}
+
+__global__ __launch_bounds__(LN_BLOCK) void kf_layernorm_bwd_kernel(
+    unsigned short* __restrict__ dx, float* __restrict__ dw_part,
+    float* __restrict__ db_part, const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ w,
+    const float* __restrict__ mu, const float* __restrict__ rstd,
+    int64_t rows, int cols) {
+  extern __shared__ float lds[];  // dw[cols] ++ db[cols] ++ scratch
+  float* dw_lds = lds;
+  float* db_lds = lds + cols;
+  float* scratch = lds + 2 * cols;
+  const int nvec = cols / LN_VEC;
+  for (int i = threadIdx.x; i < cols; i += LN_BLOCK) {
+    dw_lds[i] = 0.f;
+    db_lds[i] = 0.f;
+  }
+  __syncthreads();
+
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + row * cols;
+    const unsigned short* xr = x + row * cols;
+    unsigned short* dxr = dx + row * cols;
+    const float mu_r = mu[row], rs = rstd[row];
+    float sg = 0.f, sgx = 0.f;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK) {
+      kf_short8 dyv = *reinterpret_cast<const kf_short8*>(dyr + i * LN_VEC);
+      kf_short8 xv = *reinterpret_cast<const kf_short8*>(xr + i * LN_VEC);
+      kf_short8 wv = *reinterpret_cast<const kf_short8*>(w + i * LN_VEC);
+#pragma unroll
+      for (int j = 0; j < LN_VEC; ++j) {
+        float dyf = kf_bf16_to_f32((unsigned short)dyv[j]);
+        float xhat = (kf_bf16_to_f32((unsigned short)xv[j]) - mu_r) * rs;
+        float g = dyf * kf_bf16_to_f32((unsigned short)wv[j]);
+        sg += g;
+        sgx += g * xhat;
+        dw_lds[i * LN_VEC + j] += dyf * xhat;
+        db_lds[i * LN_VEC + j] += dyf;
+      }
+    }
+    sg = kf_block_reduce(sg, scratch, KfSum{}, 0.f) / (float)cols;
+    sgx = kf_block_reduce(sgx, scratch, KfSum{}, 0.f) / (float)cols;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK) {
+      kf_short8 dyv = *reinterpret_cast<const kf_short8*>(dyr + i * LN_VEC);
+      kf_short8 xv = *reinterpret_cast<const kf_short8*>(xr + i * LN_VEC);
+      kf_short8 wv = *reinterpret_cast<const kf_short8*>(w + i * LN_VEC);
+      kf_short8 ov;
+#pragma unroll
+      for (int j = 0; j < LN_VEC; ++j) {
+        float dyf = kf_bf16_to_f32((unsigned short)dyv[j]);
+        float xhat = (kf_bf16_to_f32((unsigned short)xv[j]) - mu_r) * rs;
+        float g = dyf * kf_bf16_to_f32((unsigned short)wv[j]);
+        ov[j] = (short)kf_f32_to_bf16(rs * (g - sg - xhat * sgx));
+      }
+      *reinterpret_cast<kf_short8*>(dxr + i * LN_VEC) = ov;
+    }
+    __syncthreads();
+  }
+  float* outw = dw_part + (int64_t)blockIdx.x * cols;
+  float* outb = db_part + (int64_t)blockIdx.x * cols;
+  for (int i = threadIdx.x; i < cols; i += LN_BLOCK) {
+    outw[i] = dw_lds[i];
+    outb[i] = db_lds[i];
+  }
+}
+
+// reuses kf_colsum_bf16_kernel from rmsnorm.hip via separate launches
+extern __global__ void kf_colsum_bf16_kernel(unsigned short*, const float*,
+                                             int, int);
+
+KF_EXPORT int kf_layernorm_fwd(void* y, float* mu, float* rstd, const void* x,
+                               const void* w, const void* b, int64_t rows,
+                               int64_t cols, float eps, void* stream) {
+  if (cols % LN_VEC) return (int)hipErrorInvalidValue;
+  hipLaunchKernelGGL(kf_layernorm_fwd_kernel, dim3(kf_grid_for(rows, 1)),
+                     dim3(LN_BLOCK), 0, (hipStream_t)stream,
+                     (unsigned short*)y, mu, rstd, (const unsigned short*)x,
+                     (const unsigned short*)w, (const unsigned short*)b, rows,
+                     (int)cols, eps);
+  return (int)hipGetLastError();
+}
+
+KF_EXPORT int64_t kf_layernorm_bwd_nparts(int64_t rows) {
+  int64_t g = rows < 1024 ? rows : 1024;
+  return g < 1 ? 1 : g;
+}
+
+KF_EXPORT int kf_layernorm_bwd(void* dx, void* dw, void* db, float* part,
+                               const void* dy, const void* x, const void* w,
+                               const float* mu, const float* rstd,
+                               int64_t rows, int64_t cols, void* stream) {
+  if (cols % LN_VEC) return (int)hipErrorInvalidValue;
+  const int grid = (int)kf_layernorm_bwd_nparts(rows);
+  float* dw_part = part;                       // [grid][cols]
+  float* db_part = part + (int64_t)grid * cols;  // [grid][cols]
+  size_t lds = (2 * cols + LN_BLOCK / KF_WAVE) * sizeof(float);
+  hipLaunchKernelGGL(kf_layernorm_bwd_kernel, dim3(grid), dim3(LN_BLOCK), lds,
+                     (hipStream_t)stream, (unsigned short*)dx, dw_part,
+                     db_part, (const unsigned short*)dy,
+                     (const unsigned short*)x, (const unsigned short*)w, mu,
+                     rstd, rows, (int)cols);
+  int err = (int)hipGetLastError();
+  if (err) return err;
+  hipLaunchKernelGGL(kf_colsum_bf16_kernel, dim3(kf_grid_for(cols, 256)),
+                     dim3(256), 0, (hipStream_t)stream, (unsigned short*)dw,
+                     dw_part, grid, (int)cols);
+  hipLaunchKernelGGL(kf_colsum_bf16_kernel, dim3(kf_grid_for(cols, 256)),
+                     dim3(256), 0, (hipStream_t)stream, (unsigned short*)db,
+                     db_part, grid, (int)cols);
+  return (int)hipGetLastError();
+}

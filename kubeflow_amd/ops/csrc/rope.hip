@@ -16,6 +16,7 @@ __global__ void kf_rope_kernel(unsigned short* __restrict__ q,
                                unsigned short* __restrict__ k,
                                const float* __restrict__ cost,
                                const float* __restrict__ sint,
+                               const int64_t* __restrict__ positions,
                                int64_t B, int S, int Hq, int Hkv, int D,
                                int64_t pos_offset, int backward) {
   const int Ht = Hq + Hkv;
@@ -36,10 +37,11 @@ __global__ void kf_rope_kernel(unsigned short* __restrict__ q,
     const int d0 = qd * ROPE_VEC;
     kf_short4 x1 = *reinterpret_cast<const kf_short4*>(base + d0);
     kf_short4 x2 = *reinterpret_cast<const kf_short4*>(base + halfD + d0);
+    const int64_t pos = (positions ? positions[b] : pos_offset) + s;
     kf_float4 c = *reinterpret_cast<const kf_float4*>(
-        cost + (pos_offset + s) * (int64_t)halfD + d0);
+        cost + pos * (int64_t)halfD + d0);
     kf_float4 sn = *reinterpret_cast<const kf_float4*>(
-        sint + (pos_offset + s) * (int64_t)halfD + d0);
+        sint + pos * (int64_t)halfD + d0);
     kf_short4 o1, o2;
 #pragma unroll
     for (int j = 0; j < ROPE_VEC; ++j) {
@@ -55,13 +57,14 @@ __global__ void kf_rope_kernel(unsigned short* __restrict__ q,
 }
 
 KF_EXPORT int kf_rope(void* q, void* k, const float* cost, const float* sint,
-                      int64_t B, int64_t S, int64_t Hq, int64_t Hkv, int64_t D,
-                      int64_t pos_offset, int backward, void* stream) {
+                      const int64_t* positions, int64_t B, int64_t S,
+                      int64_t Hq, int64_t Hkv, int64_t D, int64_t pos_offset,
+                      int backward, void* stream) {
   if ((D / 2) % ROPE_VEC) return (int)hipErrorInvalidValue;
   const int64_t total = B * S * (Hq + Hkv) * (D / 2 / ROPE_VEC);
   hipLaunchKernelGGL(kf_rope_kernel, dim3(kf_grid_for(total, 256)), dim3(256),
                      0, (hipStream_t)stream, (unsigned short*)q,
-                     (unsigned short*)k, cost, sint, B, (int)S, (int)Hq,
-                     (int)Hkv, (int)D, pos_offset, backward);
+                     (unsigned short*)k, cost, sint, positions, B, (int)S,
+                     (int)Hq, (int)Hkv, (int)D, pos_offset, backward);
   return (int)hipGetLastError();
 }

@@ -1,0 +1,284 @@
+"""InferenceService engine — continuous batching on one MI355X, no Triton.
+
+The runtime half of the KServe InferenceService the reference integrates
+with (SURVEY.md §2.12 "InferenceService: predictor spec, dynamic batching").
+
+MI355X-first design: with 288 GB of HBM3E, the KV cache is a single
+preallocated region of fixed-stride slots ([SLOTS, SMAX, Hkv, D] per layer)
+rather than a paged pool — slot granularity removes page tables and keeps
+each sequence's K/V rows contiguous for the decode kernel's streaming reads.
+(Llama-3-8B: 16 GB weights + 64 slots × 8192 tokens × 128 KB/token ≈ 80 GB —
+a third of one GPU.)
+
+Scheduling: continuous batching. Each engine iteration admits queued
+prefills (chunked into the running batch) and then runs ONE batched decode
+step for every active sequence via the kf_attn_decode kernel. Prefill uses
+the training flash-attention kernel over the padded prompt.
+"""
+from __future__ import annotations
+
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from kubeflow_amd import ops
+from kubeflow_amd.models import build_model
+from kubeflow_amd.models.llama import LlamaModel
+
+
+class KVCache:
+    """Per-layer slot cache: k/v [SLOTS, SMAX, Hkv, D] bf16."""
+
+    def __init__(self, n_layers: int, slots: int, smax: int, hkv: int,
+                 d: int, device, dtype=torch.bfloat16):
+        self.k = [torch.zeros(slots, smax, hkv, d, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.v = [torch.zeros(slots, smax, hkv, d, device=device, dtype=dtype)
+                  for _ in range(n_layers)]
+        self.slots = slots
+        self.smax = smax
+        self._free = list(range(slots))
+        self._lock = threading.Lock()
+
+    def alloc(self) -> Optional[int]:
+        with self._lock:
+            return self._free.pop() if self._free else None
+
+    def free(self, slot: int):
+        with self._lock:
+            self._free.append(slot)
+
+
+@dataclass
+class Request:
+    rid: str
+    prompt: List[int]
+    max_new_tokens: int = 32
+    temperature: float = 0.0
+    submitted: float = field(default_factory=time.time)
+    # filled by the engine:
+    slot: int = -1
+    pos: int = 0
+    generated: List[int] = field(default_factory=list)
+    done: threading.Event = field(default_factory=threading.Event)
+    first_token_at: Optional[float] = None
+    finished_at: Optional[float] = None
+    error: str = ""
+
+
+class InferenceEngine:
+    def __init__(self, model_name: str, device=None, max_slots: int = 32,
+                 smax: int = 4096, max_batch: int = 32):
+        self.device = device or (torch.device("cuda", 0)
+                                 if torch.cuda.is_available()
+                                 else torch.device("cpu"))
+        dtype = (torch.bfloat16 if self.device.type == "cuda"
+                 else torch.float32)
+        self.model: LlamaModel = build_model(model_name, device=self.device,
+                                             dtype=dtype)
+        self.model.eval()
+        cfg = self.model.cfg
+        smax = min(smax, cfg.max_seq_len)
+        self.cache = KVCache(cfg.n_layers, max_slots, smax,
+                             cfg.n_kv_heads, cfg.head_dim, self.device, dtype)
+        self.max_batch = max_batch
+        self.pending: "queue.Queue[Request]" = queue.Queue()
+        self.active: List[Request] = []
+        self._stop = False
+        self._thread: Optional[threading.Thread] = None
+        self.stats = {"requests": 0, "completed": 0, "tokens_out": 0,
+                      "prefill_tokens": 0}
+
+    # ------------------------------------------------------------- public
+    def start(self):
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="inference-engine")
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop = True
+        if self._thread:
+            self._thread.join(timeout=10)
+
+    def submit(self, req: Request) -> Request:
+        self.stats["requests"] += 1
+        self.pending.put(req)
+        return req
+
+    def generate(self, prompt: List[int], max_new_tokens: int = 32,
+                 timeout: float = 120.0) -> Request:
+        req = Request(rid=f"r{time.monotonic_ns()}", prompt=list(prompt),
+                      max_new_tokens=max_new_tokens)
+        self.submit(req)
+        if not req.done.wait(timeout):
+            req.error = req.error or "timeout"
+        return req
+
+    # -------------------------------------------------------------- engine
+    @torch.no_grad()
+    def _loop(self):
+        while not self._stop:
+            admitted = self._admit()
+            if not self.active:
+                try:
+                    req = self.pending.get(timeout=0.01)
+                    self._start_request(req)
+                except queue.Empty:
+                    continue
+                continue
+            self._decode_step()
+
+    def _admit(self) -> int:
+        n = 0
+        while len(self.active) < self.max_batch:
+            try:
+                req = self.pending.get_nowait()
+            except queue.Empty:
+                break
+            self._start_request(req)
+            n += 1
+        return n
+
+    def _start_request(self, req: Request):
+        slot = self.cache.alloc()
+        if slot is None:
+            # no slot free: push back and decode on (slots free as seqs end)
+            self.pending.put(req)
+            return
+        req.slot = slot
+        try:
+            self._prefill(req)
+            self.active.append(req)
+        except Exception as e:  # pragma: no cover
+            import traceback
+            traceback.print_exc()
+            req.error = f"{type(e).__name__}: {e}"
+            self.cache.free(slot)
+            req.finished_at = time.time()
+            req.done.set()
+
+    @torch.no_grad()
+    def _prefill(self, req: Request):
+        """Run the prompt through the model, filling this slot's cache and
+        producing the first generated token."""
+        cfg = self.model.cfg
+        keep = max(1, self.cache.smax - req.max_new_tokens - 1)
+        prompt = req.prompt[-keep:]
+        S = len(prompt)
+        tokens = torch.tensor([prompt], dtype=torch.int64, device=self.device)
+        x = self.model.embed(tokens)
+        cos, sin = self.model.rope_cos, self.model.rope_sin
+        for li, layer in enumerate(self.model.layers):
+            x = self._layer_prefill(layer, li, x, cos, sin, req.slot, S)
+        x = self.model.final_norm(x[:, -1:])
+        logits = torch.nn.functional.linear(x, self.model.lm_head.weight)
+        tok = self._sample(logits[0, -1], req.temperature)
+        req.pos = S
+        req.generated.append(tok)
+        req.first_token_at = time.time()
+        self.stats["prefill_tokens"] += S
+
+    def _layer_prefill(self, layer, li, x, cos, sin, slot, S):
+        cfg = self.model.cfg
+        B = 1
+        import torch.nn.functional as F
+        qkv = F.linear(layer.attn_norm(x), layer.wqkv.weight)
+        q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
+                             cfg.n_kv_heads * cfg.head_dim,
+                             cfg.n_kv_heads * cfg.head_dim], dim=-1)
+        q = q.view(B, S, cfg.n_heads, cfg.head_dim)
+        k = k.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        v = v.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        q, k = ops.rope(q, k, cos, sin, 0)
+        self.cache.k[li][slot, :S] = k[0]
+        self.cache.v[li][slot, :S] = v[0]
+        if self.device.type == "cuda" and S % 64 == 0:
+            o = ops.flash_attention(q, k, v, causal=True)
+        else:
+            # arbitrary prompt lengths: pad to 64 for the tile kernel; the
+            # padded rows' outputs are discarded
+            pad = (64 - S % 64) % 64
+            if pad and self.device.type == "cuda":
+                zq = torch.zeros(B, pad, cfg.n_heads, cfg.head_dim,
+                                 device=q.device, dtype=q.dtype)
+                zk = torch.zeros(B, pad, cfg.n_kv_heads, cfg.head_dim,
+                                 device=q.device, dtype=q.dtype)
+                o = ops.flash_attention(torch.cat([q, zq], 1),
+                                        torch.cat([k, zk], 1),
+                                        torch.cat([v, zk], 1),
+                                        causal=True)[:, :S]
+            else:
+                o = ops.flash_attention(q, k, v, causal=True)
+        o = layer.wo(o.reshape(B, S, cfg.n_heads * cfg.head_dim))
+        x = x + o
+        g, u = torch.nn.functional.linear(
+            layer.mlp_norm(x), layer.w13.weight).split(
+                [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
+        return x + layer.w2(torch.nn.functional.silu(g) * u)
+
+    @torch.no_grad()
+    def _decode_step(self):
+        """One token for every active sequence, batched."""
+        cfg = self.model.cfg
+        import torch.nn.functional as F
+        acts = self.active
+        N = len(acts)
+        tokens = torch.tensor([[r.generated[-1]] for r in acts],
+                              dtype=torch.int64, device=self.device)
+        positions = torch.tensor([r.pos for r in acts], dtype=torch.int64,
+                                 device=self.device)
+        slots = torch.tensor([r.slot for r in acts], dtype=torch.int32,
+                             device=self.device)
+        lens = torch.tensor([r.pos + 1 for r in acts], dtype=torch.int32,
+                            device=self.device)
+        x = self.model.embed(tokens)  # [N,1,H]
+        cos, sin = self.model.rope_cos, self.model.rope_sin
+        for li, layer in enumerate(self.model.layers):
+            qkv = F.linear(layer.attn_norm(x), layer.wqkv.weight)
+            q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
+                                 cfg.n_kv_heads * cfg.head_dim,
+                                 cfg.n_kv_heads * cfg.head_dim], dim=-1)
+            q = q.view(N, 1, cfg.n_heads, cfg.head_dim)
+            k = k.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
+            v = v.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
+            q, k = ops.rope(q, k, cos, sin, positions=positions)
+            # append to cache at pos
+            self.cache.k[li][slots.long(), positions] = k[:, 0]
+            self.cache.v[li][slots.long(), positions] = v[:, 0]
+            o = ops.attention_decode(q[:, 0], self.cache.k[li],
+                                     self.cache.v[li], slots, lens)
+            o = layer.wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim))
+            x = x + o
+            g, u = F.linear(layer.mlp_norm(x), layer.w13.weight).split(
+                [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
+            x = x + layer.w2(F.silu(g) * u)
+        x = self.model.final_norm(x)
+        logits = F.linear(x, self.model.lm_head.weight)  # [N,1,V]
+        still = []
+        toks = logits[:, -1].argmax(-1).tolist()  # greedy batch decode
+        now = time.time()
+        for i, r in enumerate(acts):
+            r.pos += 1
+            r.generated.append(int(toks[i]))
+            self.stats["tokens_out"] += 1
+            if (len(r.generated) >= r.max_new_tokens
+                    or r.pos + 1 >= self.cache.smax):
+                r.finished_at = now
+                self.cache.free(r.slot)
+                self.stats["completed"] += 1
+                r.done.set()
+            else:
+                still.append(r)
+        self.active = still
+
+    @staticmethod
+    def _sample(logits: torch.Tensor, temperature: float) -> int:
+        if temperature <= 0:
+            return int(logits.argmax().item())
+        probs = torch.softmax(logits.float() / temperature, dim=-1)
+        return int(torch.multinomial(probs, 1).item())

@@ -1,0 +1,104 @@
+"""InferenceService server process — KServe-compatible HTTP front end.
+
+Launched by the InferenceService controller as
+    python -m kubeflow_amd.runtime.serving_server --spec <dir>/spec.json
+
+Endpoints (KServe data-plane shapes, which the reference's E2E suite probes
+— testing/test_tf_serving.py hits a served model over HTTP):
+    GET  /healthz                     liveness/readiness
+    GET  /v1/models/<name>            model metadata + ready flag
+    POST /v1/models/<name>:predict    {"instances": [{"prompt_tokens": [...],
+                                       "max_new_tokens": N}, ...]}
+    POST /v2/generate                 single generate request
+    GET  /metrics                     Prometheus text format
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+from fastapi import FastAPI, Request as HttpRequest
+from fastapi.responses import JSONResponse, PlainTextResponse
+import uvicorn
+
+from kubeflow_amd.runtime.serving import InferenceEngine
+
+
+def build_app(spec: dict) -> FastAPI:
+    name = spec.get("name", "model")
+    engine = InferenceEngine(
+        spec.get("model", "llama-tiny"),
+        max_slots=int(spec.get("max_slots", 16)),
+        smax=int(spec.get("max_seq_len", 2048)),
+        max_batch=int(spec.get("max_batch", 16)),
+    ).start()
+    app = FastAPI(title=f"kubeflow-amd inference: {name}")
+    app.state.engine = engine
+    t0 = time.time()
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok", "uptime_s": time.time() - t0}
+
+    @app.get(f"/v1/models/{name}")
+    def model_meta():
+        return {"name": name, "ready": True,
+                "model": spec.get("model"),
+                "stats": engine.stats}
+
+    @app.post(f"/v1/models/{name}:predict")
+    async def predict(req: HttpRequest):
+        body = await req.json()
+        instances = body.get("instances", [])
+        preds = []
+        for inst in instances:
+            r = engine.generate(inst.get("prompt_tokens", [1]),
+                                int(inst.get("max_new_tokens", 16)))
+            preds.append({
+                "tokens": r.generated,
+                "error": r.error,
+                "ttft_ms": (None if r.first_token_at is None else
+                            round(1000 * (r.first_token_at - r.submitted), 2)),
+                "latency_ms": (None if r.finished_at is None else
+                               round(1000 * (r.finished_at - r.submitted), 2)),
+            })
+        return {"predictions": preds}
+
+    @app.post("/v2/generate")
+    async def generate(req: HttpRequest):
+        body = await req.json()
+        r = engine.generate(body.get("prompt_tokens", [1]),
+                            int(body.get("max_new_tokens", 16)))
+        return {"tokens": r.generated, "error": r.error}
+
+    @app.get("/metrics")
+    def metrics():
+        s = engine.stats
+        lines = [
+            "# TYPE kf_serving_requests_total counter",
+            f'kf_serving_requests_total{{model="{name}"}} {s["requests"]}',
+            "# TYPE kf_serving_completed_total counter",
+            f'kf_serving_completed_total{{model="{name}"}} {s["completed"]}',
+            "# TYPE kf_serving_tokens_out_total counter",
+            f'kf_serving_tokens_out_total{{model="{name}"}} {s["tokens_out"]}',
+        ]
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    return app
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--spec", required=True)
+    args = ap.parse_args(argv)
+    with open(args.spec) as f:
+        spec = json.load(f)
+    app = build_app(spec)
+    uvicorn.run(app, host="127.0.0.1", port=int(spec.get("port", 8085)),
+                log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

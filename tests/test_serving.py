@@ -1,0 +1,99 @@
+"""Serving engine + InferenceService controller tests (CPU, llama-tiny)."""
+import json
+import time
+import urllib.request
+
+import pytest
+import torch
+
+from kubeflow_amd.api import new_object
+from kubeflow_amd.api.objects import has_condition
+from kubeflow_amd.platform import Platform
+from kubeflow_amd.runtime.serving import InferenceEngine, KVCache
+
+
+def test_engine_generate_cpu():
+    eng = InferenceEngine("llama-tiny", max_slots=4, smax=256,
+                          max_batch=4).start()
+    try:
+        r = eng.generate([1, 2, 3, 4, 5], max_new_tokens=8, timeout=120)
+        assert r.error == ""
+        assert len(r.generated) == 8
+        assert all(0 <= t < eng.model.cfg.vocab_size for t in r.generated)
+        assert r.first_token_at is not None and r.finished_at is not None
+    finally:
+        eng.stop()
+
+
+def test_engine_batched_requests():
+    eng = InferenceEngine("llama-tiny", max_slots=4, smax=256,
+                          max_batch=4).start()
+    try:
+        import threading
+        results = [None] * 3
+        def run(i):
+            results[i] = eng.generate([1 + i, 2, 3], max_new_tokens=5,
+                                      timeout=120)
+        ts = [threading.Thread(target=run, args=(i,)) for i in range(3)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(130)
+        for r in results:
+            assert r is not None and r.error == "" and len(r.generated) == 5
+        assert eng.stats["completed"] >= 3
+    finally:
+        eng.stop()
+
+
+def test_decode_matches_full_forward():
+    """Incremental decode with KV cache must match a full forward pass."""
+    torch.manual_seed(0)
+    eng = InferenceEngine("llama-tiny", max_slots=2, smax=128, max_batch=2)
+    model = eng.model
+    prompt = [3, 14, 15, 9, 2, 6]
+    r = eng.generate.__self__  # no thread started; drive manually
+    req_cls = type("R", (), {})
+    from kubeflow_amd.runtime.serving import Request
+    req = Request(rid="t", prompt=list(prompt), max_new_tokens=4)
+    req.slot = eng.cache.alloc()
+    eng._prefill(req)
+    eng.active = [req]
+    eng._decode_step()
+    # reference: argmax of logits from the full sequence
+    with torch.no_grad():
+        full = model(torch.tensor([prompt], dtype=torch.int64))
+        t1 = int(full[0, -1].argmax())
+    assert req.generated[0] == t1, (req.generated, t1)
+    with torch.no_grad():
+        full2 = model(torch.tensor([prompt + [t1]], dtype=torch.int64))
+        t2 = int(full2[0, -1].argmax())
+    assert req.generated[1] == t2, (req.generated, t2)
+
+
+def test_inference_service_e2e(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        svc = new_object("InferenceService", "tiny-svc", "default", spec={
+            "predictor": {"model": "llama-tiny", "gpus": 0,
+                          "maxSlots": 2, "maxSeqLen": 256, "maxBatch": 2}},
+            api_version="serving.kserve.io/v1beta1")
+        plat.store.create(svc)
+        deadline = time.time() + 120
+        url = None
+        while time.time() < deadline:
+            obj = plat.store.get("InferenceService", "tiny-svc", "default")
+            if has_condition(obj, "Ready"):
+                url = obj["status"]["url"]
+                break
+            time.sleep(0.5)
+        assert url, obj["status"]
+        body = json.dumps({"instances": [
+            {"prompt_tokens": [1, 2, 3], "max_new_tokens": 4}]}).encode()
+        req = urllib.request.Request(
+            f"{url}/v1/models/tiny-svc:predict", data=body,
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=60) as resp:
+            out = json.loads(resp.read())
+        assert len(out["predictions"]) == 1
+        assert len(out["predictions"][0]["tokens"]) == 4
+        assert out["predictions"][0]["latency_ms"] is not None
