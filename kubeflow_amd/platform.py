@@ -44,29 +44,23 @@ class Platform:
         for rec in (self.pytorchjob, self.tfjob):
             self.manager.register(rec)
             self._controllers.append(rec)
-        try:
-            from kubeflow_amd.controllers.inference import InferenceServiceReconciler
-            self.inference = InferenceServiceReconciler(
-                self.store, self.scheduler,
-                os.path.join(self.root_dir, "serving"))
-            self.manager.register(self.inference)
-            self._controllers.append(self.inference)
-        except ImportError:
-            self.inference = None
-        try:
-            from kubeflow_amd.controllers.katib import ExperimentReconciler
-            self.katib = ExperimentReconciler(self.store)
-            self.manager.register(self.katib)
-            self._controllers.append(self.katib)
-        except ImportError:
-            self.katib = None
-        try:
-            from kubeflow_amd.controllers.pipeline import PipelineRunReconciler
-            self.pipeline = PipelineRunReconciler(self.store)
-            self.manager.register(self.pipeline)
-            self._controllers.append(self.pipeline)
-        except ImportError:
-            self.pipeline = None
+        from kubeflow_amd.controllers.inference import InferenceServiceReconciler
+        self.inference = InferenceServiceReconciler(
+            self.store, self.scheduler,
+            os.path.join(self.root_dir, "serving"))
+        self.manager.register(self.inference)
+        self._controllers.append(self.inference)
+        from kubeflow_amd.controllers.katib import (ExperimentReconciler,
+                                                    TrialReconciler)
+        self.katib = ExperimentReconciler(self.store)
+        self.trial = TrialReconciler(self.store)
+        for rec in (self.katib, self.trial):
+            self.manager.register(rec)
+            self._controllers.append(rec)
+        from kubeflow_amd.controllers.pipeline import PipelineRunReconciler
+        self.pipeline = PipelineRunReconciler(self.store)
+        self.manager.register(self.pipeline)
+        self._controllers.append(self.pipeline)
         try:
             from kubeflow_amd.controllers.notebook import NotebookReconciler
             self.notebook = NotebookReconciler(

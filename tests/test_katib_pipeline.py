@@ -1,0 +1,127 @@
+"""Katib Experiment + Pipeline DAG e2e on CPU (tiny models, world_size=1).
+
+Mirrors the reference's katib_studyjob_test (submit CR, poll conditions)
+and BASELINE config 5's DAG chaining.
+"""
+import time
+
+import pytest
+
+from kubeflow_amd.api import new_object
+from kubeflow_amd.api.objects import has_condition
+from kubeflow_amd.katib import make_suggestion
+from kubeflow_amd.platform import Platform
+
+
+def _wait(store, kind, name, ns, types=("Succeeded", "Failed"), timeout=180):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        obj = store.get(kind, name, ns)
+        for t in types:
+            if has_condition(obj, t):
+                return t, obj
+        time.sleep(0.3)
+    raise AssertionError(f"{kind}/{name} not terminal: {obj['status']}")
+
+
+PARAMS = [
+    {"name": "lr", "parameterType": "double",
+     "feasibleSpace": {"min": "0.001", "max": "0.1", "logScale": True}},
+    {"name": "micro_batch", "parameterType": "int",
+     "feasibleSpace": {"min": "8", "max": "32"}},
+]
+
+
+def test_suggestion_algorithms():
+    for alg in ("random", "grid", "bayesianoptimization"):
+        sug = make_suggestion(alg, PARAMS, seed=1)
+        props = sug.suggest([], 4)
+        assert len(props) == 4
+        for p in props:
+            assert 0.001 <= p["lr"] <= 0.1
+            assert 8 <= p["micro_batch"] <= 32
+    # bayesopt with observations proposes valid points
+    sug = make_suggestion("bayesianoptimization", PARAMS, seed=1)
+    obs = [({"lr": 0.01, "micro_batch": 16}, 0.5),
+           ({"lr": 0.05, "micro_batch": 8}, 0.9),
+           ({"lr": 0.002, "micro_batch": 32}, 0.3),
+           ({"lr": 0.09, "micro_batch": 12}, 1.1)]
+    props = sug.suggest(obs, 2)
+    assert len(props) == 2 and all(0.001 <= p["lr"] <= 0.1 for p in props)
+
+
+def test_experiment_e2e(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        exp = new_object("Experiment", "hpo-1", "default", spec={
+            "objective": {"type": "minimize",
+                          "objectiveMetricName": "loss"},
+            "algorithm": {"algorithmName": "random"},
+            "parallelTrialCount": 2,
+            "maxTrialCount": 4,
+            "maxFailedTrialCount": 2,
+            "parameters": PARAMS,
+            "trialTemplate": {
+                "model": "mnist-mlp", "steps": 4, "gpus_per_replica": 0,
+                "status_every": 2, "save_final": False, "replicas": 1,
+            },
+        }, api_version="kubeflow.org/v1beta1")
+        plat.store.create(exp)
+        state, obj = _wait(plat.store, "Experiment", "hpo-1", "default",
+                           timeout=300)
+        assert state == "Succeeded", obj["status"]
+        assert obj["status"]["trials"] == 4
+        assert obj["status"]["trialsSucceeded"] == 4
+        opt = obj["status"]["currentOptimalTrial"]
+        assert opt["observation"]["metrics"][0]["latest"] is not None
+        # trials carry assignments within the space
+        trials = plat.store.list("Trial", "default", {"experiment": "hpo-1"})
+        assert len(trials) == 4
+        for t in trials:
+            a = {x["name"]: x["value"]
+                 for x in t["spec"]["parameterAssignments"]}
+            assert 0.001 <= a["lr"] <= 0.1
+
+
+def test_pipeline_dag_e2e(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        run = new_object("PipelineRun", "pipe-1", "default", spec={
+            "tasks": [
+                {"name": "preprocess", "dependencies": [],
+                 "template": {"model": "mnist-mlp", "steps": 2,
+                              "gpus_per_replica": 0, "save_final": False}},
+                {"name": "train", "dependencies": ["preprocess"],
+                 "template": {"model": "mnist-mlp", "steps": 3,
+                              "gpus_per_replica": 0, "save_final": False}},
+                {"name": "eval", "dependencies": ["train"],
+                 "template": {"model": "mnist-mlp", "steps": 2,
+                              "gpus_per_replica": 0, "save_final": False}},
+                {"name": "deploy", "dependencies": ["eval"],
+                 "kind": "InferenceService",
+                 "template": {"model": "llama-tiny", "gpus": 0,
+                              "maxSlots": 2, "maxSeqLen": 256}},
+            ],
+        }, api_version="pipelines.kubeflow.org/v1")
+        plat.store.create(run)
+        state, obj = _wait(plat.store, "PipelineRun", "pipe-1", "default",
+                           timeout=300)
+        assert state == "Succeeded", obj["status"]
+        ts = obj["status"]["taskStates"]
+        assert all(ts[k] == "Succeeded"
+                   for k in ("preprocess", "train", "eval", "deploy"))
+        # ordering respected: train started only after preprocess succeeded
+        evs = plat.store.events_for(obj)
+        started = [e["message"] for e in evs if e["reason"] == "TaskStarted"]
+        assert started.index(
+            next(m for m in started if m.startswith("preprocess"))) < \
+            started.index(next(m for m in started if m.startswith("train")))
+
+
+def test_pipeline_invalid_dag(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        run = new_object("PipelineRun", "bad", "default", spec={
+            "tasks": [{"name": "a", "dependencies": ["ghost"],
+                       "template": {}}]})
+        plat.store.create(run)
+        state, obj = _wait(plat.store, "PipelineRun", "bad", "default",
+                           timeout=30)
+        assert state == "Failed"
