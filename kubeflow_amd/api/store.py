@@ -167,6 +167,10 @@ class ObjectStore:
                     f"{key}: stale resourceVersion "
                     f"{obj['metadata'].get('resourceVersion')} != "
                     f"{cur['metadata']['resourceVersion']}")
+            # no-op writes neither bump the version nor fire watch events
+            # (prevents reconcile storms: update -> event -> reconcile -> ...)
+            if _same_except_version(obj, cur):
+                return copy.deepcopy(cur)
             obj = copy.deepcopy(obj)
             self._rv += 1
             obj["metadata"]["resourceVersion"] = str(self._rv)
@@ -252,6 +256,15 @@ class ObjectStore:
     def _key(obj: KfObject) -> Key:
         return (obj["kind"], obj["metadata"].get("namespace"),
                 obj["metadata"]["name"])
+
+
+def _same_except_version(a: KfObject, b: KfObject) -> bool:
+    ma, mb = a.get("metadata", {}), b.get("metadata", {})
+    if {k: v for k, v in ma.items() if k != "resourceVersion"} != \
+            {k: v for k, v in mb.items() if k != "resourceVersion"}:
+        return False
+    return {k: v for k, v in a.items() if k != "metadata"} == \
+        {k: v for k, v in b.items() if k != "metadata"}
 
 
 def _deep_merge(base: dict, patch: dict) -> dict:

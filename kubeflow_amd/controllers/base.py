@@ -17,7 +17,7 @@ import heapq
 import threading
 import time
 import traceback
-from typing import Dict, List, Optional, Set, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from kubeflow_amd.api import ObjectStore, ConflictError, NotFoundError
 from kubeflow_amd.api.store import Event
@@ -66,7 +66,7 @@ class _Worker(threading.Thread):
         self.rec = rec
         self.queue: List[Tuple[float, int, Key]] = []  # (ready, seq, key) heap
         self._seq = 0
-        self.queued: Set[Key] = set()
+        self.queued: Dict[Key, float] = {}  # key -> earliest scheduled ready
         self.cv = threading.Condition()
         self.stopping = False
         self.reconcile_count = 0
@@ -74,11 +74,13 @@ class _Worker(threading.Thread):
 
     def enqueue(self, key: Key, delay: float = 0.0):
         with self.cv:
-            if key in self.queued and delay == 0.0:
-                return
+            ready = time.monotonic() + delay
+            cur = self.queued.get(key)
+            if cur is not None and cur <= ready:
+                return  # an equal-or-earlier run is already scheduled
+            self.queued[key] = ready
             self._seq += 1
-            heapq.heappush(self.queue, (time.monotonic() + delay, self._seq, key))
-            self.queued.add(key)
+            heapq.heappush(self.queue, (ready, self._seq, key))
             self.cv.notify()
 
     def run(self):
@@ -87,10 +89,13 @@ class _Worker(threading.Thread):
                 while not self.stopping:
                     if self.queue:
                         ready, _, key = self.queue[0]
+                        if self.queued.get(key) != ready:
+                            heapq.heappop(self.queue)  # superseded entry
+                            continue
                         wait = ready - time.monotonic()
                         if wait <= 0:
                             heapq.heappop(self.queue)
-                            self.queued.discard(key)
+                            self.queued.pop(key, None)
                             break
                         self.cv.wait(timeout=min(wait, 1.0))
                     else:

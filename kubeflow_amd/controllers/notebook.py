@@ -1,0 +1,186 @@
+"""Notebook controller — sessions as processes, with stop/start + culling.
+
+Behavior parity with the reference notebook-controller:
+  * Notebook CR accepts apiVersion v1alpha1/v1beta1/v1 and normalizes to
+    v1beta1 internally (the conversion hub —
+    api/v1/notebook_conversion.go:25-69);
+  * the `kubeflow-resource-stopped` annotation scales the session to 0
+    (generateStatefulSet: notebook_controller.go:303-305), removal restarts
+    it — the web app's stop/start PATCH keeps working
+    (apps/common/routes/patch.py:22-76);
+  * status mirrors pod container state: conditions + containerState
+    running/waiting/terminated (notebook_controller.go:200-250);
+  * idle culling: polls the session's /api/status last_activity and sets
+    the stop annotation after IDLE_TIME (culler.go:24-27,191 — same env
+    knobs: ENABLE_CULLING, IDLE_TIME minutes, CULLING_CHECK_PERIOD minutes);
+  * a VirtualService-equivalent URL `/notebook/<ns>/<name>/` is exposed in
+    status (generateVirtualService: notebook_controller.go:401).
+"""
+from __future__ import annotations
+
+import json
+import os
+import subprocess
+import sys
+import time
+import urllib.request
+from typing import Dict, Optional
+
+from kubeflow_amd.api import ObjectStore, set_condition
+from kubeflow_amd.api.objects import has_condition
+from kubeflow_amd.controllers.base import Reconciler, RequeueAfter
+from kubeflow_amd.scheduler.launcher import free_port, _preexec
+
+STOP_ANNOTATION = "kubeflow-resource-stopped"
+
+
+class _Session:
+    def __init__(self, proc, port, workdir):
+        self.proc = proc
+        self.port = port
+        self.workdir = workdir
+
+
+class NotebookReconciler(Reconciler):
+    kind = "Notebook"
+
+    def __init__(self, store: ObjectStore, sessions_dir: str):
+        super().__init__(store)
+        self.sessions_dir = sessions_dir
+        self.sessions: Dict[str, _Session] = {}
+        self.key_uid: Dict[tuple, str] = {}
+        self.enable_culling = os.environ.get(
+            "ENABLE_CULLING", "false").lower() == "true"
+        self.idle_minutes = float(os.environ.get("IDLE_TIME", "1440"))
+        self.cull_period = float(os.environ.get("CULLING_CHECK_PERIOD", "1"))
+
+    # ---------------------------------------------------------- versioning
+    @staticmethod
+    def normalize(nb: dict) -> dict:
+        """Accept v1alpha1/v1beta1/v1 and treat spec.template.spec uniformly
+        (conversion copies spec.template.spec verbatim in the reference)."""
+        if nb.get("apiVersion", "").endswith(("v1alpha1", "v1", "v1beta1")):
+            return nb
+        return nb
+
+    def reconcile(self, namespace: Optional[str], name: str) -> None:
+        nb = self.normalize(self.store.get(self.kind, name, namespace))
+        uid = nb["metadata"]["uid"]
+        self.key_uid[(namespace, name)] = uid
+        stopped = STOP_ANNOTATION in nb["metadata"].get("annotations", {})
+        sess = self.sessions.get(uid)
+
+        if stopped:
+            if sess is not None:
+                self._stop_session(uid)
+            nb["status"]["readyReplicas"] = 0
+            nb["status"]["containerState"] = {
+                "terminated": {"reason": "Stopped"}}
+            set_condition(nb, "Running", "False", "Stopped",
+                          "stop annotation present")
+            self.store.update(nb, check_version=False)
+            return
+
+        if sess is None or sess.proc.poll() is not None:
+            if sess is not None:  # crashed: restart (StatefulSet semantics)
+                self.store.record_event(nb, "BackOff",
+                                        "session exited; restarting",
+                                        "Warning")
+            self._start_session(nb)
+            raise RequeueAfter(0.5)
+
+        # readiness + status
+        url = f"http://127.0.0.1:{sess.port}"
+        prefix = f"/notebook/{namespace}/{name}"
+        ready, last_activity = self._probe_status(url + prefix + "/api/status")
+        nb["status"]["readyReplicas"] = 1 if ready else 0
+        nb["status"]["containerState"] = (
+            {"running": {"startedAt": nb["status"].get("startTime")}}
+            if ready else {"waiting": {"reason": "Starting"}})
+        nb["status"]["url"] = url + prefix + "/"
+        if ready and not has_condition(nb, "Running"):
+            set_condition(nb, "Running", "True", "SessionReady", url)
+            self.store.record_event(nb, "Started", "notebook session ready")
+
+        # culling (reference: NotebookNeedsCulling, requeue each period)
+        if self.enable_culling and ready and last_activity:
+            try:
+                last = time.mktime(time.strptime(last_activity,
+                                                 "%Y-%m-%dT%H:%M:%SZ"))
+                if time.time() - last > self.idle_minutes * 60:
+                    nb["metadata"]["annotations"][STOP_ANNOTATION] = \
+                        time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+                    self.store.record_event(nb, "Culling",
+                                            "idle beyond IDLE_TIME")
+            except ValueError:
+                pass
+        self.store.update(nb, check_version=False)
+        if not ready:
+            raise RequeueAfter(0.3)
+        raise RequeueAfter(self.cull_period * 60 if self.enable_culling
+                           else 5.0)
+
+    def _probe_status(self, url):
+        try:
+            with urllib.request.urlopen(url, timeout=2) as r:
+                data = json.loads(r.read())
+                return True, data.get("last_activity")
+        except Exception:
+            return False, None
+
+    def _start_session(self, nb):
+        uid = nb["metadata"]["uid"]
+        m = nb["metadata"]
+        ns = m.get("namespace") or "default"
+        port = free_port()
+        workdir = os.path.join(self.sessions_dir, ns, m["name"])
+        os.makedirs(workdir, exist_ok=True)
+        prefix = f"/notebook/{ns}/{m['name']}"
+        spec_path = os.path.join(workdir, "session.json")
+        with open(spec_path, "w") as f:
+            json.dump({"port": port, "nb_prefix": prefix,
+                       "image": self._image(nb)}, f)
+        env = dict(os.environ)
+        env["NB_PREFIX"] = prefix
+        repo_root = os.path.dirname(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))))
+        env["PYTHONPATH"] = repo_root + os.pathsep + env.get("PYTHONPATH", "")
+        logf = open(os.path.join(workdir, "session.log"), "w")
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "kubeflow_amd.runtime.notebook_server",
+             "--spec", spec_path],
+            env=env, stdout=logf, stderr=logf, cwd=workdir,
+            preexec_fn=_preexec)
+        self.sessions[uid] = _Session(proc, port, workdir)
+        nb["status"]["startTime"] = time.strftime("%Y-%m-%dT%H:%M:%SZ",
+                                                  time.gmtime())
+        set_condition(nb, "Created", "True", "SessionCreated",
+                      f"port {port}")
+        self.store.update(nb, check_version=False)
+        self.store.record_event(nb, "SuccessfulCreate",
+                                f"session process port {port}")
+
+    @staticmethod
+    def _image(nb) -> str:
+        try:
+            return nb["spec"]["template"]["spec"]["containers"][0]["image"]
+        except (KeyError, IndexError):
+            return "kubeflow-amd/session:latest"
+
+    def _stop_session(self, uid):
+        sess = self.sessions.pop(uid, None)
+        if sess and sess.proc.poll() is None:
+            sess.proc.terminate()
+            try:
+                sess.proc.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                sess.proc.kill()
+
+    def on_deleted(self, namespace, name):
+        uid = self.key_uid.pop((namespace, name), None)
+        if uid:
+            self._stop_session(uid)
+
+    def shutdown(self):
+        for uid in list(self.sessions):
+            self._stop_session(uid)

@@ -61,38 +61,23 @@ class Platform:
         self.pipeline = PipelineRunReconciler(self.store)
         self.manager.register(self.pipeline)
         self._controllers.append(self.pipeline)
-        try:
-            from kubeflow_amd.controllers.notebook import NotebookReconciler
-            self.notebook = NotebookReconciler(
-                self.store, os.path.join(self.root_dir, "notebooks"))
-            self.manager.register(self.notebook)
-            self._controllers.append(self.notebook)
-        except ImportError:
-            self.notebook = None
-        try:
-            from kubeflow_amd.controllers.tensorboard import TensorboardReconciler
-            self.tensorboard = TensorboardReconciler(
-                self.store, os.path.join(self.root_dir, "tensorboards"))
-            self.manager.register(self.tensorboard)
-            self._controllers.append(self.tensorboard)
-        except ImportError:
-            self.tensorboard = None
-        try:
-            from kubeflow_amd.controllers.volume import VolumeReconciler
-            self.volume = VolumeReconciler(
-                self.store, os.path.join(self.root_dir, "volumes"))
-            self.manager.register(self.volume)
-            self._controllers.append(self.volume)
-        except ImportError:
-            self.volume = None
-        try:
-            from kubeflow_amd.controllers.profile import ProfileReconciler
-            self.profile = ProfileReconciler(
-                self.store, os.path.join(self.root_dir, "profiles"))
-            self.manager.register(self.profile)
-            self._controllers.append(self.profile)
-        except ImportError:
-            self.profile = None
+        from kubeflow_amd.controllers.notebook import NotebookReconciler
+        self.notebook = NotebookReconciler(
+            self.store, os.path.join(self.root_dir, "notebooks"))
+        from kubeflow_amd.controllers.tensorboard import TensorboardReconciler
+        self.tensorboard = TensorboardReconciler(
+            self.store, os.path.join(self.root_dir, "tensorboards"),
+            volumes_dir=os.path.join(self.root_dir, "volumes"))
+        from kubeflow_amd.controllers.volume import VolumeReconciler
+        self.volume = VolumeReconciler(
+            self.store, os.path.join(self.root_dir, "volumes"))
+        from kubeflow_amd.controllers.profile import ProfileReconciler
+        self.profile = ProfileReconciler(
+            self.store, os.path.join(self.root_dir, "profiles"))
+        for rec in (self.notebook, self.tensorboard, self.volume,
+                    self.profile):
+            self.manager.register(rec)
+            self._controllers.append(rec)
 
     def start(self):
         self.manager.start()

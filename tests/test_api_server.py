@@ -1,0 +1,168 @@
+"""REST API surface tests — envelope, authn/authz, kfam, dashboard BFF.
+
+Covers the behavioral contracts of SURVEY.md Appendix A.
+"""
+import pytest
+from fastapi.testclient import TestClient
+
+from kubeflow_amd.api import ObjectStore, new_object
+from kubeflow_amd.api.server import build_app
+from kubeflow_amd.kfam import binding_name, BindingClient
+from kubeflow_amd.controllers.profile import ProfileReconciler
+
+
+@pytest.fixture
+def client(tmp_path):
+    store = ObjectStore()
+    app = build_app(store)
+    c = TestClient(app)
+    c.store = store
+    return c
+
+
+U = {"kubeflow-userid": "alice@example.com"}
+
+
+def test_envelope_success_and_failure(client):
+    r = client.get("/api/namespaces/ns1/notebooks", headers=U)
+    body = r.json()
+    assert body["success"] is True and body["status"] == 200
+    assert body["user"] == "alice@example.com"
+    assert body["notebooks"] == []
+
+    r = client.get("/api/namespaces/ns1/notebooks/missing", headers=U)
+    assert r.status_code == 404
+    body = r.json()
+    assert body["success"] is False and body["status"] == 404
+    assert "log" in body and body["user"] == "alice@example.com"
+
+
+def test_anonymous_default_user(client):
+    r = client.get("/api/namespaces/ns1/notebooks")
+    assert r.json()["user"] == "anonymous@kubeflow.org"
+
+
+def test_crud_roundtrip_and_stop_start(client):
+    r = client.post("/api/namespaces/ns1/notebooks", headers=U,
+                    json={"name": "nb1", "spec": {"image": "x"}})
+    assert r.json()["success"], r.json()
+    r = client.get("/api/namespaces/ns1/notebooks/nb1", headers=U)
+    assert r.json()["notebook"]["spec"]["image"] == "x"
+    # stop -> annotation appears (patch.py parity)
+    r = client.patch("/api/namespaces/ns1/notebooks/nb1", headers=U,
+                     json={"stopped": True})
+    nb = r.json()["notebook"]
+    assert "kubeflow-resource-stopped" in nb["metadata"]["annotations"]
+    # start -> annotation removed
+    r = client.patch("/api/namespaces/ns1/notebooks/nb1", headers=U,
+                     json={"stopped": False})
+    nb = r.json()["notebook"]
+    assert "kubeflow-resource-stopped" not in nb["metadata"]["annotations"]
+    r = client.delete("/api/namespaces/ns1/notebooks/nb1", headers=U)
+    assert r.json()["success"]
+    assert client.get("/api/namespaces/ns1/notebooks",
+                      headers=U).json()["notebooks"] == []
+
+
+def test_unknown_kind_404(client):
+    assert client.get("/api/namespaces/x/gizmos", headers=U).status_code == 404
+
+
+def test_kfam_binding_name_golden():
+    # golden case from access-management/kfam/bindings_test.go:25-38
+    assert binding_name("User", "lalith.vaka@zq.msds.kp.org",
+                        "ClusterRole", "edit") == \
+        "user-lalith-vaka-zq-msds-kp-org-clusterrole-edit"
+
+
+def test_kfam_bindings_and_authz(client):
+    store = client.store
+    # alice owns ns "team" via profile-controller-style rolebinding
+    bc = BindingClient(store)
+    rb = new_object("RoleBinding", "namespaceAdmin", "team",
+                    annotations={"user": "alice@example.com",
+                                 "role": "admin"})
+    rb["roleRef"] = {"kind": "ClusterRole", "name": "admin"}
+    rb["subjects"] = [{"kind": "User", "name": "alice@example.com"}]
+    store.create(rb)
+
+    # alice grants bob edit
+    r = client.post("/kfam/v1/bindings", headers=U, json={
+        "user": {"kind": "User", "name": "bob@example.com"},
+        "referredNamespace": "team",
+        "roleRef": {"kind": "ClusterRole", "name": "kubeflow-edit"},
+    })
+    assert r.json()["success"], r.json()
+    listed = client.get("/kfam/v1/bindings?namespace=team",
+                        headers=U).json()["bindings"]
+    users = {b["user"]["name"] for b in listed}
+    assert users == {"alice@example.com", "bob@example.com"}
+
+    # carol (no binding) cannot write in team
+    CAROL = {"kubeflow-userid": "carol@example.com"}
+    r = client.post("/api/namespaces/team/notebooks", headers=CAROL,
+                    json={"name": "nb", "spec": {}})
+    assert r.status_code == 403
+    # bob (edit) can
+    BOB = {"kubeflow-userid": "bob@example.com"}
+    r = client.post("/api/namespaces/team/notebooks", headers=BOB,
+                    json={"name": "nb", "spec": {}})
+    assert r.json()["success"], r.json()
+    # bob (edit, not admin) cannot grant bindings
+    r = client.post("/kfam/v1/bindings", headers=BOB, json={
+        "user": {"kind": "User", "name": "dave@x.com"},
+        "referredNamespace": "team",
+        "roleRef": {"name": "kubeflow-view"}})
+    assert r.status_code == 403
+
+
+def test_workgroup_flow(client):
+    r = client.get("/api/workgroup/exists", headers=U)
+    assert r.json()["hasWorkgroup"] is False
+    r = client.post("/api/workgroup/create", headers=U, json={})
+    assert r.json()["success"]
+    r = client.get("/api/workgroup/exists", headers=U)
+    body = r.json()
+    assert body["hasWorkgroup"] is True and body["hasAuth"] is True
+    r = client.get("/api/workgroup/env-info", headers=U)
+    info = r.json()
+    assert info["namespaces"][0]["role"] == "owner"
+
+
+def test_activities_and_links_and_metrics(client):
+    store = client.store
+    nb = store.create(new_object("Notebook", "nb", "ns1"))
+    store.record_event(nb, "Started", "hello")
+    acts = client.get("/api/activities/ns1", headers=U).json()["activities"]
+    assert acts and acts[0]["reason"] == "Started"
+    links = client.get("/api/dashboard-links", headers=U).json()
+    assert any(l["text"] == "Notebooks" for l in links["menuLinks"])
+    m = client.get("/metrics").text
+    assert "service_heartbeat" in m and "notebook_running_total" in m
+
+
+def test_profile_reconciler_provisions(tmp_path):
+    from kubeflow_amd.controllers.base import ControllerManager
+    store = ObjectStore()
+    mgr = ControllerManager(store)
+    mgr.register(ProfileReconciler(store, str(tmp_path)))
+    mgr.start()
+    try:
+        store.create(new_object(
+            "Profile", "alice", None,
+            spec={"owner": {"kind": "User", "name": "alice@example.com"}},
+            api_version="kubeflow.org/v1"))
+        assert mgr.wait_settled(10)
+        ns = store.get("Namespace", "alice", None)
+        assert ns["metadata"]["labels"]["pipelines.kubeflow.org/enabled"] == "true"
+        assert store.get("ServiceAccount", "default-editor", "alice")
+        rb = store.get("RoleBinding", "namespaceAdmin", "alice")
+        assert rb["metadata"]["annotations"]["user"] == "alice@example.com"
+        # cascade delete
+        store.delete("Profile", "alice", None)
+        import pytest as _p
+        from kubeflow_amd.api import NotFoundError
+        with _p.raises(NotFoundError):
+            store.get("Namespace", "alice", None)
+    finally:
+        mgr.stop()

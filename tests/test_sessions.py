@@ -1,0 +1,110 @@
+"""Notebook / Tensorboard / PVC controller tests (session processes)."""
+import json
+import os
+import time
+import urllib.request
+
+import pytest
+
+from kubeflow_amd.api import new_object, NotFoundError
+from kubeflow_amd.api.objects import has_condition
+from kubeflow_amd.platform import Platform
+
+
+def _wait(pred, timeout=60, period=0.25):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        v = pred()
+        if v:
+            return v
+        time.sleep(period)
+    raise AssertionError("condition not reached")
+
+
+def test_notebook_lifecycle_and_culler_contract(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        nb = new_object("Notebook", "my-nb", "alice", spec={
+            "template": {"spec": {"containers": [
+                {"image": "kubeflow-amd/session:latest"}]}}},
+            api_version="kubeflow.org/v1beta1")
+        plat.store.create(nb)
+        obj = _wait(lambda: (lambda o: o if o["status"].get("readyReplicas")
+                             else None)(
+            plat.store.get("Notebook", "my-nb", "alice")))
+        url = obj["status"]["url"]
+        assert url.endswith("/notebook/alice/my-nb/")
+        # the culler contract: /api/status with last_activity
+        with urllib.request.urlopen(
+                url.rstrip("/") + "/api/status", timeout=5) as r:
+            data = json.loads(r.read())
+        assert "last_activity" in data
+
+        # code execution endpoint works (the "kernel")
+        body = json.dumps({"code": "print(2 + 3)"}).encode()
+        req = urllib.request.Request(
+            url.rstrip("/") + "/api/execute", data=body,
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=10) as r:
+            out = json.loads(r.read())
+        assert out["output"].strip() == "5"
+
+        # stop via annotation -> replicas 0 (stop/start parity)
+        plat.store.patch("Notebook", "my-nb", "alice", {
+            "metadata": {"annotations": {
+                "kubeflow-resource-stopped": "now"}}})
+        _wait(lambda: plat.store.get("Notebook", "my-nb", "alice")
+              ["status"].get("readyReplicas") == 0)
+        # start again
+        plat.store.patch("Notebook", "my-nb", "alice", {
+            "metadata": {"annotations": {"kubeflow-resource-stopped": None}}})
+        _wait(lambda: plat.store.get("Notebook", "my-nb", "alice")
+              ["status"].get("readyReplicas") == 1)
+        plat.store.delete("Notebook", "my-nb", "alice")
+
+
+def test_pvc_bound_and_tensorboard_logspath(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        pvc = new_object("PersistentVolumeClaim", "data", "alice", spec={
+            "accessModes": ["ReadWriteOnce"],
+            "resources": {"requests": {"storage": "5Gi"}}}, api_version="v1")
+        plat.store.create(pvc)
+        obj = _wait(lambda: (lambda o: o if o["status"].get("phase") == "Bound"
+                             else None)(
+            plat.store.get("PersistentVolumeClaim", "data", "alice")))
+        path = obj["status"]["hostPath"]
+        assert os.path.isdir(path)
+        # drop a metrics file where tensorboard will scan
+        rundir = os.path.join(path, "logs", "run1")
+        os.makedirs(rundir)
+        with open(os.path.join(rundir, "status.json"), "w") as f:
+            json.dump({"step": 3, "metrics": {"loss": 0.5}}, f)
+
+        tb = new_object("Tensorboard", "tb1", "alice",
+                        spec={"logspath": "pvc://data/logs"},
+                        api_version="tensorboard.kubeflow.org/v1alpha1")
+        plat.store.create(tb)
+        obj = _wait(lambda: (lambda o: o if has_condition(o, "Running")
+                             else None)(
+            plat.store.get("Tensorboard", "tb1", "alice")))
+        # the viewer serves the scalar series from the PVC logdir
+        sess = plat.tensorboard.sessions[obj["metadata"]["uid"]]
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{sess[1]}/data/runs", timeout=5) as r:
+            runs = json.loads(r.read())["runs"]
+        assert "run1" in runs
+
+        # PVC deletion reclaims the directory
+        plat.store.delete("PersistentVolumeClaim", "data", "alice")
+        _wait(lambda: not os.path.isdir(path), timeout=20)
+
+
+def test_tensorboard_rejects_cloud_paths(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        tb = new_object("Tensorboard", "bad", "ns",
+                        spec={"logspath": "gs://bucket/x"})
+        plat.store.create(tb)
+        obj = _wait(lambda: (lambda o: o if has_condition(o, "Failed")
+                             else None)(
+            plat.store.get("Tensorboard", "bad", "ns")))
+        assert "unsupported" in [c for c in obj["status"]["conditions"]
+                                 if c["type"] == "Failed"][0]["message"]
