@@ -250,7 +250,9 @@ class ObjectStore:
     def events_for(self, involved: KfObject) -> List[KfObject]:
         evs = self.list("Event", involved["metadata"].get("namespace"))
         uid = involved["metadata"]["uid"]
-        return [e for e in evs if e.get("involvedObject", {}).get("uid") == uid]
+        mine = [e for e in evs if e.get("involvedObject", {}).get("uid") == uid]
+        mine.sort(key=lambda e: int(e["metadata"]["resourceVersion"]))
+        return mine
 
     @staticmethod
     def _key(obj: KfObject) -> Key:
