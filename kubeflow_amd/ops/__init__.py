@@ -207,10 +207,29 @@ class _FlashAttnHip(torch.autograd.Function):
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     causal: bool = True, scale: Optional[float] = None):
-    """bshd layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] -> o [B,S,Hq,D]."""
+    """bshd layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] -> o [B,S,Hq,D].
+
+    The CDNA4 kernel tiles in 64-row blocks; for causal attention,
+    sequences are zero-padded to a 64 multiple here (padded key rows are
+    causally masked for every real query, so results are exact) and the
+    output is sliced back.
+    """
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if _use_native(q):
+        S = q.shape[1]
+        pad = (64 - S % 64) % 64
+        if pad:
+            if not causal:
+                raise ValueError(
+                    "non-causal flash_attention requires seq_len % 64 == 0 "
+                    f"(got {S}); pad inputs with an attention mask upstream")
+            zq = q.new_zeros(q.shape[0], pad, q.shape[2], q.shape[3])
+            zk = k.new_zeros(k.shape[0], pad, k.shape[2], k.shape[3])
+            q = torch.cat([q, zq], dim=1)
+            k = torch.cat([k, zk], dim=1)
+            v = torch.cat([v, zk], dim=1)
+            return _FlashAttnHip.apply(q, k, v, causal, scale)[:, :S]
         return _FlashAttnHip.apply(q, k, v, causal, scale)
     # reference path works in bhsd
     qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))

@@ -197,23 +197,7 @@ class InferenceEngine:
         q, k = ops.rope(q, k, cos, sin, 0)
         self.cache.k[li][slot, :S] = k[0]
         self.cache.v[li][slot, :S] = v[0]
-        if self.device.type == "cuda" and S % 64 == 0:
-            o = ops.flash_attention(q, k, v, causal=True)
-        else:
-            # arbitrary prompt lengths: pad to 64 for the tile kernel; the
-            # padded rows' outputs are discarded
-            pad = (64 - S % 64) % 64
-            if pad and self.device.type == "cuda":
-                zq = torch.zeros(B, pad, cfg.n_heads, cfg.head_dim,
-                                 device=q.device, dtype=q.dtype)
-                zk = torch.zeros(B, pad, cfg.n_kv_heads, cfg.head_dim,
-                                 device=q.device, dtype=q.dtype)
-                o = ops.flash_attention(torch.cat([q, zq], 1),
-                                        torch.cat([k, zk], 1),
-                                        torch.cat([v, zk], 1),
-                                        causal=True)[:, :S]
-            else:
-                o = ops.flash_attention(q, k, v, causal=True)
+        o = ops.flash_attention(q, k, v, causal=True)
         o = layer.wo(o.reshape(B, S, cfg.n_heads * cfg.head_dim))
         x = x + o
         g, u = torch.nn.functional.linear(

@@ -198,3 +198,76 @@ def test_llama_tiny_train_step(gpu_device):
     losses = [float(tr.step(tokens, targets)) for _ in range(10)]
     assert all(l == l for l in losses), losses
     assert losses[-1] < losses[0], losses
+
+
+# -------------------------------------------------------------- LayerNorm
+
+@pytest.mark.parametrize("rows,cols", [(512, 768), (64, 256)])
+def test_layernorm_fwd_bwd(rows, cols, gpu_device):
+    torch.manual_seed(0)
+    x = torch.randn(rows, cols, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(cols, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(cols, device=gpu_device, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.layer_norm(x, w, b, eps=1e-6)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    br = b.detach().float().cpu().requires_grad_(True)
+    yr = R.layer_norm(xr, wr, br, eps=1e-6)
+    yr.backward(dy.float().cpu())
+    assert _relerr(y.cpu(), yr) < 2e-2
+    assert _relerr(x.grad.cpu(), xr.grad) < 5e-2
+    assert _relerr(w.grad.cpu(), wr.grad) < 3e-2
+    assert _relerr(b.grad.cpu(), br.grad) < 3e-2
+
+
+# ------------------------------------------------------- Decode attention
+
+def test_attention_decode_matches_sdpa(gpu_device):
+    torch.manual_seed(0)
+    N, Hq, Hkv, D, SMAX = 4, 8, 2, 128, 256
+    kcache = torch.randn(8, SMAX, Hkv, D, device=gpu_device,
+                         dtype=torch.bfloat16)
+    vcache = torch.randn(8, SMAX, Hkv, D, device=gpu_device,
+                         dtype=torch.bfloat16)
+    q = torch.randn(N, Hq, D, device=gpu_device, dtype=torch.bfloat16)
+    slots = torch.tensor([5, 0, 3, 7], dtype=torch.int32, device=gpu_device)
+    lens = torch.tensor([200, 1, 77, 256], dtype=torch.int32,
+                        device=gpu_device)
+    out = ops.attention_decode(q, kcache, vcache, slots, lens)
+    # reference per sequence
+    for i in range(N):
+        L, s = int(lens[i]), int(slots[i])
+        ref = R.sdpa(q[i].view(1, 1, Hq, D).float().cpu().transpose(1, 2),
+                     kcache[s, :L].unsqueeze(0).float().cpu().transpose(1, 2),
+                     vcache[s, :L].unsqueeze(0).float().cpu().transpose(1, 2),
+                     causal=False, scale=D ** -0.5)
+        assert _relerr(out[i].cpu(), ref.transpose(1, 2).reshape(Hq, D)) < 2e-2, i
+
+
+def test_serving_engine_gpu_decode_consistency(gpu_device):
+    """Engine incremental decode must match full forward on GPU (llama-tiny,
+    all HIP kernels on the path)."""
+    from kubeflow_amd.runtime.serving import InferenceEngine, Request
+    torch.manual_seed(0)
+    eng = InferenceEngine("llama-tiny", device=gpu_device, max_slots=2,
+                          smax=256, max_batch=2)
+    prompt = [3, 14, 15, 9, 2, 6, 1, 2]
+    req = Request(rid="t", prompt=list(prompt), max_new_tokens=3)
+    req.slot = eng.cache.alloc()
+    eng._prefill(req)
+    eng.active = [req]
+    eng._decode_step()
+    eng.active = [req]
+    with torch.no_grad():
+        full = eng.model(torch.tensor([prompt], device=gpu_device))
+        t1 = int(full[0, -1].argmax())
+        full2 = eng.model(torch.tensor([prompt + [t1]], device=gpu_device))
+        t2 = int(full2[0, -1].argmax())
+    assert req.generated[0] == t1
+    assert req.generated[1] == t2
