@@ -91,25 +91,28 @@ class LlamaBlock(nn.Module):
 
     def forward(self, x, cos, sin, pos_offset: int = 0, kv_cache=None):
         cfg = self.cfg
-        B, S, _ = x.shape
+        B, S, h = x.shape
         qkv = F.linear(self.attn_norm(x), self.wqkv.weight)
-        q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
-                             cfg.n_kv_heads * cfg.head_dim,
-                             cfg.n_kv_heads * cfg.head_dim], dim=-1)
-        q = q.view(B, S, cfg.n_heads, cfg.head_dim)
-        k = k.view(B, S, cfg.n_kv_heads, cfg.head_dim)
-        v = v.view(B, S, cfg.n_kv_heads, cfg.head_dim)
-        q, k = ops.rope(q, k, cos, sin, pos_offset)
         if kv_cache is not None:
+            q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
+                                 cfg.n_kv_heads * cfg.head_dim,
+                                 cfg.n_kv_heads * cfg.head_dim], dim=-1)
+            q = q.view(B, S, cfg.n_heads, cfg.head_dim)
+            k = k.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+            v = v.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+            q, k = ops.rope(q, k, cos, sin, pos_offset)
             k, v = kv_cache.update(k, v, pos_offset)
             o = ops.flash_attention(q, k, v, causal=(S > 1))
+            o = o.reshape(B, S, cfg.n_heads * cfg.head_dim)
         else:
-            o = ops.flash_attention(q, k, v, causal=True)
-        o = self.wo(o.reshape(B, S, cfg.n_heads * cfg.head_dim))
-        x = x + o
-        g, u = F.linear(self.mlp_norm(x), self.w13.weight).split(
-            [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
-        return x + self.w2(F.silu(g) * u)
+            o = ops.fused_qkv_attention(qkv, cos, sin, cfg.n_heads,
+                                        cfg.n_kv_heads, cfg.head_dim)
+        # residuals fused into the GEMM epilogue (addmm: C = input + A @ B)
+        x = torch.addmm(x.view(-1, h), o.view(B * S, -1),
+                        self.wo.weight.t()).view(B, S, h)
+        y = ops.swiglu(F.linear(self.mlp_norm(x), self.w13.weight))
+        return torch.addmm(x.view(-1, h), y.view(B * S, cfg.ffn_dim),
+                           self.w2.weight.t()).view(B, S, h)
 
 
 class LlamaModel(nn.Module):

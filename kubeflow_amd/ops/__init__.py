@@ -24,6 +24,7 @@ from . import _backend
 __all__ = [
     "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
+    "swiglu", "fused_qkv_attention",
 ]
 
 rope_cos_sin = reference.rope_cos_sin
@@ -128,7 +129,8 @@ class _RopeHip(torch.autograd.Function):
         _backend.check(
             lib.kf_rope(_p(q), _p(k), _fp(cos), _fp(sin),
                         _ip_or_null(positions), B, S, Hq, Hkv, D,
-                        pos_offset, 0, _stream()), "rope_fwd")
+                        Hq * D, Hkv * D, pos_offset, 0, _stream()),
+            "rope_fwd")
         ctx.save_for_backward(cos, sin, *(
             [positions] if positions is not None else []))
         ctx.dims = (B, S, Hq, Hkv, D, pos_offset)
@@ -145,7 +147,8 @@ class _RopeHip(torch.autograd.Function):
         _backend.check(
             lib.kf_rope(_p(dq), _p(dk), _fp(cos), _fp(sin),
                         _ip_or_null(positions), B, S, Hq, Hkv, D,
-                        pos_offset, 1, _stream()), "rope_bwd")
+                        Hq * D, Hkv * D, pos_offset, 1, _stream()),
+            "rope_bwd")
         return dq, dk, None, None, None, None
 
 
@@ -179,8 +182,8 @@ class _FlashAttnHip(torch.autograd.Function):
         lse = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
         _backend.check(
             lib.kf_attn_fwd(_p(o), _fp(lse), _p(q), _p(k), _p(v), B, S, Hq,
-                            Hkv, D, float(scale), int(causal), _stream()),
-            "attn_fwd")
+                            Hkv, D, 0, 0, float(scale), int(causal),
+                            _stream()), "attn_fwd")
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.meta = (causal, scale)
         return o
@@ -200,8 +203,8 @@ class _FlashAttnHip(torch.autograd.Function):
         _backend.check(
             lib.kf_attn_bwd(_p(dq), _p(dk), _p(dv), _p(dout), _p(q), _p(k),
                             _p(v), _p(o), _fp(lse), _fp(delta), B, S, Hq, Hkv,
-                            D, float(scale), int(causal), _stream()),
-            "attn_bwd")
+                            D, 0, 0, 0, 0, float(scale), int(causal),
+                            _stream()), "attn_bwd")
         return dq, dk, dv, None, None
 
 
@@ -209,7 +212,7 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     causal: bool = True, scale: Optional[float] = None):
     """bshd layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] -> o [B,S,Hq,D].
 
-    The CDNA4 kernel tiles in 64-row blocks; for causal attention,
+    The CDNA4 kernel tiles q in 128-row blocks; for causal attention,
     sequences are zero-padded to a 64 multiple here (padded key rows are
     causally masked for every real query, so results are exact) and the
     output is sliced back.
@@ -218,11 +221,11 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         scale = q.shape[-1] ** -0.5
     if _use_native(q):
         S = q.shape[1]
-        pad = (64 - S % 64) % 64
+        pad = (128 - S % 128) % 128
         if pad:
             if not causal:
                 raise ValueError(
-                    "non-causal flash_attention requires seq_len % 64 == 0 "
+                    "non-causal flash_attention requires seq_len % 128 == 0 "
                     f"(got {S}); pad inputs with an attention mask upstream")
             zq = q.new_zeros(q.shape[0], pad, q.shape[2], q.shape[3])
             zk = k.new_zeros(k.shape[0], pad, k.shape[2], k.shape[3])
@@ -235,6 +238,129 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))
     o = reference.sdpa(qt, kt, vt, causal=causal, scale=scale)
     return o.transpose(1, 2)
+
+
+def _p_off(t: torch.Tensor, elem_off: int) -> ctypes.c_void_p:
+    return ctypes.c_void_p(t.data_ptr() + elem_off * t.element_size())
+
+
+# ------------------------------------------------------------------ SwiGLU --
+
+class _SwigluHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, h):
+        lib = _backend.require()
+        shape = h.shape
+        F2 = shape[-1]
+        h2 = h.contiguous().view(-1, F2)
+        T = h2.shape[0]
+        y = torch.empty(T, F2 // 2, dtype=h.dtype, device=h.device)
+        _backend.check(lib.kf_swiglu_fwd(_p(y), _p(h2), T, F2 // 2,
+                                         _stream()), "swiglu_fwd")
+        ctx.save_for_backward(h2)
+        ctx.shape = shape
+        return y.view(*shape[:-1], F2 // 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = _backend.require()
+        (h2,) = ctx.saved_tensors
+        T, F2 = h2.shape
+        dy2 = dy.contiguous().view(T, F2 // 2)
+        dh = torch.empty_like(h2)
+        _backend.check(lib.kf_swiglu_bwd(_p(dh), _p(dy2), _p(h2), T, F2 // 2,
+                                         _stream()), "swiglu_bwd")
+        return dh.view(ctx.shape)
+
+
+def swiglu(h: torch.Tensor) -> torch.Tensor:
+    """h = [gate ++ up] on the last dim -> silu(gate) * up."""
+    if _use_native(h):
+        return _SwigluHip.apply(h)
+    g, u = h.chunk(2, dim=-1)
+    return torch.nn.functional.silu(g) * u
+
+
+# -------------------------------------------- Fused QKV -> RoPE -> attention
+
+class _FusedQkvAttentionHip(torch.autograd.Function):
+    """RoPE applied IN PLACE on the q/k regions of the fused qkv projection,
+    then flash attention reading q/k/v as strided views — zero copies, and
+    backward writes dq/dk/dv directly into one dqkv buffer (no torch.cat).
+
+    In-place note: qkv is the wqkv GEMM output; torch.linear's backward
+    needs its input and weight, never its output, so rotating the buffer in
+    place is safe. No other consumer may read qkv afterwards.
+    """
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, Hq, Hkv, D, causal, scale):
+        lib = _backend.require()
+        B, S, _ = qkv.shape
+        ts = (Hq + 2 * Hkv) * D  # token stride in elements
+        qkv = qkv.contiguous()
+        _backend.check(
+            lib.kf_rope(_p(qkv), _p_off(qkv, Hq * D), _fp(cos), _fp(sin),
+                        _ip_or_null(None), B, S, Hq, Hkv, D, ts, ts, 0, 0,
+                        _stream()), "rope_fwd")
+        o = torch.empty(B, S, Hq * D, dtype=qkv.dtype, device=qkv.device)
+        lse = torch.empty(B, Hq, S, dtype=torch.float32, device=qkv.device)
+        _backend.check(
+            lib.kf_attn_fwd(_p(o), _fp(lse), _p(qkv), _p_off(qkv, Hq * D),
+                            _p_off(qkv, (Hq + Hkv) * D), B, S, Hq, Hkv, D,
+                            ts, ts, float(scale), int(causal), _stream()),
+            "attn_fwd")
+        ctx.save_for_backward(qkv, cos, sin, o, lse)
+        ctx.meta = (Hq, Hkv, D, causal, scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        lib = _backend.require()
+        qkv, cos, sin, o, lse = ctx.saved_tensors
+        Hq, Hkv, D, causal, scale = ctx.meta
+        B, S, _ = qkv.shape
+        ts = (Hq + 2 * Hkv) * D
+        dout = dout.contiguous()
+        dqkv = torch.empty_like(qkv)
+        delta = torch.empty(B, Hq, S, dtype=torch.float32, device=qkv.device)
+        _backend.check(
+            lib.kf_attn_bwd(_p(dqkv), _p_off(dqkv, Hq * D),
+                            _p_off(dqkv, (Hq + Hkv) * D), _p(dout), _p(qkv),
+                            _p_off(qkv, Hq * D), _p_off(qkv, (Hq + Hkv) * D),
+                            _p(o), _fp(lse), _fp(delta), B, S, Hq, Hkv, D,
+                            ts, ts, ts, ts, float(scale), int(causal),
+                            _stream()), "attn_bwd")
+        # adjoint of the in-place rotation on the dq/dk regions
+        _backend.check(
+            lib.kf_rope(_p(dqkv), _p_off(dqkv, Hq * D), _fp(cos), _fp(sin),
+                        _ip_or_null(None), B, S, Hq, Hkv, D, ts, ts, 0, 1,
+                        _stream()), "rope_bwd")
+        return dqkv, None, None, None, None, None, None, None
+
+
+def fused_qkv_attention(qkv: torch.Tensor, cos: torch.Tensor,
+                        sin: torch.Tensor, Hq: int, Hkv: int, D: int,
+                        causal: bool = True,
+                        scale: Optional[float] = None) -> torch.Tensor:
+    """qkv [B,S,(Hq+2Hkv)*D] -> attention output [B,S,Hq*D].
+
+    Native path requires S % 128 == 0 (training shapes); otherwise (and on
+    CPU) falls back to the composed split->rope->attention ops.
+    """
+    if scale is None:
+        scale = D ** -0.5
+    B, S, _ = qkv.shape
+    if _use_native(qkv) and S % 128 == 0:
+        return _FusedQkvAttentionHip.apply(qkv, cos, sin, Hq, Hkv, D, causal,
+                                           scale)
+    q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    q = q.reshape(B, S, Hq, D)
+    k = k.reshape(B, S, Hkv, D)
+    v = v.reshape(B, S, Hkv, D)
+    q, k = rope(q, k, cos, sin)
+    o = flash_attention(q, k, v, causal=causal, scale=scale)
+    return o.reshape(B, S, Hq * D)
 
 
 # --------------------------------------------------------------- LayerNorm --

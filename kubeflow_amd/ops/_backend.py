@@ -72,7 +72,7 @@ def _configure(lib: ctypes.CDLL) -> None:
     lib.kf_rmsnorm_bwd_nparts.restype = I64
     lib.kf_rmsnorm_bwd_nparts.argtypes = [I64]
     lib.kf_rope.restype = I32
-    lib.kf_rope.argtypes = [P, P, FP, FP, PI64, I64, I64, I64, I64, I64, I64, I32, P]
+    lib.kf_rope.argtypes = [P, P, FP, FP, PI64, I64, I64, I64, I64, I64, I64, I64, I64, I32, P]
     lib.kf_adamw.restype = I32
     lib.kf_adamw.argtypes = [P, FP, P, FP, FP, FP, I64, F, F, F, F, F, I64, P]
     lib.kf_ce_fwd.restype = I32
@@ -80,17 +80,22 @@ def _configure(lib: ctypes.CDLL) -> None:
     lib.kf_ce_bwd.restype = I32
     lib.kf_ce_bwd.argtypes = [P, P, FP, PI64, FP, I64, I64, I64, P]
     lib.kf_attn_fwd.restype = I32
-    lib.kf_attn_fwd.argtypes = [P, FP, P, P, P, I64, I64, I64, I64, I64, F, I32, P]
+    lib.kf_attn_fwd.argtypes = [P, FP, P, P, P, I64, I64, I64, I64, I64, I64, I64, F, I32, P]
     if hasattr(lib, "kf_attn_bwd"):
         lib.kf_attn_bwd.restype = I32
         lib.kf_attn_bwd.argtypes = [P, P, P, P, P, P, P, P, FP, FP, I64, I64,
-                                    I64, I64, I64, F, I32, P]
+                                    I64, I64, I64, I64, I64, I64, I64, F, I32,
+                                    P]
     lib.kf_layernorm_fwd.restype = I32
     lib.kf_layernorm_fwd.argtypes = [P, FP, FP, P, P, P, I64, I64, F, P]
     lib.kf_layernorm_bwd.restype = I32
     lib.kf_layernorm_bwd.argtypes = [P, P, P, FP, P, P, P, FP, FP, I64, I64, P]
     lib.kf_layernorm_bwd_nparts.restype = I64
     lib.kf_layernorm_bwd_nparts.argtypes = [I64]
+    lib.kf_swiglu_fwd.restype = I32
+    lib.kf_swiglu_fwd.argtypes = [P, P, I64, I64, P]
+    lib.kf_swiglu_bwd.restype = I32
+    lib.kf_swiglu_bwd.argtypes = [P, P, P, I64, I64, P]
     lib.kf_attn_decode.restype = I32
     lib.kf_attn_decode.argtypes = [P, P, P, P, PI32, PI32, I64, I64, I64, I64,
                                    I64, F, P]

@@ -74,18 +74,17 @@ __global__ __launch_bounds__(256) void kf_attn_dq_kernel(
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE, lane = tid & 63, l16 = lane & 15, lg = lane >> 4;
-  const int64_t kstride = (int64_t)Hkv * AB_D;
-
-  // Q and dO fragments in registers (A-operands)
+  // Q (strided) and dO (contiguous bshd) fragments in registers
   kf_bf16x8 qfrag[4], dofrag[4];
   {
-    const int64_t base =
-        ((b * S + qt * 64 + w * 16 + l16) * (int64_t)Hq + hq) * AB_D;
+    const int64_t tok = b * S + qt * 64 + w * 16 + l16;
+    const int64_t qb = tok * qts + (int64_t)hq * AB_D;
+    const int64_t db = (tok * (int64_t)Hq + hq) * AB_D;
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      qfrag[kk] = *reinterpret_cast<const kf_bf16x8*>(q + base + kk * 32 + lg * 8);
+      qfrag[kk] = *reinterpret_cast<const kf_bf16x8*>(q + qb + kk * 32 + lg * 8);
       dofrag[kk] =
-          *reinterpret_cast<const kf_bf16x8*>(dout + base + kk * 32 + lg * 8);
+          *reinterpret_cast<const kf_bf16x8*>(dout + db + kk * 32 + lg * 8);
     }
   }
   // per-row lse/delta for the 4 rows this lane's acc regs cover
@@ -107,14 +106,14 @@ __global__ __launch_bounds__(256) void kf_attn_dq_kernel(
   for (int kt = 0; kt <= last_kt; ++kt) {
     __syncthreads();
     {  // stage K (row-major + transposed) and V (row-major)
-      const unsigned short* kg = k + ((b * S + kt * DQ_KT) * Hkv + hkv) * AB_D;
-      const unsigned short* vg = v + ((b * S + kt * DQ_KT) * Hkv + hkv) * AB_D;
+      const unsigned short* kg = k + (b * S + kt * DQ_KT) * kts + hkv * AB_D;
+      const unsigned short* vg = v + (b * S + kt * DQ_KT) * kts + hkv * AB_D;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int vi = tid + 256 * j;
         const int r = vi >> 4, c8 = vi & 15;
         kf_short8 kv8 =
-            *reinterpret_cast<const kf_short8*>(kg + r * kstride + c8 * 8);
+            *reinterpret_cast<const kf_short8*>(kg + r * kts + c8 * 8);
         *reinterpret_cast<kf_short8*>(k_lds + kf_swz2(r, c8 * 16, 256, 7)) = kv8;
 #pragma unroll
         for (int jj = 0; jj < 8; ++jj) {
@@ -124,7 +123,7 @@ __global__ __launch_bounds__(256) void kf_attn_dq_kernel(
               kt_lds + kf_swz2(dd, r * 2, 128, 7)) = (unsigned short)kv8[el];
         }
         kf_short8 vv8 =
-            *reinterpret_cast<const kf_short8*>(vg + r * kstride + c8 * 8);
+            *reinterpret_cast<const kf_short8*>(vg + r * kts + c8 * 8);
         *reinterpret_cast<kf_short8*>(v_lds + kf_swz2(r, c8 * 16, 256, 7)) = vv8;
       }
     }
@@ -185,11 +184,11 @@ __global__ __launch_bounds__(256) void kf_attn_dq_kernel(
     }
   }
 
-  // epilogue
+  // epilogue (strided output: dq may live inside a fused dqkv buffer)
   const int qrow0 = qt * 64 + w * 16 + lg * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const int64_t base = ((b * S + qrow0 + r) * (int64_t)Hq + hq) * AB_D;
+    const int64_t base = (b * S + qrow0 + r) * dqts + (int64_t)hq * AB_D;
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt)
       dq[base + nt * 16 + l16] = kf_f32_to_bf16(dqacc[nt][r]);
@@ -220,14 +219,13 @@ __global__ __launch_bounds__(256) void kf_attn_dkv_kernel(
   const int g = Hq / Hkv;
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE, lane = tid & 63, l16 = lane & 15, lg = lane >> 4;
-  const int64_t kstride = (int64_t)Hkv * AB_D;
-  const int64_t qstride = (int64_t)Hq * AB_D;
+  const int64_t dstride = (int64_t)Hq * AB_D;  // dout is contiguous bshd
 
   // K and V fragments in registers (A-operands, 16 kv rows per wave)
   kf_bf16x8 kfrag[4], vfrag[4];
   {
     const int64_t base =
-        ((b * S + kt * 64 + w * 16 + l16) * (int64_t)Hkv + hkv) * AB_D;
+        (b * S + kt * 64 + w * 16 + l16) * kts + (int64_t)hkv * AB_D;
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
       kfrag[kk] = *reinterpret_cast<const kf_bf16x8*>(k + base + kk * 32 + lg * 8);
@@ -249,18 +247,19 @@ __global__ __launch_bounds__(256) void kf_attn_dkv_kernel(
     for (int qt = qt0; qt < nqt; ++qt) {
       __syncthreads();
       {  // stage Q, dO in both orientations
-        const unsigned short* qg = q + ((b * S + qt * DKV_QT) * Hq + hq) * AB_D;
+        const unsigned short* qg =
+            q + (b * S + qt * DKV_QT) * qts + (int64_t)hq * AB_D;
         const unsigned short* dog =
-            dout + ((b * S + qt * DKV_QT) * Hq + hq) * AB_D;
+            dout + ((b * S + qt * DKV_QT) * (int64_t)Hq + hq) * AB_D;
 #pragma unroll
         for (int j = 0; j < 2; ++j) {
           const int vi = tid + 256 * j;     // 0..511 ; 32 rows × 16 chunks
           const int r = vi >> 4, c8 = vi & 15;
           kf_short8 q8 =
-              *reinterpret_cast<const kf_short8*>(qg + r * qstride + c8 * 8);
+              *reinterpret_cast<const kf_short8*>(qg + r * qts + c8 * 8);
           *reinterpret_cast<kf_short8*>(q_lds + kf_swz2(r, c8 * 16, 256, 7)) = q8;
           kf_short8 do8 =
-              *reinterpret_cast<const kf_short8*>(dog + r * qstride + c8 * 8);
+              *reinterpret_cast<const kf_short8*>(dog + r * dstride + c8 * 8);
           *reinterpret_cast<kf_short8*>(do_lds + kf_swz2(r, c8 * 16, 256, 7)) =
               do8;
 #pragma unroll
@@ -338,11 +337,11 @@ __global__ __launch_bounds__(256) void kf_attn_dkv_kernel(
     }
   }
 
-  // epilogue: write dK, dV (bshd, bf16)
+  // epilogue: write dK, dV (strided: may live inside a fused dqkv buffer)
   const int krow0 = kt * 64 + w * 16 + lg * 4;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const int64_t base = ((b * S + krow0 + r) * (int64_t)Hkv + hkv) * AB_D;
+    const int64_t base = (b * S + krow0 + r) * dkts + (int64_t)hkv * AB_D;
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt) {
       dk[base + nt * 16 + l16] = kf_f32_to_bf16(dkacc[nt][r]);
@@ -355,8 +354,14 @@ KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                           const void* q, const void* k, const void* v,
                           const void* o, const float* lse, float* delta,
                           int64_t B, int64_t S, int64_t Hq, int64_t Hkv,
-                          int64_t D, float scale, int causal, void* stream) {
+                          int64_t D, int64_t qts, int64_t kts, int64_t dqts,
+                          int64_t dkts, float scale, int causal,
+                          void* stream) {
   if (D != AB_D || S % 64 || Hq % Hkv) return (int)hipErrorInvalidValue;
+  if (qts == 0) qts = Hq * AB_D;
+  if (kts == 0) kts = Hkv * AB_D;
+  if (dqts == 0) dqts = Hq * AB_D;
+  if (dkts == 0) dkts = Hkv * AB_D;
   hipLaunchKernelGGL(kf_attn_delta2_kernel,
                      dim3(kf_grid_for(B * S * Hq, 4)), dim3(256), 0,
                      (hipStream_t)stream, delta, (const unsigned short*)dout,
@@ -368,7 +373,7 @@ KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                      (unsigned short*)dq, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, lse, delta, B, (int)S,
-                     (int)Hq, (int)Hkv, scale, causal);
+                     (int)Hq, (int)Hkv, qts, kts, dqts, scale, causal);
   err = (int)hipGetLastError();
   if (err) return err;
   dim3 gkv((unsigned)(S / 64), (unsigned)Hkv, (unsigned)B);
@@ -377,6 +382,6 @@ KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                      (unsigned short*)dv, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, lse, delta, B, (int)S,
-                     (int)Hq, (int)Hkv, scale, causal);
+                     (int)Hq, (int)Hkv, qts, kts, dkts, scale, causal);
   return (int)hipGetLastError();
 }

@@ -18,6 +18,7 @@ __global__ void kf_rope_kernel(unsigned short* __restrict__ q,
                                const float* __restrict__ sint,
                                const int64_t* __restrict__ positions,
                                int64_t B, int S, int Hq, int Hkv, int D,
+                               int64_t qts, int64_t kts,  // token strides
                                int64_t pos_offset, int backward) {
   const int Ht = Hq + Hkv;
   const int halfD = D / 2;
@@ -32,8 +33,8 @@ __global__ void kf_rope_kernel(unsigned short* __restrict__ q,
     const int s = (int)(rest % S);
     const int64_t b = rest / S;
     unsigned short* base;
-    if (h < Hq) base = q + ((b * S + s) * (int64_t)Hq + h) * D;
-    else base = k + ((b * S + s) * (int64_t)Hkv + (h - Hq)) * D;
+    if (h < Hq) base = q + (b * S + s) * qts + (int64_t)h * D;
+    else base = k + (b * S + s) * kts + (int64_t)(h - Hq) * D;
     const int d0 = qd * ROPE_VEC;
     kf_short4 x1 = *reinterpret_cast<const kf_short4*>(base + d0);
     kf_short4 x2 = *reinterpret_cast<const kf_short4*>(base + halfD + d0);
@@ -58,13 +59,15 @@ __global__ void kf_rope_kernel(unsigned short* __restrict__ q,
 
 KF_EXPORT int kf_rope(void* q, void* k, const float* cost, const float* sint,
                       const int64_t* positions, int64_t B, int64_t S,
-                      int64_t Hq, int64_t Hkv, int64_t D, int64_t pos_offset,
-                      int backward, void* stream) {
+                      int64_t Hq, int64_t Hkv, int64_t D, int64_t qts,
+                      int64_t kts, int64_t pos_offset, int backward,
+                      void* stream) {
   if ((D / 2) % ROPE_VEC) return (int)hipErrorInvalidValue;
   const int64_t total = B * S * (Hq + Hkv) * (D / 2 / ROPE_VEC);
   hipLaunchKernelGGL(kf_rope_kernel, dim3(kf_grid_for(total, 256)), dim3(256),
                      0, (hipStream_t)stream, (unsigned short*)q,
                      (unsigned short*)k, cost, sint, positions, B, (int)S,
-                     (int)Hq, (int)Hkv, (int)D, pos_offset, backward);
+                     (int)Hq, (int)Hkv, (int)D, qts, kts, pos_offset,
+                     backward);
   return (int)hipGetLastError();
 }
