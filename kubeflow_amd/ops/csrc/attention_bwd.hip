@@ -63,7 +63,7 @@ __global__ __launch_bounds__(256) void kf_attn_dq_kernel(
     const unsigned short* __restrict__ k, const unsigned short* __restrict__ v,
     const unsigned short* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, int64_t B, int S, int Hq, int Hkv,
-    float scale, int causal) {
+    int64_t qts, int64_t kts, int64_t dqts, float scale, int causal) {
   __shared__ unsigned char k_lds[DQ_KT * AB_D * 2];    // [64][128] row-major
   __shared__ unsigned char v_lds[DQ_KT * AB_D * 2];    // [64][128] row-major
   __shared__ unsigned char kt_lds[AB_D * DQ_KT * 2];   // [128][64] transposed
@@ -206,7 +206,7 @@ __global__ __launch_bounds__(256) void kf_attn_dkv_kernel(
     const unsigned short* __restrict__ v,
     const unsigned short* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, int64_t B, int S, int Hq, int Hkv,
-    float scale, int causal) {
+    int64_t qts, int64_t kts, int64_t dkts, float scale, int causal) {
   __shared__ unsigned char q_lds[DKV_QT * AB_D * 2];    // [32][128]
   __shared__ unsigned char qt_lds[AB_D * DKV_QT * 2];   // [128][32]
   __shared__ unsigned char do_lds[DKV_QT * AB_D * 2];   // [32][128]

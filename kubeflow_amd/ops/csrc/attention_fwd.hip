@@ -40,7 +40,7 @@ __global__ __launch_bounds__(AT_THREADS, 2) void kf_attn_fwd_kernel(
     unsigned short* __restrict__ o, float* __restrict__ lse,
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
-    float scale, int causal) {
+    int64_t qts, int64_t kts, float scale, int causal) {
   __shared__ unsigned char k_lds[AT_KT * AT_D * 2];        // [64][128] swz
   __shared__ unsigned char vt_lds[AT_D * AT_KT * 2];       // [128][64] swz
   __shared__ unsigned char p_lds[4][32 * AT_KT * 2];       // per-wave [32][64]
