@@ -228,6 +228,196 @@ __global__ __launch_bounds__(AT_THREADS, 2) void kf_attn_fwd_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// v3: 8-wave swapped-QK^T kernel (S % 256 == 0 — the training shapes).
+//
+// Per the CDNA4 guide's verified attention ladder (§B "8-warp 32×32"):
+//  * block = 512 threads = 8 waves; each block owns a 256-row q-tile, each
+//    wave 32 q rows; 32×32×16 MFMA.
+//  * SWAPPED QK^T: computes S^T = mfma(K, Q) so each lane's accumulator
+//    column is ONE q row (col = lane&31) — the online-softmax max/sum are
+//    register reductions plus a single shfl_xor(32), no LDS round-trip and
+//    no lgkmcnt drain on the softmax path.
+//  * P^T -> PV B-fragments in-register via v_cvt_pk_bf16_f32 +
+//    permlane32_swap (guide T12): 8 cvt_pk + 4 swaps per 32-kv tile.
+//  * K row-major and V transpose-staged in XOR-swizzled LDS as in v2;
+//    PV A-fragments (V^T) are b128 reads from Vt.
+// C/D layout for mfma_f32_32x32x16_bf16: col = lane&31,
+// row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)  [guide §3, m74/m101].
+// ---------------------------------------------------------------------------
+
+typedef float kf_f32x16 __attribute__((ext_vector_type(16)));
+
+#define A8_QT 256
+#define A8_KT 64
+#define A8_THREADS 512
+
+__device__ __forceinline__ unsigned int kf_cvt_pk_bf16(float lo, float hi) {
+  unsigned int r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+__global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
+    unsigned short* __restrict__ o, float* __restrict__ lse,
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
+    int64_t qts, int64_t kts, float scale, int causal) {
+  __shared__ unsigned char k_lds[A8_KT * AT_D * 2];   // [64][128] swz
+  __shared__ unsigned char vt_lds[AT_D * A8_KT * 2];  // [128][64] swz
+
+  const int qt = blockIdx.x, hq = blockIdx.y;
+  const int64_t b = blockIdx.z;
+  const int hkv = hq / (Hq / Hkv);
+  const int tid = threadIdx.x;
+  const int w = tid / KF_WAVE;
+  const int lane = tid & (KF_WAVE - 1);
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;  // 0 for lanes 0-31, 1 for 32-63
+
+  // ---- persistent Q B-fragments: B[k=d][n=q], lane holds q-col l31 ----
+  kf_bf16x8 qfrag[8];
+  {
+    const int64_t qbase =
+        (b * S + qt * A8_QT + w * 32 + l31) * qts + (int64_t)hq * AT_D;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      qfrag[kk] =
+          *reinterpret_cast<const kf_bf16x8*>(q + qbase + kk * 16 + hi * 8);
+  }
+
+  kf_f32x16 oacc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) oacc[i] = kf_f32x16{0.f};
+  float m_run = -INFINITY, l_run = 0.f;
+  const int qrow_g = qt * A8_QT + w * 32 + l31;
+
+  const int last_kt =
+      causal ? (qt * A8_QT + A8_QT - 1) / A8_KT : (S / A8_KT - 1);
+  for (int kt = 0; kt <= last_kt; ++kt) {
+    // ---- stage K row-major + V transposed (512 threads, 2 vecs each) ----
+    {
+      const unsigned short* kg =
+          k + (b * S + kt * A8_KT) * kts + (int64_t)hkv * AT_D;
+      const unsigned short* vg =
+          v + (b * S + kt * A8_KT) * kts + (int64_t)hkv * AT_D;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int vi = tid + A8_THREADS * j;   // 0..1023
+        const int r = vi >> 4, c8 = vi & 15;
+        kf_short8 kv8 =
+            *reinterpret_cast<const kf_short8*>(kg + r * kts + c8 * 8);
+        *reinterpret_cast<kf_short8*>(k_lds + kf_swz(r, c8 * 16, AT_D * 2)) =
+            kv8;
+        kf_short8 vv8 =
+            *reinterpret_cast<const kf_short8*>(vg + r * kts + c8 * 8);
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) {
+          const int el = (jj + tid) & 7;
+          const int dd = c8 * 8 + el;
+          *reinterpret_cast<unsigned short*>(
+              vt_lds + kf_swz(dd, r * 2, A8_KT * 2)) = (unsigned short)vv8[el];
+        }
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {  // two 32-kv M-tiles
+      // ---- S^T = mfma(K, Q): rows kv, cols q ----
+      kf_f32x16 st = kf_f32x16{0.f};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 8; ++kk) {
+        kf_bf16x8 afrag = *reinterpret_cast<const kf_bf16x8*>(
+            k_lds + kf_swz(mt * 32 + l31, kk * 32 + hi * 16, AT_D * 2));
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, qfrag[kk], st,
+                                                     0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- mask + online softmax (register-local; one shfl per reduce) --
+      const int kv0 = kt * A8_KT + mt * 32 + hi * 4;
+      float mx = -INFINITY;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv = kv0 + (r & 3) + 8 * (r >> 2);
+        float sv = st[r] * scale;
+        if (causal && kv > qrow_g) sv = -INFINITY;
+        st[r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+      mx = fmaxf(mx, __shfl_xor(mx, 32, KF_WAVE));
+      const float m_new = fmaxf(m_run, mx);
+      const float alpha =
+          (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+      float lsum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float pv =
+            (st[r] == -INFINITY) ? 0.f : __expf(st[r] - m_new);
+        st[r] = pv;
+        lsum += pv;
+      }
+      lsum += __shfl_xor(lsum, 32, KF_WAVE);
+      l_run = l_run * alpha + lsum;
+      m_run = m_new;
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[i][r] *= alpha;
+
+      // ---- P^T -> two B-fragments via cvt_pk + permlane32_swap ----
+      // regs 0-7 cover kv-local {0..3, 8..11} (+4 for hi lanes);
+      // regs 8-15 cover {16..19, 24..27} (+4 for hi lanes).
+      kf_bf16x8 pb[2];
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        const int base = step * 8;
+        unsigned int w0 = kf_cvt_pk_bf16(st[base + 0], st[base + 1]);
+        unsigned int w1 = kf_cvt_pk_bf16(st[base + 2], st[base + 3]);
+        unsigned int w2 = kf_cvt_pk_bf16(st[base + 4], st[base + 5]);
+        unsigned int w3 = kf_cvt_pk_bf16(st[base + 6], st[base + 7]);
+        auto s02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+        auto s13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+        unsigned int u[4] = {(unsigned)s02[0], (unsigned)s13[0],
+                             (unsigned)s02[1], (unsigned)s13[1]};
+        pb[step] = *reinterpret_cast<kf_bf16x8*>(u);
+      }
+
+      // ---- O^T += V^T P^T over the two 16-kv steps ----
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+#pragma unroll
+        for (int step = 0; step < 2; ++step) {
+          kf_bf16x8 vfrag = *reinterpret_cast<const kf_bf16x8*>(
+              vt_lds + kf_swz(dt * 32 + l31,
+                              mt * 64 + step * 32 + hi * 16, A8_KT * 2));
+          oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag, pb[step],
+                                                             oacc[dt], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O^T regs -> o[token][d]; lse per q row ----
+  const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+  const int64_t obase = ((b * S + qrow_g) * (int64_t)Hq + hq) * AT_D;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      o[obase + d] = kf_f32_to_bf16(oacc[dt][r] * inv_l);
+    }
+  if (hi == 0)
+    lse[(b * Hq + hq) * (int64_t)S + qrow_g] = m_run + __logf(l_run);
+}
+
 KF_EXPORT int kf_attn_fwd(void* o, float* lse, const void* q, const void* k,
                           const void* v, int64_t B, int64_t S, int64_t Hq,
                           int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
@@ -235,6 +425,15 @@ KF_EXPORT int kf_attn_fwd(void* o, float* lse, const void* q, const void* k,
   if (D != AT_D || S % AT_QT || Hq % Hkv) return (int)hipErrorInvalidValue;
   if (qts == 0) qts = Hq * AT_D;
   if (kts == 0) kts = Hkv * AT_D;
+  if (S % A8_QT == 0) {  // 8-wave swapped kernel for the training shapes
+    dim3 grid((unsigned)(S / A8_QT), (unsigned)Hq, (unsigned)B);
+    hipLaunchKernelGGL(kf_attn_fwd8_kernel, grid, dim3(A8_THREADS), 0,
+                       (hipStream_t)stream, (unsigned short*)o, lse,
+                       (const unsigned short*)q, (const unsigned short*)k,
+                       (const unsigned short*)v, B, (int)S, (int)Hq,
+                       (int)Hkv, qts, kts, scale, causal);
+    return (int)hipGetLastError();
+  }
   dim3 grid((unsigned)(S / AT_QT), (unsigned)Hq, (unsigned)B);
   hipLaunchKernelGGL(kf_attn_fwd_kernel, grid, dim3(AT_THREADS), 0,
                      (hipStream_t)stream, (unsigned short*)o, lse,
