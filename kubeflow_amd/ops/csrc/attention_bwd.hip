@@ -350,6 +350,17 @@ __global__ __launch_bounds__(256) void kf_attn_dkv_kernel(
   }
 }
 
+// 8-wave swapped backward kernels (attention_bwd8.hip)
+extern "C" int kf_attn_bwd8_dq(void*, const void*, const void*, const void*,
+                               const void*, const float*, const float*,
+                               int64_t, int64_t, int64_t, int64_t, int64_t,
+                               int64_t, int64_t, float, int, void*);
+extern "C" int kf_attn_bwd8_dkv(void*, void*, const void*, const void*,
+                                const void*, const void*, const float*,
+                                const float*, int64_t, int64_t, int64_t,
+                                int64_t, int64_t, int64_t, int64_t, float,
+                                int, void*);
+
 KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                           const void* q, const void* k, const void* v,
                           const void* o, const float* lse, float* delta,
@@ -368,6 +379,13 @@ KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                      (const unsigned short*)o, B, (int)S, (int)Hq);
   int err = (int)hipGetLastError();
   if (err) return err;
+  if (S % 256 == 0) {  // 8-wave swapped path for the training shapes
+    err = kf_attn_bwd8_dq(dq, q, k, v, dout, lse, delta, B, S, Hq, Hkv, qts,
+                          kts, dqts, scale, causal, stream);
+    if (err) return err;
+    return kf_attn_bwd8_dkv(dk, dv, q, k, v, dout, lse, delta, B, S, Hq, Hkv,
+                            qts, kts, dkts, scale, causal, stream);
+  }
   dim3 gq((unsigned)(S / 64), (unsigned)Hq, (unsigned)B);
   hipLaunchKernelGGL(kf_attn_dq_kernel, gq, dim3(256), 0, (hipStream_t)stream,
                      (unsigned short*)dq, (const unsigned short*)q,
