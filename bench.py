@@ -34,7 +34,7 @@ def main():
     ap.add_argument("--seq-len", type=int,
                     default=int(os.environ.get("KF_BENCH_SEQ", "4096")))
     ap.add_argument("--micro-batch", type=int,
-                    default=int(os.environ.get("KF_BENCH_MB", "2")))
+                    default=int(os.environ.get("KF_BENCH_MB", "4")))
     args = ap.parse_args()
 
     rank, world, device = kdist.init_distributed()

@@ -66,7 +66,8 @@ class ApiError(Exception):
         self.msg = msg
 
 
-def build_app(store: ObjectStore, scheduler=None) -> FastAPI:
+def build_app(store: ObjectStore, scheduler=None,
+              root_dir: Optional[str] = None) -> FastAPI:
     app = FastAPI(title="kubeflow-amd platform API")
     bindings = BindingClient(store)
     userid_header = os.environ.get("USERID_HEADER", "kubeflow-userid")
@@ -168,6 +169,15 @@ def build_app(store: ObjectStore, scheduler=None) -> FastAPI:
                             httponly=False)
         return resp
 
+    # ---------------------------------------------------------- app config
+    from kubeflow_amd import config as kfconfig
+    cfg = kfconfig.load()
+
+    @app.get("/api/config")
+    def get_config(request: Request):
+        me = user_of(request)
+        return ok(me, config=cfg["spawner"])
+
     # -------------------------------------------------------------- probes
     @app.get("/healthz")
     @app.get("/api/status/health")
@@ -209,7 +219,12 @@ def build_app(store: ObjectStore, scheduler=None) -> FastAPI:
             obj = tpl
         else:  # short form {"name": ..., "spec"/fields...}
             name = body.pop("name")
-            obj = new_object(kind, name, ns, spec=body.get("spec", body),
+            spec = body.get("spec", body)
+            if kind == "Notebook":
+                # admin value/readOnly enforcement (spawner config parity)
+                spec = dict(spec)
+                spec.update(kfconfig.enforce_spawner(cfg, spec))
+            obj = new_object(kind, name, ns, spec=spec,
                              api_version=api_version)
         created = store.create(obj)
         if registry and kind == "Notebook":
@@ -410,6 +425,59 @@ def build_app(store: ObjectStore, scheduler=None) -> FastAPI:
         except Exception:
             pass
         return ok(me, metric=which, scheduler=util, gpus=gpus)
+
+    # ---------------------------------------------------------------- logs
+    @app.get("/api/namespaces/{ns}/pytorchjobs/{name}/logs")
+    def job_logs(ns: str, name: str, request: Request, rank: int = 0,
+                 tail: int = 200):
+        user = user_of(request)
+        authz(user, ns, "get")
+        if root_dir is None:
+            raise ApiError(404, "no logs root configured")
+        job = store.get("PyTorchJob", name, ns)
+        uid = job["metadata"]["uid"]
+        path = os.path.join(root_dir, "jobs", ns, f"{name}-{uid[:8]}",
+                            f"rank-{rank}", "worker.log")
+        if not os.path.exists(path):
+            raise ApiError(404, f"no log for rank {rank}")
+        with open(path, errors="replace") as f:
+            lines = f.readlines()[-tail:]
+        return ok(user, logs="".join(lines))
+
+    # ------------------------------------------------ central dashboard UI
+    @app.get("/")
+    def dashboard(request: Request):
+        """Minimal server-rendered dashboard shell — the central-dashboard
+        equivalent (namespaces, workloads, activities, GPU utilization)."""
+        me = user_of(request)
+        util = scheduler.utilization() if scheduler else {}
+        sections = []
+        for plural, (kind, _) in PLURALS.items():
+            if plural in ("persistentvolumeclaims", "events"):
+                continue
+            objs = store.list(kind)
+            if not objs:
+                continue
+            rows = "".join(
+                f"<tr><td>{o['metadata'].get('namespace','')}</td>"
+                f"<td>{o['metadata']['name']}</td>"
+                f"<td>{', '.join(c['type'] for c in o.get('status', {}).get('conditions', []) if c.get('status') == 'True') or '-'}</td></tr>"
+                for o in objs[:50])
+            sections.append(
+                f"<h3>{kind}s ({len(objs)})</h3>"
+                f"<table border=1 cellpadding=4><tr><th>namespace</th>"
+                f"<th>name</th><th>conditions</th></tr>{rows}</table>")
+        html = (
+            "<html><head><title>kubeflow-amd</title></head><body>"
+            f"<h1>kubeflow-amd — MI355X platform</h1>"
+            f"<p>user: {me} · GPUs: {util.get('total_gpus', 0)} "
+            f"(busy: {util.get('exclusive_busy', 0)}) · jobs: "
+            f"{util.get('jobs', 0)}</p>"
+            + "".join(sections) +
+            "<p><a href='/docs'>REST API docs (OpenAPI)</a> · "
+            "<a href='/metrics'>metrics</a></p></body></html>")
+        from fastapi.responses import HTMLResponse
+        return HTMLResponse(html)
 
     # ------------------------------------------------------------ metrics
     @app.get("/metrics")
