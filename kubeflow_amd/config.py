@@ -76,16 +76,22 @@ def _apply_env(cfg: dict, prefix: str = "KF_") -> None:
             continue
         path = name[len(prefix):].lower().split("_")
         node = cfg
-        # greedy longest-match walk
+        # greedy longest-match walk: joined keys may be dict nodes or leaves
         while len(path) > 1:
-            two = "_".join(path[:2])
-            if two in node and isinstance(node.get(two), dict):
-                node = node[two]
-                path = path[2:]
-            elif path[0] in node and isinstance(node[path[0]], dict):
-                node = node[path[0]]
-                path = path[1:]
-            else:
+            matched = False
+            for take in range(len(path), 0, -1):
+                key = "_".join(path[:take])
+                if key in node:
+                    if isinstance(node[key], dict) and take < len(path):
+                        node = node[key]
+                        path = path[take:]
+                        matched = True
+                        break
+                    if not isinstance(node[key], dict) and take == len(path):
+                        path = [key]
+                        matched = True
+                        break
+            if not matched:
                 break
         if len(path) == 1 and path[0] in node and not isinstance(
                 node[path[0]], dict):
