@@ -16,9 +16,12 @@ import os
 import sys
 import time
 
-import torch
-
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from kubeflow_amd.ops import tunable as _kf_tunable  # noqa: E402
+_kf_tunable.enable()  # must precede the first GEMM
+
+import torch  # noqa: E402
 
 from kubeflow_amd.models import build_model  # noqa: E402
 from kubeflow_amd.parallel import dist as kdist  # noqa: E402

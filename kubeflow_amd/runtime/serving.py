@@ -17,6 +17,7 @@ the training flash-attention kernel over the padded prompt.
 """
 from __future__ import annotations
 
+import os
 import queue
 import threading
 import time
@@ -91,7 +92,15 @@ class InferenceEngine:
         self._stop = False
         self._thread: Optional[threading.Thread] = None
         self.stats = {"requests": 0, "completed": 0, "tokens_out": 0,
-                      "prefill_tokens": 0}
+                      "prefill_tokens": 0, "graph_replays": 0}
+        # hipGraph decode: capture the whole batched decode step per batch
+        # bucket (the ~10 kernels x 32 layers of launch overhead dominate
+        # small-batch decode latency otherwise). One cache slot is reserved
+        # as the padding target for bucket rows beyond the live batch.
+        self.use_graphs = (self.device.type == "cuda"
+                           and os.environ.get("KF_SERVE_GRAPH", "1") == "1")
+        self._graphs = {}
+        self._pad_slot = self.cache.alloc() if self.use_graphs else None
 
     # ------------------------------------------------------------- public
     def start(self):
@@ -206,20 +215,12 @@ class InferenceEngine:
         return x + layer.w2(torch.nn.functional.silu(g) * u)
 
     @torch.no_grad()
-    def _decode_step(self):
-        """One token for every active sequence, batched."""
+    def _decode_forward(self, tokens, positions, slots, lens):
+        """Batched single-token forward over the KV cache -> next tokens.
+        Pure function of the given (static, for graph capture) tensors."""
         cfg = self.model.cfg
         import torch.nn.functional as F
-        acts = self.active
-        N = len(acts)
-        tokens = torch.tensor([[r.generated[-1]] for r in acts],
-                              dtype=torch.int64, device=self.device)
-        positions = torch.tensor([r.pos for r in acts], dtype=torch.int64,
-                                 device=self.device)
-        slots = torch.tensor([r.slot for r in acts], dtype=torch.int32,
-                             device=self.device)
-        lens = torch.tensor([r.pos + 1 for r in acts], dtype=torch.int32,
-                            device=self.device)
+        N = tokens.shape[0]
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
@@ -231,20 +232,81 @@ class InferenceEngine:
             k = k.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
             v = v.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
             q, k = ops.rope(q, k, cos, sin, positions=positions)
-            # append to cache at pos
             self.cache.k[li][slots.long(), positions] = k[:, 0]
             self.cache.v[li][slots.long(), positions] = v[:, 0]
             o = ops.attention_decode(q[:, 0], self.cache.k[li],
                                      self.cache.v[li], slots, lens)
             o = layer.wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim))
             x = x + o
-            g, u = F.linear(layer.mlp_norm(x), layer.w13.weight).split(
-                [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
-            x = x + layer.w2(F.silu(g) * u)
+            y = ops.swiglu(F.linear(layer.mlp_norm(x), layer.w13.weight))
+            x = x + layer.w2(y)
         x = self.model.final_norm(x)
         logits = F.linear(x, self.model.lm_head.weight)  # [N,1,V]
+        return logits[:, -1].argmax(-1)  # [N] int64
+
+    def _graph_for(self, bucket: int):
+        """Capture (once) and return the decode graph for a batch bucket."""
+        if bucket in self._graphs:
+            return self._graphs[bucket]
+        dev = self.device
+        static = {
+            "tokens": torch.zeros(bucket, 1, dtype=torch.int64, device=dev),
+            "positions": torch.zeros(bucket, dtype=torch.int64, device=dev),
+            "slots": torch.full((bucket,), self._pad_slot, dtype=torch.int32,
+                                device=dev),
+            "lens": torch.ones(bucket, dtype=torch.int32, device=dev),
+        }
+        # warm-up on a side stream, then capture
+        s = torch.cuda.Stream(device=dev)
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                out = self._decode_forward(**static)
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static["next"] = self._decode_forward(
+                static["tokens"], static["positions"], static["slots"],
+                static["lens"])
+        self._graphs[bucket] = (graph, static)
+        return self._graphs[bucket]
+
+    @torch.no_grad()
+    def _decode_step(self):
+        """One token for every active sequence, batched (graph replay on
+        GPU; eager on CPU)."""
+        acts = self.active
+        N = len(acts)
+        tokens = torch.tensor([[r.generated[-1]] for r in acts],
+                              dtype=torch.int64, device=self.device)
+        positions = torch.tensor([r.pos for r in acts], dtype=torch.int64,
+                                 device=self.device)
+        slots = torch.tensor([r.slot for r in acts], dtype=torch.int32,
+                             device=self.device)
+        lens = torch.tensor([r.pos + 1 for r in acts], dtype=torch.int32,
+                            device=self.device)
+        if self.use_graphs:
+            bucket = 1
+            while bucket < N:
+                bucket *= 2
+            bucket = min(bucket, self.max_batch)
+            graph, static = self._graph_for(bucket)
+            static["tokens"][:N].copy_(tokens)
+            static["positions"][:N].copy_(positions)
+            static["slots"][:N].copy_(slots)
+            static["lens"][:N].copy_(lens)
+            if bucket > N:  # park padding rows on the reserved slot
+                static["tokens"][N:].zero_()
+                static["positions"][N:].zero_()
+                static["slots"][N:].fill_(self._pad_slot)
+                static["lens"][N:].fill_(1)
+            graph.replay()
+            self.stats["graph_replays"] += 1
+            toks = static["next"][:N].tolist()
+        else:
+            toks = self._decode_forward(tokens, positions, slots,
+                                        lens).tolist()
         still = []
-        toks = logits[:, -1].argmax(-1).tolist()  # greedy batch decode
         now = time.time()
         for i, r in enumerate(acts):
             r.pos += 1

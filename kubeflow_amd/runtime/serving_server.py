@@ -19,6 +19,8 @@ import json
 import os
 import time
 
+from kubeflow_amd.ops import tunable as _kf_tunable
+_kf_tunable.enable()
 from fastapi import FastAPI, Request as HttpRequest
 from fastapi.responses import JSONResponse, PlainTextResponse
 import uvicorn

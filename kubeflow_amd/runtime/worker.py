@@ -16,6 +16,8 @@ import signal
 import sys
 import time
 
+from kubeflow_amd.ops import tunable as _kf_tunable
+_kf_tunable.enable()
 import torch
 
 from kubeflow_amd.models import build_model

@@ -346,3 +346,31 @@ def test_llama_tiny_fused_vs_cpu_loss(gpu_device):
     lc = mc(tokens, targets)
     assert abs(lg.item() - lc.item()) / abs(lc.item()) < 2e-2, \
         (lg.item(), lc.item())
+
+
+def test_serving_graph_matches_eager(gpu_device, monkeypatch):
+    """hipGraph decode replay must produce the same greedy tokens as eager."""
+    from kubeflow_amd.runtime.serving import InferenceEngine
+    monkeypatch.setenv("KF_SERVE_GRAPH", "1")
+    torch.manual_seed(0)
+    eng_g = InferenceEngine("llama-tiny", device=gpu_device, max_slots=4,
+                            smax=256, max_batch=4)
+    assert eng_g.use_graphs
+    monkeypatch.setenv("KF_SERVE_GRAPH", "0")
+    torch.manual_seed(0)
+    eng_e = InferenceEngine("llama-tiny", device=gpu_device, max_slots=4,
+                            smax=256, max_batch=4)
+    with torch.no_grad():
+        for pg, pe in zip(eng_g.model.parameters(), eng_e.model.parameters()):
+            pe.copy_(pg)
+    eng_g.start()
+    eng_e.start()
+    try:
+        rg = eng_g.generate([5, 9, 2, 7], max_new_tokens=12, timeout=120)
+        re_ = eng_e.generate([5, 9, 2, 7], max_new_tokens=12, timeout=120)
+        assert rg.error == "" and re_.error == ""
+        assert rg.generated == re_.generated, (rg.generated, re_.generated)
+        assert eng_g.stats["graph_replays"] > 0
+    finally:
+        eng_g.stop()
+        eng_e.stop()
