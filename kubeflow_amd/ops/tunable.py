@@ -9,7 +9,9 @@ enable() must run BEFORE the first GEMM; call it at process entry
 import os
 from pathlib import Path
 
-CSV = Path(__file__).resolve().parent / "tuned" / "tunableop_gfx950.csv"
+# TunableOp expands FILENAME per device: "tunableop.csv" -> "tunableop0.csv"
+BASE = Path(__file__).resolve().parent / "tuned" / "tunableop.csv"
+CSV = Path(__file__).resolve().parent / "tuned" / "tunableop0.csv"
 
 
 def enable() -> bool:
@@ -19,5 +21,5 @@ def enable() -> bool:
         return True  # caller controls it (e.g. a re-tuning run)
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"   # use table, never re-tune
-    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = str(CSV)
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = str(BASE)
     return True
