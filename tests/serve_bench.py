@@ -1,0 +1,67 @@
+"""Serving throughput/latency benchmark (InferenceService engine, 1 GPU).
+
+Usage: python3 tests/serve_bench.py [model] [n_requests] [max_new] [prompt_len]
+Prints one JSON line: decode tokens/s out, request latency p50/p99, TTFT p50.
+"""
+import json
+import os
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from kubeflow_amd.ops import tunable as _t
+_t.enable()
+
+from kubeflow_amd.runtime.serving import InferenceEngine  # noqa: E402
+
+
+def main():
+    model = sys.argv[1] if len(sys.argv) > 1 else "llama3-8b"
+    n_req = int(sys.argv[2]) if len(sys.argv) > 2 else 16
+    max_new = int(sys.argv[3]) if len(sys.argv) > 3 else 64
+    plen = int(sys.argv[4]) if len(sys.argv) > 4 else 128
+
+    eng = InferenceEngine(model, max_slots=32, smax=4096,
+                          max_batch=32).start()
+    # warm-up (captures the decode graphs for the buckets used)
+    eng.generate(list(range(1, plen + 1)), max_new_tokens=8, timeout=300)
+
+    results = []
+    def run(i):
+        results.append(eng.generate(list(range(1 + i, plen + 1 + i)),
+                                    max_new_tokens=max_new, timeout=600))
+
+    threads = [threading.Thread(target=run, args=(i,)) for i in range(n_req)]
+    t0 = time.time()
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    el = time.time() - t0
+
+    toks = sum(len(r.generated) for r in results)
+    lat = sorted(1000 * (r.finished_at - r.submitted)
+                 for r in results if r.finished_at)
+    ttft = sorted(1000 * (r.first_token_at - r.submitted)
+                  for r in results if r.first_token_at)
+    out = {
+        "metric": "inferenceservice_tokens_per_s_out",
+        "model": model,
+        "value": round(toks / el, 1),
+        "n_requests": n_req,
+        "max_new_tokens": max_new,
+        "prompt_len": plen,
+        "latency_p50_ms": round(lat[len(lat) // 2], 1),
+        "latency_p99_ms": round(lat[min(len(lat) - 1, int(len(lat) * 0.99))], 1),
+        "ttft_p50_ms": round(ttft[len(ttft) // 2], 1),
+        "graph_replays": eng.stats.get("graph_replays", 0),
+        "errors": sum(1 for r in results if r.error),
+    }
+    eng.stop()
+    print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()
