@@ -34,6 +34,7 @@ class TrainConfig:
     grad_accum: int = 1
     bucket_mb: Optional[float] = None
     use_wd_mask: bool = True
+    grad_clip: float = 1.0  # 0 disables
 
 
 class Trainer:
@@ -85,6 +86,13 @@ class Trainer:
             else:
                 loss.backward()
         self.ddp.finalize()
+        if cfg.grad_clip > 0:
+            # one flat-buffer norm; scale in place when above the clip
+            gnorm = torch.linalg.vector_norm(self.flat.grad,
+                                             dtype=torch.float32)
+            scale = (cfg.grad_clip / (gnorm + 1e-6)).clamp(max=1.0)
+            self.flat.grad.mul_(scale.to(self.flat.grad.dtype))
+            self.last_grad_norm = gnorm
         self.step_num += 1
         lr = self.lr_at(self.step_num - 1)
         ops.fused_adamw(self.flat.data, self.p32, self.flat.grad, self.m,
