@@ -34,12 +34,11 @@ import secrets
 import time
 from typing import Optional
 
-from fastapi import FastAPI, Request, Response
+from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, PlainTextResponse
 
 from kubeflow_amd.api import (ObjectStore, new_object, NotFoundError,
                               AlreadyExistsError, ConflictError)
-from kubeflow_amd.api.store import StoreError
 from kubeflow_amd.kfam import BindingClient
 
 PLURALS = {

@@ -19,10 +19,9 @@ from __future__ import annotations
 import copy
 import json
 import os
-import queue
 import threading
 import time
-from typing import Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 from .objects import KfObject, match_labels, new_object, now_iso
 

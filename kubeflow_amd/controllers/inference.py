@@ -22,9 +22,7 @@ like Deployment-backed predictors).
 """
 from __future__ import annotations
 
-import json
 import os
-import time
 import urllib.request
 from typing import Dict, Optional
 

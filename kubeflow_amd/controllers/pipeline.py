@@ -20,10 +20,10 @@ GPU demands overlap — gang-aware chaining on one node.
 from __future__ import annotations
 
 import time
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
-from kubeflow_amd.api import ObjectStore, new_object, set_condition
-from kubeflow_amd.api.objects import has_condition, owner_ref, get_condition
+from kubeflow_amd.api import new_object, set_condition
+from kubeflow_amd.api.objects import has_condition, owner_ref
 from kubeflow_amd.controllers.base import Reconciler, RequeueAfter
 
 

@@ -17,7 +17,6 @@ import json
 import os
 import subprocess
 import sys
-import time
 import urllib.request
 from typing import Dict, Optional
 

@@ -19,11 +19,10 @@ from __future__ import annotations
 
 import json
 import os
-import shutil
 import time
 from typing import Dict, Optional
 
-from kubeflow_amd.api import ObjectStore, set_condition, get_condition
+from kubeflow_amd.api import ObjectStore, set_condition
 from kubeflow_amd.api.objects import has_condition
 from kubeflow_amd.controllers.base import Reconciler, RequeueAfter
 from kubeflow_amd.scheduler import (GangScheduler, InsufficientResources,

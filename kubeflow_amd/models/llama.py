@@ -16,7 +16,7 @@ ffn 14336, vocab 128256).
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 import torch.nn as nn

@@ -6,7 +6,6 @@ small, fast, runs on CPU with gloo world_size=1..N.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

@@ -16,13 +16,12 @@ from __future__ import annotations
 
 import argparse
 import json
-import os
 import time
 
 from kubeflow_amd.ops import tunable as _kf_tunable
 _kf_tunable.enable()
 from fastapi import FastAPI, Request as HttpRequest
-from fastapi.responses import JSONResponse, PlainTextResponse
+from fastapi.responses import PlainTextResponse
 import uvicorn
 
 from kubeflow_amd.runtime.serving import InferenceEngine

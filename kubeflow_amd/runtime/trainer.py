@@ -9,16 +9,13 @@ a PyTorchJob-compatible directory layout (checkpoint.py).
 from __future__ import annotations
 
 import math
-import os
-import time
-from dataclasses import dataclass, field
-from typing import Callable, Optional
+from dataclasses import dataclass
+from typing import Optional
 
 import torch
 
 from kubeflow_amd import ops
 from kubeflow_amd.parallel import FlatParamSpace, BucketedDDP
-from kubeflow_amd.parallel import dist as kdist
 
 
 @dataclass

@@ -11,7 +11,7 @@ InferenceServices) can share the node with a big training job.
 from __future__ import annotations
 
 import threading
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 from .inventory import GpuInventory

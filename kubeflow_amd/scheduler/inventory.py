@@ -9,8 +9,8 @@ from __future__ import annotations
 
 import os
 import subprocess
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import List
 
 
 @dataclass
