@@ -142,6 +142,14 @@ class NotebookReconciler(Reconciler):
                        "image": self._image(nb)}, f)
         env = dict(os.environ)
         env["NB_PREFIX"] = prefix
+        from kubeflow_amd.scheduler.launcher import merge_poddefaults
+        try:
+            env = merge_poddefaults(
+                env, m.get("labels", {}),
+                self.store.list("PodDefault", ns))
+        except ValueError as e:  # conflicting defaults -> surface, keep going
+            self.store.record_event(nb, "PodDefaultConflict", str(e),
+                                    "Warning")
         repo_root = os.path.dirname(os.path.dirname(os.path.dirname(
             os.path.abspath(__file__))))
         env["PYTHONPATH"] = repo_root + os.pathsep + env.get("PYTHONPATH", "")

@@ -108,3 +108,51 @@ def test_tensorboard_rejects_cloud_paths(tmp_path):
             plat.store.get("Tensorboard", "bad", "ns")))
         assert "unsupported" in [c for c in obj["status"]["conditions"]
                                  if c["type"] == "Failed"][0]["message"]
+
+
+def test_culler_stops_idle_notebook(tmp_path, monkeypatch):
+    """ENABLE_CULLING with a tiny IDLE_TIME stops an idle session
+    (culler.go parity: poll /api/status, set stop annotation)."""
+    monkeypatch.setenv("ENABLE_CULLING", "true")
+    monkeypatch.setenv("IDLE_TIME", "0.0001")          # ~6 ms idle budget
+    monkeypatch.setenv("CULLING_CHECK_PERIOD", "0.01")  # 0.6 s requeue
+    with Platform(root_dir=str(tmp_path)) as plat:
+        assert plat.notebook.enable_culling
+        plat.store.create(new_object("Notebook", "idle-nb", "ns", spec={}))
+        obj = _wait(lambda: (lambda o: o if o["status"].get("readyReplicas")
+                             else None)(
+            plat.store.get("Notebook", "idle-nb", "ns")), timeout=60)
+        # idle -> culled: stop annotation appears and session stops
+        def culled():
+            o = plat.store.get("Notebook", "idle-nb", "ns")
+            ann = o["metadata"].get("annotations", {})
+            return o if ("kubeflow-resource-stopped" in ann
+                         and o["status"].get("readyReplicas") == 0) else None
+        obj = _wait(culled, timeout=60)
+        reasons = {e["reason"] for e in plat.store.events_for(obj)}
+        assert "Culling" in reasons
+
+
+def test_notebook_poddefault_env(tmp_path):
+    """PodDefault env reaches notebook session processes too (the webhook
+    mutates every pod in the reference)."""
+    with Platform(root_dir=str(tmp_path)) as plat:
+        pd = new_object("PodDefault", "nb-env", "ns", spec={
+            "selector": {"matchLabels": {"wants-env": "yes"}},
+            "env": [{"name": "KF_NB_INJECTED", "value": "hello"}]},
+            api_version="kubeflow.org/v1alpha1")
+        plat.store.create(pd)
+        nb = new_object("Notebook", "envy", "ns", spec={},
+                        labels={"wants-env": "yes"})
+        plat.store.create(nb)
+        obj = _wait(lambda: (lambda o: o if o["status"].get("readyReplicas")
+                             else None)(
+            plat.store.get("Notebook", "envy", "ns")), timeout=60)
+        # session process can see the injected env via /api/execute
+        url = obj["status"]["url"].rstrip("/")
+        body = json.dumps({"code": "import os; print(os.environ.get('KF_NB_INJECTED'))"}).encode()
+        req = urllib.request.Request(url + "/api/execute", data=body,
+                                     headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=10) as r:
+            out = json.loads(r.read())
+        assert out["output"].strip() == "hello"
