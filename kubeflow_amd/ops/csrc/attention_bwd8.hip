@@ -54,7 +54,7 @@ __device__ __forceinline__ kf_bf16x8 kf_exchange8(const kf_f32x16& a,
 
 // ---------------------------------------------------------------- pass dQ --
 #define DQ8_QT 256
-#define DQ8_KT 64
+#define DQ8_KT 64   // kv rows per LDS tile (128 measured slower)
 
 __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
     unsigned short* __restrict__ dq, const unsigned short* __restrict__ q,
@@ -103,7 +103,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
       const unsigned short* vg =
           v + (b * S + kt * DQ8_KT) * kts + (int64_t)hkv * AB_D;
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
+      for (int j = 0; j < DQ8_KT * 16 / 512; ++j) {
         const int vi = tid + 512 * j;
         const int r = vi >> 4, c8 = vi & 15;
         kf_short8 kv8 =
@@ -114,7 +114,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
           const int el = (jj + tid) & 7;
           const int dd = c8 * 8 + el;
           *reinterpret_cast<unsigned short*>(
-              kt_lds + kf_swz8(dd, r * 2, 128)) = (unsigned short)kv8[el];
+              kt_lds + kf_swz8(dd, r * 2, DQ8_KT * 2)) =
+              (unsigned short)kv8[el];
         }
         kf_short8 vv8 =
             *reinterpret_cast<const kf_short8*>(vg + r * kts + c8 * 8);
@@ -123,8 +124,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
     }
     __syncthreads();
 
-#pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll 1  // dynamic (full unroll spills at KT=128)
+    for (int mt = 0; mt < DQ8_KT / 32; ++mt) {
       kf_f32x16 st = kf_f32x16{0.f}, dpt = kf_f32x16{0.f};
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -156,7 +157,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
         for (int step = 0; step < 2; ++step) {
           kf_bf16x8 kta = *reinterpret_cast<const kf_bf16x8*>(
               kt_lds + kf_swz8(dt * 32 + l31,
-                               mt * 64 + step * 32 + hi * 16, 128));
+                               mt * 64 + step * 32 + hi * 16, DQ8_KT * 2));
           dqacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kta, pb[step],
                                                               dqacc[dt], 0, 0,
                                                               0);
@@ -183,7 +184,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
 // one kernel spilled 62 VGPRs at the 2-waves/SIMD budget; recomputing S in
 // a second kernel (+25% MFMA) is far cheaper than scratch traffic.
 #define DKV8_KT 256   // kv rows per block (8 waves x 32)
-#define DKV8_QT 64    // q rows staged per iteration
+#define DV8_QT 64     // dv8 q-tile (128 measured slower)
+#define DK8_QT 64     // dk8 q-tile (128 spilled 29 VGPRs -> regressed)
 
 __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
     unsigned short* __restrict__ dv, const unsigned short* __restrict__ q,
@@ -191,9 +193,9 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
     const unsigned short* __restrict__ dout, const float* __restrict__ lse,
     int64_t B, int S, int Hq, int Hkv, int64_t qts, int64_t kts,
     int64_t dkts, float scale, int causal) {
-  __shared__ unsigned char q_lds[DKV8_QT * AB_D * 2];    // [64][128]
-  __shared__ unsigned char dot_lds[AB_D * DKV8_QT * 2];  // [128][64]
-  __shared__ float lse_s[DKV8_QT];
+  __shared__ unsigned char q_lds[DV8_QT * AB_D * 2];    // [64][128]
+  __shared__ unsigned char dot_lds[AB_D * DV8_QT * 2];  // [128][64]
+  __shared__ float lse_s[DV8_QT];
 
   const int kt = blockIdx.x, hkv = blockIdx.y;
   const int64_t b = blockIdx.z;
@@ -210,21 +212,21 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
 #pragma unroll
   for (int i = 0; i < 4; ++i) dvacc[i] = kf_f32x16{0.f};
 
-  const int qt0 = causal ? (kt * DKV8_KT) / DKV8_QT : 0;
-  const int nqt = S / DKV8_QT;
+  const int qt0 = causal ? (kt * DKV8_KT) / DV8_QT : 0;
+  const int nqt = S / DV8_QT;
   for (int hg = 0; hg < g; ++hg) {
     const int hq = hkv * g + hg;
     for (int qt = qt0; qt < nqt; ++qt) {
       {  // stage Q row-major + dO transposed + the q-tile's lse row
         const unsigned short* qg =
-            q + (b * S + qt * DKV8_QT) * qts + (int64_t)hq * AB_D;
+            q + (b * S + qt * DV8_QT) * qts + (int64_t)hq * AB_D;
         const unsigned short* dog =
-            dout + ((b * S + qt * DKV8_QT) * (int64_t)Hq + hq) * AB_D;
-        if (tid < DKV8_QT)
+            dout + ((b * S + qt * DV8_QT) * (int64_t)Hq + hq) * AB_D;
+        if (tid < DV8_QT)
           lse_s[tid] =
-              lse[(b * Hq + hq) * (int64_t)S + qt * DKV8_QT + tid];
+              lse[(b * Hq + hq) * (int64_t)S + qt * DV8_QT + tid];
 #pragma unroll
-        for (int j = 0; j < 2; ++j) {
+        for (int j = 0; j < DV8_QT * 16 / 512; ++j) {
           const int vi = tid + 512 * j;
           const int r = vi >> 4, c8 = vi & 15;
           kf_short8 q8 =
@@ -237,14 +239,15 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
             const int el = (jj + tid) & 7;
             const int dd = c8 * 8 + el;
             *reinterpret_cast<unsigned short*>(
-                dot_lds + kf_swz8(dd, r * 2, 128)) = (unsigned short)do8[el];
+                dot_lds + kf_swz8(dd, r * 2, DV8_QT * 2)) =
+                (unsigned short)do8[el];
           }
         }
       }
       __syncthreads();
 
-#pragma unroll
-      for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll 1
+      for (int mt = 0; mt < DV8_QT / 32; ++mt) {
         kf_f32x16 sacc = kf_f32x16{0.f};
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -261,7 +264,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int ql = q0 + (r & 3) + 8 * (r >> 2);
-          const int qq = qt * DKV8_QT + ql;
+          const int qq = qt * DV8_QT + ql;
           sacc[r] = (causal && kvrow_g > qq)
                         ? 0.f
                         : __expf(sacc[r] * scale - lse_s[ql]);
@@ -274,7 +277,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
           for (int step = 0; step < 2; ++step) {
             kf_bf16x8 dob = *reinterpret_cast<const kf_bf16x8*>(
                 dot_lds + kf_swz8(dt * 32 + l31,
-                                  mt * 64 + step * 32 + hi * 16, 128));
+                                  mt * 64 + step * 32 + hi * 16,
+                                  DV8_QT * 2));
             dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 pa[step], dob, dvacc[dt], 0, 0, 0);
           }
@@ -302,10 +306,10 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
     const unsigned short* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, int64_t B, int S, int Hq, int Hkv,
     int64_t qts, int64_t kts, int64_t dkts, float scale, int causal) {
-  __shared__ unsigned char q_lds[DKV8_QT * AB_D * 2];    // [64][128]
-  __shared__ unsigned char do_lds[DKV8_QT * AB_D * 2];   // [64][128]
-  __shared__ unsigned char qt_lds[AB_D * DKV8_QT * 2];   // [128][64]
-  __shared__ float lse_s2[DKV8_QT], dlt_s2[DKV8_QT];
+  __shared__ unsigned char q_lds[DK8_QT * AB_D * 2];    // [64][128]
+  __shared__ unsigned char do_lds[DK8_QT * AB_D * 2];   // [64][128]
+  __shared__ unsigned char qt_lds[AB_D * DK8_QT * 2];   // [128][64]
+  __shared__ float lse_s2[DK8_QT], dlt_s2[DK8_QT];
 
   const int kt = blockIdx.x, hkv = blockIdx.y;
   const int64_t b = blockIdx.z;
@@ -324,25 +328,25 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
 #pragma unroll
   for (int i = 0; i < 4; ++i) dkacc[i] = kf_f32x16{0.f};
 
-  const int qt0 = causal ? (kt * DKV8_KT) / DKV8_QT : 0;
-  const int nqt = S / DKV8_QT;
+  const int qt0 = causal ? (kt * DKV8_KT) / DK8_QT : 0;
+  const int nqt = S / DK8_QT;
   for (int hg = 0; hg < g; ++hg) {
     const int hq = hkv * g + hg;
     for (int qt = qt0; qt < nqt; ++qt) {
       {  // stage Q (row-major + transposed), dO row-major, lse/delta rows
         const unsigned short* qg =
-            q + (b * S + qt * DKV8_QT) * qts + (int64_t)hq * AB_D;
+            q + (b * S + qt * DK8_QT) * qts + (int64_t)hq * AB_D;
         const unsigned short* dog =
-            dout + ((b * S + qt * DKV8_QT) * (int64_t)Hq + hq) * AB_D;
-        if (tid < DKV8_QT)
+            dout + ((b * S + qt * DK8_QT) * (int64_t)Hq + hq) * AB_D;
+        if (tid < DK8_QT)
           lse_s2[tid] =
-              lse[(b * Hq + hq) * (int64_t)S + qt * DKV8_QT + tid];
-        else if (tid < 2 * DKV8_QT)
-          dlt_s2[tid - DKV8_QT] =
-              delta[(b * Hq + hq) * (int64_t)S + qt * DKV8_QT + tid -
-                    DKV8_QT];
+              lse[(b * Hq + hq) * (int64_t)S + qt * DK8_QT + tid];
+        else if (tid < 2 * DK8_QT)
+          dlt_s2[tid - DK8_QT] =
+              delta[(b * Hq + hq) * (int64_t)S + qt * DK8_QT + tid -
+                    DK8_QT];
 #pragma unroll
-        for (int j = 0; j < 2; ++j) {
+        for (int j = 0; j < DK8_QT * 16 / 512; ++j) {
           const int vi = tid + 512 * j;
           const int r = vi >> 4, c8 = vi & 15;
           kf_short8 q8 =
@@ -357,14 +361,15 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
             const int el = (jj + tid) & 7;
             const int dd = c8 * 8 + el;
             *reinterpret_cast<unsigned short*>(
-                qt_lds + kf_swz8(dd, r * 2, 128)) = (unsigned short)q8[el];
+                qt_lds + kf_swz8(dd, r * 2, DK8_QT * 2)) =
+                (unsigned short)q8[el];
           }
         }
       }
       __syncthreads();
 
-#pragma unroll
-      for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll 1
+      for (int mt = 0; mt < DK8_QT / 32; ++mt) {
         kf_f32x16 sacc = kf_f32x16{0.f}, dpacc = kf_f32x16{0.f};
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -387,7 +392,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int ql = q0 + (r & 3) + 8 * (r >> 2);
-          const int qq = qt * DKV8_QT + ql;
+          const int qq = qt * DK8_QT + ql;
           const float p = (causal && kvrow_g > qq)
                               ? 0.f
                               : __expf(sacc[r] * scale - lse_s2[ql]);
@@ -401,7 +406,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
           for (int step = 0; step < 2; ++step) {
             kf_bf16x8 qb = *reinterpret_cast<const kf_bf16x8*>(
                 qt_lds + kf_swz8(dt * 32 + l31,
-                                 mt * 64 + step * 32 + hi * 16, 128));
+                                 mt * 64 + step * 32 + hi * 16,
+                                 DK8_QT * 2));
             dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 dsa[step], qb, dkacc[dt], 0, 0, 0);
           }

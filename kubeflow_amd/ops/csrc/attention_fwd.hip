@@ -251,7 +251,7 @@ __global__ __launch_bounds__(AT_THREADS, 2) void kf_attn_fwd_kernel(
 typedef float kf_f32x16 __attribute__((ext_vector_type(16)));
 
 #define A8_QT 256
-#define A8_KT 64
+#define A8_KT 128   // kv rows per LDS tile (barrier pair)
 #define A8_THREADS 512
 
 __device__ __forceinline__ unsigned int kf_cvt_pk_bf16(float lo, float hi) {
@@ -265,8 +265,8 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
     int64_t qts, int64_t kts, float scale, int causal) {
-  __shared__ unsigned char k_lds[A8_KT * AT_D * 2];   // [64][128] swz
-  __shared__ unsigned char vt_lds[AT_D * A8_KT * 2];  // [128][64] swz
+  __shared__ unsigned char k_lds[A8_KT * AT_D * 2];   // [A8_KT][128] swz
+  __shared__ unsigned char vt_lds[AT_D * A8_KT * 2];  // [128][A8_KT] swz
 
   const int qt = blockIdx.x, hq = blockIdx.y;
   const int64_t b = blockIdx.z;
@@ -304,8 +304,8 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
       const unsigned short* vg =
           v + (b * S + kt * A8_KT) * kts + (int64_t)hkv * AT_D;
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
-        const int vi = tid + A8_THREADS * j;   // 0..1023
+      for (int j = 0; j < A8_KT * 16 / A8_THREADS; ++j) {
+        const int vi = tid + A8_THREADS * j;
         const int r = vi >> 4, c8 = vi & 15;
         kf_short8 kv8 =
             *reinterpret_cast<const kf_short8*>(kg + r * kts + c8 * 8);
@@ -324,8 +324,8 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
     }
     __syncthreads();
 
-#pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {  // two 32-kv M-tiles
+#pragma unroll 1  // dynamic: full unroll at KT=128 spilled 127 VGPRs
+    for (int mt = 0; mt < A8_KT / 32; ++mt) {  // 32-kv M-tiles
       // ---- S^T = mfma(K, Q): rows kv, cols q ----
       kf_f32x16 st = kf_f32x16{0.f};
       __builtin_amdgcn_s_setprio(1);
