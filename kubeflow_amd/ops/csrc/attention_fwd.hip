@@ -251,7 +251,7 @@ __global__ __launch_bounds__(AT_THREADS, 2) void kf_attn_fwd_kernel(
 typedef float kf_f32x16 __attribute__((ext_vector_type(16)));
 
 #define A8_QT 256
-#define A8_KT 128   // kv rows per LDS tile (barrier pair)
+#define A8_KT 256   // kv rows per LDS tile (barrier pair)
 #define A8_THREADS 512
 
 __device__ __forceinline__ unsigned int kf_cvt_pk_bf16(float lo, float hi) {
@@ -435,13 +435,14 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
 //    for the V transpose-scatter visibility.
 // ---------------------------------------------------------------------------
 
+#define A8B_KT 64  // the async variant keeps 64-row tiles (double-buffered)
 __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
     unsigned short* __restrict__ o, float* __restrict__ lse,
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
     int64_t qts, int64_t kts, float scale, int causal) {
-  __shared__ unsigned char k_lds[2][A8_KT * AT_D * 2];  // double-buffered
-  __shared__ unsigned char vt_lds[AT_D * A8_KT * 2];
+  __shared__ unsigned char k_lds[2][A8B_KT * AT_D * 2];  // double-buffered
+  __shared__ unsigned char vt_lds[AT_D * A8B_KT * 2];
 
   const int qt = blockIdx.x, hq = blockIdx.y;
   const int64_t b = blockIdx.z;
@@ -469,7 +470,7 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
   const int qrow_g = qt * A8_QT + w * 32 + l31;
 
   const int last_kt =
-      causal ? (qt * A8_QT + A8_QT - 1) / A8_KT : (S / A8_KT - 1);
+      causal ? (qt * A8_QT + A8_QT - 1) / A8B_KT : (S / A8B_KT - 1);
   const unsigned short* kbase0 = k + b * S * kts + (int64_t)hkv * AT_D;
   const unsigned short* vbase0 = v + b * S * kts + (int64_t)hkv * AT_D;
 
@@ -481,7 +482,7 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
       const int ci = w * 128 + j * 64 + lane;
       const int r = ci >> 4, c8 = ci & 15;
       const unsigned short* src =
-          kbase0 + (int64_t)(kt_i * A8_KT + r) * kts + (c8 ^ (r & 7)) * 8;
+          kbase0 + (int64_t)(kt_i * A8B_KT + r) * kts + (c8 ^ (r & 7)) * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)
@@ -494,7 +495,7 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
   for (int kt = 0; kt <= last_kt; ++kt) {
     // ---- V load + transpose-scatter (single buffer) ----
     {
-      const unsigned short* vg = vbase0 + (int64_t)(kt * A8_KT) * kts;
+      const unsigned short* vg = vbase0 + (int64_t)(kt * A8B_KT) * kts;
 #pragma unroll
       for (int j = 0; j < 2; ++j) {
         const int vi = tid + A8_THREADS * j;
@@ -506,7 +507,7 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
           const int el = (jj + tid) & 7;
           const int dd = c8 * 8 + el;
           *reinterpret_cast<unsigned short*>(
-              vt_lds + kf_swz(dd, r * 2, A8_KT * 2)) = (unsigned short)vv8[el];
+              vt_lds + kf_swz(dd, r * 2, A8B_KT * 2)) = (unsigned short)vv8[el];
         }
       }
     }
@@ -534,7 +535,7 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
 
-      const int kv0 = kt * A8_KT + mt * 32 + hi * 4;
+      const int kv0 = kt * A8B_KT + mt * 32 + hi * 4;
       float mx = -INFINITY;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -586,7 +587,7 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
         for (int step = 0; step < 2; ++step) {
           kf_bf16x8 vfrag = *reinterpret_cast<const kf_bf16x8*>(
               vt_lds + kf_swz(dt * 32 + l31,
-                              mt * 64 + step * 32 + hi * 16, A8_KT * 2));
+                              mt * 64 + step * 32 + hi * 16, A8B_KT * 2));
           oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag, pb[step],
                                                              oacc[dt], 0, 0, 0);
         }

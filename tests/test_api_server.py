@@ -166,3 +166,54 @@ def test_profile_reconciler_provisions(tmp_path):
             store.get("Namespace", "alice", None)
     finally:
         mgr.stop()
+
+
+def test_csrf_double_submit(monkeypatch):
+    """KF_CSRF=1: mutations need matching cookie+header (csrf.py parity)."""
+    monkeypatch.setenv("KF_CSRF", "1")
+    store = ObjectStore()
+    c = TestClient(build_app(store))
+    # GET is exempt and seeds the cookie
+    r = c.get("/api/namespaces/ns/notebooks", headers=U)
+    assert r.json()["success"]
+    token = r.cookies.get("XSRF-TOKEN")
+    assert token
+    # mutation without the header -> 403
+    r = c.post("/api/namespaces/ns/notebooks", headers=U,
+               json={"name": "nb", "spec": {}})
+    assert r.status_code == 403
+    # with the matching header -> ok
+    r = c.post("/api/namespaces/ns/notebooks",
+               headers={**U, "X-XSRF-TOKEN": token},
+               json={"name": "nb", "spec": {}})
+    assert r.json()["success"], r.json()
+
+
+def test_store_concurrent_writers():
+    """Optimistic concurrency under racing writers (the
+    reconcile-conflict discipline)."""
+    import threading
+    store = ObjectStore()
+    store.create(new_object("Counter", "c", "ns", spec={"n": 0}))
+    errors = []
+
+    def bump(k):
+        for _ in range(50):
+            store.patch("Counter", "c", "ns", {"spec": {f"w{k}": True}})
+            while True:
+                cur = store.get("Counter", "c", "ns")
+                cur["spec"]["n"] += 1
+                try:
+                    store.update(cur)
+                    break
+                except Exception as e:
+                    if "stale" not in str(e):
+                        errors.append(e)
+                        break
+    ts = [threading.Thread(target=bump, args=(k,)) for k in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(60)
+    assert not errors
+    assert store.get("Counter", "c", "ns")["spec"]["n"] == 200
