@@ -251,7 +251,7 @@ __global__ __launch_bounds__(AT_THREADS, 2) void kf_attn_fwd_kernel(
 typedef float kf_f32x16 __attribute__((ext_vector_type(16)));
 
 #define A8_QT 256
-#define A8_KT 256   // kv rows per LDS tile (barrier pair)
+#define A8_KT 128   // kv rows per LDS tile (256 measured slower, 64 was 236 TF)
 #define A8_THREADS 512
 
 __device__ __forceinline__ unsigned int kf_cvt_pk_bf16(float lo, float hi) {
