@@ -1,0 +1,58 @@
+"""Parallelism strategy descriptor — the launcher's extension seam.
+
+SURVEY.md §2.14: DDP is the only v1 strategy (the only one any BASELINE
+config requires); TP/PP/SP/CP/EP/Ulysses are declared here as enums so
+PyTorchJob specs stay forward-compatible, and rejected with a clear error
+until a config demands them. The seam is the per-rank environment the gang
+launcher already provides (RANK/WORLD_SIZE/LOCAL_RANK + this descriptor
+serialized into the worker spec as `parallelism`).
+
+Spec form (PyTorchJob template):
+    parallelism: {strategy: ddp}                      # default
+    parallelism: {strategy: tp, degree: 4}            # reserved, v2+
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from enum import Enum
+
+
+class Strategy(str, Enum):
+    DDP = "ddp"      # implemented: bucketed all-reduce over RCCL/xGMI
+    TP = "tp"        # reserved: tensor parallel (xGMI all-reduce per layer)
+    PP = "pp"        # reserved: pipeline parallel
+    SP = "sp"        # reserved: sequence/context parallel (ring attention)
+    EP = "ep"        # reserved: expert parallel (all-to-all)
+    ULYSSES = "ulysses"  # reserved: attention head-scatter SP
+
+
+IMPLEMENTED = {Strategy.DDP}
+
+
+@dataclass
+class ParallelismSpec:
+    strategy: Strategy = Strategy.DDP
+    degree: int = 1
+
+    @classmethod
+    def from_spec(cls, template: dict) -> "ParallelismSpec":
+        p = template.get("parallelism") or {}
+        if isinstance(p, str):
+            p = {"strategy": p}
+        try:
+            strategy = Strategy(p.get("strategy", "ddp"))
+        except ValueError:
+            raise ValueError(
+                f"unknown parallelism strategy {p.get('strategy')!r}; "
+                f"known: {[s.value for s in Strategy]}")
+        spec = cls(strategy=strategy, degree=int(p.get("degree", 1)))
+        spec.validate()
+        return spec
+
+    def validate(self):
+        if self.strategy not in IMPLEMENTED:
+            raise NotImplementedError(
+                f"parallelism strategy {self.strategy.value!r} is reserved "
+                "but not implemented in v1 — DDP (one process per MI355X, "
+                "RCCL over xGMI) is the supported strategy; see "
+                "SURVEY.md §2.14 for the extension seam")

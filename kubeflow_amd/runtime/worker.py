@@ -82,6 +82,8 @@ def main(argv=None):
     signal.signal(signal.SIGTERM, lambda *_: stop.update(flag=True))
 
     try:
+        from kubeflow_amd.parallel.strategy import ParallelismSpec
+        ParallelismSpec.from_spec(spec)  # DDP-only v1; loud on reserved ones
         rank, world, device = kdist.init_distributed()
         torch.manual_seed(int(spec.get("seed", 0)) + rank)
         dtype = (torch.bfloat16 if device.type == "cuda"

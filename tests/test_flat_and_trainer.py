@@ -94,3 +94,15 @@ def test_grad_accum_equivalence():
     p1 = torch.cat([p.flatten() for p in m1.parameters()])
     p2 = torch.cat([p.flatten() for p in m2.parameters()])
     assert torch.allclose(p1, p2, atol=1e-4)
+
+
+def test_parallelism_strategy_seam():
+    from kubeflow_amd.parallel.strategy import ParallelismSpec, Strategy
+    assert ParallelismSpec.from_spec({}).strategy == Strategy.DDP
+    assert ParallelismSpec.from_spec(
+        {"parallelism": "ddp"}).strategy == Strategy.DDP
+    with pytest.raises(NotImplementedError):
+        ParallelismSpec.from_spec({"parallelism": {"strategy": "tp",
+                                                   "degree": 4}})
+    with pytest.raises(ValueError):
+        ParallelismSpec.from_spec({"parallelism": {"strategy": "magic"}})
