@@ -36,11 +36,12 @@ class TrainingJobReconciler(Reconciler):
     replica_field = "pytorchReplicaSpecs"
 
     def __init__(self, store: ObjectStore, scheduler: GangScheduler,
-                 jobs_dir: str, poll_period: float = 1.0):
+                 jobs_dir: str, poll_period: float = 0.25, warm_pool=None):
         super().__init__(store)
         self.scheduler = scheduler
         self.jobs_dir = jobs_dir
         self.poll_period = poll_period
+        self.warm_pool = warm_pool
         self.gangs: Dict[str, ProcessGang] = {}
         self.restarts: Dict[str, int] = {}
         self.key_uid: Dict[tuple, str] = {}
@@ -121,7 +122,7 @@ class TrainingJobReconciler(Reconciler):
         gang = launch_gang(uid, workdir, spec, gpu_indices,
                            poddefaults=poddefaults,
                            labels=job["metadata"].get("labels", {}),
-                           numa_nodes=numa)
+                           numa_nodes=numa, warm_pool=self.warm_pool)
         self.gangs[uid] = gang
         set_condition(job, "Created", "True", "JobCreated", "gang launched")
         set_condition(job, "Running", "True", "JobRunning",

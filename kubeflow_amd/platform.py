@@ -30,6 +30,12 @@ class Platform:
             if persist else None)
         self.inventory = GpuInventory()
         self.scheduler = GangScheduler(self.inventory)
+        from kubeflow_amd.scheduler.warmpool import WarmPool
+        pool_size = int(os.environ.get(
+            "KF_WARM_POOL", str(max(2, self.inventory.n_gpus))))
+        self.warm_pool = (WarmPool(pool_size,
+                                   os.path.join(self.root_dir, "warmpool"))
+                          if pool_size > 0 else None)
         self.manager = ControllerManager(self.store)
         self._controllers = []
         self._register_all()
@@ -39,8 +45,10 @@ class Platform:
             TrainingJobReconciler, TFJobReconciler)
         jobs_dir = os.path.join(self.root_dir, "jobs")
         self.pytorchjob = TrainingJobReconciler(self.store, self.scheduler,
-                                                jobs_dir)
-        self.tfjob = TFJobReconciler(self.store, self.scheduler, jobs_dir)
+                                                jobs_dir,
+                                                warm_pool=self.warm_pool)
+        self.tfjob = TFJobReconciler(self.store, self.scheduler, jobs_dir,
+                                     warm_pool=self.warm_pool)
         for rec in (self.pytorchjob, self.tfjob):
             self.manager.register(rec)
             self._controllers.append(rec)
@@ -85,6 +93,8 @@ class Platform:
 
     def stop(self):
         self.manager.stop()
+        if self.warm_pool is not None:
+            self.warm_pool.shutdown()
         for rec in self._controllers:
             if hasattr(rec, "shutdown"):
                 try:

@@ -137,9 +137,11 @@ def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
                 poddefaults: Optional[List[dict]] = None,
                 labels: Optional[Dict[str, str]] = None,
                 numa_nodes: Optional[Dict[int, int]] = None,
-                entry_module: str = "kubeflow_amd.runtime.worker") -> ProcessGang:
+                entry_module: str = "kubeflow_amd.runtime.worker",
+                warm_pool=None) -> ProcessGang:
     """Spawn one process per rank. CPU jobs pass gpu_indices=[] and
-    spec['world_size'] ranks run on CPU (gloo)."""
+    spec['world_size'] ranks run on CPU (gloo). With a WarmPool, ranks are
+    handed to pre-forked workers (torch already imported) when available."""
     os.makedirs(workdir, exist_ok=True)
     world = max(1, len(gpu_indices) or int(spec.get("world_size", 1)))
     port = free_port()
@@ -150,7 +152,25 @@ def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
     gang = ProcessGang(job_uid, workdir)
     repo_root = os.path.dirname(os.path.dirname(
         os.path.dirname(os.path.abspath(__file__))))
+    injected = merge_poddefaults({}, labels or {}, poddefaults or [])
     for rank in range(world):
+        if warm_pool is not None and entry_module == \
+                "kubeflow_amd.runtime.worker":
+            got = warm_pool.take()
+            if got is not None:
+                proc, log_path = got
+                warm_pool.assign(proc, {
+                    "spec_path": spec_path,
+                    "rank": rank,
+                    "world_size": world,
+                    "master_port": port,
+                    "workdir": workdir,
+                    "job_uid": job_uid,
+                    "gpu": gpu_indices[rank] if gpu_indices else None,
+                    "env": injected,
+                })
+                gang.ranks.append(RankProc(rank, proc, log_path))
+                continue
         env = dict(os.environ)
         env.update({
             "RANK": str(rank),

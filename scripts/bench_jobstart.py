@@ -22,6 +22,8 @@ from kubeflow_amd.platform import Platform
 def run(n=5, model="mnist-mlp", gpus=0):
     lat_start, lat_done = [], []
     with Platform(root_dir=tempfile.mkdtemp(prefix="jobstart-")) as plat:
+        if plat.warm_pool is not None:
+            plat.warm_pool.wait_ready(1, timeout=60)
         for i in range(n):
             name = f"js-{i}"
             spec = {"pytorchReplicaSpecs": {"Worker": {
@@ -48,6 +50,8 @@ def run(n=5, model="mnist-mlp", gpus=0):
                 time.sleep(0.05)
             lat_start.append(started)
             lat_done.append(done)
+            if plat.warm_pool is not None:  # measure the warm-hit path
+                plat.warm_pool.wait_ready(1, timeout=60)
     lat_start.sort()
     lat_done.sort()
     out = {
