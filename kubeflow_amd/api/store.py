@@ -186,12 +186,14 @@ class ObjectStore:
         Retries internally on conflict — patch semantics are last-writer-wins
         per field, like the k8s PATCH verb the web apps use
         (crud-web-apps .../patch.py:58-66)."""
-        for _ in range(10):
+        for attempt in range(100):
             cur = self.get(kind, name, namespace)
             merged = _deep_merge(cur, patch)
             try:
                 return self.update(merged)
             except ConflictError:
+                if attempt > 10:
+                    time.sleep(0.001 * (attempt - 10))
                 continue
         raise ConflictError(f"patch {kind} {namespace}/{name}: retries exhausted")
 
