@@ -66,7 +66,8 @@ class ApiError(Exception):
 
 
 def build_app(store: ObjectStore, scheduler=None,
-              root_dir: Optional[str] = None) -> FastAPI:
+              root_dir: Optional[str] = None,
+              volume_controller=None) -> FastAPI:
     app = FastAPI(title="kubeflow-amd platform API")
     bindings = BindingClient(store)
     userid_header = os.environ.get("USERID_HEADER", "kubeflow-userid")

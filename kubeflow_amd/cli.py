@@ -58,7 +58,8 @@ def serve(root: str = typer.Option("/tmp/kubeflow-amd", help="state dir"),
     from kubeflow_amd.platform import Platform
     from kubeflow_amd.api.server import build_app
     plat = Platform(root_dir=root, persist=persist).start()
-    api = build_app(plat.store, scheduler=plat.scheduler, root_dir=root)
+    api = build_app(plat.store, scheduler=plat.scheduler, root_dir=root,
+                    volume_controller=plat.volume)
     typer.secho(f"kubeflow-amd platform on http://127.0.0.1:{port} "
                 f"(state: {root}, GPUs: {plat.inventory.n_gpus})", fg="green")
     try:

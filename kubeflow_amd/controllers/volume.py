@@ -35,6 +35,10 @@ class VolumeReconciler(Reconciler):
     def reconcile(self, namespace: Optional[str], name: str) -> None:
         pvc = self.store.get(self.kind, name, namespace)
         path = self.path_for(namespace, name)
+        origin = pvc["metadata"].get("annotations", {}).get("rok/origin")
+        if origin and not pvc.get("status", {}).get("phase"):
+            if self.restore_from(namespace, name, origin):
+                self.store.record_event(pvc, "SnapshotRestored", origin)
         os.makedirs(path, exist_ok=True)
         req = (pvc["spec"].get("resources", {}).get("requests", {})
                .get("storage", "1Gi"))
@@ -51,6 +55,34 @@ class VolumeReconciler(Reconciler):
     def on_deleted(self, namespace, name):
         if self.reclaim_policy == "Delete":
             shutil.rmtree(self.path_for(namespace, name), ignore_errors=True)
+
+    # ---- snapshot provider seam (the reference's "rok" flavor:
+    # jupyter/backend/apps/rok — PVCs restored from snapshot URLs carried in
+    # `rok/origin` annotations). Single-node provider = directory copies.
+    def snapshot(self, namespace: Optional[str], name: str) -> str:
+        """Snapshot a PVC's contents; returns the snapshot URL."""
+        import time as _t
+        src = self.path_for(namespace, name)
+        snap_id = f"{name}-{int(_t.time() * 1000)}"
+        dst = os.path.join(self.volumes_dir, "_snapshots", snap_id)
+        os.makedirs(os.path.dirname(dst), exist_ok=True)
+        shutil.copytree(src, dst)
+        return f"rok://{snap_id}"
+
+    def restore_from(self, namespace: Optional[str], name: str,
+                     origin: str) -> bool:
+        """Materialize a new PVC directory from a rok:// snapshot URL."""
+        if not origin.startswith("rok://"):
+            return False
+        src = os.path.join(self.volumes_dir, "_snapshots",
+                           origin[len("rok://"):])
+        if not os.path.isdir(src):
+            return False
+        dst = self.path_for(namespace, name)
+        if os.path.exists(dst):
+            shutil.rmtree(dst)
+        shutil.copytree(src, dst)
+        return True
 
     @staticmethod
     def parse_quantity(q: str) -> int:

@@ -156,3 +156,30 @@ def test_notebook_poddefault_env(tmp_path):
         with urllib.request.urlopen(req, timeout=10) as r:
             out = json.loads(r.read())
         assert out["output"].strip() == "hello"
+
+
+def test_pvc_snapshot_and_restore(tmp_path):
+    """rok-flavor parity: snapshot a PVC, restore a new PVC from the
+    rok:// origin annotation (apps/rok/routes/post.py seam)."""
+    with Platform(root_dir=str(tmp_path)) as plat:
+        plat.store.create(new_object("PersistentVolumeClaim", "src", "ns",
+                                     spec={}, api_version="v1"))
+        obj = _wait(lambda: (lambda o: o if o["status"].get("phase") == "Bound"
+                             else None)(
+            plat.store.get("PersistentVolumeClaim", "src", "ns")))
+        with open(os.path.join(obj["status"]["hostPath"], "data.txt"),
+                  "w") as f:
+            f.write("precious")
+        url = plat.volume.snapshot("ns", "src")
+        assert url.startswith("rok://")
+        restored = new_object("PersistentVolumeClaim", "copy", "ns", spec={},
+                              api_version="v1",
+                              annotations={"rok/origin": url})
+        plat.store.create(restored)
+        obj = _wait(lambda: (lambda o: o if o["status"].get("phase") == "Bound"
+                             else None)(
+            plat.store.get("PersistentVolumeClaim", "copy", "ns")))
+        with open(os.path.join(obj["status"]["hostPath"], "data.txt")) as f:
+            assert f.read() == "precious"
+        evs = {e["reason"] for e in plat.store.events_for(obj)}
+        assert "SnapshotRestored" in evs
