@@ -22,7 +22,6 @@
 //    swizzled LDS round-trip per wave to reorient acc -> A-fragment.
 
 #include "kf_common.h"
-#include <cstdlib>
 
 typedef __bf16 kf_bf16x8 __attribute__((ext_vector_type(8)));
 typedef float kf_f32x4 __attribute__((ext_vector_type(4)));
@@ -250,8 +249,8 @@ __global__ __launch_bounds__(AT_THREADS, 2) void kf_attn_fwd_kernel(
 
 typedef float kf_f32x16 __attribute__((ext_vector_type(16)));
 
-#define A8_QT 256
-#define A8_KT 128   // kv rows per LDS tile (256 measured slower, 64 was 236 TF)
+#define A8_QT 256  // q rows per block (4-wave/128-row blocks measured 193 TF)
+#define A8_KT 128  // kv rows per LDS tile (64 -> 236 TF, 128 -> 296, 256 -> 240)
 #define A8_THREADS 512
 
 __device__ __forceinline__ unsigned int kf_cvt_pk_bf16(float lo, float hi) {
@@ -420,197 +419,6 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
 }
 
 
-// ---------------------------------------------------------------------------
-// v4 (A/B variant, KF_ATTN_FWD_ASYNC=1): v3 + async double-buffered K staging
-// via global_load_lds with RAW s_barriers and counted vmcnt (guide T3/T4):
-//  * K tile kt+1's direct-to-LDS loads are issued at the top of iteration kt
-//    and only waited (vmcnt(2), the 2 newest stay in flight) right before
-//    the barrier that publishes tile kt — HBM latency hides under tile
-//    kt-1's MFMA.
-//  * LDS dest is linear (global_load_lds is base+lane*16); the XOR swizzle
-//    is applied by PRE-SWIZZLING the per-lane global source chunk
-//    (c8_src = c8 ^ (r&7)) — both-sides-or-neither (guide rule 21).
-//  * __syncthreads() would drain vmcnt to 0 and kill the pipeline, so the
-//    loop uses raw __builtin_amdgcn_s_barrier() plus explicit lgkmcnt(0)
-//    for the V transpose-scatter visibility.
-// ---------------------------------------------------------------------------
-
-#define A8B_KT 64  // the async variant keeps 64-row tiles (double-buffered)
-__global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8b_kernel(
-    unsigned short* __restrict__ o, float* __restrict__ lse,
-    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
-    const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
-    int64_t qts, int64_t kts, float scale, int causal) {
-  __shared__ unsigned char k_lds[2][A8B_KT * AT_D * 2];  // double-buffered
-  __shared__ unsigned char vt_lds[AT_D * A8B_KT * 2];
-
-  const int qt = blockIdx.x, hq = blockIdx.y;
-  const int64_t b = blockIdx.z;
-  const int hkv = hq / (Hq / Hkv);
-  const int tid = threadIdx.x;
-  const int w = tid / KF_WAVE;
-  const int lane = tid & (KF_WAVE - 1);
-  const int l31 = lane & 31;
-  const int hi = lane >> 5;
-
-  kf_bf16x8 qfrag[8];
-  {
-    const int64_t qbase =
-        (b * S + qt * A8_QT + w * 32 + l31) * qts + (int64_t)hq * AT_D;
-#pragma unroll
-    for (int kk = 0; kk < 8; ++kk)
-      qfrag[kk] =
-          *reinterpret_cast<const kf_bf16x8*>(q + qbase + kk * 16 + hi * 8);
-  }
-
-  kf_f32x16 oacc[4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i) oacc[i] = kf_f32x16{0.f};
-  float m_run = -INFINITY, l_run = 0.f;
-  const int qrow_g = qt * A8_QT + w * 32 + l31;
-
-  const int last_kt =
-      causal ? (qt * A8_QT + A8_QT - 1) / A8B_KT : (S / A8B_KT - 1);
-  const unsigned short* kbase0 = k + b * S * kts + (int64_t)hkv * AT_D;
-  const unsigned short* vbase0 = v + b * S * kts + (int64_t)hkv * AT_D;
-
-  // per-wave gll geometry: 2 instructions x 1 KiB; this thread's chunk
-  // for instruction j is ci = w*128 + j*64 + lane -> dest row/col.
-  auto issue_k_tile = [&](int kt_i, int buf) {
-#pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      const int ci = w * 128 + j * 64 + lane;
-      const int r = ci >> 4, c8 = ci & 15;
-      const unsigned short* src =
-          kbase0 + (int64_t)(kt_i * A8B_KT + r) * kts + (c8 ^ (r & 7)) * 8;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)src,
-          (__attribute__((address_space(3))) unsigned int*)
-              &k_lds[buf][(w * 128 + j * 64) * 16],
-          16, 0, 0);
-    }
-  };
-
-  issue_k_tile(0, 0);
-  for (int kt = 0; kt <= last_kt; ++kt) {
-    // ---- V load + transpose-scatter (single buffer) ----
-    {
-      const unsigned short* vg = vbase0 + (int64_t)(kt * A8B_KT) * kts;
-#pragma unroll
-      for (int j = 0; j < 2; ++j) {
-        const int vi = tid + A8_THREADS * j;
-        const int r = vi >> 4, c8 = vi & 15;
-        kf_short8 vv8 =
-            *reinterpret_cast<const kf_short8*>(vg + r * kts + c8 * 8);
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {
-          const int el = (jj + tid) & 7;
-          const int dd = c8 * 8 + el;
-          *reinterpret_cast<unsigned short*>(
-              vt_lds + kf_swz(dd, r * 2, A8B_KT * 2)) = (unsigned short)vv8[el];
-        }
-      }
-    }
-    if (kt < last_kt) issue_k_tile(kt + 1, (kt + 1) & 1);
-    // my K(kt) loads done (2 newest = kt+1 may stay in flight); V retired by
-    // register use; my vt scatter writes complete:
-    if (kt < last_kt)
-      asm volatile("s_waitcnt vmcnt(2) lgkmcnt(0)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
-    __builtin_amdgcn_s_barrier();  // publish vt(kt) + everyone's K(kt)
-
-    const unsigned char* kbuf = k_lds[kt & 1];
-#pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
-      kf_f32x16 st = kf_f32x16{0.f};
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        kf_bf16x8 afrag = *reinterpret_cast<const kf_bf16x8*>(
-            kbuf + kf_swz(mt * 32 + l31, kk * 32 + hi * 16, AT_D * 2));
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, qfrag[kk], st,
-                                                     0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
-
-      const int kv0 = kt * A8B_KT + mt * 32 + hi * 4;
-      float mx = -INFINITY;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kv = kv0 + (r & 3) + 8 * (r >> 2);
-        float sv = st[r] * scale;
-        if (causal && kv > qrow_g) sv = -INFINITY;
-        st[r] = sv;
-        mx = fmaxf(mx, sv);
-      }
-      mx = fmaxf(mx, __shfl_xor(mx, 32, KF_WAVE));
-      const float m_new = fmaxf(m_run, mx);
-      const float alpha =
-          (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
-      float lsum = 0.f;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float pv =
-            (st[r] == -INFINITY) ? 0.f : __expf(st[r] - m_new);
-        st[r] = pv;
-        lsum += pv;
-      }
-      lsum += __shfl_xor(lsum, 32, KF_WAVE);
-      l_run = l_run * alpha + lsum;
-      m_run = m_new;
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) oacc[i][r] *= alpha;
-
-      kf_bf16x8 pb[2];
-#pragma unroll
-      for (int step = 0; step < 2; ++step) {
-        const int base = step * 8;
-        unsigned int w0 = kf_cvt_pk_bf16(st[base + 0], st[base + 1]);
-        unsigned int w1 = kf_cvt_pk_bf16(st[base + 2], st[base + 3]);
-        unsigned int w2 = kf_cvt_pk_bf16(st[base + 4], st[base + 5]);
-        unsigned int w3 = kf_cvt_pk_bf16(st[base + 6], st[base + 7]);
-        auto s02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
-        auto s13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
-        unsigned int u[4] = {(unsigned)s02[0], (unsigned)s13[0],
-                             (unsigned)s02[1], (unsigned)s13[1]};
-        pb[step] = *reinterpret_cast<kf_bf16x8*>(u);
-      }
-
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-#pragma unroll
-        for (int step = 0; step < 2; ++step) {
-          kf_bf16x8 vfrag = *reinterpret_cast<const kf_bf16x8*>(
-              vt_lds + kf_swz(dt * 32 + l31,
-                              mt * 64 + step * 32 + hi * 16, A8B_KT * 2));
-          oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag, pb[step],
-                                                             oacc[dt], 0, 0, 0);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
-    }
-    // end barrier: all waves done reading vt(kt)/kbuf before next overwrite
-    __builtin_amdgcn_s_barrier();
-  }
-
-  const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
-  const int64_t obase = ((b * S + qrow_g) * (int64_t)Hq + hq) * AT_D;
-#pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int d = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      o[obase + d] = kf_f32_to_bf16(oacc[dt][r] * inv_l);
-    }
-  if (hi == 0)
-    lse[(b * Hq + hq) * (int64_t)S + qrow_g] = m_run + __logf(l_run);
-}
-
 KF_EXPORT int kf_attn_fwd(void* o, float* lse, const void* q, const void* k,
                           const void* v, int64_t B, int64_t S, int64_t Hq,
                           int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
@@ -620,22 +428,8 @@ KF_EXPORT int kf_attn_fwd(void* o, float* lse, const void* q, const void* k,
   if (kts == 0) kts = Hkv * AT_D;
   if (S % A8_QT == 0) {  // 8-wave swapped kernel for the training shapes
     dim3 grid((unsigned)(S / A8_QT), (unsigned)Hq, (unsigned)B);
-    static int use_async = -1;
-    if (use_async < 0) {
-      const char* e = getenv("KF_ATTN_FWD_ASYNC");
-      // A/B on MI355X: async variant is within noise of the simple one
-      // (230.6 vs 231.6 TF) — single-stage K prefetch is not this kernel's
-      // bottleneck. Default OFF; kept for round-2 pipeline work.
-      use_async = (e != nullptr && e[0] == '1') ? 1 : 0;
-    }
-    if (use_async) {
-      hipLaunchKernelGGL(kf_attn_fwd8b_kernel, grid, dim3(A8_THREADS), 0,
-                         (hipStream_t)stream, (unsigned short*)o, lse,
-                         (const unsigned short*)q, (const unsigned short*)k,
-                         (const unsigned short*)v, B, (int)S, (int)Hq,
-                         (int)Hkv, qts, kts, scale, causal);
-      return (int)hipGetLastError();
-    }
+    // (an async global_load_lds double-buffer variant was measured NULL
+    // here — see docs/ROUND1.md negative results — and removed)
     hipLaunchKernelGGL(kf_attn_fwd8_kernel, grid, dim3(A8_THREADS), 0,
                        (hipStream_t)stream, (unsigned short*)o, lse,
                        (const unsigned short*)q, (const unsigned short*)k,
