@@ -120,9 +120,9 @@ class InferenceEngine:
         return req
 
     def generate(self, prompt: List[int], max_new_tokens: int = 32,
-                 timeout: float = 120.0) -> Request:
+                 timeout: float = 120.0, temperature: float = 0.0) -> Request:
         req = Request(rid=f"r{time.monotonic_ns()}", prompt=list(prompt),
-                      max_new_tokens=max_new_tokens)
+                      max_new_tokens=max_new_tokens, temperature=temperature)
         self.submit(req)
         if not req.done.wait(timeout):
             req.error = req.error or "timeout"
@@ -242,7 +242,7 @@ class InferenceEngine:
             x = x + layer.w2(y)
         x = self.model.final_norm(x)
         logits = F.linear(x, self.model.lm_head.weight)  # [N,1,V]
-        return logits[:, -1].argmax(-1)  # [N] int64
+        return logits[:, -1]  # [N,V] (sampling happens outside the graph)
 
     def _graph_for(self, bucket: int):
         """Capture (once) and return the decode graph for a batch bucket."""
@@ -265,7 +265,7 @@ class InferenceEngine:
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            static["next"] = self._decode_forward(
+            static["logits"] = self._decode_forward(
                 static["tokens"], static["positions"], static["slots"],
                 static["lens"])
         self._graphs[bucket] = (graph, static)
@@ -302,10 +302,11 @@ class InferenceEngine:
                 static["lens"][N:].fill_(1)
             graph.replay()
             self.stats["graph_replays"] += 1
-            toks = static["next"][:N].tolist()
+            logits = static["logits"][:N]
+            toks = self._sample_batch(logits, acts)
         else:
-            toks = self._decode_forward(tokens, positions, slots,
-                                        lens).tolist()
+            logits = self._decode_forward(tokens, positions, slots, lens)
+            toks = self._sample_batch(logits, acts)
         still = []
         now = time.time()
         for i, r in enumerate(acts):
@@ -328,3 +329,18 @@ class InferenceEngine:
             return int(logits.argmax().item())
         probs = torch.softmax(logits.float() / temperature, dim=-1)
         return int(torch.multinomial(probs, 1).item())
+
+    @staticmethod
+    def _sample_batch(logits: torch.Tensor, acts) -> list:
+        """Per-request temperature sampling on [N,V] logits (outside the
+        captured graph so RNG stays ordinary)."""
+        temps = [r.temperature for r in acts]
+        if all(t <= 0 for t in temps):
+            return logits.argmax(-1).tolist()
+        t = torch.tensor([max(tt, 1e-6) if tt > 0 else 1.0 for tt in temps],
+                         device=logits.device).unsqueeze(1)
+        probs = torch.softmax(logits.float() / t, dim=-1)
+        sampled = torch.multinomial(probs, 1).squeeze(1)
+        greedy = logits.argmax(-1)
+        pick = torch.tensor([tt > 0 for tt in temps], device=logits.device)
+        return torch.where(pick, sampled, greedy).tolist()

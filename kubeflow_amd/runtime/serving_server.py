@@ -56,7 +56,9 @@ def build_app(spec: dict) -> FastAPI:
         preds = []
         for inst in instances:
             r = engine.generate(inst.get("prompt_tokens", [1]),
-                                int(inst.get("max_new_tokens", 16)))
+                                int(inst.get("max_new_tokens", 16)),
+                                temperature=float(
+                                    inst.get("temperature", 0.0)))
             preds.append({
                 "tokens": r.generated,
                 "error": r.error,
@@ -71,7 +73,8 @@ def build_app(spec: dict) -> FastAPI:
     async def generate(req: HttpRequest):
         body = await req.json()
         r = engine.generate(body.get("prompt_tokens", [1]),
-                            int(body.get("max_new_tokens", 16)))
+                            int(body.get("max_new_tokens", 16)),
+                            temperature=float(body.get("temperature", 0.0)))
         return {"tokens": r.generated, "error": r.error}
 
     @app.get("/metrics")

@@ -97,3 +97,17 @@ def test_inference_service_e2e(tmp_path):
         assert len(out["predictions"]) == 1
         assert len(out["predictions"][0]["tokens"]) == 4
         assert out["predictions"][0]["latency_ms"] is not None
+
+
+def test_sampling_temperature_cpu():
+    eng = InferenceEngine("llama-tiny", max_slots=2, smax=256,
+                          max_batch=2).start()
+    try:
+        torch.manual_seed(0)
+        r0 = eng.generate([1, 2, 3], max_new_tokens=6, temperature=0.0)
+        r1 = eng.generate([1, 2, 3], max_new_tokens=6, temperature=0.0)
+        assert r0.generated == r1.generated  # greedy deterministic
+        r2 = eng.generate([1, 2, 3], max_new_tokens=20, temperature=5.0)
+        assert len(r2.generated) == 20 and r2.error == ""
+    finally:
+        eng.stop()
