@@ -35,6 +35,9 @@ def write_status(rank_dir: str, state: str, step: int = 0, loss=None,
     with open(tmp, "w") as f:
         json.dump(payload, f)
     os.replace(tmp, os.path.join(rank_dir, "status.json"))
+    if metrics:  # history series for the tensorboard viewer
+        with open(os.path.join(rank_dir, "metrics.jsonl"), "a") as f:
+            f.write(json.dumps({"step": step, "metrics": metrics}) + "\n")
 
 
 def synthetic_batch(spec: dict, cfg, device, rank: int, step: int):

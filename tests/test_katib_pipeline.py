@@ -125,3 +125,27 @@ def test_pipeline_invalid_dag(tmp_path):
         state, obj = _wait(plat.store, "PipelineRun", "bad", "default",
                            timeout=30)
         assert state == "Failed"
+
+
+def test_experiment_goal_early_stop(tmp_path):
+    """Objective goal reached -> Succeeded with GoalReached, no extra trials
+    beyond the running ones."""
+    with Platform(root_dir=str(tmp_path)) as plat:
+        exp = new_object("Experiment", "goal-exp", "default", spec={
+            "objective": {"type": "minimize", "objectiveMetricName": "loss",
+                          "goal": 1e9},  # any finite loss satisfies it
+            "algorithm": {"algorithmName": "random"},
+            "parallelTrialCount": 1,
+            "maxTrialCount": 10,
+            "parameters": PARAMS,
+            "trialTemplate": {"model": "mnist-mlp", "steps": 3,
+                              "gpus_per_replica": 0, "status_every": 1,
+                              "save_final": False, "replicas": 1},
+        })
+        plat.store.create(exp)
+        state, obj = _wait(plat.store, "Experiment", "goal-exp", "default",
+                           timeout=180)
+        assert state == "Succeeded"
+        conds = {c["type"]: c for c in obj["status"]["conditions"]}
+        assert conds["Succeeded"]["reason"] == "GoalReached"
+        assert obj["status"]["trials"] <= 2  # stopped early, not 10
