@@ -389,6 +389,7 @@ def build_app(store: ObjectStore, scheduler=None,
     @app.get("/api/activities/{ns}")
     def activities(ns: str, request: Request):
         me = user_of(request)
+        authz(me, ns, "get")
         evs = store.list("Event", ns)
         evs.sort(key=lambda e: e.get("lastTimestamp", ""), reverse=True)
         return ok(me, activities=evs[:100])

@@ -105,8 +105,9 @@ class NotebookReconciler(Reconciler):
         # culling (reference: NotebookNeedsCulling, requeue each period)
         if self.enable_culling and ready and last_activity:
             try:
-                last = time.mktime(time.strptime(last_activity,
-                                                 "%Y-%m-%dT%H:%M:%SZ"))
+                import calendar
+                last = calendar.timegm(time.strptime(last_activity,
+                                                     "%Y-%m-%dT%H:%M:%SZ"))
                 if time.time() - last > self.idle_minutes * 60:
                     nb["metadata"]["annotations"][STOP_ANNOTATION] = \
                         time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
