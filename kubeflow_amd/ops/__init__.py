@@ -98,9 +98,7 @@ class _RMSNormHip(torch.autograd.Function):
         dy2 = dy.contiguous().view(rows, cols)
         dx = torch.empty_like(x2)
         dw = torch.empty_like(weight)
-        nparts = lib.kf_rmsnorm_bwd_nparts(rows)
-        dw_part = torch.empty(nparts * cols, dtype=torch.float32,
-                              device=x2.device)
+        dw_part = torch.zeros(cols, dtype=torch.float32, device=x2.device)
         _backend.check(
             lib.kf_rmsnorm_bwd(_p(dx), _p(dw), _fp(dw_part), _p(dy2), _p(x2),
                                _p(weight), _fp(rstd), rows, cols, _stream()),
@@ -392,9 +390,7 @@ class _LayerNormHip(torch.autograd.Function):
         dx = torch.empty_like(x2)
         dw = torch.empty_like(weight)
         db = torch.empty_like(weight)
-        nparts = lib.kf_layernorm_bwd_nparts(rows)
-        part = torch.empty(2 * nparts * cols, dtype=torch.float32,
-                           device=x2.device)
+        part = torch.zeros(2 * cols, dtype=torch.float32, device=x2.device)
         _backend.check(
             lib.kf_layernorm_bwd(_p(dx), _p(dw), _p(db), _fp(part), _p(dy2),
                                  _p(x2), _p(weight), _fp(mu), _fp(rstd), rows,

@@ -116,17 +116,14 @@ __global__ __launch_bounds__(LN_BLOCK) void kf_layernorm_bwd_kernel(
     }
     __syncthreads();
   }
-  float* outw = dw_part + (int64_t)blockIdx.x * cols;
-  float* outb = db_part + (int64_t)blockIdx.x * cols;
   for (int i = threadIdx.x; i < cols; i += LN_BLOCK) {
-    outw[i] = dw_lds[i];
-    outb[i] = db_lds[i];
+    atomicAdd(&dw_part[i], dw_lds[i]);
+    atomicAdd(&db_part[i], db_lds[i]);
   }
 }
 
-// reuses kf_colsum_bf16_kernel from rmsnorm.hip via separate launches
-extern __global__ void kf_colsum_bf16_kernel(unsigned short*, const float*,
-                                             int, int);
+extern __global__ void kf_cast_bf16_kernel(unsigned short*, const float*,
+                                           int64_t);
 
 KF_EXPORT int kf_layernorm_fwd(void* y, float* mu, float* rstd, const void* x,
                                const void* w, const void* b, int64_t rows,
@@ -141,31 +138,33 @@ KF_EXPORT int kf_layernorm_fwd(void* y, float* mu, float* rstd, const void* x,
 }
 
 KF_EXPORT int64_t kf_layernorm_bwd_nparts(int64_t rows) {
-  int64_t g = rows < 1024 ? rows : 1024;
-  return g < 1 ? 1 : g;
+  (void)rows;
+  return 1;  // accumulators are [2][cols] now (kept for ABI compat)
 }
 
+// part must be a ZEROED fp32 buffer of 2*cols floats (dw ++ db).
 KF_EXPORT int kf_layernorm_bwd(void* dx, void* dw, void* db, float* part,
                                const void* dy, const void* x, const void* w,
                                const float* mu, const float* rstd,
                                int64_t rows, int64_t cols, void* stream) {
   if (cols % LN_VEC) return (int)hipErrorInvalidValue;
-  const int grid = (int)kf_layernorm_bwd_nparts(rows);
-  float* dw_part = part;                       // [grid][cols]
-  float* db_part = part + (int64_t)grid * cols;  // [grid][cols]
+  int grid = (int)(rows < 1024 ? rows : 1024);
+  if (grid < 1) grid = 1;
+  float* dw_acc = part;
+  float* db_acc = part + cols;
   size_t lds = (2 * cols + LN_BLOCK / KF_WAVE) * sizeof(float);
   hipLaunchKernelGGL(kf_layernorm_bwd_kernel, dim3(grid), dim3(LN_BLOCK), lds,
-                     (hipStream_t)stream, (unsigned short*)dx, dw_part,
-                     db_part, (const unsigned short*)dy,
+                     (hipStream_t)stream, (unsigned short*)dx, dw_acc,
+                     db_acc, (const unsigned short*)dy,
                      (const unsigned short*)x, (const unsigned short*)w, mu,
                      rstd, rows, (int)cols);
   int err = (int)hipGetLastError();
   if (err) return err;
-  hipLaunchKernelGGL(kf_colsum_bf16_kernel, dim3(kf_grid_for(cols, 256)),
+  hipLaunchKernelGGL(kf_cast_bf16_kernel, dim3(kf_grid_for(cols, 256)),
                      dim3(256), 0, (hipStream_t)stream, (unsigned short*)dw,
-                     dw_part, grid, (int)cols);
-  hipLaunchKernelGGL(kf_colsum_bf16_kernel, dim3(kf_grid_for(cols, 256)),
+                     dw_acc, cols);
+  hipLaunchKernelGGL(kf_cast_bf16_kernel, dim3(kf_grid_for(cols, 256)),
                      dim3(256), 0, (hipStream_t)stream, (unsigned short*)db,
-                     db_part, grid, (int)cols);
+                     db_acc, cols);
   return (int)hipGetLastError();
 }

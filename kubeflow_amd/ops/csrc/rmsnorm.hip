@@ -9,8 +9,8 @@
 //
 // Backward (per row, s = sum(dy*w*x)):
 //   dx = rstd * dy*w - rstd^3/C * s * x
-//   dw_j = sum_rows(dy_j * x_j * rstd)   — accumulated as fp32 partials per
-//   block, reduced by a second kernel (deterministic, no atomics).
+//   dw_j = sum_rows(dy_j * x_j * rstd)   — block-local LDS accumulation,
+//   one fp32 atomicAdd pass per block, then a cast kernel.
 
 #include "kf_common.h"
 
@@ -53,9 +53,11 @@ __global__ __launch_bounds__(RN_BLOCK) void kf_rmsnorm_fwd_kernel(
   }
 }
 
-// dw partials: dw_part has shape [gridDim.x][cols] fp32. Each block keeps its
-// running dw in LDS (cols * 4 bytes; cols <= 8192 fits the 160 KiB LDS at
-// block=256 with margin for hidden sizes used here: 4096 -> 16 KiB).
+// dw: each block accumulates its rows' contribution in LDS (cols * 4 B;
+// hidden 4096 -> 16 KiB) and atomicAdds it once into a zeroed fp32 buffer
+// (guide G12: block-level pre-reduction, one atomic pass per block — the
+// earlier partials+column-sum design ran the reduction on cols/256 = 16
+// blocks, 6% of the chip, ~16 ms/step in the r01 profile).
 __global__ __launch_bounds__(RN_BLOCK) void kf_rmsnorm_bwd_kernel(
     unsigned short* __restrict__ dx, float* __restrict__ dw_part,
     const unsigned short* __restrict__ dy, const unsigned short* __restrict__ x,
@@ -105,20 +107,17 @@ __global__ __launch_bounds__(RN_BLOCK) void kf_rmsnorm_bwd_kernel(
     }
     __syncthreads();  // dw_lds writes of this row done before next row reuse
   }
-  float* out = dw_part + (int64_t)blockIdx.x * cols;
-  for (int i = threadIdx.x; i < cols; i += RN_BLOCK) out[i] = dw_lds[i];
+  for (int i = threadIdx.x; i < cols; i += RN_BLOCK)
+    atomicAdd(&dw_part[i], dw_lds[i]);
 }
 
-// Reduce [npart][cols] fp32 partials into bf16 dw.
-__global__ void kf_colsum_bf16_kernel(unsigned short* __restrict__ dw,
-                                      const float* __restrict__ part,
-                                      int npart, int cols) {
-  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
-       c += gridDim.x * blockDim.x) {
-    float s = 0.f;
-    for (int p = 0; p < npart; ++p) s += part[(int64_t)p * cols + c];
-    dw[c] = kf_f32_to_bf16(s);
-  }
+// fp32 accumulator -> bf16 output.
+__global__ void kf_cast_bf16_kernel(unsigned short* __restrict__ dst,
+                                    const float* __restrict__ src,
+                                    int64_t n) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * (int64_t)blockDim.x)
+    dst[i] = kf_f32_to_bf16(src[i]);
 }
 
 KF_EXPORT int kf_rmsnorm_fwd(void* y, float* rstd, const void* x, const void* w,
@@ -133,27 +132,28 @@ KF_EXPORT int kf_rmsnorm_fwd(void* y, float* rstd, const void* x, const void* w,
   return (int)hipGetLastError();
 }
 
-// dw_part must hold kf_rmsnorm_bwd_nparts(rows) * cols floats.
+// dw_acc must be a ZEROED fp32 buffer of `cols` floats.
 KF_EXPORT int64_t kf_rmsnorm_bwd_nparts(int64_t rows) {
-  int64_t g = rows < 1024 ? rows : 1024;
-  return g < 1 ? 1 : g;
+  (void)rows;
+  return 1;  // accumulator is [cols] now (kept for ABI compat)
 }
 
-KF_EXPORT int kf_rmsnorm_bwd(void* dx, void* dw, float* dw_part,
+KF_EXPORT int kf_rmsnorm_bwd(void* dx, void* dw, float* dw_acc,
                              const void* dy, const void* x, const void* w,
                              const float* rstd, int64_t rows, int64_t cols,
                              void* stream) {
   if (cols % RN_VEC) return (int)hipErrorInvalidValue;
-  int grid = (int)kf_rmsnorm_bwd_nparts(rows);
+  int grid = (int)(rows < 1024 ? rows : 1024);
+  if (grid < 1) grid = 1;
   size_t lds = (cols + RN_BLOCK / KF_WAVE) * sizeof(float);
   hipLaunchKernelGGL(kf_rmsnorm_bwd_kernel, dim3(grid), dim3(RN_BLOCK), lds,
-                     (hipStream_t)stream, (unsigned short*)dx, dw_part,
+                     (hipStream_t)stream, (unsigned short*)dx, dw_acc,
                      (const unsigned short*)dy, (const unsigned short*)x,
                      (const unsigned short*)w, rstd, rows, (int)cols);
   int err = (int)hipGetLastError();
   if (err) return err;
-  hipLaunchKernelGGL(kf_colsum_bf16_kernel, dim3(kf_grid_for(cols, 256)),
+  hipLaunchKernelGGL(kf_cast_bf16_kernel, dim3(kf_grid_for(cols, 256)),
                      dim3(256), 0, (hipStream_t)stream, (unsigned short*)dw,
-                     dw_part, grid, (int)cols);
+                     dw_acc, cols);
   return (int)hipGetLastError();
 }
