@@ -49,6 +49,10 @@ class KVCache:
         with self._lock:
             return self._free.pop() if self._free else None
 
+    def has_free(self) -> bool:
+        with self._lock:
+            return bool(self._free)
+
     def free(self, slot: int):
         with self._lock:
             self._free.append(slot)
@@ -132,19 +136,19 @@ class InferenceEngine:
     @torch.no_grad()
     def _loop(self):
         while not self._stop:
-            admitted = self._admit()
+            self._admit()
             if not self.active:
-                try:
-                    req = self.pending.get(timeout=0.01)
-                    self._start_request(req)
-                except queue.Empty:
-                    continue
+                time.sleep(0.002)  # queued-but-unadmittable or truly idle
                 continue
             self._decode_step()
 
     def _admit(self) -> int:
         n = 0
         while len(self.active) < self.max_batch:
+            if not self.cache.has_free():
+                break  # no KV slot: leave requests queued — decode frees
+                # slots as sequences finish (an unconditional loop here
+                # re-took requeued requests forever and starved decode)
             try:
                 req = self.pending.get_nowait()
             except queue.Empty:

@@ -111,3 +111,45 @@ def test_sampling_temperature_cpu():
         assert len(r2.generated) == 20 and r2.error == ""
     finally:
         eng.stop()
+
+
+def test_slot_exhaustion_queues_and_completes():
+    """More concurrent requests than KV slots: the overflow waits in the
+    queue and completes when slots free (continuous-batching admission)."""
+    import threading
+    eng = InferenceEngine("llama-tiny", max_slots=2, smax=128,
+                          max_batch=8).start()
+    try:
+        results = [None] * 6
+        def run(i):
+            results[i] = eng.generate([1 + i, 2, 3], max_new_tokens=4,
+                                      timeout=180)
+        ts = [threading.Thread(target=run, args=(i,)) for i in range(6)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(200)
+        for i, r in enumerate(results):
+            assert r is not None and r.error == "", (i, r and r.error)
+            assert len(r.generated) == 4
+        assert eng.stats["completed"] == 6
+        # pad slot (if GPU graphs) excluded; on CPU all slots returned free
+        assert len(eng.cache._free) == eng.cache.slots - (
+            1 if eng._pad_slot is not None else 0)
+    finally:
+        eng.stop()
+
+
+def test_context_length_cap():
+    """A request that would overrun the slot capacity stops at smax."""
+    eng = InferenceEngine("llama-tiny", max_slots=2, smax=64,
+                          max_batch=2).start()
+    try:
+        r = eng.generate(list(range(1, 60)), max_new_tokens=50, timeout=120)
+        assert r.error == ""
+        # prompt is truncated to smax - max_new - 1 and generation stops
+        # before exceeding the slot
+        assert len(r.generated) <= 50
+        assert r.pos < 64
+    finally:
+        eng.stop()
