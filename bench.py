@@ -37,7 +37,9 @@ def main():
     ap.add_argument("--seq-len", type=int,
                     default=int(os.environ.get("KF_BENCH_SEQ", "4096")))
     ap.add_argument("--micro-batch", type=int,
-                    default=int(os.environ.get("KF_BENCH_MB", "4")))
+                    default=int(os.environ.get("KF_BENCH_MB", "6")))
+    # mb sweep on MI355X (profiles/r01_mb_sweep.log): mb4 16.4k, mb6 16.8k
+    # tok/s, mb8 OOM (280 GB activations) — 6 is the per-GPU sweet spot.
     args = ap.parse_args()
 
     rank, world, device = kdist.init_distributed()
