@@ -109,6 +109,10 @@ def main():
             },
         }
         print(json.dumps(out), flush=True)
+    if os.environ.get("KF_BENCH_MEM") and device.type == "cuda":
+        peak = torch.cuda.max_memory_allocated(device) / (1 << 30)
+        print(f"# rank {rank} peak allocated {peak:.1f} GiB", file=sys.stderr,
+              flush=True)
 
 
 if __name__ == "__main__":
