@@ -6,7 +6,8 @@ from __future__ import annotations
 
 from typing import Callable, Dict
 
-from .llama import (LlamaConfig, LlamaModel, llama3_8b, llama3_1b, llama_tiny)
+from .llama import (LlamaConfig, LlamaModel, llama3_8b, llama3_1b,
+                    llama_tiny, llama_tiny_mha)
 from .bert import (BertConfig, BertClassifier, bert_base, bert_base_hd128,
                    bert_tiny)
 from .mlp import MnistMLP
@@ -22,7 +23,7 @@ def _llama(cfg_fn):
     def build(device=None, dtype=None, **kw):
         import torch
         return LlamaModel(cfg_fn(), device=device,
-                          dtype=dtype or torch.bfloat16)
+                          dtype=dtype or torch.bfloat16, tp=kw.get("tp"))
     return build
 
 
@@ -45,6 +46,7 @@ MODEL_REGISTRY: Dict[str, Callable] = {
     "llama3-8b": _llama(llama3_8b),
     "llama3-1b": _llama(llama3_1b),
     "llama-tiny": _llama(llama_tiny),
+    "llama-tiny-mha": _llama(llama_tiny_mha),
     "bert-base": _bert(bert_base),
     "bert-base-hd128": _bert(bert_base_hd128),
     "bert-tiny": _bert(bert_tiny),
@@ -55,4 +57,7 @@ MODEL_REGISTRY: Dict[str, Callable] = {
 def build_model(name: str, device=None, dtype=None, **kw):
     if name not in MODEL_REGISTRY:
         raise KeyError(f"unknown model {name!r}; known: {sorted(MODEL_REGISTRY)}")
+    if kw.get("tp") is not None and not name.startswith("llama"):
+        raise ValueError(f"tensor parallelism is implemented for the llama "
+                         f"family only, not {name!r}")
     return MODEL_REGISTRY[name](device=device, dtype=dtype, **kw)

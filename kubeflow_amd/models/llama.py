@@ -23,6 +23,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from kubeflow_amd import ops
+from kubeflow_amd.parallel import tp as tpmod
 
 
 @dataclass
@@ -66,6 +67,14 @@ def llama_tiny(vocab: int = 512) -> LlamaConfig:
                        max_seq_len=512)
 
 
+def llama_tiny_mha(vocab: int = 512) -> LlamaConfig:
+    """Tiny MHA config (kv == q heads) whose head/ffn counts divide by 2 —
+    used by the tensor-parallel tests and examples."""
+    return LlamaConfig(name="llama-tiny-mha", vocab_size=vocab,
+                       hidden_size=256, n_layers=2, n_heads=2, n_kv_heads=2,
+                       ffn_dim=512, max_seq_len=512)
+
+
 class RMSNorm(nn.Module):
     def __init__(self, dim: int, eps: float):
         super().__init__()
@@ -77,36 +86,62 @@ class RMSNorm(nn.Module):
 
 
 class LlamaBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp=None):
         super().__init__()
         h, d = cfg.hidden_size, cfg.head_dim
         self.cfg = cfg
+        self.tp = tp if (tp is not None and tp.world > 1) else None
+        n = self.tp.world if self.tp else 1
+        if cfg.n_heads % n or cfg.n_kv_heads % n or cfg.ffn_dim % n:
+            raise ValueError(f"TP degree {n} must divide heads "
+                             f"({cfg.n_heads}/{cfg.n_kv_heads}) and ffn "
+                             f"({cfg.ffn_dim})")
+        # local (per-TP-rank) shard sizes; n=1 reproduces the full model
+        self.hq, self.hkv = cfg.n_heads // n, cfg.n_kv_heads // n
+        self.ffn = cfg.ffn_dim // n
         self.attn_norm = RMSNorm(h, cfg.norm_eps)
-        self.wqkv = nn.Linear(h, (cfg.n_heads + 2 * cfg.n_kv_heads) * d,
-                              bias=False)
-        self.wo = nn.Linear(cfg.n_heads * d, h, bias=False)
+        self.wqkv = nn.Linear(h, (self.hq + 2 * self.hkv) * d, bias=False)
+        self.wo = nn.Linear(self.hq * d, h, bias=False)
         self.mlp_norm = RMSNorm(h, cfg.norm_eps)
-        self.w13 = nn.Linear(h, 2 * cfg.ffn_dim, bias=False)  # gate ++ up
-        self.w2 = nn.Linear(cfg.ffn_dim, h, bias=False)
+        self.w13 = nn.Linear(h, 2 * self.ffn, bias=False)  # gate ++ up
+        self.w2 = nn.Linear(self.ffn, h, bias=False)
+        if self.tp:
+            for lin in (self.wqkv, self.wo, self.w13, self.w2):
+                lin.weight._tp_sharded = True
 
     def forward(self, x, cos, sin, pos_offset: int = 0, kv_cache=None):
         cfg = self.cfg
         B, S, h = x.shape
-        qkv = F.linear(self.attn_norm(x), self.wqkv.weight)
+        nx = self.attn_norm(x)
+        if self.tp:
+            nx = tpmod.copy_to(nx, self.tp)
+        qkv = F.linear(nx, self.wqkv.weight)
         if kv_cache is not None:
-            q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
-                                 cfg.n_kv_heads * cfg.head_dim,
-                                 cfg.n_kv_heads * cfg.head_dim], dim=-1)
-            q = q.view(B, S, cfg.n_heads, cfg.head_dim)
-            k = k.view(B, S, cfg.n_kv_heads, cfg.head_dim)
-            v = v.view(B, S, cfg.n_kv_heads, cfg.head_dim)
+            q, k, v = qkv.split([self.hq * cfg.head_dim,
+                                 self.hkv * cfg.head_dim,
+                                 self.hkv * cfg.head_dim], dim=-1)
+            q = q.view(B, S, self.hq, cfg.head_dim)
+            k = k.view(B, S, self.hkv, cfg.head_dim)
+            v = v.view(B, S, self.hkv, cfg.head_dim)
             q, k = ops.rope(q, k, cos, sin, pos_offset)
             k, v = kv_cache.update(k, v, pos_offset)
             o = ops.flash_attention(q, k, v, causal=(S > 1))
-            o = o.reshape(B, S, cfg.n_heads * cfg.head_dim)
+            o = o.reshape(B, S, self.hq * cfg.head_dim)
         else:
-            o = ops.fused_qkv_attention(qkv, cos, sin, cfg.n_heads,
-                                        cfg.n_kv_heads, cfg.head_dim)
+            o = ops.fused_qkv_attention(qkv, cos, sin, self.hq,
+                                        self.hkv, cfg.head_dim)
+        if self.tp:
+            # row-parallel wo: reduce the partial BEFORE the residual add
+            # (the addmm fusion below would add the residual tp.world times)
+            attn = tpmod.reduce_from(
+                torch.mm(o.view(B * S, -1), self.wo.weight.t()), self.tp)
+            x = (x.view(-1, h) + attn).view(B, S, h)
+            ny = tpmod.copy_to(self.mlp_norm(x), self.tp)
+            y = ops.swiglu(F.linear(ny, self.w13.weight))
+            mlp = tpmod.reduce_from(
+                torch.mm(y.view(B * S, self.ffn), self.w2.weight.t()),
+                self.tp)
+            return (x.view(-1, h) + mlp).view(B, S, h)
         # residuals fused into the GEMM epilogue (addmm: C = input + A @ B)
         x = torch.addmm(x.view(-1, h), o.view(B * S, -1),
                         self.wo.weight.t()).view(B, S, h)
@@ -116,14 +151,16 @@ class LlamaBlock(nn.Module):
 
 
 class LlamaModel(nn.Module):
-    def __init__(self, cfg: LlamaConfig, device=None, dtype=torch.bfloat16):
+    def __init__(self, cfg: LlamaConfig, device=None, dtype=torch.bfloat16,
+                 tp=None):
         super().__init__()
         self.cfg = cfg
+        self.tp = tp
         factory = dict(device=device, dtype=dtype)
         with torch.device(device if device is not None else "cpu"):
             self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
             self.layers = nn.ModuleList(
-                [LlamaBlock(cfg) for _ in range(cfg.n_layers)])
+                [LlamaBlock(cfg, tp=tp) for _ in range(cfg.n_layers)])
             self.final_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
             self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
                                      bias=False)

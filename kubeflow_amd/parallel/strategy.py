@@ -19,14 +19,15 @@ from enum import Enum
 
 class Strategy(str, Enum):
     DDP = "ddp"      # implemented: bucketed all-reduce over RCCL/xGMI
-    TP = "tp"        # reserved: tensor parallel (xGMI all-reduce per layer)
+    TP = "tp"        # implemented: Megatron-style head/ffn sharding (tp.py);
+                     # pure TP (degree == world_size) in v1
     PP = "pp"        # reserved: pipeline parallel
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
     EP = "ep"        # reserved: expert parallel (all-to-all)
     ULYSSES = "ulysses"  # reserved: attention head-scatter SP
 
 
-IMPLEMENTED = {Strategy.DDP}
+IMPLEMENTED = {Strategy.DDP, Strategy.TP}
 
 
 @dataclass

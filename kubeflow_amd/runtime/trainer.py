@@ -34,14 +34,31 @@ class TrainConfig:
     grad_clip: float = 1.0  # 0 disables
 
 
+class _NullDDP:
+    """Comm no-op stand-in when the data-parallel degree is 1 (pure TP)."""
+
+    def prepare_step(self):
+        pass
+
+    def finalize(self):
+        pass
+
+
 class Trainer:
     """Owns the flat parameter space, DDP overlap and the fused optimizer."""
 
-    def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig()):
+    def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig(),
+                 tp_ctx=None):
         self.model = model
         self.cfg = cfg
         self.flat = FlatParamSpace(model)
-        self.ddp = BucketedDDP(self.flat, bucket_mb=cfg.bucket_mb)
+        self.tp = tp_ctx
+        if tp_ctx is not None:
+            # pure TP: ranks hold distinct shards — no grad all-reduce; the
+            # two per-block TP all-reduces are the only comm per step.
+            self.ddp = _NullDDP()
+        else:
+            self.ddp = BucketedDDP(self.flat, bucket_mb=cfg.bucket_mb)
         dev = self.flat.device
         self.p32 = self.flat.data.float()
         self.m = torch.zeros_like(self.p32)
@@ -85,8 +102,11 @@ class Trainer:
         self.ddp.finalize()
         if cfg.grad_clip > 0:
             # one flat-buffer norm; scale in place when above the clip
-            gnorm = torch.linalg.vector_norm(self.flat.grad,
-                                             dtype=torch.float32)
+            if self.tp is not None:
+                gnorm = self.tp.global_grad_norm(self.flat)
+            else:
+                gnorm = torch.linalg.vector_norm(self.flat.grad,
+                                                 dtype=torch.float32)
             scale = (cfg.grad_clip / (gnorm + 1e-6)).clamp(max=1.0)
             self.flat.grad.mul_(scale.to(self.flat.grad.dtype))
             self.last_grad_norm = gnorm

@@ -85,13 +85,24 @@ def main(argv=None):
     signal.signal(signal.SIGTERM, lambda *_: stop.update(flag=True))
 
     try:
-        from kubeflow_amd.parallel.strategy import ParallelismSpec
-        ParallelismSpec.from_spec(spec)  # DDP-only v1; loud on reserved ones
+        from kubeflow_amd.parallel.strategy import ParallelismSpec, Strategy
+        pspec = ParallelismSpec.from_spec(spec)  # loud on reserved ones
         rank, world, device = kdist.init_distributed()
         torch.manual_seed(int(spec.get("seed", 0)) + rank)
         dtype = (torch.bfloat16 if device.type == "cuda"
                  and spec.get("dtype", "bf16") == "bf16" else torch.float32)
-        model = build_model(spec["model"], device=device, dtype=dtype)
+        tp_ctx = None
+        if pspec.strategy == Strategy.TP and world > 1:
+            from kubeflow_amd.parallel.tp import TpContext
+            if pspec.degree not in (1, world):
+                raise ValueError(
+                    f"pure TP requires degree == world_size ({world}); "
+                    f"got {pspec.degree} — TPxDP meshes are a v2 seam")
+            tp_ctx = TpContext.from_group(None)  # TP group = WORLD
+        model = build_model(spec["model"], device=device, dtype=dtype,
+                            tp=tp_ctx)
+        if tp_ctx is not None:
+            tp_ctx.sync_replicated(model)
         cfg = getattr(model, "cfg", None)
         tcfg = TrainConfig(
             lr=float(spec.get("lr", 3e-4)),
@@ -100,7 +111,7 @@ def main(argv=None):
             lr_decay_steps=int(spec.get("steps", 100)),
             grad_accum=int(spec.get("grad_accum", 1)),
         )
-        trainer = Trainer(model, tcfg)
+        trainer = Trainer(model, tcfg, tp_ctx=tp_ctx)
 
         ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir, "checkpoints")
         save_every = int(spec.get("save_every", 0))
@@ -118,7 +129,9 @@ def main(argv=None):
                 write_status(rank_dir, "failed", step, loss,
                              error="terminated")
                 return 143
-            x, y = synthetic_batch(spec, cfg, device, rank, step)
+            # TP ranks are one data-parallel replica: identical batches
+            x, y = synthetic_batch(spec, cfg, device,
+                                   0 if tp_ctx is not None else rank, step)
             loss = trainer.step(x, y)
             if (step + 1) % status_every == 0 or step + 1 == steps:
                 lval = float(loss)
