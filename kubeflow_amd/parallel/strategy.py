@@ -1,15 +1,17 @@
 """Parallelism strategy descriptor — the launcher's extension seam.
 
-SURVEY.md §2.14: DDP is the only v1 strategy (the only one any BASELINE
-config requires); TP/PP/SP/CP/EP/Ulysses are declared here as enums so
-PyTorchJob specs stay forward-compatible, and rejected with a clear error
-until a config demands them. The seam is the per-rank environment the gang
-launcher already provides (RANK/WORLD_SIZE/LOCAL_RANK + this descriptor
-serialized into the worker spec as `parallelism`).
+SURVEY.md §2.14: DDP is the only strategy any BASELINE config requires.
+Implemented: DDP (bucketed all-reduce, ddp.py) and TP (Megatron-style
+head/ffn sharding, tp.py — pure TP, degree == world_size, llama family).
+PP/SP/CP/EP/Ulysses remain declared-but-reserved enums so PyTorchJob specs
+stay forward-compatible, rejected with a clear error until a config
+demands them. The seam is the per-rank environment the gang launcher
+already provides (RANK/WORLD_SIZE/LOCAL_RANK + this descriptor serialized
+into the worker spec as `parallelism`).
 
 Spec form (PyTorchJob template):
     parallelism: {strategy: ddp}                      # default
-    parallelism: {strategy: tp, degree: 4}            # reserved, v2+
+    parallelism: {strategy: tp, degree: 8}            # pure TP
 """
 from __future__ import annotations
 
