@@ -42,6 +42,25 @@ def _mlp(device=None, dtype=None, **kw):
     return m
 
 
+MODEL_CONFIGS: Dict[str, Callable] = {
+    "llama3-8b": llama3_8b,
+    "llama3-1b": llama3_1b,
+    "llama-tiny": llama_tiny,
+    "llama-tiny-mha": llama_tiny_mha,
+    "bert-base": bert_base,
+    "bert-base-hd128": bert_base_hd128,
+    "bert-tiny": bert_tiny,
+}
+
+
+def model_config(name: str):
+    """The config object a registered model is built from (no weights)."""
+    if name not in MODEL_CONFIGS:
+        raise KeyError(f"no config for model {name!r}; "
+                       f"known: {sorted(MODEL_CONFIGS)}")
+    return MODEL_CONFIGS[name]()
+
+
 MODEL_REGISTRY: Dict[str, Callable] = {
     "llama3-8b": _llama(llama3_8b),
     "llama3-1b": _llama(llama3_1b),

@@ -205,3 +205,75 @@ class LlamaModel(nn.Module):
         attn = (3 * 2 * 2 * self.cfg.n_layers * self.cfg.n_heads *
                 self.cfg.head_dim * seq_len / 2)
         return 6 * n + attn
+
+
+class LlamaStage(nn.Module):
+    """One pipeline stage of LlamaModel (parallel/pp.py contract): stage 0
+    owns the embedding, the last stage owns final_norm + lm_head + loss,
+    every stage owns a contiguous slice of the blocks."""
+
+    def __init__(self, cfg: LlamaConfig, stage: int, n_stages: int,
+                 device=None, dtype=torch.bfloat16):
+        super().__init__()
+        from kubeflow_amd.parallel.pp import layer_range
+        self.cfg = cfg
+        self.stage_idx, self.n_stages = stage, n_stages
+        self.layer_lo, self.layer_hi = layer_range(cfg.n_layers, stage,
+                                                   n_stages)
+        self.is_first = stage == 0
+        self.is_last = stage == n_stages - 1
+        factory = dict(device=device, dtype=dtype)
+        with torch.device(device if device is not None else "cpu"):
+            if self.is_first:
+                self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+            self.layers = nn.ModuleList(
+                [LlamaBlock(cfg)
+                 for _ in range(self.layer_lo, self.layer_hi)])
+            if self.is_last:
+                self.final_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
+                self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
+                                         bias=False)
+        self.to(**{k: v for k, v in factory.items() if v is not None})
+        cos, sin = ops.rope_cos_sin(cfg.max_seq_len, cfg.head_dim,
+                                    cfg.rope_theta, device=device)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        LlamaModel.init_weights(self)  # same init rule, stage-local params
+
+    def forward(self, x, targets: torch.Tensor = None):
+        if self.is_first:
+            x = self.embed(x)  # x is the token ids here
+        for layer in self.layers:
+            x = layer(x, self.rope_cos, self.rope_sin)
+        if not self.is_last:
+            return x
+        x = self.final_norm(x)
+        logits = F.linear(x, self.lm_head.weight)
+        if targets is None:
+            return logits
+        T = logits.shape[0] * logits.shape[1]
+        return ops.cross_entropy(logits.view(T, -1), targets.view(T))
+
+
+def stage_state_dict(full_sd: dict, cfg: LlamaConfig, stage: int,
+                     n_stages: int) -> dict:
+    """Slice a full LlamaModel state dict to one LlamaStage (tests,
+    checkpoint import)."""
+    from kubeflow_amd.parallel.pp import layer_range
+    lo, hi = layer_range(cfg.n_layers, stage, n_stages)
+    out = {}
+    for k, v in full_sd.items():
+        if k.startswith("layers."):
+            idx = int(k.split(".")[1])
+            if lo <= idx < hi:
+                out[k.replace(f"layers.{idx}.", f"layers.{idx - lo}.", 1)] = \
+                    v.clone()
+        elif k.startswith("embed."):
+            if stage == 0:
+                out[k] = v.clone()
+        elif k.startswith(("final_norm.", "lm_head.")):
+            if stage == n_stages - 1:
+                out[k] = v.clone()
+        else:
+            out[k] = v.clone()
+    return out

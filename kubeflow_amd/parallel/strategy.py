@@ -1,11 +1,11 @@
 """Parallelism strategy descriptor — the launcher's extension seam.
 
 SURVEY.md §2.14: DDP is the only strategy any BASELINE config requires.
-Implemented: DDP (bucketed all-reduce, ddp.py) and TP (Megatron-style
-head/ffn sharding, tp.py — pure TP, degree == world_size, llama family).
-PP/SP/CP/EP/Ulysses remain declared-but-reserved enums so PyTorchJob specs
-stay forward-compatible, rejected with a clear error until a config
-demands them. The seam is the per-rank environment the gang launcher
+Implemented: DDP (bucketed all-reduce, ddp.py), TP (Megatron-style
+head/ffn sharding, tp.py) and PP (GPipe fill-drain stages, pp.py) — TP/PP
+in pure form (degree == world_size, llama family). SP/CP/EP/Ulysses remain
+declared-but-reserved enums so PyTorchJob specs stay forward-compatible,
+rejected with a clear error until a config demands them. The seam is the per-rank environment the gang launcher
 already provides (RANK/WORLD_SIZE/LOCAL_RANK + this descriptor serialized
 into the worker spec as `parallelism`).
 
@@ -23,13 +23,14 @@ class Strategy(str, Enum):
     DDP = "ddp"      # implemented: bucketed all-reduce over RCCL/xGMI
     TP = "tp"        # implemented: Megatron-style head/ffn sharding (tp.py);
                      # pure TP (degree == world_size) in v1
-    PP = "pp"        # reserved: pipeline parallel
+    PP = "pp"        # implemented: GPipe fill-drain stages (pp.py);
+                     # pure PP (degree == world_size) in v1
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
     EP = "ep"        # reserved: expert parallel (all-to-all)
     ULYSSES = "ulysses"  # reserved: attention head-scatter SP
 
 
-IMPLEMENTED = {Strategy.DDP, Strategy.TP}
+IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP}
 
 
 @dataclass

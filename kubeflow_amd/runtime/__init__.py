@@ -1,3 +1,3 @@
-from .trainer import Trainer, TrainConfig
+from .trainer import Trainer, TrainConfig, PpTrainer
 
-__all__ = ["Trainer", "TrainConfig"]
+__all__ = ["Trainer", "TrainConfig", "PpTrainer"]

@@ -48,14 +48,16 @@ class Trainer:
     """Owns the flat parameter space, DDP overlap and the fused optimizer."""
 
     def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig(),
-                 tp_ctx=None):
+                 tp_ctx=None, pp_ctx=None):
         self.model = model
         self.cfg = cfg
         self.flat = FlatParamSpace(model)
         self.tp = tp_ctx
-        if tp_ctx is not None:
-            # pure TP: ranks hold distinct shards — no grad all-reduce; the
-            # two per-block TP all-reduces are the only comm per step.
+        self.pp = pp_ctx
+        if tp_ctx is not None or pp_ctx is not None:
+            # pure TP/PP: ranks hold distinct shards/stages — no grad
+            # all-reduce; comm happens inside the model (TP block
+            # all-reduces) or the schedule (PP sends).
             self.ddp = _NullDDP()
         else:
             self.ddp = BucketedDDP(self.flat, bucket_mb=cfg.bucket_mb)
@@ -100,10 +102,18 @@ class Trainer:
             else:
                 loss.backward()
         self.ddp.finalize()
+        self._clip_and_update()
+        return loss.detach()
+
+    # ---------------------------------------------------------- optimizer
+    def _clip_and_update(self):
+        """Shared grad-clip + fused-AdamW tail of a step (grads already in
+        the flat buffer, comm done)."""
+        cfg = self.cfg
         if cfg.grad_clip > 0:
-            # one flat-buffer norm; scale in place when above the clip
-            if self.tp is not None:
-                gnorm = self.tp.global_grad_norm(self.flat)
+            comm = self.tp if self.tp is not None else self.pp
+            if comm is not None:
+                gnorm = comm.global_grad_norm(self.flat)
             else:
                 gnorm = torch.linalg.vector_norm(self.flat.grad,
                                                  dtype=torch.float32)
@@ -115,7 +125,6 @@ class Trainer:
         ops.fused_adamw(self.flat.data, self.p32, self.flat.grad, self.m,
                         self.v, self.wd_mask, lr, cfg.beta1, cfg.beta2,
                         cfg.eps, cfg.weight_decay, self.step_num)
-        return loss.detach()
 
     # ---------------------------------------------------------- checkpoint
     def state_dict(self) -> dict:
@@ -140,3 +149,24 @@ class Trainer:
         torch.set_rng_state(sd["rng"].cpu() if hasattr(sd["rng"], "cpu") else sd["rng"])
         if sd.get("cuda_rng") is not None and torch.cuda.is_available():
             torch.cuda.set_rng_state(sd["cuda_rng"])
+
+
+class PpTrainer(Trainer):
+    """Trainer over one pipeline stage: forward/backward run through the
+    GPipe schedule (parallel/pp.py); clip + fused AdamW are stage-local with
+    a pipeline-global grad norm. `microbatches` plays grad_accum's role."""
+
+    def __init__(self, stage: torch.nn.Module, cfg: TrainConfig,
+                 pp_ctx, microbatches: int):
+        super().__init__(stage, cfg, pp_ctx=pp_ctx)
+        from kubeflow_amd.parallel.pp import PipelineRunner
+        self.runner = PipelineRunner(
+            stage, pp_ctx, microbatches,
+            act_dtype=self.flat.data.dtype,
+            hidden_size=stage.cfg.hidden_size)
+
+    def step(self, tokens, targets) -> torch.Tensor:
+        self.flat.zero_grad()
+        loss = self.runner.step(tokens, targets)
+        self._clip_and_update()
+        return loss

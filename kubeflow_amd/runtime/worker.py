@@ -92,6 +92,7 @@ def main(argv=None):
         dtype = (torch.bfloat16 if device.type == "cuda"
                  and spec.get("dtype", "bf16") == "bf16" else torch.float32)
         tp_ctx = None
+        pp_ctx = None
         if pspec.strategy == Strategy.TP and world > 1:
             from kubeflow_amd.parallel.tp import TpContext
             if pspec.degree not in (1, world):
@@ -99,8 +100,24 @@ def main(argv=None):
                     f"pure TP requires degree == world_size ({world}); "
                     f"got {pspec.degree} — TPxDP meshes are a v2 seam")
             tp_ctx = TpContext.from_group(None)  # TP group = WORLD
-        model = build_model(spec["model"], device=device, dtype=dtype,
-                            tp=tp_ctx)
+        elif pspec.strategy == Strategy.PP and world > 1:
+            from kubeflow_amd.parallel.pp import PpContext
+            if pspec.degree not in (1, world):
+                raise ValueError(
+                    f"pure PP requires degree == world_size ({world}); "
+                    f"got {pspec.degree}")
+            if not spec["model"].startswith("llama"):
+                raise ValueError("pipeline parallelism is implemented for "
+                                 f"the llama family only, not {spec['model']!r}")
+            pp_ctx = PpContext.from_group(None)
+        if pp_ctx is not None:
+            from kubeflow_amd.models import model_config
+            from kubeflow_amd.models.llama import LlamaStage
+            model = LlamaStage(model_config(spec["model"]), rank, world,
+                               device=device, dtype=dtype)
+        else:
+            model = build_model(spec["model"], device=device, dtype=dtype,
+                                tp=tp_ctx)
         if tp_ctx is not None:
             tp_ctx.sync_replicated(model)
         cfg = getattr(model, "cfg", None)
@@ -111,7 +128,16 @@ def main(argv=None):
             lr_decay_steps=int(spec.get("steps", 100)),
             grad_accum=int(spec.get("grad_accum", 1)),
         )
-        trainer = Trainer(model, tcfg, tp_ctx=tp_ctx)
+        if pp_ctx is not None:
+            from kubeflow_amd.runtime import PpTrainer
+            mb = int(spec.get("micro_batch", 2))
+            # default: one microbatch per stage keeps the pipe full and
+            # always divides; callers can override via pp_microbatches
+            micros = int(spec.get("pp_microbatches", 0)) or (
+                world if mb % world == 0 else 1)
+            trainer = PpTrainer(model, tcfg, pp_ctx, micros)
+        else:
+            trainer = Trainer(model, tcfg, tp_ctx=tp_ctx)
 
         ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir, "checkpoints")
         save_every = int(spec.get("save_every", 0))
@@ -129,9 +155,9 @@ def main(argv=None):
                 write_status(rank_dir, "failed", step, loss,
                              error="terminated")
                 return 143
-            # TP ranks are one data-parallel replica: identical batches
-            x, y = synthetic_batch(spec, cfg, device,
-                                   0 if tp_ctx is not None else rank, step)
+            # TP/PP ranks are one data-parallel replica: identical batches
+            dr = 0 if (tp_ctx is not None or pp_ctx is not None) else rank
+            x, y = synthetic_batch(spec, cfg, device, dr, step)
             loss = trainer.step(x, y)
             if (step + 1) % status_every == 0 or step + 1 == steps:
                 lval = float(loss)

