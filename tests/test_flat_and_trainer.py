@@ -101,12 +101,15 @@ def test_parallelism_strategy_seam():
     assert ParallelismSpec.from_spec({}).strategy == Strategy.DDP
     assert ParallelismSpec.from_spec(
         {"parallelism": "ddp"}).strategy == Strategy.DDP
-    # TP graduated from reserved to implemented
+    # TP and PP graduated from reserved to implemented
     tp = ParallelismSpec.from_spec({"parallelism": {"strategy": "tp",
                                                     "degree": 4}})
     assert tp.strategy == Strategy.TP and tp.degree == 4
+    pp = ParallelismSpec.from_spec({"parallelism": {"strategy": "pp",
+                                                    "degree": 2}})
+    assert pp.strategy == Strategy.PP and pp.degree == 2
     with pytest.raises(NotImplementedError):
-        ParallelismSpec.from_spec({"parallelism": {"strategy": "pp",
+        ParallelismSpec.from_spec({"parallelism": {"strategy": "ep",
                                                    "degree": 2}})
     with pytest.raises(ValueError):
         ParallelismSpec.from_spec({"parallelism": {"strategy": "magic"}})
