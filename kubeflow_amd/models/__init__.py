@@ -23,7 +23,8 @@ def _llama(cfg_fn):
     def build(device=None, dtype=None, **kw):
         import torch
         return LlamaModel(cfg_fn(), device=device,
-                          dtype=dtype or torch.bfloat16, tp=kw.get("tp"))
+                          dtype=dtype or torch.bfloat16, tp=kw.get("tp"),
+                          sp=kw.get("sp"))
     return build
 
 
@@ -76,7 +77,8 @@ MODEL_REGISTRY: Dict[str, Callable] = {
 def build_model(name: str, device=None, dtype=None, **kw):
     if name not in MODEL_REGISTRY:
         raise KeyError(f"unknown model {name!r}; known: {sorted(MODEL_REGISTRY)}")
-    if kw.get("tp") is not None and not name.startswith("llama"):
-        raise ValueError(f"tensor parallelism is implemented for the llama "
-                         f"family only, not {name!r}")
+    if (kw.get("tp") is not None or kw.get("sp") is not None) \
+            and not name.startswith("llama"):
+        raise ValueError(f"tensor/sequence parallelism is implemented for "
+                         f"the llama family only, not {name!r}")
     return MODEL_REGISTRY[name](device=device, dtype=dtype, **kw)

@@ -86,11 +86,19 @@ class RMSNorm(nn.Module):
 
 
 class LlamaBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig, tp=None):
+    def __init__(self, cfg: LlamaConfig, tp=None, sp=None):
         super().__init__()
         h, d = cfg.hidden_size, cfg.head_dim
         self.cfg = cfg
         self.tp = tp if (tp is not None and tp.world > 1) else None
+        self.sp = sp if (sp is not None and sp.world > 1) else None
+        if self.tp is not None and self.sp is not None:
+            raise ValueError("tp and sp (ulysses) are mutually exclusive")
+        if self.sp is not None and (cfg.n_heads % self.sp.world or
+                                    cfg.n_kv_heads % self.sp.world):
+            raise ValueError(
+                f"ulysses degree {self.sp.world} must divide heads "
+                f"({cfg.n_heads}/{cfg.n_kv_heads})")
         n = self.tp.world if self.tp else 1
         if cfg.n_heads % n or cfg.n_kv_heads % n or cfg.ffn_dim % n:
             raise ValueError(f"TP degree {n} must divide heads "
@@ -127,6 +135,17 @@ class LlamaBlock(nn.Module):
             k, v = kv_cache.update(k, v, pos_offset)
             o = ops.flash_attention(q, k, v, causal=(S > 1))
             o = o.reshape(B, S, self.hq * cfg.head_dim)
+        elif self.sp is not None:
+            # ulysses: seq-sharded in, full-seq/local-heads attention,
+            # seq-sharded out (parallel/sp.py); weights stay replicated so
+            # the residual addmm fusion below applies unchanged
+            from kubeflow_amd.parallel import sp as spmod
+            n = self.sp.world
+            qkv = spmod.scatter_heads_gather_seq(qkv, self.sp, self.hq,
+                                                 self.hkv, cfg.head_dim)
+            o = ops.fused_qkv_attention(qkv, cos, sin, self.hq // n,
+                                        self.hkv // n, cfg.head_dim)
+            o = spmod.gather_heads_scatter_seq(o, self.sp)
         else:
             o = ops.fused_qkv_attention(qkv, cos, sin, self.hq,
                                         self.hkv, cfg.head_dim)
@@ -152,15 +171,17 @@ class LlamaBlock(nn.Module):
 
 class LlamaModel(nn.Module):
     def __init__(self, cfg: LlamaConfig, device=None, dtype=torch.bfloat16,
-                 tp=None):
+                 tp=None, sp=None):
         super().__init__()
         self.cfg = cfg
         self.tp = tp
+        self.sp = sp
         factory = dict(device=device, dtype=dtype)
         with torch.device(device if device is not None else "cpu"):
             self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
             self.layers = nn.ModuleList(
-                [LlamaBlock(cfg, tp=tp) for _ in range(cfg.n_layers)])
+                [LlamaBlock(cfg, tp=tp, sp=sp)
+                 for _ in range(cfg.n_layers)])
             self.final_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
             self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
                                      bias=False)

@@ -27,10 +27,11 @@ class Strategy(str, Enum):
                      # pure PP (degree == world_size) in v1
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
     EP = "ep"        # reserved: expert parallel (all-to-all)
-    ULYSSES = "ulysses"  # reserved: attention head-scatter SP
+    ULYSSES = "ulysses"  # implemented: attention head-scatter SP (sp.py);
+                         # pure form (degree == world_size) in v1
 
 
-IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP}
+IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP, Strategy.ULYSSES}
 
 
 @dataclass

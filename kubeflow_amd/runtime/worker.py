@@ -93,6 +93,7 @@ def main(argv=None):
                  and spec.get("dtype", "bf16") == "bf16" else torch.float32)
         tp_ctx = None
         pp_ctx = None
+        sp_ctx = None
         if pspec.strategy == Strategy.TP and world > 1:
             from kubeflow_amd.parallel.tp import TpContext
             if pspec.degree not in (1, world):
@@ -100,6 +101,13 @@ def main(argv=None):
                     f"pure TP requires degree == world_size ({world}); "
                     f"got {pspec.degree} — TPxDP meshes are a v2 seam")
             tp_ctx = TpContext.from_group(None)  # TP group = WORLD
+        elif pspec.strategy == Strategy.ULYSSES and world > 1:
+            from kubeflow_amd.parallel.sp import SpContext
+            if pspec.degree not in (1, world):
+                raise ValueError(
+                    f"pure ulysses requires degree == world_size ({world}); "
+                    f"got {pspec.degree}")
+            sp_ctx = SpContext.from_group(None)
         elif pspec.strategy == Strategy.PP and world > 1:
             from kubeflow_amd.parallel.pp import PpContext
             if pspec.degree not in (1, world):
@@ -117,7 +125,7 @@ def main(argv=None):
                                device=device, dtype=dtype)
         else:
             model = build_model(spec["model"], device=device, dtype=dtype,
-                                tp=tp_ctx)
+                                tp=tp_ctx, sp=sp_ctx)
         if tp_ctx is not None:
             tp_ctx.sync_replicated(model)
         cfg = getattr(model, "cfg", None)
@@ -155,10 +163,24 @@ def main(argv=None):
                 write_status(rank_dir, "failed", step, loss,
                              error="terminated")
                 return 143
-            # TP/PP ranks are one data-parallel replica: identical batches
-            dr = 0 if (tp_ctx is not None or pp_ctx is not None) else rank
-            x, y = synthetic_batch(spec, cfg, device, dr, step)
+            # TP/PP/SP ranks are one data-parallel replica: identical batches
+            model_par = (tp_ctx is not None or pp_ctx is not None
+                         or sp_ctx is not None)
+            x, y = synthetic_batch(spec, cfg, device,
+                                   0 if model_par else rank, step)
+            if sp_ctx is not None:  # ulysses: each rank takes its seq shard
+                if x.shape[1] % world:
+                    raise ValueError(f"seq_len {x.shape[1]} not divisible "
+                                     f"by ulysses degree {world}")
+                s = x.shape[1] // world
+                x = x[:, rank * s:(rank + 1) * s].contiguous()
+                y = y[:, rank * s:(rank + 1) * s].contiguous()
             loss = trainer.step(x, y)
+            if sp_ctx is not None:  # status shows the global-mean loss
+                import torch.distributed as tdist
+                lt = loss.to(torch.float32).clone()
+                tdist.all_reduce(lt)
+                loss = lt / world
             if (step + 1) % status_every == 0 or step + 1 == steps:
                 lval = float(loss)
                 ema = lval if ema is None else 0.9 * ema + 0.1 * lval
