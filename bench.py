@@ -52,7 +52,10 @@ def main():
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     model = build_model(args.model, device=device, dtype=dtype)
     cfg = model.cfg
-    trainer = Trainer(model, TrainConfig(warmup_steps=2, lr=3e-4))
+    # KF_ZERO=1 opts into the ZeRO-1 sharded optimizer (parallel/zero.py);
+    # default stays bucketed-overlap DDP pending a measured comparison
+    trainer = Trainer(model, TrainConfig(warmup_steps=2, lr=3e-4),
+                      zero=os.environ.get("KF_ZERO") == "1")
 
     B, S = args.micro_batch, args.seq_len
     tokens = torch.randint(0, cfg.vocab_size, (B, S), device=device)

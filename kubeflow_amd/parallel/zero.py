@@ -1,0 +1,89 @@
+"""ZeRO-1 sharded optimizer state over the flat parameter space.
+
+The flat layout (flat.py) makes ZeRO stage 1 nearly free to express: the
+optimizer state (fp32 master + m + v) for an equal 1/N slice of the flat
+buffer lives on each rank instead of being replicated, cutting optimizer
+memory from 12 bytes/param to 12/N (Llama-3-8B on 8 GPUs: 96 GB -> 12 GB
+per GPU). Per step:
+
+    reduce-scatter(flat.grad) -> averaged grad shard on each rank
+    fused AdamW on the shard (one kernel, 1/N the work)
+    all-gather(flat.data)     -> updated bf16 params everywhere
+
+Communication volume equals DDP's all-reduce (RS + AG move the same
+bytes), but v1 runs both as single unoverlapped collectives — measured
+trade-off vs BucketedDDP's backward overlap is a round-2 item, so ZeRO is
+OPT-IN: `zero: true` in the PyTorchJob template (or KF_ZERO=1 for
+bench.py).
+
+gloo (CPU CI) lacks reduce_scatter_tensor — fall back to all-reduce +
+local slice (same numerics, no memory win; the win is GPU-only anyway).
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+from .flat import FlatParamSpace
+
+
+class ZeroShard:
+    def __init__(self, flat: FlatParamSpace, process_group=None):
+        if not dist.is_initialized():
+            raise RuntimeError("ZeroShard requires torch.distributed init")
+        self.flat = flat
+        self.pg = process_group
+        self.world = dist.get_world_size(process_group)
+        self.rank = dist.get_rank(process_group)
+        if flat.numel % self.world:
+            # flat.py aligns every param slice to 64 elements; pad the shard
+            # boundary the same way by requiring divisibility (ALIGN=64 and
+            # world in {2,4,8} make numel % world == 0 in practice)
+            raise ValueError(f"flat numel {flat.numel} not divisible by "
+                             f"world {self.world}")
+        self.chunk = flat.numel // self.world
+        self.lo = self.rank * self.chunk
+        self.hi = self.lo + self.chunk
+        self._rs_supported = dist.get_backend(process_group) == "nccl"
+
+    # shard views -----------------------------------------------------------
+    def data_shard(self) -> torch.Tensor:
+        return self.flat.data[self.lo:self.hi]
+
+    def new_state(self) -> torch.Tensor:
+        """A zero fp32 buffer of shard size (master/m/v allocations)."""
+        return torch.zeros(self.chunk, dtype=torch.float32,
+                           device=self.flat.device)
+
+    def shard_of(self, full: torch.Tensor) -> torch.Tensor:
+        return full[self.lo:self.hi]
+
+    # step ------------------------------------------------------------------
+    def reduce_scatter_grads(self) -> torch.Tensor:
+        """Average grads across ranks; return this rank's grad shard."""
+        g = self.flat.grad
+        if self._rs_supported:
+            shard = torch.empty(self.chunk, dtype=g.dtype, device=g.device)
+            dist.reduce_scatter_tensor(shard, g, op=dist.ReduceOp.AVG,
+                                       group=self.pg)
+            return shard
+        dist.all_reduce(g, group=self.pg)
+        g.div_(self.world)
+        return g[self.lo:self.hi]
+
+    def global_grad_norm(self, grad_shard: torch.Tensor) -> torch.Tensor:
+        s = torch.linalg.vector_norm(grad_shard, dtype=torch.float32) ** 2
+        dist.all_reduce(s, group=self.pg)
+        return torch.sqrt(s)
+
+    def all_gather_params(self):
+        """Publish this rank's updated bf16 shard to every rank."""
+        if self._rs_supported:
+            dist.all_gather_into_tensor(self.flat.data, self.data_shard(),
+                                        group=self.pg)
+        else:
+            chunks = list(self.flat.data.chunk(self.world))
+            dist.all_gather(chunks, self.data_shard().clone(), group=self.pg)
+            # all_gather into chunk views writes in place for gloo only when
+            # chunks are contiguous views of flat.data — they are (equal
+            # split of a 1-D tensor), but clone the src to avoid aliasing

@@ -48,12 +48,33 @@ class Trainer:
     """Owns the flat parameter space, DDP overlap and the fused optimizer."""
 
     def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig(),
-                 tp_ctx=None, pp_ctx=None):
+                 tp_ctx=None, pp_ctx=None, zero: bool = False):
+        import torch.distributed as dist
         self.model = model
         self.cfg = cfg
         self.flat = FlatParamSpace(model)
         self.tp = tp_ctx
         self.pp = pp_ctx
+        self.zero = None
+        if (zero and dist.is_initialized() and dist.get_world_size() > 1
+                and tp_ctx is None and pp_ctx is None):
+            # ZeRO-1: optimizer state sharded 1/N; RS + shard-AdamW + AG
+            # replace the DDP all-reduce (parallel/zero.py)
+            from kubeflow_amd.parallel.zero import ZeroShard
+            self.zero = ZeroShard(self.flat)
+            self.ddp = _NullDDP()
+            dist.broadcast(self.flat.data, src=0)
+            self.p32 = self.zero.data_shard().float()
+            self.m = self.zero.new_state()
+            self.v = self.zero.new_state()
+            if cfg.use_wd_mask:
+                full = self.flat.build_wd_mask()
+                self.wd_mask = self.zero.shard_of(full).clone()
+                del full
+            else:
+                self.wd_mask = None
+            self.step_num = 0
+            return
         if tp_ctx is not None or pp_ctx is not None:
             # pure TP/PP: ranks hold distinct shards/stages — no grad
             # all-reduce; comm happens inside the model (TP block
@@ -61,7 +82,6 @@ class Trainer:
             self.ddp = _NullDDP()
         else:
             self.ddp = BucketedDDP(self.flat, bucket_mb=cfg.bucket_mb)
-        dev = self.flat.device
         self.p32 = self.flat.data.float()
         self.m = torch.zeros_like(self.p32)
         self.v = torch.zeros_like(self.p32)
@@ -102,7 +122,10 @@ class Trainer:
             else:
                 loss.backward()
         self.ddp.finalize()
-        self._clip_and_update()
+        if self.zero is not None:
+            self._zero_update()
+        else:
+            self._clip_and_update()
         return loss.detach()
 
     # ---------------------------------------------------------- optimizer
@@ -125,6 +148,23 @@ class Trainer:
         ops.fused_adamw(self.flat.data, self.p32, self.flat.grad, self.m,
                         self.v, self.wd_mask, lr, cfg.beta1, cfg.beta2,
                         cfg.eps, cfg.weight_decay, self.step_num)
+
+    def _zero_update(self):
+        """ZeRO-1 step tail: averaged grad shard -> clip -> shard AdamW ->
+        param all-gather."""
+        cfg = self.cfg
+        gs = self.zero.reduce_scatter_grads()
+        if cfg.grad_clip > 0:
+            gnorm = self.zero.global_grad_norm(gs)
+            scale = (cfg.grad_clip / (gnorm + 1e-6)).clamp(max=1.0)
+            gs.mul_(scale.to(gs.dtype))
+            self.last_grad_norm = gnorm
+        self.step_num += 1
+        lr = self.lr_at(self.step_num - 1)
+        ops.fused_adamw(self.zero.data_shard(), self.p32, gs, self.m,
+                        self.v, self.wd_mask, lr, cfg.beta1, cfg.beta2,
+                        cfg.eps, cfg.weight_decay, self.step_num)
+        self.zero.all_gather_params()
 
     # ---------------------------------------------------------- checkpoint
     def state_dict(self) -> dict:

@@ -145,7 +145,8 @@ def main(argv=None):
                 world if mb % world == 0 else 1)
             trainer = PpTrainer(model, tcfg, pp_ctx, micros)
         else:
-            trainer = Trainer(model, tcfg, tp_ctx=tp_ctx)
+            trainer = Trainer(model, tcfg, tp_ctx=tp_ctx,
+                              zero=bool(spec.get("zero", False)))
 
         ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir, "checkpoints")
         save_every = int(spec.get("save_every", 0))
