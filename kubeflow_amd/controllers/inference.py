@@ -50,6 +50,8 @@ class InferenceServiceReconciler(Reconciler):
         svc = self.store.get(self.kind, name, namespace)
         uid = svc["metadata"]["uid"]
         self.key_uid[(namespace, name)] = uid
+        if has_condition(svc, "Failed"):  # invalid spec — terminal
+            return
         gang = self.gangs.get(uid)
 
         if gang is None:
@@ -94,6 +96,18 @@ class InferenceServiceReconciler(Reconciler):
         uid = svc["metadata"]["uid"]
         pred = svc["spec"].get("predictor", {})
         gpus = int(pred.get("gpus", 1))
+        if gpus > 1:
+            # BASELINE serving config is TP=1; multi-GPU serving (TP decode
+            # + broadcast-coordinated batching) is a declared v2 seam —
+            # fail loudly rather than strand an allocated-but-idle GPU.
+            set_condition(svc, "Failed", "True", "InvalidSpec",
+                          f"predictor.gpus={gpus}: multi-GPU serving is not "
+                          "implemented in v1 (TP=1 per BASELINE config)")
+            self.store.update(svc, check_version=False)
+            self.store.record_event(svc, "InvalidSpec",
+                                    "multi-GPU serving not implemented",
+                                    "Warning")
+            return
         import torch
         want_gpu = gpus > 0 and (torch.cuda.is_available()
                                  or os.environ.get("KF_FAKE_GPUS"))

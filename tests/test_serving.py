@@ -153,3 +153,28 @@ def test_context_length_cap():
         assert r.pos < 64
     finally:
         eng.stop()
+
+
+def test_multi_gpu_serving_rejected(tmp_path):
+    """predictor.gpus > 1 is a declared v2 seam — fails loudly, once."""
+    import time
+    from kubeflow_amd.api import new_object
+    from kubeflow_amd.api.objects import has_condition
+    from kubeflow_amd.platform import Platform
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        plat.store.create(new_object(
+            "InferenceService", "big", "default",
+            spec={"predictor": {"model": "llama-tiny", "gpus": 2}}))
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            obj = plat.store.get("InferenceService", "big", "default")
+            if has_condition(obj, "Failed"):
+                break
+            time.sleep(0.2)
+        assert has_condition(obj, "Failed")
+        time.sleep(1.5)  # terminal: no event storm, no gang
+        evs = [e for e in plat.store.events_for(obj)
+               if e["reason"] == "InvalidSpec"]
+        assert len(evs) == 1
+        assert not plat.inference.gangs
