@@ -23,6 +23,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from kubeflow_amd import ops
+from kubeflow_amd.parallel import sp as spmod
 from kubeflow_amd.parallel import tp as tpmod
 
 
@@ -139,7 +140,6 @@ class LlamaBlock(nn.Module):
             # ulysses: seq-sharded in, full-seq/local-heads attention,
             # seq-sharded out (parallel/sp.py); weights stay replicated so
             # the residual addmm fusion below applies unchanged
-            from kubeflow_amd.parallel import sp as spmod
             n = self.sp.world
             qkv = spmod.scatter_heads_gather_seq(qkv, self.sp, self.hq,
                                                  self.hkv, cfg.head_dim)

@@ -51,7 +51,10 @@ def _exchange(chunks: List[torch.Tensor], group) -> List[torch.Tensor]:
     shapes). Falls back to isend/irecv for backends without all_to_all."""
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
-    out = [torch.empty_like(chunks[i]) for i in range(world)]
+    # empty() not empty_like(): chunks are slices (non-contiguous) and the
+    # receive buffers must be contiguous for the collective
+    out = [torch.empty(c.shape, dtype=c.dtype, device=c.device)
+           for c in chunks]
     backend = dist.get_backend(group)
     if backend == "nccl":
         dist.all_to_all(out, [c.contiguous() for c in chunks], group=group)

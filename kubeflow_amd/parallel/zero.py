@@ -18,6 +18,10 @@ bench.py).
 
 gloo (CPU CI) lacks reduce_scatter_tensor — fall back to all-reduce +
 local slice (same numerics, no memory win; the win is GPU-only anyway).
+
+Checkpoint note: optimizer state is shard-local, so resume requires the
+SAME world size; checkpoint.py's rank-0 fallback (elastic world changes)
+applies to DDP's replicated state only.
 """
 from __future__ import annotations
 
