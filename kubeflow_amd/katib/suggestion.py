@@ -65,6 +65,29 @@ class _Space:
         return vec
 
 
+    def from_unit(self, vec: List[float]) -> Dict[str, object]:
+        """Decode a [0,1]^d point back into an assignment (inverse of
+        to_unit; ints rounded, categoricals to the nearest index)."""
+        out = {}
+        for u, p in zip(vec, self.params):
+            fs = p.get("feasibleSpace", {})
+            t = p.get("parameterType", "double")
+            u = min(1.0, max(0.0, float(u)))
+            if t == "categorical":
+                lst = fs["list"]
+                out[p["name"]] = lst[round(u * (len(lst) - 1))]
+                continue
+            lo, hi = float(fs["min"]), float(fs["max"])
+            if fs.get("logScale"):
+                v = math.exp(math.log(lo) + u * (math.log(hi) - math.log(lo)))
+            else:
+                v = lo + u * (hi - lo)
+            out[p["name"]] = (min(int(fs["max"]), max(int(fs["min"]),
+                                                      round(v)))
+                              if t == "int" else v)
+        return out
+
+
 class RandomSuggestion:
     def __init__(self, parameters: List[dict], seed: int = 0):
         self.space = _Space(parameters)
@@ -152,6 +175,59 @@ class BayesOptSuggestion:
         return out
 
 
+class TpeSuggestion:
+    """Tree-structured Parzen estimator (minimization). Completed trials
+    split at the gamma quantile into good/bad sets; each is modeled as a
+    Gaussian mixture (kernel centered on every observation, fixed unit-cube
+    bandwidth); candidates sampled around the good set are ranked by
+    log l(x) - log g(x). Mirrors Katib's `tpe` algorithm (hyperopt-style)
+    without the hyperopt dependency."""
+
+    def __init__(self, parameters: List[dict], seed: int = 0,
+                 gamma: float = 0.25, n_initial: int = 4,
+                 n_candidates: int = 128, bandwidth: float = 0.15):
+        self.space = _Space(parameters)
+        self.rng = random.Random(seed)
+        self.seed = seed
+        self.gamma = gamma
+        self.n_initial = n_initial
+        self.n_candidates = n_candidates
+        self.bw = bandwidth
+
+    def suggest(self, trials, n):
+        done = [(a, v) for a, v in trials if v is not None]
+        if len(done) < max(self.n_initial, 2):
+            return [self.space.sample(self.rng) for _ in range(n)]
+        import numpy as np
+        npr = np.random.default_rng(self.seed + len(done))
+        X = np.array([self.space.to_unit(a) for a, _ in done])
+        y = np.array([v for _, v in done], dtype=float)
+        order = np.argsort(y)
+        n_good = max(1, int(math.ceil(self.gamma * len(done))))
+        good, bad = X[order[:n_good]], X[order[n_good:]]
+        if not len(bad):
+            return [self.space.sample(self.rng) for _ in range(n)]
+        d = X.shape[1]
+
+        def log_mix(pts, data):
+            diff = (pts[:, None, :] - data[None, :, :]) / self.bw
+            ll = (-0.5 * (diff ** 2).sum(-1)
+                  - d * math.log(self.bw * math.sqrt(2 * math.pi)))
+            m = ll.max(axis=1)
+            return m + np.log(np.exp(ll - m[:, None]).mean(axis=1))
+
+        out = []
+        for _ in range(n):
+            centers = good[npr.integers(0, len(good), self.n_candidates)]
+            cand = np.clip(centers + npr.normal(0, self.bw, centers.shape),
+                           0.0, 1.0)
+            cand[:self.n_candidates // 4] = npr.random(
+                (self.n_candidates // 4, d))  # keep exploring
+            score = log_mix(cand, good) - log_mix(cand, bad)
+            out.append(self.space.from_unit(cand[int(score.argmax())]))
+        return out
+
+
 def make_suggestion(algorithm: str, parameters: List[dict], seed: int = 0):
     alg = (algorithm or "random").lower()
     if alg in ("random",):
@@ -160,4 +236,6 @@ def make_suggestion(algorithm: str, parameters: List[dict], seed: int = 0):
         return GridSuggestion(parameters, seed)
     if alg in ("bayesianoptimization", "bayesopt", "skopt"):
         return BayesOptSuggestion(parameters, seed)
+    if alg in ("tpe", "hyperopt-tpe"):
+        return TpeSuggestion(parameters, seed)
     raise ValueError(f"unknown suggestion algorithm {algorithm!r}")
