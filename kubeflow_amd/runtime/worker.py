@@ -159,6 +159,7 @@ def main(argv=None):
         write_status(rank_dir, "running", start_step)
         loss = None
         ema = None
+        win_t0, win_steps, win_items = time.time(), 0, 0
         for step in range(start_step, steps):
             if stop["flag"]:
                 write_status(rank_dir, "failed", step, loss,
@@ -177,6 +178,8 @@ def main(argv=None):
                 x = x[:, rank * s:(rank + 1) * s].contiguous()
                 y = y[:, rank * s:(rank + 1) * s].contiguous()
             loss = trainer.step(x, y)
+            win_steps += 1
+            win_items += x.numel()  # tokens (lm) or features processed
             if sp_ctx is not None:  # status shows the global-mean loss
                 import torch.distributed as tdist
                 lt = loss.to(torch.float32).clone()
@@ -185,9 +188,15 @@ def main(argv=None):
             if (step + 1) % status_every == 0 or step + 1 == steps:
                 lval = float(loss)
                 ema = lval if ema is None else 0.9 * ema + 0.1 * lval
+                dt = max(1e-9, time.time() - win_t0)
                 write_status(rank_dir, "running", step + 1, lval,
                              metrics={"loss": lval, "loss_ema": ema,
-                                      "lr": trainer.lr_at(step)})
+                                      "lr": trainer.lr_at(step),
+                                      "step_ms": round(dt / win_steps * 1e3,
+                                                       2),
+                                      "items_per_s": round(win_items / dt,
+                                                           1)})
+                win_t0, win_steps, win_items = time.time(), 0, 0
             if save_every and (step + 1) % save_every == 0:
                 kdist.barrier()
                 ckpt.save(trainer, ckpt_dir, spec["model"], rank, world)
