@@ -138,10 +138,17 @@ class InferenceServiceReconciler(Reconciler):
             "world_size": 1,
         }
         poddefaults = self.store.list("PodDefault", m.get("namespace"))
-        gang = launch_gang(uid, workdir, spec, gpu_indices,
-                           poddefaults=poddefaults,
-                           labels=m.get("labels", {}),
-                           entry_module="kubeflow_amd.runtime.serving_server")
+        try:
+            gang = launch_gang(
+                uid, workdir, spec, gpu_indices, poddefaults=poddefaults,
+                labels=m.get("labels", {}),
+                entry_module="kubeflow_amd.runtime.serving_server")
+        except ValueError as e:  # spec-level launch error — terminal
+            self.scheduler.release(uid)
+            set_condition(svc, "Failed", "True", "InvalidSpec", str(e))
+            self.store.update(svc, check_version=False)
+            self.store.record_event(svc, "InvalidSpec", str(e), "Warning")
+            return
         self.gangs[uid] = gang
         set_condition(svc, "Created", "True", "PredictorCreated",
                       f"port {port}")

@@ -119,10 +119,19 @@ class TrainingJobReconciler(Reconciler):
         poddefaults = self.store.list("PodDefault",
                                       job["metadata"].get("namespace"))
         numa = {g.index: g.numa_node for g in self.scheduler.inv.gpus}
-        gang = launch_gang(uid, workdir, spec, gpu_indices,
-                           poddefaults=poddefaults,
-                           labels=job["metadata"].get("labels", {}),
-                           numa_nodes=numa, warm_pool=self.warm_pool)
+        try:
+            gang = launch_gang(uid, workdir, spec, gpu_indices,
+                               poddefaults=poddefaults,
+                               labels=job["metadata"].get("labels", {}),
+                               numa_nodes=numa, warm_pool=self.warm_pool)
+        except ValueError as e:
+            # spec-level launch error (e.g. PodDefault env conflict) —
+            # terminal, not retryable: mark Failed instead of hot-looping
+            self.scheduler.release(uid)
+            set_condition(job, "Failed", "True", "InvalidSpec", str(e))
+            self.store.update(job, check_version=False)
+            self.store.record_event(job, "InvalidSpec", str(e), "Warning")
+            return
         self.gangs[uid] = gang
         set_condition(job, "Created", "True", "JobCreated", "gang launched")
         set_condition(job, "Running", "True", "JobRunning",

@@ -141,3 +141,37 @@ def test_gang_insufficient_resources_queues(tmp_path, monkeypatch):
             time.sleep(0.2)
         assert seen_event
         assert not has_condition(obj, "Running")
+
+
+def test_poddefault_conflict_fails_job(tmp_path):
+    """Two selected PodDefaults setting the same env differently must mark
+    the job Failed/InvalidSpec once — not hot-loop the reconciler
+    (admission-webhook conflict semantics, main.go:98-132)."""
+    with Platform(root_dir=str(tmp_path)) as plat:
+        for name, val in (("pd-a", "1"), ("pd-b", "2")):
+            plat.store.create(new_object("PodDefault", name, "default", spec={
+                "selector": {"matchLabels": {"team": "x"}},
+                "env": [{"name": "SHARED_KEY", "value": val}]}))
+        job = new_object("PyTorchJob", "conflicted", "default", spec={
+            "pytorchReplicaSpecs": {"Worker": {
+                "replicas": 1, "restartPolicy": "Never",
+                "template": {"model": "mnist-mlp", "steps": 2,
+                             "gpus_per_replica": 0,
+                             "save_final": False}}}})
+        job["metadata"]["labels"] = {"team": "x"}
+        plat.store.create(job)
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            obj = plat.store.get("PyTorchJob", "conflicted", "default")
+            if has_condition(obj, "Failed"):
+                break
+            time.sleep(0.2)
+        assert has_condition(obj, "Failed"), obj["status"]
+        conds = {c["type"]: c for c in obj["status"]["conditions"]}
+        assert conds["Failed"]["reason"] == "InvalidSpec"
+        assert "SHARED_KEY" in conds["Failed"]["message"]
+        time.sleep(1.5)  # terminal — no retry storm
+        evs = [e for e in plat.store.events_for(obj)
+               if e["reason"] == "InvalidSpec"]
+        assert len(evs) == 1
+        assert not plat.pytorchjob.gangs
