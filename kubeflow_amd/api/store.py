@@ -66,10 +66,14 @@ class ObjectStore:
         self._watchers: List[Tuple[Optional[str], Callable[[Event], None]]] = []
         self._persist_path = persist_path
         self._persist_f = None
+        self._persist_lines = 0
         if persist_path:
             self._load(persist_path)
             os.makedirs(os.path.dirname(persist_path) or ".", exist_ok=True)
-            self._persist_f = open(persist_path, "a", encoding="utf-8")
+            # the log is append-only (etcd-WAL analog) and every update adds
+            # a line — compact to a snapshot of live objects on load, and
+            # again whenever the log grows well past the live set
+            self._compact_locked()
 
     # ------------------------------------------------------------ persist
     def _load(self, path: str):
@@ -91,10 +95,27 @@ class ObjectStore:
                     self._rv = max(self._rv,
                                    int(rec["obj"]["metadata"]["resourceVersion"]))
 
+    def _compact_locked(self):
+        """Rewrite the log as one PUT per live object (atomic rename).
+        Caller must hold _lock (or be in __init__ before threads exist)."""
+        path = self._persist_path
+        if self._persist_f:
+            self._persist_f.close()
+        tmp = path + ".tmp"
+        with open(tmp, "w", encoding="utf-8") as f:
+            for obj in self._objs.values():
+                f.write(json.dumps({"op": "PUT", "obj": obj}) + "\n")
+        os.replace(tmp, path)
+        self._persist_f = open(path, "a", encoding="utf-8")
+        self._persist_lines = len(self._objs)
+
     def _persist(self, op: str, obj: KfObject):
         if self._persist_f:
             self._persist_f.write(json.dumps({"op": op, "obj": obj}) + "\n")
             self._persist_f.flush()
+            self._persist_lines += 1
+            if self._persist_lines > max(4096, 8 * len(self._objs)):
+                self._compact_locked()
 
     # -------------------------------------------------------------- watch
     def watch(self, callback: Callable[[Event], None],

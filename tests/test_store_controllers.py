@@ -160,3 +160,25 @@ def test_requeue_after():
         assert rec.count >= 3
     finally:
         mgr.stop()
+
+
+def test_store_log_compaction(tmp_path):
+    """The JSONL log compacts instead of growing unboundedly (etcd-WAL
+    analog); reload after heavy update churn restores the live set."""
+    path = str(tmp_path / "store.jsonl")
+    s = ObjectStore(persist_path=path)
+    for i in range(8):
+        s.create(new_object("PyTorchJob", f"j{i}", "default"))
+    o = s.get("PyTorchJob", "j0", "default")
+    for k in range(6000):  # way past the 4096-line floor
+        o["status"]["k"] = k
+        s.update(o, check_version=False)
+    with open(path) as f:
+        lines = sum(1 for _ in f)
+    assert lines < 6000, lines  # compaction happened
+    s2 = ObjectStore(persist_path=path)
+    assert len(s2.list("PyTorchJob", "default")) == 8
+    assert s2.get("PyTorchJob", "j0", "default")["status"]["k"] == 5999
+    # compact-on-load leaves exactly the live set
+    with open(path) as f:
+        assert sum(1 for _ in f) == 8
