@@ -43,6 +43,10 @@ def build_app(spec: dict) -> FastAPI:
     def healthz():
         return {"status": "ok", "uptime_s": time.time() - t0}
 
+    @app.get("/v1/models")
+    def models_list():  # KServe model-list surface
+        return {"models": [name]}
+
     @app.get(f"/v1/models/{name}")
     def model_meta():
         return {"name": name, "ready": True,

@@ -97,6 +97,8 @@ def test_inference_service_e2e(tmp_path):
         assert len(out["predictions"]) == 1
         assert len(out["predictions"][0]["tokens"]) == 4
         assert out["predictions"][0]["latency_ms"] is not None
+        with urllib.request.urlopen(f"{url}/v1/models", timeout=10) as r:
+            assert json.loads(r.read())["models"] == ["tiny-svc"]
 
 
 def test_sampling_temperature_cpu():
