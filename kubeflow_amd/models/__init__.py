@@ -7,7 +7,7 @@ from __future__ import annotations
 from typing import Callable, Dict
 
 from .llama import (LlamaConfig, LlamaModel, llama3_8b, llama3_1b,
-                    llama_tiny, llama_tiny_mha)
+                    llama3_70b, llama_tiny, llama_tiny_mha)
 from .bert import (BertConfig, BertClassifier, bert_base, bert_base_hd128,
                    bert_tiny)
 from .mlp import MnistMLP
@@ -46,6 +46,7 @@ def _mlp(device=None, dtype=None, **kw):
 MODEL_CONFIGS: Dict[str, Callable] = {
     "llama3-8b": llama3_8b,
     "llama3-1b": llama3_1b,
+    "llama3-70b": llama3_70b,
     "llama-tiny": llama_tiny,
     "llama-tiny-mha": llama_tiny_mha,
     "bert-base": bert_base,
@@ -65,6 +66,7 @@ def model_config(name: str):
 MODEL_REGISTRY: Dict[str, Callable] = {
     "llama3-8b": _llama(llama3_8b),
     "llama3-1b": _llama(llama3_1b),
+    "llama3-70b": _llama(llama3_70b),
     "llama-tiny": _llama(llama_tiny),
     "llama-tiny-mha": _llama(llama_tiny_mha),
     "bert-base": _bert(bert_base),

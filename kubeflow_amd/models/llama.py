@@ -61,6 +61,17 @@ def llama3_1b() -> LlamaConfig:
                        n_heads=16, n_kv_heads=8, ffn_dim=8192)
 
 
+def llama3_70b() -> LlamaConfig:
+    """Llama-3-70B (70.6B params): needs the model-parallel strategies.
+    Replicated training state is 131 GiB bf16 weights + 131 GiB grads +
+    788 GiB fp32 master/moments — far past 288 GB/GPU for DDP and still
+    ~360 GiB/GPU under ZeRO-1. TP8 or PP8 shard ALL of it 1/8 per GPU
+    (~131 GiB total + activations), which fits with room for long
+    sequences."""
+    return LlamaConfig(name="llama3-70b", hidden_size=8192, n_layers=80,
+                       n_heads=64, n_kv_heads=8, ffn_dim=28672)
+
+
 def llama_tiny(vocab: int = 512) -> LlamaConfig:
     """Tiny config for tests. head_dim stays 128 (HIP kernel contract)."""
     return LlamaConfig(name="llama-tiny", vocab_size=vocab, hidden_size=256,
