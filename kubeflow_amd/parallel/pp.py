@@ -55,6 +55,12 @@ class PpContext:
     def send(self, t: torch.Tensor, to_rank: int):
         dist.send(t.contiguous(), dst=self._peer(to_rank), group=self.group)
 
+    def isend(self, t: torch.Tensor, to_rank: int):
+        """Non-blocking send; returns (work, tensor) — the tensor must stay
+        referenced until work.wait()."""
+        t = t.contiguous()
+        return (dist.isend(t, dst=self._peer(to_rank), group=self.group), t)
+
     def recv(self, shape, dtype, device, from_rank: int) -> torch.Tensor:
         t = torch.empty(shape, dtype=dtype, device=device)
         dist.recv(t, src=self._peer(from_rank), group=self.group)
@@ -94,19 +100,28 @@ class PipelineRunner:
 
     def __init__(self, stage: torch.nn.Module, ctx: PpContext,
                  microbatches: int, act_dtype: torch.dtype,
-                 hidden_size: int):
+                 hidden_size: int, schedule: str = "1f1b"):
         if microbatches < 1:
             raise ValueError("microbatches must be >= 1")
+        if schedule not in ("1f1b", "gpipe"):
+            raise ValueError(f"unknown pp schedule {schedule!r}")
         self.stage = stage
         self.ctx = ctx
         self.m = microbatches
         self.act_dtype = act_dtype
         self.h = hidden_size
+        self.schedule = schedule
 
     def step(self, tokens: torch.Tensor, targets: torch.Tensor):
         """Forward+backward all microbatches (grads accumulate into the
         stage's params, pre-divided by M like Trainer's grad_accum).
-        Returns the mean microbatch loss, broadcast to every rank."""
+        Returns the mean microbatch loss, broadcast to every rank.
+
+        Schedules (identical math, different memory):
+          * gpipe: all forwards, then all backwards — M activations live.
+          * 1f1b: warmup forwards then fwd/bwd interleave — at most
+            (stages - rank) activations live, so M can grow freely.
+        """
         ctx = self.ctx
         B, S = tokens.shape
         if B % self.m:
@@ -116,10 +131,12 @@ class PipelineRunner:
         tgt_micro = targets.split(b)
         device = next(self.stage.parameters()).device
 
-        saved: List[tuple] = []
+        from collections import deque
+        outstanding = deque()  # (x_in, y) FIFO — backward oldest first
+        pending = []           # in-flight isend (work, tensor) pairs
         losses = []
-        # ---- fill: forwards
-        for i in range(self.m):
+
+        def fwd(i):
             if ctx.is_first:
                 x_in = None
                 y = self.stage(tok_micro[i].to(device))
@@ -130,13 +147,12 @@ class PipelineRunner:
                      if ctx.is_last else self.stage(x_in))
             if ctx.is_last:
                 losses.append(y)  # y is the loss
-                saved.append((x_in, y))
             else:
-                ctx.send(y.detach(), ctx.rank + 1)
-                saved.append((x_in, y))
-        # ---- drain: backwards in reverse order
-        for i in reversed(range(self.m)):
-            x_in, y = saved[i]
+                pending.append(ctx.isend(y.detach(), ctx.rank + 1))
+            outstanding.append((x_in, y))
+
+        def bwd():
+            x_in, y = outstanding.popleft()
             if ctx.is_last:
                 (y / self.m).backward()
             else:
@@ -144,8 +160,24 @@ class PipelineRunner:
                               ctx.rank + 1)
                 y.backward(dy)
             if not ctx.is_first:
-                ctx.send(x_in.grad, ctx.rank - 1)
-        saved.clear()
+                pending.append(ctx.isend(x_in.grad, ctx.rank - 1))
+
+        if self.schedule == "gpipe":
+            for i in range(self.m):
+                fwd(i)
+            for _ in range(self.m):
+                bwd()
+        else:  # 1f1b
+            warm = min(self.m, ctx.world - 1 - ctx.rank)
+            for i in range(warm):
+                fwd(i)
+            for i in range(warm, self.m):
+                fwd(i)
+                bwd()
+            while outstanding:
+                bwd()
+        for work, _t in pending:
+            work.wait()
         mean_loss = (torch.stack([l.detach() for l in losses]).mean()
                      if ctx.is_last else None)
         return ctx.broadcast_scalar(mean_loss, ctx.world - 1, device)

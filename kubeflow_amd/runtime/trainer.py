@@ -197,13 +197,13 @@ class PpTrainer(Trainer):
     a pipeline-global grad norm. `microbatches` plays grad_accum's role."""
 
     def __init__(self, stage: torch.nn.Module, cfg: TrainConfig,
-                 pp_ctx, microbatches: int):
+                 pp_ctx, microbatches: int, schedule: str = "1f1b"):
         super().__init__(stage, cfg, pp_ctx=pp_ctx)
         from kubeflow_amd.parallel.pp import PipelineRunner
         self.runner = PipelineRunner(
             stage, pp_ctx, microbatches,
             act_dtype=self.flat.data.dtype,
-            hidden_size=stage.cfg.hidden_size)
+            hidden_size=stage.cfg.hidden_size, schedule=schedule)
 
     def step(self, tokens, targets) -> torch.Tensor:
         self.flat.zero_grad()

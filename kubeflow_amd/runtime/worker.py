@@ -143,7 +143,8 @@ def main(argv=None):
             # always divides; callers can override via pp_microbatches
             micros = int(spec.get("pp_microbatches", 0)) or (
                 world if mb % world == 0 else 1)
-            trainer = PpTrainer(model, tcfg, pp_ctx, micros)
+            trainer = PpTrainer(model, tcfg, pp_ctx, micros,
+                                schedule=spec.get("pp_schedule", "1f1b"))
         else:
             trainer = Trainer(model, tcfg, tp_ctx=tp_ctx,
                               zero=bool(spec.get("zero", False)))

@@ -52,7 +52,7 @@ def test_stage_state_dict_partitions():
         stage.load_state_dict(s)
 
 
-def _pp_worker(rank, world, port, results):
+def _pp_worker(rank, world, port, results, schedule="1f1b", M=2):
     os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
                       RANK=str(rank), WORLD_SIZE=str(world))
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -68,9 +68,8 @@ def _pp_worker(rank, world, port, results):
         toks = torch.randint(0, cfg.vocab_size, (4, 64))
         tgts = torch.randint(0, cfg.vocab_size, (4, 64))
 
-        M = 2
         tr_pp = PpTrainer(stage, TrainConfig(lr=1e-3, warmup_steps=1),
-                          ctx, M)
+                          ctx, M, schedule=schedule)
         # oracle: the full model with grad_accum=M over the same micros
         tr_full = Trainer(_full_model(),
                           TrainConfig(lr=1e-3, warmup_steps=1, grad_accum=M))
@@ -94,13 +93,16 @@ def _pp_worker(rank, world, port, results):
         dist.destroy_process_group()
 
 
-def test_pp_two_stages_matches_grad_accum():
+@pytest.mark.parametrize("schedule,M,port", [("gpipe", 2, 29571),
+                                             ("1f1b", 2, 29572),
+                                             ("1f1b", 4, 29573)])
+def test_pp_two_stages_matches_grad_accum(schedule, M, port):
     world = 2
     ctx = mp.get_context("spawn")
     with ctx.Manager() as mgr:
         results = mgr.dict()
         procs = [ctx.Process(target=_pp_worker,
-                             args=(r, world, 29571, results))
+                             args=(r, world, port, results, schedule, M))
                  for r in range(world)]
         for p in procs:
             p.start()
