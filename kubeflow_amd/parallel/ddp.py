@@ -49,7 +49,11 @@ class BucketedDDP:
         if self.enabled:
             self._register_hooks()
             # broadcast initial parameters so every rank starts identical
-            dist.broadcast(flat.data, src=0, group=self.pg)
+            # (src = the group's first member — a global rank, which is not
+            # 0 for non-leading DP groups of a TP x DP mesh)
+            src = (dist.get_global_rank(self.pg, 0)
+                   if self.pg is not None else 0)
+            dist.broadcast(flat.data, src=src, group=self.pg)
 
     def _build_buckets(self, bucket_numel: int):
         self.buckets: List[_Bucket] = []

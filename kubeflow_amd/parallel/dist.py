@@ -51,6 +51,29 @@ def init_distributed(backend: str | None = None,
     return rank, world, device
 
 
+def build_mesh(tp_degree: int):
+    """Split WORLD into a TP x DP mesh. TP groups are CONTIGUOUS rank
+    blocks — the gang scheduler allocates xGMI-adjacent GPUs to adjacent
+    ranks, so the latency-critical per-block TP all-reduces stay on
+    neighboring links while DP's overlappable grad traffic strides across.
+
+    Returns (tp_group, dp_group, tp_rank, dp_rank). Every rank must call
+    this (new_group is collective).
+    """
+    world, rank = dist.get_world_size(), dist.get_rank()
+    if world % tp_degree:
+        raise ValueError(f"world {world} not divisible by tp degree "
+                         f"{tp_degree}")
+    dp_degree = world // tp_degree
+    tp_groups = [dist.new_group(list(range(i * tp_degree,
+                                           (i + 1) * tp_degree)))
+                 for i in range(dp_degree)]
+    dp_groups = [dist.new_group(list(range(j, world, tp_degree)))
+                 for j in range(tp_degree)]
+    tp_rank, dp_rank = rank % tp_degree, rank // tp_degree
+    return tp_groups[dp_rank], dp_groups[tp_rank], tp_rank, dp_rank
+
+
 def get_rank() -> int:
     return dist.get_rank() if dist.is_initialized() else 0
 

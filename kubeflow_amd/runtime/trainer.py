@@ -48,7 +48,8 @@ class Trainer:
     """Owns the flat parameter space, DDP overlap and the fused optimizer."""
 
     def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig(),
-                 tp_ctx=None, pp_ctx=None, zero: bool = False):
+                 tp_ctx=None, pp_ctx=None, zero: bool = False,
+                 dp_group=None):
         import torch.distributed as dist
         self.model = model
         self.cfg = cfg
@@ -75,7 +76,13 @@ class Trainer:
                 self.wd_mask = None
             self.step_num = 0
             return
-        if tp_ctx is not None or pp_ctx is not None:
+        if tp_ctx is not None and dp_group is not None \
+                and dist.get_world_size(dp_group) > 1:
+            # TP x DP mesh: DP peers hold the SAME shard, so bucketed
+            # all-reduce over the dp group is exactly DDP on each shard
+            self.ddp = BucketedDDP(self.flat, bucket_mb=cfg.bucket_mb,
+                                   process_group=dp_group)
+        elif tp_ctx is not None or pp_ctx is not None:
             # pure TP/PP: ranks hold distinct shards/stages — no grad
             # all-reduce; comm happens inside the model (TP block
             # all-reduces) or the schedule (PP sends).

@@ -94,13 +94,21 @@ def main(argv=None):
         tp_ctx = None
         pp_ctx = None
         sp_ctx = None
+        dp_group = None
+        data_rank = rank
         if pspec.strategy == Strategy.TP and world > 1:
             from kubeflow_amd.parallel.tp import TpContext
-            if pspec.degree not in (1, world):
-                raise ValueError(
-                    f"pure TP requires degree == world_size ({world}); "
-                    f"got {pspec.degree} — TPxDP meshes are a v2 seam")
-            tp_ctx = TpContext.from_group(None)  # TP group = WORLD
+            deg = pspec.degree if pspec.degree > 1 else world
+            if world % deg:
+                raise ValueError(f"world_size {world} not divisible by "
+                                 f"tp degree {deg}")
+            if deg == world:
+                tp_ctx = TpContext.from_group(None)  # pure TP
+                data_rank = 0
+            else:  # TP x DP mesh: contiguous TP blocks, strided DP
+                tp_group, dp_group, _tp_rank, dp_rank = kdist.build_mesh(deg)
+                tp_ctx = TpContext.from_group(tp_group)
+                data_rank = dp_rank
         elif pspec.strategy == Strategy.ULYSSES and world > 1:
             from kubeflow_amd.parallel.sp import SpContext
             if pspec.degree not in (1, world):
@@ -146,7 +154,7 @@ def main(argv=None):
             trainer = PpTrainer(model, tcfg, pp_ctx, micros,
                                 schedule=spec.get("pp_schedule", "1f1b"))
         else:
-            trainer = Trainer(model, tcfg, tp_ctx=tp_ctx,
+            trainer = Trainer(model, tcfg, tp_ctx=tp_ctx, dp_group=dp_group,
                               zero=bool(spec.get("zero", False)))
 
         ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir, "checkpoints")
@@ -166,11 +174,13 @@ def main(argv=None):
                 write_status(rank_dir, "failed", step, loss,
                              error="terminated")
                 return 143
-            # TP/PP/SP ranks are one data-parallel replica: identical batches
-            model_par = (tp_ctx is not None or pp_ctx is not None
-                         or sp_ctx is not None)
-            x, y = synthetic_batch(spec, cfg, device,
-                                   0 if model_par else rank, step)
+            # model-parallel peers form one data replica: same batch (for
+            # the TP x DP mesh, data_rank is the dp index)
+            if pp_ctx is not None or sp_ctx is not None:
+                dr = 0
+            else:
+                dr = data_rank
+            x, y = synthetic_batch(spec, cfg, device, dr, step)
             if sp_ctx is not None:  # ulysses: each rank takes its seq shard
                 if x.shape[1] % world:
                     raise ValueError(f"seq_len {x.shape[1]} not divisible "
