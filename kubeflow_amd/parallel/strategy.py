@@ -2,8 +2,9 @@
 
 SURVEY.md §2.14: DDP is the only strategy any BASELINE config requires.
 Implemented: DDP (bucketed all-reduce, ddp.py), TP (Megatron-style
-head/ffn sharding, tp.py) and PP (GPipe fill-drain stages, pp.py) — TP/PP
-in pure form (degree == world_size, llama family). SP/CP/EP/Ulysses remain
+head/ffn sharding, tp.py; composes with DP as a TP x DP mesh when
+degree < world_size), PP (1F1B/GPipe stages, pp.py) and Ulysses SP
+(sp.py) — llama family. Ring-SP/CP and EP remain
 declared-but-reserved enums so PyTorchJob specs stay forward-compatible,
 rejected with a clear error until a config demands them. The seam is the per-rank environment the gang launcher
 already provides (RANK/WORLD_SIZE/LOCAL_RANK + this descriptor serialized
@@ -12,6 +13,7 @@ into the worker spec as `parallelism`).
 Spec form (PyTorchJob template):
     parallelism: {strategy: ddp}                      # default
     parallelism: {strategy: tp, degree: 8}            # pure TP
+    parallelism: {strategy: tp, degree: 2}            # tp2 x dp(world/2)
 """
 from __future__ import annotations
 
@@ -22,7 +24,7 @@ from enum import Enum
 class Strategy(str, Enum):
     DDP = "ddp"      # implemented: bucketed all-reduce over RCCL/xGMI
     TP = "tp"        # implemented: Megatron-style head/ffn sharding (tp.py);
-                     # pure TP (degree == world_size) in v1
+                     # degree == world -> pure TP, degree < world -> TP x DP
     PP = "pp"        # implemented: GPipe fill-drain stages (pp.py);
                      # pure PP (degree == world_size) in v1
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
