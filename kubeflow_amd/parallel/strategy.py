@@ -25,7 +25,7 @@ class Strategy(str, Enum):
     DDP = "ddp"      # implemented: bucketed all-reduce over RCCL/xGMI
     TP = "tp"        # implemented: Megatron-style head/ffn sharding (tp.py);
                      # degree == world -> pure TP, degree < world -> TP x DP
-    PP = "pp"        # implemented: GPipe fill-drain stages (pp.py);
+    PP = "pp"        # implemented: 1F1B/GPipe stages (pp.py);
                      # pure PP (degree == world_size) in v1
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
     EP = "ep"        # reserved: expert parallel (all-to-all)
