@@ -143,8 +143,15 @@ class InferenceEngine:
             self._decode_step()
 
     def _admit(self) -> int:
+        # Fairness: each admit runs a full inline prefill (~10s of ms for a
+        # 4k prompt), so while sequences are decoding admit at most ONE new
+        # request per iteration — a burst of long prompts then stalls the
+        # running decodes by one prefill, not the whole burst. With nothing
+        # active, drain the queue freely. (True chunked prefill needs a
+        # rectangular-causal flash kernel — round-2, docs/ROUND1.md item 3.)
+        limit = 1 if self.active else self.max_batch
         n = 0
-        while len(self.active) < self.max_batch:
+        while len(self.active) < self.max_batch and n < limit:
             if not self.cache.has_free():
                 break  # no KV slot: leave requests queued — decode frees
                 # slots as sequences finish (an unconditional loop here

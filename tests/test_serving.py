@@ -180,3 +180,27 @@ def test_multi_gpu_serving_rejected(tmp_path):
                if e["reason"] == "InvalidSpec"]
         assert len(evs) == 1
         assert not plat.inference.gangs
+
+
+def test_admission_fairness_bounds_prefill_burst():
+    """With sequences decoding, at most one new prefill is admitted per
+    engine iteration, so a burst of prompts cannot stall active decodes by
+    the whole burst's prefill time."""
+    from kubeflow_amd.runtime.serving import InferenceEngine, Request
+
+    eng = InferenceEngine("llama-tiny", max_slots=8, smax=128, max_batch=8)
+    try:
+        # one active sequence decoding
+        r0 = Request(rid="r0", prompt=[1, 2, 3], max_new_tokens=64)
+        eng.pending.put(r0)
+        assert eng._admit() == 3 - 2  # idle engine: drains freely (1 here)
+        assert len(eng.active) == 1
+        # burst of 4 queued requests while r0 decodes
+        for i in range(4):
+            eng.pending.put(Request(rid=f"b{i}", prompt=[1], max_new_tokens=4))
+        assert eng._admit() == 1  # fairness cap
+        assert eng.pending.qsize() == 3
+        assert eng._admit() == 1
+        assert eng.pending.qsize() == 2
+    finally:
+        eng._stop = True
