@@ -37,6 +37,8 @@ def main():
     ap.add_argument("--micro-batch", type=int, default=0)  # 0 = per-strategy
     ap.add_argument("--seq", type=int, default=4096)
     ap.add_argument("--pp-microbatches", type=int, default=0)
+    ap.add_argument("--degree", type=int, default=0,
+                    help="tp only: degree < world builds a TP x DP mesh")
     args = ap.parse_args()
 
     rank, world, device = kdist.init_distributed()
@@ -48,11 +50,17 @@ def main():
 
     if args.strategy == "tp" and world > 1:
         from kubeflow_amd.parallel.tp import TpContext
-        ctx = TpContext.from_group(None)
+        deg = args.degree or world
+        if deg == world:
+            ctx, dp_group, dp_degree = TpContext.from_group(None), None, 1
+        else:
+            tp_group, dp_group, _tr, _dr = kdist.build_mesh(deg)
+            ctx, dp_degree = TpContext.from_group(tp_group), world // deg
         model = build_model(args.model, device=device, dtype=dtype, tp=ctx)
         ctx.sync_replicated(model)
-        trainer = Trainer(model, TrainConfig(warmup_steps=2), tp_ctx=ctx)
-        per_rank_tokens, job_tokens = B * S, B * S
+        trainer = Trainer(model, TrainConfig(warmup_steps=2), tp_ctx=ctx,
+                          dp_group=dp_group)
+        per_rank_tokens, job_tokens = B * S, B * S * dp_degree
     elif args.strategy == "ulysses" and world > 1:
         from kubeflow_amd.parallel.sp import SpContext
         ctx = SpContext.from_group(None)
@@ -107,7 +115,8 @@ def main():
             "strategy": args.strategy, "n_gpus": world,
             "value": round(job_tokens * args.steps / el, 2),
             "ms_per_step": round(el / args.steps * 1e3, 2),
-            "global_batch": B, "seq_len": S, "model": args.model,
+            "global_batch": job_tokens // S, "seq_len": S,
+            "model": args.model,
             "loss": round(float(loss), 4), "dtype": str(dtype).split(".")[-1],
         }), flush=True)
 
