@@ -169,3 +169,28 @@ def test_experiment_goal_early_stop(tmp_path):
         conds = {c["type"]: c for c in obj["status"]["conditions"]}
         assert conds["Succeeded"]["reason"] == "GoalReached"
         assert obj["status"]["trials"] <= 2  # stopped early, not 10
+
+
+def test_experiment_failure_budget(tmp_path):
+    """Trials that keep failing exhaust maxFailedTrialCount -> Experiment
+    Failed (katib failure-budget semantics)."""
+    with Platform(root_dir=str(tmp_path)) as plat:
+        exp = new_object("Experiment", "doomed", "default", spec={
+            "objective": {"type": "minimize", "objectiveMetricName": "loss"},
+            "algorithm": {"algorithmName": "random"},
+            "parallelTrialCount": 2,
+            "maxTrialCount": 8,
+            "maxFailedTrialCount": 2,
+            "parameters": PARAMS,
+            # nonexistent model -> every trial's worker exits 1
+            "trialTemplate": {"model": "no-such-model", "steps": 2,
+                              "gpus_per_replica": 0, "save_final": False,
+                              "replicas": 1},
+        })
+        plat.store.create(exp)
+        state, obj = _wait(plat.store, "Experiment", "doomed", "default",
+                           timeout=120)
+        assert state == "Failed", obj["status"]
+        conds = {c["type"]: c for c in obj["status"]["conditions"]}
+        assert conds["Failed"]["reason"] == "TooManyFailedTrials"
+        assert obj["status"]["trialsFailed"] >= 2
