@@ -450,7 +450,8 @@ def build_app(store: ObjectStore, scheduler=None,
     def dashboard(request: Request):
         """Minimal server-rendered dashboard shell — the central-dashboard
         equivalent (namespaces, workloads, activities, GPU utilization)."""
-        me = user_of(request)
+        import html as _html
+        me = _html.escape(user_of(request))
         util = scheduler.utilization() if scheduler else {}
         sections = []
         for plural, (kind, _) in PLURALS.items():
@@ -460,8 +461,8 @@ def build_app(store: ObjectStore, scheduler=None,
             if not objs:
                 continue
             rows = "".join(
-                f"<tr><td>{o['metadata'].get('namespace','')}</td>"
-                f"<td>{o['metadata']['name']}</td>"
+                f"<tr><td>{_html.escape(o['metadata'].get('namespace') or '')}</td>"
+                f"<td>{_html.escape(o['metadata']['name'])}</td>"
                 f"<td>{', '.join(c['type'] for c in o.get('status', {}).get('conditions', []) if c.get('status') == 'True') or '-'}</td></tr>"
                 for o in objs[:50])
             sections.append(
