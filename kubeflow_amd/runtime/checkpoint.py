@@ -24,7 +24,15 @@ def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int):
     step = trainer.step_num
     d = os.path.join(ckpt_dir, f"step-{step}")
     os.makedirs(d, exist_ok=True)
-    if rank == 0:
+    # TP/PP flat spaces hold DISTINCT shards/stages per rank: every rank
+    # must persist its own; replicated (DDP/ZeRO-1) keeps rank 0 canonical
+    sharded = (getattr(trainer, "tp", None) is not None
+               or getattr(trainer, "pp", None) is not None)
+    if sharded:
+        torch.save({"flat_data": trainer.flat.data,
+                    "param_names": trainer.flat.names},
+                   os.path.join(d, f"model-rank{rank}.pt"))
+    elif rank == 0:
         torch.save({"flat_data": trainer.flat.data,
                     "param_names": trainer.flat.names}, os.path.join(d, "model.pt"))
     torch.save({"step": step,
@@ -36,7 +44,7 @@ def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int):
     if rank == 0:
         with open(os.path.join(d, "meta.json"), "w") as f:
             json.dump({"step": step, "world_size": world, "model": model_name,
-                       "timestamp": time.time()}, f)
+                       "sharded": sharded, "timestamp": time.time()}, f)
         tmp = os.path.join(ckpt_dir, ".latest.tmp")
         with open(tmp, "w") as f:
             f.write(f"step-{step}")
@@ -60,8 +68,16 @@ def load(trainer, ckpt_dir: str, rank: int) -> int:
     d = latest_dir(ckpt_dir)
     if d is None:
         return 0
-    model = torch.load(os.path.join(d, "model.pt"), map_location="cpu",
-                       weights_only=False)
+    shard_path = os.path.join(d, f"model-rank{rank}.pt")
+    if os.path.exists(shard_path):  # TP/PP: this rank's own shard/stage
+        model_path = shard_path
+    elif os.path.exists(os.path.join(d, "model.pt")):
+        model_path = os.path.join(d, "model.pt")  # replicated (DDP/ZeRO)
+    else:
+        raise FileNotFoundError(
+            f"{d}: no model-rank{rank}.pt — sharded (TP/PP) checkpoints "
+            "require resuming at the same world size")
+    model = torch.load(model_path, map_location="cpu", weights_only=False)
     trainer.flat.data.copy_(model["flat_data"].to(trainer.flat.device))
     opt_path = os.path.join(d, f"optim-rank{rank}.pt")
     if not os.path.exists(opt_path):  # elastic restart with different world
