@@ -57,7 +57,15 @@ class WarmPool:
                             and len(self._idle) < self.size)
                 if not need:
                     break
-                self._spawn()
+                try:
+                    self._spawn()
+                except Exception:
+                    # a spawn failure (e.g. transient fork/OOM) must not
+                    # kill this thread: its death SIGKILLs every warm
+                    # worker via their PDEATHSIG
+                    import traceback
+                    traceback.print_exc()
+                    time.sleep(1.0)
 
     def _spawn(self):
         with self._lock:
