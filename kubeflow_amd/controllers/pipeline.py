@@ -22,7 +22,7 @@ from __future__ import annotations
 import time
 from typing import Dict, Optional
 
-from kubeflow_amd.api import new_object, set_condition
+from kubeflow_amd.api import AlreadyExistsError, new_object, set_condition
 from kubeflow_amd.api.objects import has_condition, owner_ref
 from kubeflow_amd.controllers.base import Reconciler, RequeueAfter
 
@@ -130,7 +130,7 @@ class PipelineRunReconciler(Reconciler):
         obj["metadata"]["ownerReferences"] = [owner_ref(run)]
         try:
             self.store.create(obj)
-        except Exception:
-            pass
+        except AlreadyExistsError:
+            pass  # re-reconcile after a crash mid-update: child exists
         self.store.record_event(run, "TaskStarted",
                                 f"{task['name']} -> {child_kind}/{child_name}")
