@@ -45,7 +45,27 @@ def needs_build() -> bool:
     return any(p.stat().st_mtime > out_mtime for p in deps)
 
 
+CSRC_CPP = Path(__file__).resolve().parent / "csrc_cpp"
+OUT_IO = Path(__file__).resolve().parent / "libkfio.so"
+CXX = os.environ.get("KF_CXX", "g++")
+
+
+def build_io(verbose: bool = True, force: bool = False) -> Path:
+    """Plain-C++ IO library (no HIP): threaded checkpoint reads/writes."""
+    src = CSRC_CPP / "kfio.cpp"
+    if (OUT_IO.exists() and not force
+            and OUT_IO.stat().st_mtime > src.stat().st_mtime):
+        return OUT_IO
+    cmd = [CXX, "-O2", "-std=c++17", "-shared", "-fPIC",
+           "-fvisibility=hidden", "-pthread", str(src), "-o", str(OUT_IO)]
+    if verbose:
+        print("[kfio]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return OUT_IO
+
+
 def build(verbose: bool = True, force: bool = False) -> Path:
+    build_io(verbose=verbose, force=force)
     if not force and not needs_build():
         return OUT
     BUILD.mkdir(exist_ok=True)

@@ -1,12 +1,22 @@
 """Checkpoint save/load with a PyTorchJob-compatible directory layout.
 
-Layout (BASELINE.json requires layout compat with PyTorchJob conventions):
+Layout (format 2 — BASELINE.json requires layout compat with PyTorchJob
+conventions):
     <ckpt_dir>/step-<K>/
-        meta.json            {step, world_size, model, timestamp}
-        model.pt             bf16 flat parameter space (rank 0)
-        optim-rank0.pt       fp32 master + moments + RNG (per rank; DDP keeps
-                             them replicated so rank 0's copy is canonical)
-    <ckpt_dir>/latest        text file: "step-<K>"
+        meta.json              {step, world_size, model, sharded, format,
+                                flat_dtype, timestamp}
+        model.bin              raw flat bf16/fp32 parameter bytes (rank 0;
+                               replicated layouts: DDP / ZeRO-1)
+        model-rank<r>.bin      per-rank flat shards instead, when the model
+                               is TP/PP-sharded
+        optim-<t>-rank<r>.bin  raw fp32 master/m/v (per rank)
+        optim-rank<r>.pt       small torch.save blob: step + RNG states
+    <ckpt_dir>/latest          text file: "step-<K>"
+
+The flat buffers go through kubeflow_amd.utils.fastio (libkfio: threaded
+pwrite/pread) — torch.save's single-threaded pickle stream is the wrong
+tool for the ~112 GB of flat state an 8B model checkpoints per node.
+Format-1 checkpoints (model.pt / monolithic optim-rank<r>.pt) still load.
 
 The reference platform itself delegates checkpointing to workloads + PVCs
 (SURVEY.md §5 checkpoint/resume); here it is a worker-runtime feature.
@@ -19,24 +29,30 @@ import time
 
 import torch
 
+from kubeflow_amd.utils import fastio
+
+
+def _is_sharded(trainer) -> bool:
+    return (getattr(trainer, "tp", None) is not None
+            or getattr(trainer, "pp", None) is not None)
+
 
 def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int):
     step = trainer.step_num
     d = os.path.join(ckpt_dir, f"step-{step}")
     os.makedirs(d, exist_ok=True)
-    # TP/PP flat spaces hold DISTINCT shards/stages per rank: every rank
-    # must persist its own; replicated (DDP/ZeRO-1) keeps rank 0 canonical
-    sharded = (getattr(trainer, "tp", None) is not None
-               or getattr(trainer, "pp", None) is not None)
+    sharded = _is_sharded(trainer)
+    # model flat buffer: every rank owns a distinct shard under TP/PP;
+    # replicated (DDP/ZeRO-1) keeps rank 0 canonical
     if sharded:
-        torch.save({"flat_data": trainer.flat.data,
-                    "param_names": trainer.flat.names},
-                   os.path.join(d, f"model-rank{rank}.pt"))
+        fastio.write_tensor(os.path.join(d, f"model-rank{rank}.bin"),
+                            trainer.flat.data)
     elif rank == 0:
-        torch.save({"flat_data": trainer.flat.data,
-                    "param_names": trainer.flat.names}, os.path.join(d, "model.pt"))
+        fastio.write_tensor(os.path.join(d, "model.bin"), trainer.flat.data)
+    # optimizer state (per rank; ZeRO-1 holds 1/world shards)
+    for tag, t in (("p32", trainer.p32), ("m", trainer.m), ("v", trainer.v)):
+        fastio.write_tensor(os.path.join(d, f"optim-{tag}-rank{rank}.bin"), t)
     torch.save({"step": step,
-                "p32": trainer.p32, "m": trainer.m, "v": trainer.v,
                 "rng": torch.get_rng_state(),
                 "cuda_rng": (torch.cuda.get_rng_state()
                              if torch.cuda.is_available() else None)},
@@ -44,7 +60,10 @@ def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int):
     if rank == 0:
         with open(os.path.join(d, "meta.json"), "w") as f:
             json.dump({"step": step, "world_size": world, "model": model_name,
-                       "sharded": sharded, "timestamp": time.time()}, f)
+                       "sharded": sharded, "format": 2,
+                       "flat_dtype": str(trainer.flat.data.dtype),
+                       "param_names": trainer.flat.names,
+                       "timestamp": time.time()}, f)
         tmp = os.path.join(ckpt_dir, ".latest.tmp")
         with open(tmp, "w") as f:
             f.write(f"step-{step}")
@@ -62,30 +81,52 @@ def latest_dir(ckpt_dir: str):
     return d if os.path.isdir(d) else None
 
 
+def _model_path(d: str, rank: int, ext: str):
+    shard = os.path.join(d, f"model-rank{rank}{ext}")
+    if os.path.exists(shard):
+        return shard
+    full = os.path.join(d, f"model{ext}")
+    if os.path.exists(full):
+        return full
+    raise FileNotFoundError(
+        f"{d}: no model-rank{rank}{ext} — sharded (TP/PP) checkpoints "
+        "require resuming at the same world size")
+
+
 def load(trainer, ckpt_dir: str, rank: int) -> int:
     """Restore trainer state from the latest checkpoint; returns step (0 if
     no checkpoint)."""
     d = latest_dir(ckpt_dir)
     if d is None:
         return 0
-    shard_path = os.path.join(d, f"model-rank{rank}.pt")
-    if os.path.exists(shard_path):  # TP/PP: this rank's own shard/stage
-        model_path = shard_path
-    elif os.path.exists(os.path.join(d, "model.pt")):
-        model_path = os.path.join(d, "model.pt")  # replicated (DDP/ZeRO)
-    else:
-        raise FileNotFoundError(
-            f"{d}: no model-rank{rank}.pt — sharded (TP/PP) checkpoints "
-            "require resuming at the same world size")
-    model = torch.load(model_path, map_location="cpu", weights_only=False)
-    trainer.flat.data.copy_(model["flat_data"].to(trainer.flat.device))
-    opt_path = os.path.join(d, f"optim-rank{rank}.pt")
-    if not os.path.exists(opt_path):  # elastic restart with different world
-        opt_path = os.path.join(d, "optim-rank0.pt")
-    opt = torch.load(opt_path, map_location="cpu", weights_only=False)
-    trainer.p32.copy_(opt["p32"].to(trainer.p32.device))
-    trainer.m.copy_(opt["m"].to(trainer.m.device))
-    trainer.v.copy_(opt["v"].to(trainer.v.device))
+    fmt = 1
+    meta_path = os.path.join(d, "meta.json")
+    if os.path.exists(meta_path):
+        with open(meta_path) as f:
+            fmt = json.load(f).get("format", 1)
+    if fmt >= 2:
+        fastio.read_into(_model_path(d, rank, ".bin"), trainer.flat.data)
+        for tag, t in (("p32", trainer.p32), ("m", trainer.m),
+                       ("v", trainer.v)):
+            path = os.path.join(d, f"optim-{tag}-rank{rank}.bin")
+            if not os.path.exists(path):  # elastic: replicated state only
+                path = os.path.join(d, f"optim-{tag}-rank0.bin")
+            fastio.read_into(path, t)
+        opt_path = os.path.join(d, f"optim-rank{rank}.pt")
+        if not os.path.exists(opt_path):
+            opt_path = os.path.join(d, "optim-rank0.pt")
+        opt = torch.load(opt_path, map_location="cpu", weights_only=False)
+    else:  # format-1 (torch.save monolith) compatibility
+        model = torch.load(_model_path(d, rank, ".pt"), map_location="cpu",
+                           weights_only=False)
+        trainer.flat.data.copy_(model["flat_data"].to(trainer.flat.device))
+        opt_path = os.path.join(d, f"optim-rank{rank}.pt")
+        if not os.path.exists(opt_path):  # elastic restart, different world
+            opt_path = os.path.join(d, "optim-rank0.pt")
+        opt = torch.load(opt_path, map_location="cpu", weights_only=False)
+        trainer.p32.copy_(opt["p32"].to(trainer.p32.device))
+        trainer.m.copy_(opt["m"].to(trainer.m.device))
+        trainer.v.copy_(opt["v"].to(trainer.v.device))
     trainer.step_num = opt["step"]
     rng = opt["rng"]
     torch.set_rng_state(rng if isinstance(rng, torch.Tensor) else rng)
