@@ -113,3 +113,40 @@ def test_parallelism_strategy_seam():
                                                    "degree": 2}})
     with pytest.raises(ValueError):
         ParallelismSpec.from_spec({"parallelism": {"strategy": "magic"}})
+
+
+def test_checkpoint_format1_compat(tmp_path):
+    """Legacy (torch.save monolith) checkpoints written before the raw
+    format-2 layout still load."""
+    import json
+    import os
+    import torch
+    from kubeflow_amd.models import build_model
+    from kubeflow_amd.runtime import Trainer, TrainConfig
+    from kubeflow_amd.runtime import checkpoint as ckpt
+
+    torch.manual_seed(0)
+    m = build_model("mnist-mlp", dtype=torch.float32)
+    tr = Trainer(m, TrainConfig(lr=1e-3, warmup_steps=1))
+    x = torch.randn(8, 784)
+    y = torch.randint(0, 10, (8,))
+    tr.step(x, y)
+
+    # hand-write a format-1 checkpoint from tr's state
+    d = tmp_path / "step-1"
+    d.mkdir(parents=True)
+    torch.save({"flat_data": tr.flat.data, "param_names": tr.flat.names},
+               d / "model.pt")
+    torch.save({"step": 1, "p32": tr.p32, "m": tr.m, "v": tr.v,
+                "rng": torch.get_rng_state(), "cuda_rng": None},
+               d / "optim-rank0.pt")
+    (d / "meta.json").write_text(json.dumps(
+        {"step": 1, "world_size": 1, "model": "mnist-mlp"}))  # no "format"
+    (tmp_path / "latest").write_text("step-1")
+
+    torch.manual_seed(99)
+    m2 = build_model("mnist-mlp", dtype=torch.float32)
+    tr2 = Trainer(m2, TrainConfig(lr=1e-3, warmup_steps=1))
+    assert ckpt.load(tr2, str(tmp_path), 0) == 1
+    assert torch.equal(tr.flat.data, tr2.flat.data)
+    assert torch.equal(tr.p32, tr2.p32)
