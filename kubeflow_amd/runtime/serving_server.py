@@ -57,12 +57,21 @@ def build_app(spec: dict) -> FastAPI:
     async def predict(req: HttpRequest):
         body = await req.json()
         instances = body.get("instances", [])
-        preds = []
+        # submit ALL instances first so they decode as one continuous
+        # batch; waiting per-instance would serialize the request
+        from kubeflow_amd.runtime.serving import Request
+        reqs = []
         for inst in instances:
-            r = engine.generate(inst.get("prompt_tokens", [1]),
-                                int(inst.get("max_new_tokens", 16)),
-                                temperature=float(
-                                    inst.get("temperature", 0.0)))
+            r = Request(rid=f"p{time.monotonic_ns()}-{len(reqs)}",
+                        prompt=list(inst.get("prompt_tokens", [1])),
+                        max_new_tokens=int(inst.get("max_new_tokens", 16)),
+                        temperature=float(inst.get("temperature", 0.0)))
+            engine.submit(r)
+            reqs.append(r)
+        preds = []
+        for r in reqs:
+            if not r.done.wait(120.0):
+                r.error = r.error or "timeout"
             preds.append({
                 "tokens": r.generated,
                 "error": r.error,
