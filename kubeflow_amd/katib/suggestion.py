@@ -82,6 +82,7 @@ class _Space:
                 v = math.exp(math.log(lo) + u * (math.log(hi) - math.log(lo)))
             else:
                 v = lo + u * (hi - lo)
+            v = min(hi, max(lo, v))  # exp/log round-trip can overshoot 1 ulp
             out[p["name"]] = (min(int(fs["max"]), max(int(fs["min"]),
                                                       round(v)))
                               if t == "int" else v)

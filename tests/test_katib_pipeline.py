@@ -194,3 +194,33 @@ def test_experiment_failure_budget(tmp_path):
         conds = {c["type"]: c for c in obj["status"]["conditions"]}
         assert conds["Failed"]["reason"] == "TooManyFailedTrials"
         assert obj["status"]["trialsFailed"] >= 2
+
+
+def test_space_unit_roundtrip_property():
+    """to_unit/from_unit roundtrip across the parameter space (hypothesis)."""
+    from hypothesis import given, settings, strategies as st
+    from kubeflow_amd.katib.suggestion import _Space
+
+    params = [
+        {"name": "lr", "parameterType": "double",
+         "feasibleSpace": {"min": "0.001", "max": "0.1", "logScale": True}},
+        {"name": "bs", "parameterType": "int",
+         "feasibleSpace": {"min": "8", "max": "64"}},
+        {"name": "opt", "parameterType": "categorical",
+         "feasibleSpace": {"list": ["adamw", "sgd", "lion"]}},
+    ]
+    space = _Space(params)
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.floats(0, 1), st.floats(0, 1), st.floats(0, 1))
+    def check(u1, u2, u3):
+        a = space.from_unit([u1, u2, u3])
+        assert 0.001 <= a["lr"] <= 0.1
+        assert 8 <= a["bs"] <= 64 and isinstance(a["bs"], int)
+        assert a["opt"] in ("adamw", "sgd", "lion")
+        # encode-decode is a projection: applying it twice is stable
+        b = space.from_unit(space.to_unit(a))
+        assert abs(b["lr"] - a["lr"]) < 1e-9 * max(1, abs(a["lr"]))
+        assert b["bs"] == a["bs"] and b["opt"] == a["opt"]
+
+    check()
