@@ -138,14 +138,26 @@ def submit_train(model: str = typer.Option("llama3-8b"),
                  micro_batch: int = typer.Option(4),
                  seq_len: int = typer.Option(4096),
                  lr: float = typer.Option(3e-4),
+                 strategy: str = typer.Option(
+                     "ddp", help="ddp | tp | pp | ulysses"),
+                 degree: int = typer.Option(
+                     0, help="tp: degree<replicas builds a TPxDP mesh"),
+                 zero: bool = typer.Option(
+                     False, help="ZeRO-1 sharded optimizer (ddp only)"),
                  namespace: str = typer.Option("default", "-n")):
-    """Submit a PyTorchJob (DDP over RCCL/xGMI, one process per GPU)."""
+    """Submit a PyTorchJob (one process per GPU over RCCL/xGMI)."""
     name = name or f"train-{model.replace('.', '-')}-{int(time.time()) % 100000}"
+    template = {"model": model, "steps": steps,
+                "micro_batch": micro_batch, "seq_len": seq_len,
+                "lr": lr, "gpus_per_replica": 1 if gpus else 0}
+    if strategy != "ddp":
+        template["parallelism"] = {"strategy": strategy,
+                                   "degree": degree or gpus}
+    if zero:
+        template["zero"] = True
     spec = {"pytorchReplicaSpecs": {"Worker": {
         "replicas": gpus, "restartPolicy": "OnFailure",
-        "template": {"model": model, "steps": steps,
-                     "micro_batch": micro_batch, "seq_len": seq_len,
-                     "lr": lr, "gpus_per_replica": 1 if gpus else 0}}}}
+        "template": template}}}
     _req("POST", f"/api/namespaces/{namespace}/pytorchjobs",
          {"name": name, "spec": spec})
     typer.secho(f"pytorchjob/{name} submitted", fg="green")
