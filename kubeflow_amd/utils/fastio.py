@@ -49,22 +49,27 @@ def _nthreads() -> int:
 
 
 def write_tensor(path: str, t: torch.Tensor):
-    """Write a tensor's raw bytes (host-staged if on GPU)."""
+    """Write a tensor's raw bytes (host-staged if on GPU). Atomic: bytes go
+    to a .tmp sibling and are renamed in — ftruncate preallocates the full
+    size, so a crash mid-write would otherwise leave a full-size file of
+    partial content that passes read_into's size check."""
     t = t.detach()
     if t.device.type != "cpu":
         t = t.cpu()
     t = t.contiguous()
     n = t.numel() * t.element_size()
+    tmp = path + ".tmp"
     lib = _load()
     if lib is not None:
-        rc = lib.kf_write_file(path.encode(), ctypes.c_void_p(t.data_ptr()),
+        rc = lib.kf_write_file(tmp.encode(), ctypes.c_void_p(t.data_ptr()),
                                n, _nthreads())
         if rc != 0:
-            raise OSError(rc, f"kf_write_file({path}): {os.strerror(rc)}")
-        return
-    with open(path, "wb") as f:  # fallback: plain write, same format
-        f.write(t.numpy().tobytes() if t.dtype != torch.bfloat16
-                else t.view(torch.uint8).numpy().tobytes())
+            raise OSError(rc, f"kf_write_file({tmp}): {os.strerror(rc)}")
+    else:
+        with open(tmp, "wb") as f:  # fallback: plain write, same format
+            f.write(t.numpy().tobytes() if t.dtype != torch.bfloat16
+                    else t.view(torch.uint8).numpy().tobytes())
+    os.replace(tmp, path)
 
 
 def file_size(path: str) -> int:
