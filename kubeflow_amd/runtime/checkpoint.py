@@ -37,7 +37,13 @@ def _is_sharded(trainer) -> bool:
             or getattr(trainer, "pp", None) is not None)
 
 
-def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int):
+def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int,
+         commit: bool = True):
+    """Write this rank's checkpoint files. With commit=True (single-rank
+    callers) rank 0 also publishes the `latest` marker; multi-rank callers
+    pass commit=False, barrier so every rank's files are durable, then call
+    commit() from rank 0 — otherwise a crash between rank 0 finishing and a
+    peer finishing leaves `latest` pointing at an incomplete step."""
     step = trainer.step_num
     d = os.path.join(ckpt_dir, f"step-{step}")
     os.makedirs(d, exist_ok=True)
@@ -64,11 +70,18 @@ def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int):
                        "flat_dtype": str(trainer.flat.data.dtype),
                        "param_names": trainer.flat.names,
                        "timestamp": time.time()}, f)
-        tmp = os.path.join(ckpt_dir, ".latest.tmp")
-        with open(tmp, "w") as f:
-            f.write(f"step-{step}")
-        os.replace(tmp, os.path.join(ckpt_dir, "latest"))
+        if commit:
+            commit_latest(ckpt_dir, step)
     return d
+
+
+def commit_latest(ckpt_dir: str, step: int):
+    """Atomically point `latest` at step-<step> (call after ALL ranks'
+    files are written)."""
+    tmp = os.path.join(ckpt_dir, ".latest.tmp")
+    with open(tmp, "w") as f:
+        f.write(f"step-{step}")
+    os.replace(tmp, os.path.join(ckpt_dir, "latest"))
 
 
 def latest_dir(ckpt_dir: str):

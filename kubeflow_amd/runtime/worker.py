@@ -67,6 +67,17 @@ def synthetic_batch(spec: dict, cfg, device, rank: int, step: int):
     return x.to(device), y.to(device)
 
 
+def _save_all_ranks(trainer, ckpt_dir, spec, rank, world):
+    """Checkpoint with the `latest` marker committed only after EVERY
+    rank's files are durable (checkpoint.py:save commit semantics)."""
+    kdist.barrier()
+    from kubeflow_amd.runtime import checkpoint as _ckpt
+    _ckpt.save(trainer, ckpt_dir, spec["model"], rank, world, commit=False)
+    kdist.barrier()
+    if rank == 0:
+        _ckpt.commit_latest(ckpt_dir, trainer.step_num)
+
+
 def main(argv=None):
     ap = argparse.ArgumentParser()
     ap.add_argument("--spec", required=True)
@@ -209,11 +220,9 @@ def main(argv=None):
                                                            1)})
                 win_t0, win_steps, win_items = time.time(), 0, 0
             if save_every and (step + 1) % save_every == 0:
-                kdist.barrier()
-                ckpt.save(trainer, ckpt_dir, spec["model"], rank, world)
+                _save_all_ranks(trainer, ckpt_dir, spec, rank, world)
         if spec.get("save_final", True):
-            kdist.barrier()
-            ckpt.save(trainer, ckpt_dir, spec["model"], rank, world)
+            _save_all_ranks(trainer, ckpt_dir, spec, rank, world)
         write_status(rank_dir, "succeeded", steps, loss,
                      metrics={"loss": None if loss is None else float(loss),
                               "loss_ema": ema})
