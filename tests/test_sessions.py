@@ -121,14 +121,14 @@ def test_culler_stops_idle_notebook(tmp_path, monkeypatch):
         plat.store.create(new_object("Notebook", "idle-nb", "ns", spec={}))
         obj = _wait(lambda: (lambda o: o if o["status"].get("readyReplicas")
                              else None)(
-            plat.store.get("Notebook", "idle-nb", "ns")), timeout=60)
+            plat.store.get("Notebook", "idle-nb", "ns")), timeout=120)
         # idle -> culled: stop annotation appears and session stops
         def culled():
             o = plat.store.get("Notebook", "idle-nb", "ns")
             ann = o["metadata"].get("annotations", {})
             return o if ("kubeflow-resource-stopped" in ann
                          and o["status"].get("readyReplicas") == 0) else None
-        obj = _wait(culled, timeout=60)
+        obj = _wait(culled, timeout=120)
         reasons = {e["reason"] for e in plat.store.events_for(obj)}
         assert "Culling" in reasons
 
