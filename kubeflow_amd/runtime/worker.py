@@ -211,9 +211,12 @@ def main(argv=None):
                 lval = float(loss)
                 ema = lval if ema is None else 0.9 * ema + 0.1 * lval
                 dt = max(1e-9, time.time() - win_t0)
+                gn = getattr(trainer, "last_grad_norm", None)
                 write_status(rank_dir, "running", step + 1, lval,
                              metrics={"loss": lval, "loss_ema": ema,
                                       "lr": trainer.lr_at(step),
+                                      "grad_norm": (None if gn is None
+                                                    else round(float(gn), 4)),
                                       "step_ms": round(dt / win_steps * 1e3,
                                                        2),
                                       "items_per_s": round(win_items / dt,
