@@ -7,7 +7,7 @@ from __future__ import annotations
 from typing import Callable, Dict
 
 from .llama import (LlamaConfig, LlamaModel, llama3_8b, llama3_1b,
-                    llama3_70b, llama_tiny, llama_tiny_mha)
+                    llama3_70b, llama_moe_tiny, llama_tiny, llama_tiny_mha)
 from .bert import (BertConfig, BertClassifier, bert_base, bert_base_hd128,
                    bert_tiny)
 from .mlp import MnistMLP
@@ -24,7 +24,7 @@ def _llama(cfg_fn):
         import torch
         return LlamaModel(cfg_fn(), device=device,
                           dtype=dtype or torch.bfloat16, tp=kw.get("tp"),
-                          sp=kw.get("sp"))
+                          sp=kw.get("sp"), ep=kw.get("ep"))
     return build
 
 
@@ -49,6 +49,7 @@ MODEL_CONFIGS: Dict[str, Callable] = {
     "llama3-70b": llama3_70b,
     "llama-tiny": llama_tiny,
     "llama-tiny-mha": llama_tiny_mha,
+    "llama-moe-tiny": llama_moe_tiny,
     "bert-base": bert_base,
     "bert-base-hd128": bert_base_hd128,
     "bert-tiny": bert_tiny,
@@ -69,6 +70,7 @@ MODEL_REGISTRY: Dict[str, Callable] = {
     "llama3-70b": _llama(llama3_70b),
     "llama-tiny": _llama(llama_tiny),
     "llama-tiny-mha": _llama(llama_tiny_mha),
+    "llama-moe-tiny": _llama(llama_moe_tiny),
     "bert-base": _bert(bert_base),
     "bert-base-hd128": _bert(bert_base_hd128),
     "bert-tiny": _bert(bert_tiny),
@@ -79,8 +81,8 @@ MODEL_REGISTRY: Dict[str, Callable] = {
 def build_model(name: str, device=None, dtype=None, **kw):
     if name not in MODEL_REGISTRY:
         raise KeyError(f"unknown model {name!r}; known: {sorted(MODEL_REGISTRY)}")
-    if (kw.get("tp") is not None or kw.get("sp") is not None) \
-            and not name.startswith("llama"):
-        raise ValueError(f"tensor/sequence parallelism is implemented for "
-                         f"the llama family only, not {name!r}")
+    if (kw.get("tp") is not None or kw.get("sp") is not None
+            or kw.get("ep") is not None) and not name.startswith("llama"):
+        raise ValueError(f"tensor/sequence/expert parallelism is implemented "
+                         f"for the llama family only, not {name!r}")
     return MODEL_REGISTRY[name](device=device, dtype=dtype, **kw)

@@ -41,14 +41,17 @@ class LlamaConfig:
     max_seq_len: int = 8192
     norm_eps: float = 1e-5
     init_std: float = 0.02
+    n_experts: int = 0   # 0 = dense MLP; >0 = mixture-of-experts blocks
+    top_k: int = 2       # experts per token (MoE only)
 
     @property
     def n_params(self) -> int:
         h, v = self.hidden_size, self.vocab_size
         qkv = h * (self.n_heads + 2 * self.n_kv_heads) * self.head_dim
         o = self.n_heads * self.head_dim * h
-        mlp = 3 * h * self.ffn_dim
-        return 2 * v * h + self.n_layers * (qkv + o + mlp + 2 * h) + h
+        mlp = 3 * h * self.ffn_dim * max(1, self.n_experts)
+        gate = h * self.n_experts
+        return 2 * v * h + self.n_layers * (qkv + o + mlp + gate + 2 * h) + h
 
 
 def llama3_8b() -> LlamaConfig:
@@ -79,6 +82,13 @@ def llama_tiny(vocab: int = 512) -> LlamaConfig:
                        max_seq_len=512)
 
 
+def llama_moe_tiny(vocab: int = 512) -> LlamaConfig:
+    """Tiny mixture-of-experts config (4 experts, top-2) for the EP tests."""
+    return LlamaConfig(name="llama-moe-tiny", vocab_size=vocab,
+                       hidden_size=256, n_layers=2, n_heads=2, n_kv_heads=1,
+                       ffn_dim=512, max_seq_len=512, n_experts=4, top_k=2)
+
+
 def llama_tiny_mha(vocab: int = 512) -> LlamaConfig:
     """Tiny MHA config (kv == q heads) whose head/ffn counts divide by 2 —
     used by the tensor-parallel tests and examples."""
@@ -97,8 +107,58 @@ class RMSNorm(nn.Module):
         return ops.rms_norm(x, self.weight, self.eps)
 
 
+class MoEMLP(nn.Module):
+    """Top-k gated mixture of SwiGLU experts (replaces the dense MLP when
+    cfg.n_experts > 0). Expert banks are [E_local, ...] parameter tensors;
+    under expert parallelism (parallel/ep.py) each rank owns a contiguous
+    E/N block and tokens travel by all-gather / reduce-scatter."""
+
+    def __init__(self, cfg: LlamaConfig, ep=None):
+        super().__init__()
+        h, f, E = cfg.hidden_size, cfg.ffn_dim, cfg.n_experts
+        self.cfg = cfg
+        self.ep = ep if (ep is not None and ep.world > 1) else None
+        n = self.ep.world if self.ep else 1
+        if E % n:
+            raise ValueError(f"EP degree {n} must divide n_experts ({E})")
+        self.local_e = E // n
+        self.e_lo = (self.ep.rank if self.ep else 0) * self.local_e
+        self.gate = nn.Linear(h, E, bias=False)  # replicated router
+        self.experts_w13 = nn.Parameter(torch.empty(self.local_e, 2 * f, h))
+        self.experts_w2 = nn.Parameter(torch.empty(self.local_e, h, f))
+        if self.ep:
+            self.experts_w13._ep_local = True
+            self.experts_w2._ep_local = True
+
+    def forward(self, x):
+        from kubeflow_amd.parallel import ep as epmod
+        cfg = self.cfg
+        B, S, h = x.shape
+        xf = x.reshape(B * S, h)
+        if self.ep:
+            xf = epmod.all_gather_cat(xf, self.ep)
+        # router runs (replicated weights) on whatever token set is local
+        logits = torch.nn.functional.linear(xf, self.gate.weight)
+        topv, topi = logits.topk(cfg.top_k, dim=-1)
+        weights = torch.softmax(topv.float(), dim=-1).to(x.dtype)
+        out = torch.zeros_like(xf)
+        for el in range(self.local_e):
+            eg = self.e_lo + el
+            sel, slot = (topi == eg).nonzero(as_tuple=True)
+            if sel.numel() == 0:
+                continue
+            rows = xf[sel]
+            y = ops.swiglu(
+                torch.nn.functional.linear(rows, self.experts_w13[el]))
+            y = torch.nn.functional.linear(y, self.experts_w2[el])
+            out.index_add_(0, sel, y * weights[sel, slot].unsqueeze(1))
+        if self.ep:
+            out = epmod.reduce_scatter_sum(out, self.ep)
+        return out.view(B, S, h)
+
+
 class LlamaBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig, tp=None, sp=None):
+    def __init__(self, cfg: LlamaConfig, tp=None, sp=None, ep=None):
         super().__init__()
         h, d = cfg.hidden_size, cfg.head_dim
         self.cfg = cfg
@@ -106,6 +166,11 @@ class LlamaBlock(nn.Module):
         self.sp = sp if (sp is not None and sp.world > 1) else None
         if self.tp is not None and self.sp is not None:
             raise ValueError("tp and sp (ulysses) are mutually exclusive")
+        if ep is not None and cfg.n_experts <= 0:
+            raise ValueError("expert parallelism requires an MoE config "
+                             "(n_experts > 0)")
+        if ep is not None and (self.tp is not None or self.sp is not None):
+            raise ValueError("ep composes with DP only in v1")
         if self.sp is not None and (cfg.n_heads % self.sp.world or
                                     cfg.n_kv_heads % self.sp.world):
             raise ValueError(
@@ -123,8 +188,13 @@ class LlamaBlock(nn.Module):
         self.wqkv = nn.Linear(h, (self.hq + 2 * self.hkv) * d, bias=False)
         self.wo = nn.Linear(self.hq * d, h, bias=False)
         self.mlp_norm = RMSNorm(h, cfg.norm_eps)
-        self.w13 = nn.Linear(h, 2 * self.ffn, bias=False)  # gate ++ up
-        self.w2 = nn.Linear(self.ffn, h, bias=False)
+        if cfg.n_experts > 0:
+            self.moe = MoEMLP(cfg, ep=ep)
+            self.w13 = self.w2 = None
+        else:
+            self.moe = None
+            self.w13 = nn.Linear(h, 2 * self.ffn, bias=False)  # gate ++ up
+            self.w2 = nn.Linear(self.ffn, h, bias=False)
         if self.tp:
             for lin in (self.wqkv, self.wo, self.w13, self.w2):
                 lin.weight._tp_sharded = True
@@ -175,6 +245,8 @@ class LlamaBlock(nn.Module):
         # residuals fused into the GEMM epilogue (addmm: C = input + A @ B)
         x = torch.addmm(x.view(-1, h), o.view(B * S, -1),
                         self.wo.weight.t()).view(B, S, h)
+        if self.moe is not None:
+            return x + self.moe(self.mlp_norm(x))
         y = ops.swiglu(F.linear(self.mlp_norm(x), self.w13.weight))
         return torch.addmm(x.view(-1, h), y.view(B * S, cfg.ffn_dim),
                            self.w2.weight.t()).view(B, S, h)
@@ -182,16 +254,17 @@ class LlamaBlock(nn.Module):
 
 class LlamaModel(nn.Module):
     def __init__(self, cfg: LlamaConfig, device=None, dtype=torch.bfloat16,
-                 tp=None, sp=None):
+                 tp=None, sp=None, ep=None):
         super().__init__()
         self.cfg = cfg
         self.tp = tp
         self.sp = sp
+        self.ep = ep
         factory = dict(device=device, dtype=dtype)
         with torch.device(device if device is not None else "cpu"):
             self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
             self.layers = nn.ModuleList(
-                [LlamaBlock(cfg, tp=tp, sp=sp)
+                [LlamaBlock(cfg, tp=tp, sp=sp, ep=ep)
                  for _ in range(cfg.n_layers)])
             self.final_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
             self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
@@ -210,7 +283,7 @@ class LlamaModel(nn.Module):
             if p.dim() >= 2:
                 p.normal_(0.0, std)
                 # scaled init for residual-out projections (GPT-2 style)
-                if name.endswith(("wo.weight", "w2.weight")):
+                if name.endswith(("wo.weight", "w2.weight", "experts_w2")):
                     p.mul_(1.0 / math.sqrt(2 * self.cfg.n_layers))
             else:
                 p.fill_(1.0)

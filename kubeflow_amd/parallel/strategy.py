@@ -28,12 +28,14 @@ class Strategy(str, Enum):
     PP = "pp"        # implemented: 1F1B/GPipe stages (pp.py);
                      # pure PP (degree == world_size) in v1
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
-    EP = "ep"        # reserved: expert parallel (all-to-all)
+    EP = "ep"        # implemented: expert parallel for MoE configs (ep.py);
+                     # gather-compute-scatter, pure form in v1
     ULYSSES = "ulysses"  # implemented: attention head-scatter SP (sp.py);
                          # pure form (degree == world_size) in v1
 
 
-IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP, Strategy.ULYSSES}
+IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP, Strategy.ULYSSES,
+               Strategy.EP}
 
 
 @dataclass

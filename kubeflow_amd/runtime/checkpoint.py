@@ -33,8 +33,9 @@ from kubeflow_amd.utils import fastio
 
 
 def _is_sharded(trainer) -> bool:
-    return (getattr(trainer, "tp", None) is not None
-            or getattr(trainer, "pp", None) is not None)
+    # TP/PP/EP flat spaces hold rank-distinct shards/stages/experts
+    return any(getattr(trainer, a, None) is not None
+               for a in ("tp", "pp", "ep"))
 
 
 def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int,

@@ -49,14 +49,26 @@ class Trainer:
 
     def __init__(self, model: torch.nn.Module, cfg: TrainConfig = TrainConfig(),
                  tp_ctx=None, pp_ctx=None, zero: bool = False,
-                 dp_group=None):
+                 dp_group=None, ep_ctx=None):
         import torch.distributed as dist
         self.model = model
         self.cfg = cfg
         self.flat = FlatParamSpace(model)
         self.tp = tp_ctx
         self.pp = pp_ctx
+        self.ep = ep_ctx
         self.zero = None
+        if ep_ctx is not None:
+            # EP: expert params are rank-local (a DDP all-reduce would
+            # corrupt them); EpContext.sync_grads runs after backward
+            self.ddp = _NullDDP()
+            self.p32 = self.flat.data.float()
+            self.m = torch.zeros_like(self.p32)
+            self.v = torch.zeros_like(self.p32)
+            self.wd_mask = (self.flat.build_wd_mask() if cfg.use_wd_mask
+                            else None)
+            self.step_num = 0
+            return
         if (zero and dist.is_initialized() and dist.get_world_size() > 1
                 and tp_ctx is None and pp_ctx is None):
             # ZeRO-1: optimizer state sharded 1/N; RS + shard-AdamW + AG
@@ -129,6 +141,8 @@ class Trainer:
             else:
                 loss.backward()
         self.ddp.finalize()
+        if self.ep is not None:
+            self.ep.sync_grads(self.flat)
         if self.zero is not None:
             self._zero_update()
         else:
@@ -141,7 +155,8 @@ class Trainer:
         the flat buffer, comm done)."""
         cfg = self.cfg
         if cfg.grad_clip > 0:
-            comm = self.tp if self.tp is not None else self.pp
+            comm = next((c for c in (self.tp, self.pp, self.ep)
+                         if c is not None), None)
             if comm is not None:
                 gnorm = comm.global_grad_norm(self.flat)
             else:

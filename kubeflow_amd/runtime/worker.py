@@ -105,6 +105,7 @@ def main(argv=None):
         tp_ctx = None
         pp_ctx = None
         sp_ctx = None
+        ep_ctx = None
         dp_group = None
         data_rank = rank
         if pspec.strategy == Strategy.TP and world > 1:
@@ -120,6 +121,13 @@ def main(argv=None):
                 tp_group, dp_group, _tp_rank, dp_rank = kdist.build_mesh(deg)
                 tp_ctx = TpContext.from_group(tp_group)
                 data_rank = dp_rank
+        elif pspec.strategy == Strategy.EP and world > 1:
+            from kubeflow_amd.parallel.ep import EpContext
+            if pspec.degree not in (1, world):
+                raise ValueError(
+                    f"pure EP requires degree == world_size ({world}); "
+                    f"got {pspec.degree}")
+            ep_ctx = EpContext.from_group(None)
         elif pspec.strategy == Strategy.ULYSSES and world > 1:
             from kubeflow_amd.parallel.sp import SpContext
             if pspec.degree not in (1, world):
@@ -144,9 +152,11 @@ def main(argv=None):
                                device=device, dtype=dtype)
         else:
             model = build_model(spec["model"], device=device, dtype=dtype,
-                                tp=tp_ctx, sp=sp_ctx)
+                                tp=tp_ctx, sp=sp_ctx, ep=ep_ctx)
         if tp_ctx is not None:
             tp_ctx.sync_replicated(model)
+        if ep_ctx is not None:
+            ep_ctx.sync_replicated(model)
         cfg = getattr(model, "cfg", None)
         tcfg = TrainConfig(
             lr=float(spec.get("lr", 3e-4)),
@@ -166,6 +176,7 @@ def main(argv=None):
                                 schedule=spec.get("pp_schedule", "1f1b"))
         else:
             trainer = Trainer(model, tcfg, tp_ctx=tp_ctx, dp_group=dp_group,
+                              ep_ctx=ep_ctx,
                               zero=bool(spec.get("zero", False)))
 
         ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir, "checkpoints")

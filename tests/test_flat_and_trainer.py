@@ -108,8 +108,11 @@ def test_parallelism_strategy_seam():
     pp = ParallelismSpec.from_spec({"parallelism": {"strategy": "pp",
                                                     "degree": 2}})
     assert pp.strategy == Strategy.PP and pp.degree == 2
-    with pytest.raises(NotImplementedError):
-        ParallelismSpec.from_spec({"parallelism": {"strategy": "ep",
+    ep = ParallelismSpec.from_spec({"parallelism": {"strategy": "ep",
+                                                    "degree": 2}})
+    assert ep.strategy == Strategy.EP
+    with pytest.raises(NotImplementedError):  # ring-SP still reserved
+        ParallelismSpec.from_spec({"parallelism": {"strategy": "sp",
                                                    "degree": 2}})
     with pytest.raises(ValueError):
         ParallelismSpec.from_spec({"parallelism": {"strategy": "magic"}})
