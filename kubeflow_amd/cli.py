@@ -139,7 +139,7 @@ def submit_train(model: str = typer.Option("llama3-8b"),
                  seq_len: int = typer.Option(4096),
                  lr: float = typer.Option(3e-4),
                  strategy: str = typer.Option(
-                     "ddp", help="ddp | tp | pp | ulysses"),
+                     "ddp", help="ddp | tp | pp | ulysses | ep"),
                  degree: int = typer.Option(
                      0, help="tp: degree<replicas builds a TPxDP mesh"),
                  zero: bool = typer.Option(
