@@ -3,17 +3,19 @@
 SURVEY.md §2.14: DDP is the only strategy any BASELINE config requires.
 Implemented: DDP (bucketed all-reduce, ddp.py), TP (Megatron-style
 head/ffn sharding, tp.py; composes with DP as a TP x DP mesh when
-degree < world_size), PP (1F1B/GPipe stages, pp.py) and Ulysses SP
-(sp.py) — llama family. Ring-SP/CP and EP remain
-declared-but-reserved enums so PyTorchJob specs stay forward-compatible,
-rejected with a clear error until a config demands them. The seam is the per-rank environment the gang launcher
-already provides (RANK/WORLD_SIZE/LOCAL_RANK + this descriptor serialized
-into the worker spec as `parallelism`).
+degree < world_size), PP (1F1B/GPipe stages, pp.py), Ulysses SP (sp.py)
+and EP (expert parallel for MoE configs, ep.py) — llama family. Ring-SP/CP
+remains a declared-but-reserved enum so PyTorchJob specs stay
+forward-compatible, rejected with a clear error until a config demands it
+(it needs a ring-attention kernel). The seam is the per-rank environment
+the gang launcher already provides (RANK/WORLD_SIZE/LOCAL_RANK + this
+descriptor serialized into the worker spec as `parallelism`).
 
 Spec form (PyTorchJob template):
     parallelism: {strategy: ddp}                      # default
     parallelism: {strategy: tp, degree: 8}            # pure TP
     parallelism: {strategy: tp, degree: 2}            # tp2 x dp(world/2)
+    parallelism: {strategy: ep, degree: 8}            # MoE expert parallel
 """
 from __future__ import annotations
 
