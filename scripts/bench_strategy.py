@@ -30,7 +30,7 @@ from kubeflow_amd.runtime import PpTrainer, Trainer, TrainConfig  # noqa: E402
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--strategy", default="tp",
-                    choices=["ddp", "zero", "tp", "ulysses", "pp"])
+                    choices=["ddp", "zero", "tp", "ulysses", "pp", "ep"])
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
@@ -46,7 +46,7 @@ def main():
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     S = args.seq
     B = args.micro_batch or {"ddp": 6, "zero": 6, "tp": 8, "ulysses": 2,
-                             "pp": 8}[args.strategy]
+                             "pp": 8, "ep": 4}[args.strategy]
 
     if args.strategy == "tp" and world > 1:
         from kubeflow_amd.parallel.tp import TpContext
@@ -68,6 +68,13 @@ def main():
         trainer = Trainer(model, TrainConfig(warmup_steps=2))
         S_local = S // world
         per_rank_tokens, job_tokens = B * S_local, B * S
+    elif args.strategy == "ep" and world > 1:
+        from kubeflow_amd.parallel.ep import EpContext
+        ctx = EpContext.from_group(None)
+        model = build_model(args.model, device=device, dtype=dtype, ep=ctx)
+        ctx.sync_replicated(model)
+        trainer = Trainer(model, TrainConfig(warmup_steps=2), ep_ctx=ctx)
+        per_rank_tokens, job_tokens = B * S, B * S * world
     elif args.strategy == "pp" and world > 1:
         from kubeflow_amd.models.llama import LlamaStage
         from kubeflow_amd.parallel.pp import PpContext
