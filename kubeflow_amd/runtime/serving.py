@@ -220,6 +220,8 @@ class InferenceEngine:
         o = ops.flash_attention(q, k, v, causal=True)
         o = layer.wo(o.reshape(B, S, cfg.n_heads * cfg.head_dim))
         x = x + o
+        if layer.moe is not None:  # MoE predictor (dense experts, EP=1)
+            return x + layer.moe(layer.mlp_norm(x))
         g, u = torch.nn.functional.linear(
             layer.mlp_norm(x), layer.w13.weight).split(
                 [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
@@ -249,8 +251,11 @@ class InferenceEngine:
                                      self.cache.v[li], slots, lens)
             o = layer.wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim))
             x = x + o
-            y = ops.swiglu(F.linear(layer.mlp_norm(x), layer.w13.weight))
-            x = x + layer.w2(y)
+            if layer.moe is not None:
+                x = x + layer.moe(layer.mlp_norm(x))
+            else:
+                y = ops.swiglu(F.linear(layer.mlp_norm(x), layer.w13.weight))
+                x = x + layer.w2(y)
         x = self.model.final_norm(x)
         logits = F.linear(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)

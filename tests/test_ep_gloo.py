@@ -129,3 +129,28 @@ def test_ep_pytorchjob_e2e(tmp_path):
                 break
             time.sleep(0.5)
         assert has_condition(obj, "Succeeded"), obj["status"]
+
+
+def test_moe_serving_decode_matches_full_forward():
+    """The serving engine's incremental decode path supports MoE models
+    (dense experts) and matches the full forward pass."""
+    from kubeflow_amd.runtime.serving import InferenceEngine, Request
+
+    torch.manual_seed(0)
+    eng = InferenceEngine("llama-moe-tiny", max_slots=2, smax=128,
+                          max_batch=2)
+    model = eng.model
+    prompt = [3, 14, 15, 9, 2, 6]
+    req = Request(rid="t", prompt=list(prompt), max_new_tokens=3)
+    req.slot = eng.cache.alloc()
+    eng._prefill(req)
+    eng.active = [req]
+    eng._decode_step()
+    with torch.no_grad():
+        t1 = int(model(torch.tensor([prompt], dtype=torch.int64))
+                 [0, -1].argmax())
+    assert req.generated[0] == t1, (req.generated, t1)
+    with torch.no_grad():
+        t2 = int(model(torch.tensor([prompt + [t1]], dtype=torch.int64))
+                 [0, -1].argmax())
+    assert req.generated[1] == t2, (req.generated, t2)
