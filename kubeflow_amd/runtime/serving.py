@@ -101,8 +101,13 @@ class InferenceEngine:
         # bucket (the ~10 kernels x 32 layers of launch overhead dominate
         # small-batch decode latency otherwise). One cache slot is reserved
         # as the padding target for bucket rows beyond the live batch.
+        # MoE routing (topk/nonzero/index_add) is data-dependent control
+        # flow — a captured graph would replay the CAPTURED batch's routing
+        # on every decode, silently corrupting outputs. Eager decode for
+        # MoE until a capture-safe dispatch (dense masked einsum) exists.
         self.use_graphs = (self.device.type == "cuda"
-                           and os.environ.get("KF_SERVE_GRAPH", "1") == "1")
+                           and os.environ.get("KF_SERVE_GRAPH", "1") == "1"
+                           and getattr(cfg, "n_experts", 0) == 0)
         self._graphs = {}
         self._pad_slot = self.cache.alloc() if self.use_graphs else None
 
