@@ -428,22 +428,33 @@ def build_app(store: ObjectStore, scheduler=None,
         return ok(me, metric=which, scheduler=util, gpus=gpus)
 
     # ---------------------------------------------------------------- logs
-    @app.get("/api/namespaces/{ns}/pytorchjobs/{name}/logs")
-    def job_logs(ns: str, name: str, request: Request, rank: int = 0,
-                 tail: int = 200):
-        user = user_of(request)
-        authz(user, ns, "get")
+    def _tail_log(user, subdir, kind, ns, name, rank, tail):
         if root_dir is None:
             raise ApiError(404, "no logs root configured")
-        job = store.get("PyTorchJob", name, ns)
-        uid = job["metadata"]["uid"]
-        path = os.path.join(root_dir, "jobs", ns, f"{name}-{uid[:8]}",
+        obj = store.get(kind, name, ns)
+        uid = obj["metadata"]["uid"]
+        path = os.path.join(root_dir, subdir, ns, f"{name}-{uid[:8]}",
                             f"rank-{rank}", "worker.log")
         if not os.path.exists(path):
             raise ApiError(404, f"no log for rank {rank}")
         with open(path, errors="replace") as f:
             lines = f.readlines()[-tail:]
         return ok(user, logs="".join(lines))
+
+    @app.get("/api/namespaces/{ns}/pytorchjobs/{name}/logs")
+    def job_logs(ns: str, name: str, request: Request, rank: int = 0,
+                 tail: int = 200):
+        user = user_of(request)
+        authz(user, ns, "get")
+        return _tail_log(user, "jobs", "PyTorchJob", ns, name, rank, tail)
+
+    @app.get("/api/namespaces/{ns}/inferenceservices/{name}/logs")
+    def svc_logs(ns: str, name: str, request: Request, rank: int = 0,
+                 tail: int = 200):
+        user = user_of(request)
+        authz(user, ns, "get")
+        return _tail_log(user, "serving", "InferenceService", ns, name,
+                         rank, tail)
 
     # ------------------------------------------------ central dashboard UI
     @app.get("/")

@@ -99,6 +99,13 @@ def test_inference_service_e2e(tmp_path):
         assert out["predictions"][0]["latency_ms"] is not None
         with urllib.request.urlopen(f"{url}/v1/models", timeout=10) as r:
             assert json.loads(r.read())["models"] == ["tiny-svc"]
+        # predictor logs via the platform API
+        from fastapi.testclient import TestClient
+        from kubeflow_amd.api.server import build_app
+        api = TestClient(build_app(plat.store, scheduler=plat.scheduler,
+                                   root_dir=plat.root_dir))
+        r = api.get("/api/namespaces/default/inferenceservices/tiny-svc/logs")
+        assert r.status_code == 200 and "logs" in r.json()
 
 
 def test_sampling_temperature_cpu():
