@@ -75,7 +75,7 @@ def _pp_worker(rank, world, port, results, schedule="1f1b", M=2):
                           TrainConfig(lr=1e-3, warmup_steps=1, grad_accum=M))
         micros = list(zip(toks.split(4 // M), tgts.split(4 // M)))
         losses_pp = []
-        for _ in range(3):
+        for _ in range(2):
             losses_pp.append(float(tr_pp.step(toks, tgts)))
             tr_full.step(lambda i: micros[i])
         # the real invariant: identical math => stage params track the
