@@ -64,6 +64,5 @@ class ParallelismSpec:
         if self.strategy not in IMPLEMENTED:
             raise NotImplementedError(
                 f"parallelism strategy {self.strategy.value!r} is reserved "
-                "but not implemented in v1 — DDP (one process per MI355X, "
-                "RCCL over xGMI) is the supported strategy; see "
-                "SURVEY.md §2.14 for the extension seam")
+                "but not implemented (it needs a ring-attention kernel) — "
+                f"implemented: {sorted(s.value for s in IMPLEMENTED)}")
