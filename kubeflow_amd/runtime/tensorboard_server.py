@@ -40,8 +40,12 @@ def scan_runs(logdir: str):
         try:
             with open(path) as f:
                 for line in f:
-                    if line.strip():
+                    if not line.strip():
+                        continue
+                    try:
                         series.append(json.loads(line))
+                    except json.JSONDecodeError:
+                        continue  # torn concurrent append: skip the line
         except OSError:
             continue
         runs.setdefault(run, []).extend(series)
