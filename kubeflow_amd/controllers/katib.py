@@ -36,9 +36,16 @@ class ExperimentReconciler(Reconciler):
     def __init__(self, store: ObjectStore):
         super().__init__(store)
         self._suggesters: Dict[str, object] = {}
+        self._key_uid: Dict[tuple, str] = {}
+
+    def on_deleted(self, namespace, name):
+        uid = self._key_uid.pop((namespace, name), None)
+        if uid:
+            self._suggesters.pop(uid, None)
 
     def reconcile(self, namespace: Optional[str], name: str) -> None:
         exp = self.store.get(self.kind, name, namespace)
+        self._key_uid[(namespace, name)] = exp["metadata"]["uid"]
         if any(has_condition(exp, t) for t in ("Succeeded", "Failed")):
             return
         spec = exp["spec"]
@@ -95,7 +102,9 @@ class ExperimentReconciler(Reconciler):
                     ((minimize and best[1] <= goal) or
                      (not minimize and best[1] >= goal)))
 
-        if len(failed) > max_failed:
+        # upstream semantics: experiment fails when the failure count
+        # REACHES maxFailedTrialCount (>=, not >)
+        if len(failed) >= max_failed and max_failed > 0:
             set_condition(exp, "Failed", "True", "TooManyFailedTrials",
                           f"{len(failed)} trials failed")
             self.store.update(exp, check_version=False)
