@@ -119,10 +119,9 @@ def test_culler_stops_idle_notebook(tmp_path, monkeypatch):
     with Platform(root_dir=str(tmp_path)) as plat:
         assert plat.notebook.enable_culling
         plat.store.create(new_object("Notebook", "idle-nb", "ns", spec={}))
-        obj = _wait(lambda: (lambda o: o if o["status"].get("readyReplicas")
-                             else None)(
-            plat.store.get("Notebook", "idle-nb", "ns")), timeout=120)
-        # idle -> culled: stop annotation appears and session stops
+        # NOTE: with a ~6 ms idle budget the ready (readyReplicas==1) state
+        # can be culled away between store polls — don't wait on it, it is
+        # incidental; the invariant under test is the cull itself.
         def culled():
             o = plat.store.get("Notebook", "idle-nb", "ns")
             ann = o["metadata"].get("annotations", {})
