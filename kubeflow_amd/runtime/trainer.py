@@ -215,13 +215,25 @@ class Trainer:
 
 class PpTrainer(Trainer):
     """Trainer over one pipeline stage: forward/backward run through the
-    GPipe schedule (parallel/pp.py); clip + fused AdamW are stage-local with
-    a pipeline-global grad norm. `microbatches` plays grad_accum's role."""
+    1F1B/GPipe schedule (parallel/pp.py); clip + fused AdamW are stage-local
+    with a pipeline-global grad norm. `microbatches` plays grad_accum's
+    role. With a `dp_group` (PP x DP mesh: same-stage peers across data
+    replicas) stage grads are averaged over it after the schedule."""
 
     def __init__(self, stage: torch.nn.Module, cfg: TrainConfig,
-                 pp_ctx, microbatches: int, schedule: str = "1f1b"):
+                 pp_ctx, microbatches: int, schedule: str = "1f1b",
+                 dp_group=None):
         super().__init__(stage, cfg, pp_ctx=pp_ctx)
+        import torch.distributed as dist
         from kubeflow_amd.parallel.pp import PipelineRunner
+        self.dp_group = dp_group
+        self._dp_world = (dist.get_world_size(dp_group)
+                          if dp_group is not None else 1)
+        if dp_group is not None and self._dp_world > 1:
+            # same-stage peers start identical
+            dist.broadcast(self.flat.data,
+                           src=dist.get_global_rank(dp_group, 0),
+                           group=dp_group)
         self.runner = PipelineRunner(
             stage, pp_ctx, microbatches,
             act_dtype=self.flat.data.dtype,
@@ -230,5 +242,9 @@ class PpTrainer(Trainer):
     def step(self, tokens, targets) -> torch.Tensor:
         self.flat.zero_grad()
         loss = self.runner.step(tokens, targets)
+        if self.dp_group is not None and self._dp_world > 1:
+            import torch.distributed as dist
+            dist.all_reduce(self.flat.grad, group=self.dp_group)
+            self.flat.grad.div_(self._dp_world)
         self._clip_and_update()
         return loss

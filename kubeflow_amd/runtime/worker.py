@@ -135,21 +135,28 @@ def main(argv=None):
                     f"pure ulysses requires degree == world_size ({world}); "
                     f"got {pspec.degree}")
             sp_ctx = SpContext.from_group(None)
+            data_rank = 0  # ranks share the batch (each takes a seq shard)
         elif pspec.strategy == Strategy.PP and world > 1:
             from kubeflow_amd.parallel.pp import PpContext
-            if pspec.degree not in (1, world):
-                raise ValueError(
-                    f"pure PP requires degree == world_size ({world}); "
-                    f"got {pspec.degree}")
+            deg = pspec.degree if pspec.degree > 1 else world
+            if world % deg:
+                raise ValueError(f"world_size {world} not divisible by "
+                                 f"pp degree {deg}")
             if not spec["model"].startswith("llama"):
                 raise ValueError("pipeline parallelism is implemented for "
                                  f"the llama family only, not {spec['model']!r}")
-            pp_ctx = PpContext.from_group(None)
+            if deg == world:
+                pp_ctx = PpContext.from_group(None)  # pure PP
+                data_rank = 0
+            else:  # PP x DP mesh: contiguous stage chains, strided DP
+                pp_group, dp_group, _st, dp_rank = kdist.build_mesh(deg)
+                pp_ctx = PpContext.from_group(pp_group)
+                data_rank = dp_rank
         if pp_ctx is not None:
             from kubeflow_amd.models import model_config
             from kubeflow_amd.models.llama import LlamaStage
-            model = LlamaStage(model_config(spec["model"]), rank, world,
-                               device=device, dtype=dtype)
+            model = LlamaStage(model_config(spec["model"]), pp_ctx.rank,
+                               pp_ctx.world, device=device, dtype=dtype)
         else:
             model = build_model(spec["model"], device=device, dtype=dtype,
                                 tp=tp_ctx, sp=sp_ctx, ep=ep_ctx)
@@ -171,9 +178,10 @@ def main(argv=None):
             # default: one microbatch per stage keeps the pipe full and
             # always divides; callers can override via pp_microbatches
             micros = int(spec.get("pp_microbatches", 0)) or (
-                world if mb % world == 0 else 1)
+                pp_ctx.world if mb % pp_ctx.world == 0 else 1)
             trainer = PpTrainer(model, tcfg, pp_ctx, micros,
-                                schedule=spec.get("pp_schedule", "1f1b"))
+                                schedule=spec.get("pp_schedule", "1f1b"),
+                                dp_group=dp_group)
         else:
             trainer = Trainer(model, tcfg, tp_ctx=tp_ctx, dp_group=dp_group,
                               ep_ctx=ep_ctx,
@@ -196,13 +204,9 @@ def main(argv=None):
                 write_status(rank_dir, "failed", step, loss,
                              error="terminated")
                 return 143
-            # model-parallel peers form one data replica: same batch (for
-            # the TP x DP mesh, data_rank is the dp index)
-            if pp_ctx is not None or sp_ctx is not None:
-                dr = 0
-            else:
-                dr = data_rank
-            x, y = synthetic_batch(spec, cfg, device, dr, step)
+            # model-parallel peers form one data replica: data_rank is 0
+            # (pure) or the dp index (TP/PP x DP meshes)
+            x, y = synthetic_batch(spec, cfg, device, data_rank, step)
             if sp_ctx is not None:  # ulysses: each rank takes its seq shard
                 if x.shape[1] % world:
                     raise ValueError(f"seq_len {x.shape[1]} not divisible "
