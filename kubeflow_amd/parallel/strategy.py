@@ -28,7 +28,7 @@ class Strategy(str, Enum):
     TP = "tp"        # implemented: Megatron-style head/ffn sharding (tp.py);
                      # degree == world -> pure TP, degree < world -> TP x DP
     PP = "pp"        # implemented: 1F1B/GPipe stages (pp.py);
-                     # pure PP (degree == world_size) in v1
+                     # degree == world -> pure PP, degree < world -> PP x DP
     SP = "sp"        # reserved: sequence/context parallel (ring attention)
     EP = "ep"        # implemented: expert parallel for MoE configs (ep.py);
                      # gather-compute-scatter, pure form in v1
