@@ -130,12 +130,21 @@ def main(argv=None):
             ep_ctx = EpContext.from_group(None)
         elif pspec.strategy == Strategy.ULYSSES and world > 1:
             from kubeflow_amd.parallel.sp import SpContext
-            if pspec.degree not in (1, world):
-                raise ValueError(
-                    f"pure ulysses requires degree == world_size ({world}); "
-                    f"got {pspec.degree}")
-            sp_ctx = SpContext.from_group(None)
-            data_rank = 0  # ranks share the batch (each takes a seq shard)
+            deg = pspec.degree if pspec.degree > 1 else world
+            if world % deg:
+                raise ValueError(f"world_size {world} not divisible by "
+                                 f"ulysses degree {deg}")
+            if deg == world:
+                sp_ctx = SpContext.from_group(None)  # pure ulysses
+                data_rank = 0
+            else:  # SP x DP mesh: contiguous seq groups, strided DP.
+                # Params stay replicated across ALL ranks and every rank
+                # holds a distinct token subset, so the default WORLD-wide
+                # DDP grad average is exactly right — only the sequence
+                # group and batch index change.
+                sp_group, _dp_g, _sr, dp_rank = kdist.build_mesh(deg)
+                sp_ctx = SpContext.from_group(sp_group)
+                data_rank = dp_rank
         elif pspec.strategy == Strategy.PP and world > 1:
             from kubeflow_amd.parallel.pp import PpContext
             deg = pspec.degree if pspec.degree > 1 else world
@@ -207,13 +216,13 @@ def main(argv=None):
             # model-parallel peers form one data replica: data_rank is 0
             # (pure) or the dp index (TP/PP x DP meshes)
             x, y = synthetic_batch(spec, cfg, device, data_rank, step)
-            if sp_ctx is not None:  # ulysses: each rank takes its seq shard
-                if x.shape[1] % world:
+            if sp_ctx is not None:  # ulysses: take this rank's seq shard
+                if x.shape[1] % sp_ctx.world:
                     raise ValueError(f"seq_len {x.shape[1]} not divisible "
-                                     f"by ulysses degree {world}")
-                s = x.shape[1] // world
-                x = x[:, rank * s:(rank + 1) * s].contiguous()
-                y = y[:, rank * s:(rank + 1) * s].contiguous()
+                                     f"by ulysses degree {sp_ctx.world}")
+                s = x.shape[1] // sp_ctx.world
+                x = x[:, sp_ctx.rank * s:(sp_ctx.rank + 1) * s].contiguous()
+                y = y[:, sp_ctx.rank * s:(sp_ctx.rank + 1) * s].contiguous()
             loss = trainer.step(x, y)
             win_steps += 1
             win_items += x.numel()  # tokens (lm) or features processed
