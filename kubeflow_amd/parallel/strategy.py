@@ -33,7 +33,7 @@ class Strategy(str, Enum):
     EP = "ep"        # implemented: expert parallel for MoE configs (ep.py);
                      # gather-compute-scatter, pure form in v1
     ULYSSES = "ulysses"  # implemented: attention head-scatter SP (sp.py);
-                         # pure form (degree == world_size) in v1
+                         # degree == world -> pure, degree < world -> SP x DP
 
 
 IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP, Strategy.ULYSSES,
