@@ -118,3 +118,34 @@ def test_tp_mesh_pytorchjob_e2e(tmp_path):
                 break
             time.sleep(0.5)
         assert has_condition(obj, "Succeeded"), obj["status"]
+
+
+@pytest.mark.parametrize("strategy,model", [("pp", "llama-tiny"),
+                                            ("ulysses", "llama-tiny-mha")])
+def test_mesh_pytorchjob_e2e_other_strategies(tmp_path, strategy, model):
+    """4-replica PyTorchJob at degree 2 -> (pp|sp)2 x dp2 mesh through the
+    real worker path."""
+    import time
+    from kubeflow_amd.api import new_object
+    from kubeflow_amd.api.objects import has_condition
+    from kubeflow_amd.platform import Platform
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        plat.store.create(new_object("PyTorchJob", f"{strategy}-mesh",
+                                     "default",
+                                     spec={"pytorchReplicaSpecs": {"Worker": {
+            "replicas": 4, "restartPolicy": "Never",
+            "template": {"model": model, "steps": 3,
+                         "micro_batch": 2, "seq_len": 64,
+                         "gpus_per_replica": 0, "status_every": 1,
+                         "save_final": False,
+                         "parallelism": {"strategy": strategy,
+                                         "degree": 2}}}}}))
+        deadline = time.time() + 240
+        while time.time() < deadline:
+            obj = plat.store.get("PyTorchJob", f"{strategy}-mesh", "default")
+            assert not has_condition(obj, "Failed"), obj["status"]
+            if has_condition(obj, "Succeeded"):
+                break
+            time.sleep(0.5)
+        assert has_condition(obj, "Succeeded"), obj["status"]
