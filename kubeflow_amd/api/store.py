@@ -19,6 +19,7 @@ from __future__ import annotations
 import copy
 import json
 import os
+import re
 import threading
 import time
 from typing import Callable, Dict, List, Optional, Tuple
@@ -56,6 +57,46 @@ class Event:
 
 
 Key = Tuple[str, Optional[str], str]  # (kind, namespace, name)
+
+
+class InvalidNameError(StoreError):
+    status = 422
+
+
+# Kube name validation (the reference gets this from the kube-apiserver;
+# here the store is the apiserver). Names are DNS-1123 subdomains (dots
+# allowed — Events use dotted names), namespaces are DNS-1123 labels.
+# Without this, a crafted name like '../../x' flows into os.path.join in
+# controllers that derive workdirs/log paths from metadata.
+_DNS1123_LABEL = re.compile(r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?$")
+_DNS1123_SUBDOMAIN = re.compile(
+    r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?(\.[a-z0-9]([-a-z0-9]*[a-z0-9])?)*$")
+# RBAC object names in kube are path-segment validated, not DNS-1123
+# (the reference's owner RoleBinding is literally "namespaceAdmin",
+# profile_controller.go:223-244).
+_PATH_SEGMENT_KINDS = frozenset({"RoleBinding", "Role", "ClusterRole",
+                                 "ClusterRoleBinding"})
+
+
+def validate_metadata(obj: KfObject) -> None:
+    m = obj.get("metadata") or {}
+    name = m.get("name")
+    if obj.get("kind") in _PATH_SEGMENT_KINDS:
+        if not isinstance(name, str) or not name or len(name) > 253 or \
+                "/" in name or "%" in name or name in (".", ".."):
+            raise InvalidNameError(
+                f"invalid name {name!r}: must be a valid path segment")
+    elif not isinstance(name, str) or len(name) > 253 or \
+            not _DNS1123_SUBDOMAIN.match(name):
+        raise InvalidNameError(
+            f"invalid name {name!r}: must be a DNS-1123 subdomain "
+            "([a-z0-9-.], start/end alphanumeric, <=253 chars)")
+    ns = m.get("namespace")
+    if ns is not None and (not isinstance(ns, str) or len(ns) > 63 or
+                           not _DNS1123_LABEL.match(ns)):
+        raise InvalidNameError(
+            f"invalid namespace {ns!r}: must be a DNS-1123 label "
+            "([a-z0-9-], start/end alphanumeric, <=63 chars)")
 
 
 class ObjectStore:
@@ -137,6 +178,7 @@ class ObjectStore:
 
     # --------------------------------------------------------------- CRUD
     def create(self, obj: KfObject) -> KfObject:
+        validate_metadata(obj)
         with self._lock:
             key = self._key(obj)
             if key in self._objs:

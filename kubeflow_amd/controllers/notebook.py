@@ -141,16 +141,21 @@ class NotebookReconciler(Reconciler):
         with open(spec_path, "w") as f:
             json.dump({"port": port, "nb_prefix": prefix,
                        "image": self._image(nb)}, f)
-        env = dict(os.environ)
-        env["NB_PREFIX"] = prefix
+        # Merge PodDefaults against the EXPLICIT session env only, then
+        # overlay on the inherited environment — the webhook compares
+        # PodDefault env against pod-spec env, not the controller's own
+        # environment (admission-webhook/main.go:152-187).
         from kubeflow_amd.scheduler.launcher import merge_poddefaults
+        explicit = {"NB_PREFIX": prefix}
         try:
-            env = merge_poddefaults(
-                env, m.get("labels", {}),
+            explicit = merge_poddefaults(
+                explicit, m.get("labels", {}),
                 self.store.list("PodDefault", ns))
         except ValueError as e:  # conflicting defaults -> surface, keep going
             self.store.record_event(nb, "PodDefaultConflict", str(e),
                                     "Warning")
+        env = dict(os.environ)
+        env.update(explicit)
         repo_root = os.path.dirname(os.path.dirname(os.path.dirname(
             os.path.abspath(__file__))))
         env["PYTHONPATH"] = repo_root + os.pathsep + env.get("PYTHONPATH", "")

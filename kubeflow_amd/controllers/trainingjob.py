@@ -136,6 +136,12 @@ class TrainingJobReconciler(Reconciler):
         set_condition(job, "Created", "True", "JobCreated", "gang launched")
         set_condition(job, "Running", "True", "JobRunning",
                       f"{n} replicas on GPUs {gpu_indices or 'cpu'}")
+        # an OnFailure relaunch is no longer restarting once Running flips
+        # back on (training-operator flips Restarting=False on the
+        # Restarting->Running transition)
+        if has_condition(job, "Restarting"):
+            set_condition(job, "Restarting", "False", "JobRunning",
+                          "restart complete")
         job["status"]["startTime"] = job["status"].get("startTime") or time.time()
         self.store.update(job, check_version=False)
         self.store.record_event(job, "SuccessfulCreate",

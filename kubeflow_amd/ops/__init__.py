@@ -211,9 +211,10 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     """bshd layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] -> o [B,S,Hq,D].
 
     The CDNA4 kernel tiles q in 128-row blocks; for causal attention,
-    sequences are zero-padded to a 64 multiple here (padded key rows are
+    sequences are zero-padded to a 128 multiple here (padded key rows are
     causally masked for every real query, so results are exact) and the
-    output is sliced back.
+    output is sliced back. Non-causal input must already satisfy
+    seq_len % 128 == 0.
     """
     if scale is None:
         scale = q.shape[-1] ** -0.5

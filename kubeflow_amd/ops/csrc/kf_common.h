@@ -28,10 +28,13 @@ __device__ __forceinline__ float kf_bf16_to_f32(unsigned short u) {
 }
 
 // Round-to-nearest-even f32 -> bf16 (matches PyTorch's conversion).
+// NaN guard: RNE add would overflow high-payload NaNs (e.g. 0x7FFFFFFF +
+// 0x8000 -> -0.0) — propagate a quiet NaN instead, like torch does.
 __device__ __forceinline__ unsigned short kf_f32_to_bf16(float f) {
   union { float f; unsigned int i; } w;
   w.f = f;
   unsigned int x = w.i;
+  if ((x & 0x7FFFFFFFu) > 0x7F800000u) return 0x7FC0;
   unsigned int rounding = 0x7fff + ((x >> 16) & 1);
   x += rounding;
   return (unsigned short)(x >> 16);

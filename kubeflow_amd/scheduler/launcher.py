@@ -49,25 +49,17 @@ def free_port() -> int:
 
 def merge_poddefaults(env: Dict[str, str], labels: Dict[str, str],
                       poddefaults: List[dict]) -> Dict[str, str]:
-    """Merge env from PodDefaults whose selector matches `labels`.
-    Conflicting values across defaults raise, mirroring
-    safeToApplyPodDefaultsOnPod (admission-webhook/main.go:98-132)."""
-    out = dict(env)
-    origin: Dict[str, str] = {}
-    for pd in poddefaults:
-        sel = pd.get("spec", {}).get("selector", {}).get("matchLabels", {})
-        if not all(labels.get(k) == v for k, v in sel.items()):
-            continue
-        name = pd["metadata"]["name"]
-        for e in pd.get("spec", {}).get("env", []):
-            k, v = e["name"], str(e.get("value", ""))
-            if k in origin and out.get(k) != v:
-                raise ValueError(
-                    f"PodDefault conflict on env {k}: {origin[k]} vs {name}")
-            if k not in env:  # explicit job env wins over defaults
-                out[k] = v
-                origin[k] = name
-    return out
+    """Env-only PodDefault merge for callers that have no volume/toleration
+    surface (notebook sessions). `env` must be the EXPLICITLY constructed
+    env for the workload — never the inherited os.environ: the webhook
+    compares PodDefault env only against pod-spec env
+    (admission-webhook/main.go:152-187), so a PodDefault var that happens
+    to collide with a controller environment variable must still inject.
+    Conflicting values raise, mirroring safeToApplyPodDefaultsOnPod."""
+    from kubeflow_amd.scheduler.poddefaults import apply_poddefaults
+    merged = apply_poddefaults({"env": dict(env), "labels": dict(labels)},
+                               poddefaults, labels=labels)
+    return merged["env"]
 
 
 @dataclass
@@ -138,10 +130,19 @@ def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
                 labels: Optional[Dict[str, str]] = None,
                 numa_nodes: Optional[Dict[int, int]] = None,
                 entry_module: str = "kubeflow_amd.runtime.worker",
-                warm_pool=None) -> ProcessGang:
+                warm_pool=None, pvc_root: Optional[str] = None,
+                configmaps: Optional[Dict[str, Dict[str, str]]] = None
+                ) -> ProcessGang:
     """Spawn one process per rank. CPU jobs pass gpu_indices=[] and
     spec['world_size'] ranks run on CPU (gloo). With a WarmPool, ranks are
-    handed to pre-forked workers (torch already imported) when available."""
+    handed to pre-forked workers (torch already imported) when available.
+
+    PodDefault injection covers the webhook's full merge surface
+    (env/envFrom/volumes/volumeMounts/tolerations/labels/annotations —
+    scheduler/poddefaults.py); volumes materialize as per-rank symlinks
+    under rank-N/mnt plus a KF_VOLUME_MOUNTS env table."""
+    from kubeflow_amd.scheduler.poddefaults import (
+        apply_poddefaults, materialize_mounts, mounts_env, resolve_env_from)
     os.makedirs(workdir, exist_ok=True)
     world = max(1, len(gpu_indices) or int(spec.get("world_size", 1)))
     port = free_port()
@@ -152,8 +153,24 @@ def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
     gang = ProcessGang(job_uid, workdir)
     repo_root = os.path.dirname(os.path.dirname(
         os.path.dirname(os.path.abspath(__file__))))
-    injected = merge_poddefaults({}, labels or {}, poddefaults or [])
+    merged = apply_poddefaults(
+        {"env": {k: str(v) for k, v in (spec.get("env") or {}).items()},
+         "env_from": spec.get("env_from") or [],
+         "volumes": spec.get("volumes") or [],
+         "volume_mounts": spec.get("volume_mounts") or [],
+         "tolerations": spec.get("tolerations") or [],
+         "labels": dict(labels or {})},
+        poddefaults or [], labels=labels or {})
+    injected = resolve_env_from(merged["env_from"], configmaps or {},
+                                env=merged["env"])
     for rank in range(world):
+        rank_dir = os.path.join(workdir, f"rank-{rank}")
+        os.makedirs(rank_dir, exist_ok=True)
+        mounts = materialize_mounts(rank_dir, merged["volumes"],
+                                    merged["volume_mounts"],
+                                    pvc_root=pvc_root)
+        rank_injected = dict(injected)
+        rank_injected.update(mounts_env(mounts))
         if warm_pool is not None and entry_module == \
                 "kubeflow_amd.runtime.worker":
             got = warm_pool.take()
@@ -167,11 +184,12 @@ def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
                     "workdir": workdir,
                     "job_uid": job_uid,
                     "gpu": gpu_indices[rank] if gpu_indices else None,
-                    "env": injected,
+                    "env": rank_injected,
                 })
                 gang.ranks.append(RankProc(rank, proc, log_path))
                 continue
         env = dict(os.environ)
+        env.update(rank_injected)  # PodDefault/envFrom merge (explicit-only)
         env.update({
             "RANK": str(rank),
             "WORLD_SIZE": str(world),
@@ -180,21 +198,19 @@ def launch_gang(job_uid: str, workdir: str, spec: dict, gpu_indices: List[int],
             "MASTER_PORT": str(port),
             "KF_JOB_UID": job_uid,
             "KF_JOB_WORKDIR": workdir,
-            "PYTHONPATH": repo_root + os.pathsep + env.get("PYTHONPATH", ""),
+            "PYTHONPATH": repo_root + os.pathsep +
+                          os.environ.get("PYTHONPATH", ""),
         })
         env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
         if gpu_indices:
             env["HIP_VISIBLE_DEVICES"] = str(gpu_indices[rank])
             env["LOCAL_RANK"] = "0"  # each proc sees exactly one GPU
-        env = merge_poddefaults(env, labels or {}, poddefaults or [])
 
         cmd = [sys.executable, "-m", entry_module, "--spec", spec_path]
         if gpu_indices and numa_nodes and shutil.which("numactl"):
             node = numa_nodes.get(gpu_indices[rank], 0)
             cmd = ["numactl", f"--cpunodebind={node}",
                    f"--preferred={node}"] + cmd
-        rank_dir = os.path.join(workdir, f"rank-{rank}")
-        os.makedirs(rank_dir, exist_ok=True)
         log_path = os.path.join(rank_dir, "worker.log")
         logf = open(log_path, "w")
         proc = subprocess.Popen(cmd, env=env, stdout=logf, stderr=logf,

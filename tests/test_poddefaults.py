@@ -1,0 +1,213 @@
+"""PodDefault full-merge semantics + store name validation.
+
+Table-driven parity tests mirroring the reference webhook's unit tests
+(admission-webhook/main_test.go:12-80: mergeMap and apply fixtures) over
+the process-model merge (kubeflow_amd/scheduler/poddefaults.py).
+"""
+import os
+
+import pytest
+
+from kubeflow_amd.api.store import (InvalidNameError, ObjectStore,
+                                    validate_metadata)
+from kubeflow_amd.api.objects import new_object
+from kubeflow_amd.scheduler.poddefaults import (PodDefaultConflict,
+                                                apply_poddefaults,
+                                                filter_poddefaults,
+                                                materialize_mounts,
+                                                resolve_env_from,
+                                                selector_matches)
+
+
+def pd(name, sel=None, **spec):
+    spec = dict(spec)
+    spec["selector"] = {"matchLabels": sel or {}}
+    return {"metadata": {"name": name, "namespace": "ns1",
+                         "resourceVersion": "7"},
+            "spec": spec}
+
+
+# ---------------------------------------------------------- selector
+
+@pytest.mark.parametrize("selector,labels,want", [
+    ({"matchLabels": {"a": "1"}}, {"a": "1"}, True),
+    ({"matchLabels": {"a": "1"}}, {"a": "2"}, False),
+    ({"matchLabels": {}}, {}, True),
+    ({"matchExpressions": [{"key": "a", "operator": "In",
+                            "values": ["1", "2"]}]}, {"a": "2"}, True),
+    ({"matchExpressions": [{"key": "a", "operator": "In",
+                            "values": ["1"]}]}, {"a": "3"}, False),
+    ({"matchExpressions": [{"key": "a", "operator": "NotIn",
+                            "values": ["1"]}]}, {"a": "1"}, False),
+    ({"matchExpressions": [{"key": "a", "operator": "Exists"}]},
+     {"a": "x"}, True),
+    ({"matchExpressions": [{"key": "a", "operator": "DoesNotExist"}]},
+     {}, True),
+    ({"matchExpressions": [{"key": "a", "operator": "Bogus"}]},
+     {"a": "x"}, False),  # unknown operator fails closed
+])
+def test_selector_matches(selector, labels, want):
+    assert selector_matches(selector, labels) is want
+
+
+def test_filter_by_namespace():
+    pds = [pd("one", {"k": "v"})]
+    assert filter_poddefaults(pds, {"k": "v"}, "ns1")
+    assert not filter_poddefaults(pds, {"k": "v"}, "other-ns")
+
+
+# ---------------------------------------------------------- merge table
+# mirrors main_test.go's TestMergeMap shape: existing + defaults -> want
+
+def test_merge_env_inject_and_identical_ok():
+    pds = [pd("a", {}, env=[{"name": "X", "value": "1"}]),
+           pd("b", {}, env=[{"name": "X", "value": "1"},
+                            {"name": "Y", "value": "2"}])]
+    out = apply_poddefaults({"env": {"Z": "0"}}, pds, labels={})
+    assert out["env"] == {"Z": "0", "X": "1", "Y": "2"}
+
+
+def test_merge_env_conflict_between_defaults():
+    pds = [pd("a", {}, env=[{"name": "X", "value": "1"}]),
+           pd("b", {}, env=[{"name": "X", "value": "2"}])]
+    with pytest.raises(PodDefaultConflict):
+        apply_poddefaults({}, pds, labels={})
+
+
+def test_merge_env_conflict_with_explicit_env():
+    # reference mergeEnv errors when the container already defines the
+    # var differently (main.go:170-175) — explicit env is not silently won
+    pds = [pd("a", {}, env=[{"name": "X", "value": "1"}])]
+    with pytest.raises(PodDefaultConflict):
+        apply_poddefaults({"env": {"X": "other"}}, pds, labels={})
+
+
+def test_selector_gates_merge():
+    pds = [pd("a", {"team": "ml"}, env=[{"name": "X", "value": "1"}])]
+    out = apply_poddefaults({}, pds, labels={"team": "infra"})
+    assert "X" not in out["env"]
+    out = apply_poddefaults({}, pds, labels={"team": "ml"})
+    assert out["env"]["X"] == "1"
+
+
+def test_merge_envfrom_appends():
+    pds = [pd("a", {}, envFrom=[{"configMapRef": {"name": "cm1"}}])]
+    out = apply_poddefaults(
+        {"env_from": [{"configMapRef": {"name": "cm0"}}]}, pds, labels={})
+    assert [e["configMapRef"]["name"] for e in out["env_from"]] == \
+        ["cm0", "cm1"]
+
+
+def test_merge_volumes_and_mounts():
+    pds = [pd("a", {},
+              volumes=[{"name": "data", "persistentVolumeClaim":
+                        {"claimName": "pvc1"}}],
+              volumeMounts=[{"name": "data", "mountPath": "/data"}])]
+    out = apply_poddefaults({}, pds, labels={})
+    assert out["volumes"][0]["name"] == "data"
+    assert out["volume_mounts"][0]["mountPath"] == "/data"
+
+
+def test_merge_volume_name_conflict():
+    pds = [pd("a", {}, volumes=[{"name": "v", "emptyDir": {}}]),
+           pd("b", {}, volumes=[{"name": "v", "persistentVolumeClaim":
+                                 {"claimName": "x"}}])]
+    with pytest.raises(PodDefaultConflict):
+        apply_poddefaults({}, pds, labels={})
+
+
+def test_merge_mountpath_conflict():
+    # distinct names, same mountPath with differing definitions ->
+    # conflict (mergeVolumeMounts' volumeMountsByPath check)
+    pds = [pd("a", {}, volumeMounts=[{"name": "v1", "mountPath": "/m"}]),
+           pd("b", {}, volumeMounts=[{"name": "v2", "mountPath": "/m"}])]
+    with pytest.raises(PodDefaultConflict):
+        apply_poddefaults({}, pds, labels={})
+
+
+def test_merge_tolerations_keyed_by_key():
+    pds = [pd("a", {}, tolerations=[{"key": "gpu", "operator": "Exists"}]),
+           pd("b", {}, tolerations=[{"key": "gpu", "operator": "Exists"}])]
+    out = apply_poddefaults({}, pds, labels={})
+    assert len(out["tolerations"]) == 1
+    pds.append(pd("c", {}, tolerations=[{"key": "gpu", "value": "other"}]))
+    with pytest.raises(PodDefaultConflict):
+        apply_poddefaults({}, pds, labels={})
+
+
+def test_merge_labels_annotations_and_marker():
+    pds = [pd("a", {}, labels={"l1": "v1"}, annotations={"an1": "av1"})]
+    out = apply_poddefaults({"labels": {"mine": "x"}}, pds, labels={})
+    assert out["labels"] == {"mine": "x", "l1": "v1"}
+    assert out["annotations"]["an1"] == "av1"
+    # mutation marker annotation (applyPodDefaultsOnPod main.go:418-421)
+    assert out["annotations"][
+        "poddefault.admission.kubeflow.org/poddefault-a"] == "7"
+
+
+def test_merge_annotation_conflict():
+    pds = [pd("a", {}, annotations={"k": "1"})]
+    with pytest.raises(PodDefaultConflict):
+        apply_poddefaults({"annotations": {"k": "2"}}, pds, labels={})
+
+
+def test_conflicts_aggregate_all_fields():
+    # safeToApplyPodDefaultsOnPod collects errors across fields before
+    # failing — both the env and the volume conflict must be reported
+    pds = [pd("a", {}, env=[{"name": "X", "value": "1"}],
+              volumes=[{"name": "v", "emptyDir": {}}]),
+           pd("b", {}, env=[{"name": "X", "value": "2"}],
+              volumes=[{"name": "v", "hostPath": {"path": "/x"}}])]
+    with pytest.raises(PodDefaultConflict) as ei:
+        apply_poddefaults({}, pds, labels={})
+    assert len(ei.value.errors) == 2
+
+
+# ---------------------------------------------------------- envFrom + mounts
+
+def test_resolve_env_from_prefix_and_precedence():
+    out = resolve_env_from(
+        [{"configMapRef": {"name": "cm"}, "prefix": "P_"}],
+        {"cm": {"A": "1", "B": "2"}}, env={"P_A": "explicit"})
+    assert out == {"P_A": "explicit", "P_B": "2"}
+
+
+def test_materialize_mounts(tmp_path):
+    pvc_root = tmp_path / "pvcs"
+    (pvc_root / "pvc1").mkdir(parents=True)
+    rank_dir = tmp_path / "rank-0"
+    rank_dir.mkdir()
+    table = materialize_mounts(
+        str(rank_dir),
+        [{"name": "data", "persistentVolumeClaim": {"claimName": "pvc1"}},
+         {"name": "scratch", "emptyDir": {}}],
+        [{"name": "data", "mountPath": "/data", "readOnly": True},
+         {"name": "scratch", "mountPath": "/tmp/scratch"}],
+        pvc_root=str(pvc_root))
+    assert len(table) == 2
+    link = os.path.join(str(rank_dir), "mnt", "data")
+    assert os.path.islink(link)
+    assert os.path.realpath(link) == os.path.realpath(
+        str(pvc_root / "pvc1"))
+    assert table[0]["readOnly"] is True
+
+
+# ---------------------------------------------------------- name validation
+
+def test_store_rejects_path_traversal_names(tmp_path):
+    store = ObjectStore(persist_path=str(tmp_path / "s.jsonl"))
+    for bad in ("../../x", "a/b", "UPPER", "has_underscore", "-lead",
+                "trail-", ""):
+        with pytest.raises(InvalidNameError):
+            store.create(new_object("PyTorchJob", bad, "default"))
+    with pytest.raises(InvalidNameError):
+        store.create(new_object("PyTorchJob", "ok", "Bad_NS"))
+    store.create(new_object("PyTorchJob", "ok-name.v1", "default"))
+
+
+def test_rbac_names_are_path_segment_validated():
+    validate_metadata(new_object("RoleBinding", "namespaceAdmin", "ns"))
+    with pytest.raises(InvalidNameError):
+        validate_metadata(new_object("RoleBinding", "a/b", "ns"))
+    with pytest.raises(InvalidNameError):
+        validate_metadata(new_object("RoleBinding", "..", "ns"))
