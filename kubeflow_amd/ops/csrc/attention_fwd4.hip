@@ -1,0 +1,320 @@
+// attention_fwd4.hip — v4 causal flash-attention forward (bf16, GQA, gfx950).
+//
+// Round-2 rework of the 8-wave swapped-QK^T kernel (attention_fwd.hip v3,
+// 296 TF causal) applying the CDNA4 guide's full verified technique stack
+// (§B 8-warp ladder, §5.5 T3/T10/T13/T14):
+//  * KVBLK=64 K/V tiles DOUBLE-BUFFERED in LDS (64 KiB total), with the
+//    async-STAGE split: next tile's global loads issue at the top of the
+//    iteration and land during compute; LDS writes + one barrier per tile
+//    (T3 minimum 2-phase recipe + T14) — replaces the serial
+//    sync/load/scatter/sync staging of v3.
+//  * V is stored in a [d/16-slab][kv/4-tile][4][16] subtiled layout and the
+//    PV A-operand (V^T) is read with ds_read_b64_tr_b16, the gfx950
+//    hardware transpose read (T10) — replaces v3's 8-scalar-ds_write
+//    transpose-scatter per staged vector (the top staging cost).
+//  * defer-max online softmax (T13, THR=8): the O-rescale is skipped while
+//    the running max grows by <8, bounding P by e^8 which f32 accumulation
+//    absorbs.
+//  * swapped QK^T (S^T = mfma(K, Q)) with fully in-register softmax and
+//    cvt_pk_bf16 + permlane32_swap P repack, K XOR-swizzled LDS, setprio
+//    around MFMA clusters — carried over from v3.
+//  * per-wave causal skip of fully-masked 32-kv sub-blocks.
+//
+// Layout: bshd q [B,S,Hq,D], k/v [B,S,Hkv,D], o [B,S,Hq,D], lse [B,Hq,S]
+// fp32. D == 128, S % 256 == 0 (wrapper pads). Reference behavior anchor:
+// SURVEY.md §2.13 attention_fwd row.
+
+#include "kf_common.h"
+
+typedef __bf16 kf_bf16x8v4 __attribute__((ext_vector_type(8)));
+typedef float kf_f32x16v4 __attribute__((ext_vector_type(16)));
+
+#define A4_D 128
+#define A4_QT 256      // q rows per block (8 waves x 32)
+#define A4_KT 64       // kv rows per LDS tile (double-buffered)
+#define A4_THREADS 512
+#define A4_RESCALE_THR 8.0f
+
+// row-major K tile swizzle (guide G4): byte ^= (row&7)<<4
+__device__ __forceinline__ int kf_swz4(int row, int byte_in_row) {
+  return row * (A4_D * 2) + (byte_in_row ^ ((row & 7) << 4));
+}
+
+// V subtile layout for ds_read_b64_tr_b16 (T10): element offset of V[kv][d]
+// within one tile = slab(d>>4)*1024 + (kv>>2)*64 + (kv&3)*16 + (d&15).
+// Each [4][16] subtile is 128 contiguous bytes; a 16-lane group's tr_read
+// covers one subtile and delivers column (lane&15) — a free 4x4-per-lane
+// transpose, so lane l receives V^T[d = base_d + (l&15)][kv = base_kv + j].
+__device__ __forceinline__ int kf_vsub4(int kv, int d) {
+  return ((d >> 4) << 10) + ((kv >> 2) << 6) + ((kv & 3) << 4) + (d & 15);
+}
+
+__device__ __forceinline__ unsigned int kf_cvt_pk_bf16_v4(float lo, float hi) {
+  unsigned int r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+typedef short kf_short4v4 __attribute__((ext_vector_type(4)));
+
+#define KF_TR16(dst, addr, OFFLIT)                                      \
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:" OFFLIT               \
+               : "=v"(dst) : "v"(addr))
+
+__global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
+    unsigned short* __restrict__ o, float* __restrict__ lse,
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
+    int64_t qts, int64_t kts, float scale, int causal) {
+  __shared__ unsigned char k_lds[2][A4_KT * A4_D * 2];  // 16 KiB x2, swizzled
+  __shared__ unsigned char v_lds[2][A4_KT * A4_D * 2];  // 16 KiB x2, subtiled
+
+  const int qt = blockIdx.x, hq = blockIdx.y;
+  const int64_t b = blockIdx.z;
+  const int hkv = hq / (Hq / Hkv);
+  const int tid = threadIdx.x;
+  const int w = tid / KF_WAVE;
+  const int lane = tid & (KF_WAVE - 1);
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  // ---- persistent Q B-fragments: lane holds q-col l31 ----
+  kf_bf16x8v4 qfrag[8];
+  {
+    const int64_t qbase =
+        (b * S + qt * A4_QT + w * 32 + l31) * qts + (int64_t)hq * A4_D;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      qfrag[kk] =
+          *reinterpret_cast<const kf_bf16x8v4*>(q + qbase + kk * 16 + hi * 8);
+  }
+
+  kf_f32x16v4 oacc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) oacc[i] = kf_f32x16v4{0.f};
+  float m_run = -INFINITY, l_run = 0.f;
+  const int qrow_g = qt * A4_QT + w * 32 + l31;
+  const int wave_qmax = qt * A4_QT + w * 32 + 31;
+
+  const int last_kt =
+      causal ? (qt * A4_QT + A4_QT - 1) / A4_KT : (S / A4_KT - 1);
+
+  // staging: 512 threads x 2 vectors cover one 64x128 tile for K and V.
+  // rows r = vi>>4, 8-elem column chunk c8 = vi&15.
+  const unsigned short* kg0 = k + (b * S) * kts + (int64_t)hkv * A4_D;
+  const unsigned short* vg0 = v + (b * S) * kts + (int64_t)hkv * A4_D;
+  const int r0 = tid >> 4, c8 = tid & 15;          // j=0 row
+  const int r1 = (tid + A4_THREADS) >> 4;          // j=1 row
+
+  // per-lane tr_read base byte address inside a V buffer: group g covers
+  // slab (g&1), kv-subtile 2*(g>>1), column lane&15 (see kf_vsub4).
+  const int g = lane >> 4;
+  const unsigned v_lane_off =
+      (unsigned)(((g & 1) << 11) + ((g >> 1) << 8) + ((lane & 15) << 3));
+
+  // ---- prologue: stage tile 0 into buffer 0 ----
+  kf_short8 kst0, kst1, vst0, vst1;
+  kst0 = *reinterpret_cast<const kf_short8*>(kg0 + r0 * kts + c8 * 8);
+  kst1 = *reinterpret_cast<const kf_short8*>(kg0 + r1 * kts + c8 * 8);
+  vst0 = *reinterpret_cast<const kf_short8*>(vg0 + r0 * kts + c8 * 8);
+  vst1 = *reinterpret_cast<const kf_short8*>(vg0 + r1 * kts + c8 * 8);
+  *reinterpret_cast<kf_short8*>(k_lds[0] + kf_swz4(r0, c8 * 16)) = kst0;
+  *reinterpret_cast<kf_short8*>(k_lds[0] + kf_swz4(r1, c8 * 16)) = kst1;
+  *reinterpret_cast<kf_short8*>(v_lds[0] + 2 * kf_vsub4(r0, c8 * 8)) = vst0;
+  *reinterpret_cast<kf_short8*>(v_lds[0] + 2 * kf_vsub4(r1, c8 * 8)) = vst1;
+  __syncthreads();
+
+  for (int kt = 0; kt <= last_kt; ++kt) {
+    const int cur = kt & 1;
+    // ---- async-STAGE: issue next tile's global loads now; they land
+    // under this tile's compute (T14) ----
+    const bool have_next = kt < last_kt;
+    if (have_next) {
+      const unsigned short* kg = kg0 + (int64_t)(kt + 1) * A4_KT * kts;
+      const unsigned short* vg = vg0 + (int64_t)(kt + 1) * A4_KT * kts;
+      kst0 = *reinterpret_cast<const kf_short8*>(kg + r0 * kts + c8 * 8);
+      kst1 = *reinterpret_cast<const kf_short8*>(kg + r1 * kts + c8 * 8);
+      vst0 = *reinterpret_cast<const kf_short8*>(vg + r0 * kts + c8 * 8);
+      vst1 = *reinterpret_cast<const kf_short8*>(vg + r1 * kts + c8 * 8);
+    }
+
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {  // 32-kv sub-blocks
+      const int kv_lo = kt * A4_KT + mt * 32;
+      if (causal && kv_lo > wave_qmax) continue;  // wave-uniform skip
+
+      // ---- S^T = mfma(K, Q) ----
+      kf_f32x16v4 st = kf_f32x16v4{0.f};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 8; ++kk) {
+        kf_bf16x8v4 afrag = *reinterpret_cast<const kf_bf16x8v4*>(
+            k_lds[cur] + kf_swz4(mt * 32 + l31, kk * 32 + hi * 16));
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, qfrag[kk], st,
+                                                     0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- mask + defer-max online softmax (T13) ----
+      const int kv0 = kv_lo + hi * 4;
+      float mx = -INFINITY;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv = kv0 + (r & 3) + 8 * (r >> 2);
+        float sv = st[r] * scale;
+        if (causal && kv > qrow_g) sv = -INFINITY;
+        st[r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+      mx = fmaxf(mx, __shfl_xor(mx, 32, KF_WAVE));
+      if (!__all(mx <= m_run + A4_RESCALE_THR)) {
+        const float m_new = fmaxf(m_run, mx);
+        const float alpha =
+            (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) oacc[i][r] *= alpha;
+        l_run *= alpha;
+        m_run = m_new;
+      }
+      float lsum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float pv =
+            (st[r] == -INFINITY) ? 0.f : __expf(st[r] - m_run);
+        st[r] = pv;
+        lsum += pv;
+      }
+      lsum += __shfl_xor(lsum, 32, KF_WAVE);
+      l_run += lsum;
+
+      // ---- P^T -> two B-fragments via cvt_pk + permlane32_swap (T12) ----
+      kf_bf16x8v4 pb[2];
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        const int base = step * 8;
+        unsigned int w0 = kf_cvt_pk_bf16_v4(st[base + 0], st[base + 1]);
+        unsigned int w1 = kf_cvt_pk_bf16_v4(st[base + 2], st[base + 3]);
+        unsigned int w2 = kf_cvt_pk_bf16_v4(st[base + 4], st[base + 5]);
+        unsigned int w3 = kf_cvt_pk_bf16_v4(st[base + 6], st[base + 7]);
+        auto s02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+        auto s13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+        unsigned int u[4] = {(unsigned)s02[0], (unsigned)s13[0],
+                             (unsigned)s02[1], (unsigned)s13[1]};
+        pb[step] = *reinterpret_cast<kf_bf16x8v4*>(u);
+      }
+
+      // ---- O^T += V^T P^T; V^T fragments via hardware transpose read ----
+      const unsigned vbase =
+          (unsigned)(size_t)(v_lds[cur]) + v_lane_off + (mt << 10);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        const unsigned vaddr = vbase + (dt << 12);
+        kf_short4v4 t0, t1, t2, t3;
+        KF_TR16(t0, vaddr, "0");    // step 0, kv j=0..3
+        KF_TR16(t1, vaddr, "128");  // step 0, kv j=4..7
+        KF_TR16(t2, vaddr, "512");  // step 1, kv j=0..3
+        KF_TR16(t3, vaddr, "640");  // step 1, kv j=4..7
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);  // guide rule 18
+        kf_short8 f0, f1;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          f0[j] = t0[j]; f0[j + 4] = t1[j];
+          f1[j] = t2[j]; f1[j + 4] = t3[j];
+        }
+        __builtin_amdgcn_s_setprio(1);
+        oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<kf_bf16x8v4*>(&f0), pb[0], oacc[dt], 0, 0, 0);
+        oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<kf_bf16x8v4*>(&f1), pb[1], oacc[dt], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+
+    // ---- write the staged tile into the other buffer; the compiler
+    // inserts the vmcnt wait on the staged registers here, after the
+    // compute phase has covered the HBM latency ----
+    if (have_next) {
+      const int nxt = cur ^ 1;
+      *reinterpret_cast<kf_short8*>(k_lds[nxt] + kf_swz4(r0, c8 * 16)) = kst0;
+      *reinterpret_cast<kf_short8*>(k_lds[nxt] + kf_swz4(r1, c8 * 16)) = kst1;
+      *reinterpret_cast<kf_short8*>(v_lds[nxt] + 2 * kf_vsub4(r0, c8 * 8)) =
+          vst0;
+      *reinterpret_cast<kf_short8*>(v_lds[nxt] + 2 * kf_vsub4(r1, c8 * 8)) =
+          vst1;
+    }
+    __syncthreads();  // readers of [cur] done + writes to [nxt] visible
+  }
+
+  // ---- epilogue: O^T regs -> o[token][d]; lse per q row ----
+  const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+  const int64_t obase = ((b * S + qrow_g) * (int64_t)Hq + hq) * A4_D;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int rq = 0; rq < 4; ++rq) {  // 4 consecutive-d quads per dt
+      const int d0 = dt * 32 + 8 * rq + 4 * hi;
+      unsigned short q4[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        q4[j] = kf_f32_to_bf16(oacc[dt][rq * 4 + j] * inv_l);
+      *reinterpret_cast<kf_short4v4*>(o + obase + d0) =
+          *reinterpret_cast<kf_short4v4*>(q4);
+    }
+  if (hi == 0)
+    lse[(b * Hq + hq) * (int64_t)S + qrow_g] = m_run + __logf(l_run);
+}
+
+KF_EXPORT int kf_attn_fwd4(void* o, float* lse, const void* q, const void* k,
+                           const void* v, int64_t B, int64_t S, int64_t Hq,
+                           int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
+                           float scale, int causal, void* stream) {
+  if (D != A4_D || S % A4_QT || Hq % Hkv) return (int)hipErrorInvalidValue;
+  if (qts == 0) qts = Hq * A4_D;
+  if (kts == 0) kts = Hkv * A4_D;
+  dim3 grid((unsigned)(S / A4_QT), (unsigned)Hq, (unsigned)B);
+  hipLaunchKernelGGL(kf_attn_fwd4_kernel, grid, dim3(A4_THREADS), 0,
+                     (hipStream_t)stream, (unsigned short*)o, lse,
+                     (const unsigned short*)q, (const unsigned short*)k,
+                     (const unsigned short*)v, B, (int)S, (int)Hq, (int)Hkv,
+                     qts, kts, scale, causal);
+  return (int)hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 semantics probe: fills LDS with element indices and
+// dumps what each lane receives for three addressing modes. Used once on
+// hardware to pin the lane->element mapping the v4 kernel assumes
+// (guide T10, m156/m162); kept for regression if the layout ever changes.
+// ---------------------------------------------------------------------------
+__global__ void kf_tr16_probe_kernel(short* out, const short* in, int mode) {
+  __shared__ short lds[1024];
+  const int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 1024; i += 64) lds[i] = in[i];
+  __syncthreads();
+  unsigned base = (unsigned)(size_t)(&lds[0]);
+  unsigned addr;
+  if (mode == 0) addr = base + 8u * lane;          // 4 tiles, one per group
+  else if (mode == 1) addr = base;                 // uniform
+  else addr = base + 8u * (lane & 15);             // same tile all groups
+  kf_short4v4 r0, r1;
+  KF_TR16(r0, addr, "0");
+  KF_TR16(r1, addr, "128");
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    out[lane * 8 + j] = r0[j];
+    out[lane * 8 + 4 + j] = r1[j];
+  }
+}
+
+KF_EXPORT int kf_tr16_probe(void* out, const void* in, int mode,
+                            void* stream) {
+  hipLaunchKernelGGL(kf_tr16_probe_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (short*)out, (const short*)in,
+                     mode);
+  return (int)hipGetLastError();
+}
