@@ -2,7 +2,7 @@
 //
 // Round-2 rework of the 8-wave swapped-QK^T kernel (attention_fwd.hip v3,
 // 296 TF causal) applying the CDNA4 guide's full verified technique stack
-// (§B 8-warp ladder, §5.5 T3/T10/T13/T14):
+// (§B 8-warp ladder, §5.5 T3/T10/T13/T14) plus a VALU-overlap pipeline:
 //  * KVBLK=64 K/V tiles DOUBLE-BUFFERED in LDS (64 KiB total), with the
 //    async-STAGE split: next tile's global loads issue at the top of the
 //    iteration and land during compute; LDS writes + one barrier per tile
@@ -11,14 +11,19 @@
 //  * V is stored in a [d/16-slab][kv/4-tile][4][16] subtiled layout and the
 //    PV A-operand (V^T) is read with ds_read_b64_tr_b16, the gfx950
 //    hardware transpose read (T10) — replaces v3's 8-scalar-ds_write
-//    transpose-scatter per staged vector (the top staging cost).
-//  * defer-max online softmax (T13, THR=8): the O-rescale is skipped while
-//    the running max grows by <8, bounding P by e^8 which f32 accumulation
-//    absorbs.
+//    transpose-scatter per staged vector. Lane mapping HW-verified by
+//    scripts/probe_tr16.py (each lane reads 8 B at its own address; each
+//    4-lane subgroup supplies one 16-elem tile row; lane gets column l&15).
+//  * both 32-kv sub-blocks' QK^T MFMA clusters issue back-to-back so the
+//    first block's softmax VALU overlaps the second's MFMA tail; PV
+//    tr_reads are 1-deep prefetched with counted lgkmcnt.
+//  * softmax runs in the exp2 domain (one v_exp_f32 per element, scale and
+//    log2e folded into one multiply), with defer-max (T13, THR=8/ln2) and
+//    causal masking only on diagonal sub-blocks (wave-uniform template
+//    branch); fully-masked sub-blocks are skipped per wave.
 //  * swapped QK^T (S^T = mfma(K, Q)) with fully in-register softmax and
 //    cvt_pk_bf16 + permlane32_swap P repack, K XOR-swizzled LDS, setprio
 //    around MFMA clusters — carried over from v3.
-//  * per-wave causal skip of fully-masked 32-kv sub-blocks.
 //
 // Layout: bshd q [B,S,Hq,D], k/v [B,S,Hkv,D], o [B,S,Hq,D], lse [B,Hq,S]
 // fp32. D == 128, S % 256 == 0 (wrapper pads). Reference behavior anchor:
@@ -28,12 +33,15 @@
 
 typedef __bf16 kf_bf16x8v4 __attribute__((ext_vector_type(8)));
 typedef float kf_f32x16v4 __attribute__((ext_vector_type(16)));
+typedef short kf_short4v4 __attribute__((ext_vector_type(4)));
 
 #define A4_D 128
 #define A4_QT 256      // q rows per block (8 waves x 32)
 #define A4_KT 64       // kv rows per LDS tile (double-buffered)
 #define A4_THREADS 512
-#define A4_RESCALE_THR 8.0f
+#define A4_LOG2E 1.44269504f
+#define A4_LN2 0.69314718f
+#define A4_RESCALE_THR2 11.5415603f  // 8 * log2(e): defer-max in exp2 units
 
 // row-major K tile swizzle (guide G4): byte ^= (row&7)<<4
 __device__ __forceinline__ int kf_swz4(int row, int byte_in_row) {
@@ -41,10 +49,9 @@ __device__ __forceinline__ int kf_swz4(int row, int byte_in_row) {
 }
 
 // V subtile layout for ds_read_b64_tr_b16 (T10): element offset of V[kv][d]
-// within one tile = slab(d>>4)*1024 + (kv>>2)*64 + (kv&3)*16 + (d&15).
-// Each [4][16] subtile is 128 contiguous bytes; a 16-lane group's tr_read
-// covers one subtile and delivers column (lane&15) — a free 4x4-per-lane
-// transpose, so lane l receives V^T[d = base_d + (l&15)][kv = base_kv + j].
+// = slab(d>>4)*1024 + (kv>>2)*64 + (kv&3)*16 + (d&15). Each [4][16] subtile
+// is 128 contiguous bytes; a 16-lane group's tr_read covers one subtile and
+// delivers column (lane&15): lane l receives V^T[d=base_d+(l&15)][kv=base+j].
 __device__ __forceinline__ int kf_vsub4(int kv, int d) {
   return ((d >> 4) << 10) + ((kv >> 2) << 6) + ((kv & 3) << 4) + (d & 15);
 }
@@ -55,7 +62,11 @@ __device__ __forceinline__ unsigned int kf_cvt_pk_bf16_v4(float lo, float hi) {
   return r;
 }
 
-typedef short kf_short4v4 __attribute__((ext_vector_type(4)));
+__device__ __forceinline__ float kf_exp2(float x) {
+  float r;
+  asm volatile("v_exp_f32 %0, %1" : "=v"(r) : "v"(x));
+  return r;
+}
 
 #define KF_TR16(dst, addr, OFFLIT)                                      \
   asm volatile("ds_read_b64_tr_b16 %0, %1 offset:" OFFLIT               \
@@ -77,6 +88,7 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   const int lane = tid & (KF_WAVE - 1);
   const int l31 = lane & 31;
   const int hi = lane >> 5;
+  const float scale2 = scale * A4_LOG2E;  // QK^T scaled into exp2 domain
 
   // ---- persistent Q B-fragments: lane holds q-col l31 ----
   kf_bf16x8v4 qfrag[8];
@@ -92,15 +104,15 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   kf_f32x16v4 oacc[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) oacc[i] = kf_f32x16v4{0.f};
-  float m_run = -INFINITY, l_run = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;  // m_run in exp2 units
   const int qrow_g = qt * A4_QT + w * 32 + l31;
-  const int wave_qmax = qt * A4_QT + w * 32 + 31;
+  const int wave_qmin = qt * A4_QT + w * 32;
+  const int wave_qmax = wave_qmin + 31;
 
   const int last_kt =
       causal ? (qt * A4_QT + A4_QT - 1) / A4_KT : (S / A4_KT - 1);
 
   // staging: 512 threads x 2 vectors cover one 64x128 tile for K and V.
-  // rows r = vi>>4, 8-elem column chunk c8 = vi&15.
   const unsigned short* kg0 = k + (b * S) * kts + (int64_t)hkv * A4_D;
   const unsigned short* vg0 = v + (b * S) * kts + (int64_t)hkv * A4_D;
   const int r0 = tid >> 4, c8 = tid & 15;          // j=0 row
@@ -111,6 +123,111 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   const int g = lane >> 4;
   const unsigned v_lane_off =
       (unsigned)(((g & 1) << 11) + ((g >> 1) << 8) + ((lane & 15) << 3));
+
+  // ---- S^T = mfma(K, Q) over one 32-kv sub-block ----
+  auto qk_block = [&](int cur, int mt) {
+    kf_f32x16v4 st = kf_f32x16v4{0.f};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) {
+      kf_bf16x8v4 afrag = *reinterpret_cast<const kf_bf16x8v4*>(
+          k_lds[cur] + kf_swz4(mt * 32 + l31, kk * 32 + hi * 16));
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, qfrag[kk], st,
+                                                   0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    return st;
+  };
+
+  // ---- defer-max online softmax (exp2 domain) + P repack (T12/T13) ----
+  auto sm_block = [&](kf_f32x16v4& st, int kv_lo, bool need_mask,
+                      kf_bf16x8v4* pb) {
+    const int kv0 = kv_lo + hi * 4;
+    float mx = -INFINITY;
+    if (need_mask) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kv = kv0 + (r & 3) + 8 * (r >> 2);
+        float sv = st[r] * scale2;
+        if (kv > qrow_g) sv = -INFINITY;
+        st[r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float sv = st[r] * scale2;
+        st[r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+    }
+    mx = fmaxf(mx, __shfl_xor(mx, 32, KF_WAVE));
+    if (!__all(mx <= m_run + A4_RESCALE_THR2)) {
+      const float m_new = fmaxf(m_run, mx);
+      const float alpha = kf_exp2(m_run - m_new);  // 0 on first block
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[i][r] *= alpha;
+      l_run *= alpha;
+      m_run = m_new;
+    }
+    float lsum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const float pv = kf_exp2(st[r] - m_run);  // masked -inf -> 0
+      st[r] = pv;
+      lsum += pv;
+    }
+    lsum += __shfl_xor(lsum, 32, KF_WAVE);
+    l_run += lsum;
+#pragma unroll
+    for (int step = 0; step < 2; ++step) {
+      const int base = step * 8;
+      unsigned int w0 = kf_cvt_pk_bf16_v4(st[base + 0], st[base + 1]);
+      unsigned int w1 = kf_cvt_pk_bf16_v4(st[base + 2], st[base + 3]);
+      unsigned int w2 = kf_cvt_pk_bf16_v4(st[base + 4], st[base + 5]);
+      unsigned int w3 = kf_cvt_pk_bf16_v4(st[base + 6], st[base + 7]);
+      auto s02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+      auto s13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+      unsigned int u[4] = {(unsigned)s02[0], (unsigned)s13[0],
+                           (unsigned)s02[1], (unsigned)s13[1]};
+      pb[step] = *reinterpret_cast<kf_bf16x8v4*>(u);
+    }
+  };
+
+  // ---- O^T += V^T P^T with 1-deep tr_read prefetch ----
+  auto pv_block = [&](const kf_bf16x8v4* pb, unsigned vbase) {
+    kf_short4v4 t[2][4];
+    KF_TR16(t[0][0], vbase, "0");
+    KF_TR16(t[0][1], vbase, "128");
+    KF_TR16(t[0][2], vbase, "512");
+    KF_TR16(t[0][3], vbase, "640");
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      if (dt < 3) {
+        const unsigned va = vbase + ((dt + 1) << 12);
+        KF_TR16(t[(dt + 1) & 1][0], va, "0");
+        KF_TR16(t[(dt + 1) & 1][1], va, "128");
+        KF_TR16(t[(dt + 1) & 1][2], va, "512");
+        KF_TR16(t[(dt + 1) & 1][3], va, "640");
+        asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_sched_barrier(0);  // guide rule 18
+      kf_short8 f0 = __builtin_shufflevector(t[dt & 1][0], t[dt & 1][1],
+                                             0, 1, 2, 3, 4, 5, 6, 7);
+      kf_short8 f1 = __builtin_shufflevector(t[dt & 1][2], t[dt & 1][3],
+                                             0, 1, 2, 3, 4, 5, 6, 7);
+      __builtin_amdgcn_s_setprio(1);
+      oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<kf_bf16x8v4*>(&f0), pb[0], oacc[dt], 0, 0, 0);
+      oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<kf_bf16x8v4*>(&f1), pb[1], oacc[dt], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
 
   // ---- prologue: stage tile 0 into buffer 0 ----
   kf_short8 kst0, kst1, vst0, vst1;
@@ -138,99 +255,23 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
       vst1 = *reinterpret_cast<const kf_short8*>(vg + r1 * kts + c8 * 8);
     }
 
-#pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {  // 32-kv sub-blocks
-      const int kv_lo = kt * A4_KT + mt * 32;
-      if (causal && kv_lo > wave_qmax) continue;  // wave-uniform skip
-
-      // ---- S^T = mfma(K, Q) ----
-      kf_f32x16v4 st = kf_f32x16v4{0.f};
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        kf_bf16x8v4 afrag = *reinterpret_cast<const kf_bf16x8v4*>(
-            k_lds[cur] + kf_swz4(mt * 32 + l31, kk * 32 + hi * 16));
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, qfrag[kk], st,
-                                                     0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
-
-      // ---- mask + defer-max online softmax (T13) ----
-      const int kv0 = kv_lo + hi * 4;
-      float mx = -INFINITY;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kv = kv0 + (r & 3) + 8 * (r >> 2);
-        float sv = st[r] * scale;
-        if (causal && kv > qrow_g) sv = -INFINITY;
-        st[r] = sv;
-        mx = fmaxf(mx, sv);
-      }
-      mx = fmaxf(mx, __shfl_xor(mx, 32, KF_WAVE));
-      if (!__all(mx <= m_run + A4_RESCALE_THR)) {
-        const float m_new = fmaxf(m_run, mx);
-        const float alpha =
-            (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-#pragma unroll
-          for (int r = 0; r < 16; ++r) oacc[i][r] *= alpha;
-        l_run *= alpha;
-        m_run = m_new;
-      }
-      float lsum = 0.f;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float pv =
-            (st[r] == -INFINITY) ? 0.f : __expf(st[r] - m_run);
-        st[r] = pv;
-        lsum += pv;
-      }
-      lsum += __shfl_xor(lsum, 32, KF_WAVE);
-      l_run += lsum;
-
-      // ---- P^T -> two B-fragments via cvt_pk + permlane32_swap (T12) ----
-      kf_bf16x8v4 pb[2];
-#pragma unroll
-      for (int step = 0; step < 2; ++step) {
-        const int base = step * 8;
-        unsigned int w0 = kf_cvt_pk_bf16_v4(st[base + 0], st[base + 1]);
-        unsigned int w1 = kf_cvt_pk_bf16_v4(st[base + 2], st[base + 3]);
-        unsigned int w2 = kf_cvt_pk_bf16_v4(st[base + 4], st[base + 5]);
-        unsigned int w3 = kf_cvt_pk_bf16_v4(st[base + 6], st[base + 7]);
-        auto s02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
-        auto s13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
-        unsigned int u[4] = {(unsigned)s02[0], (unsigned)s13[0],
-                             (unsigned)s02[1], (unsigned)s13[1]};
-        pb[step] = *reinterpret_cast<kf_bf16x8v4*>(u);
-      }
-
-      // ---- O^T += V^T P^T; V^T fragments via hardware transpose read ----
-      const unsigned vbase =
-          (unsigned)(size_t)(v_lds[cur]) + v_lane_off + (mt << 10);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const unsigned vaddr = vbase + (dt << 12);
-        kf_short4v4 t0, t1, t2, t3;
-        KF_TR16(t0, vaddr, "0");    // step 0, kv j=0..3
-        KF_TR16(t1, vaddr, "128");  // step 0, kv j=4..7
-        KF_TR16(t2, vaddr, "512");  // step 1, kv j=0..3
-        KF_TR16(t3, vaddr, "640");  // step 1, kv j=4..7
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_sched_barrier(0);  // guide rule 18
-        kf_short8 f0, f1;
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          f0[j] = t0[j]; f0[j + 4] = t1[j];
-          f1[j] = t2[j]; f1[j + 4] = t3[j];
-        }
-        __builtin_amdgcn_s_setprio(1);
-        oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *reinterpret_cast<kf_bf16x8v4*>(&f0), pb[0], oacc[dt], 0, 0, 0);
-        oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *reinterpret_cast<kf_bf16x8v4*>(&f1), pb[1], oacc[dt], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
-      }
+    // ---- compute: both QK^T clusters issue first so sub-block 0's
+    // softmax VALU overlaps sub-block 1's MFMA tail ----
+    const int kv_lo0 = kt * A4_KT, kv_lo1 = kt * A4_KT + 32;
+    const bool do0 = !causal || kv_lo0 <= wave_qmax;
+    const bool do1 = !causal || kv_lo1 <= wave_qmax;
+    const unsigned vbase = (unsigned)(size_t)(v_lds[cur]) + v_lane_off;
+    kf_f32x16v4 st0, st1;
+    if (do0) st0 = qk_block(cur, 0);
+    if (do1) st1 = qk_block(cur, 1);
+    kf_bf16x8v4 pb[2];
+    if (do0) {
+      sm_block(st0, kv_lo0, causal && kv_lo0 + 31 > wave_qmin, pb);
+      pv_block(pb, vbase);
+    }
+    if (do1) {
+      sm_block(st1, kv_lo1, causal && kv_lo1 + 31 > wave_qmin, pb);
+      pv_block(pb, vbase + 1024);
     }
 
     // ---- write the staged tile into the other buffer; the compiler
@@ -264,7 +305,8 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
           *reinterpret_cast<kf_short4v4*>(q4);
     }
   if (hi == 0)
-    lse[(b * Hq + hq) * (int64_t)S + qrow_g] = m_run + __logf(l_run);
+    lse[(b * Hq + hq) * (int64_t)S + qrow_g] =
+        m_run * A4_LN2 + __logf(l_run);
 }
 
 KF_EXPORT int kf_attn_fwd4(void* o, float* lse, const void* q, const void* k,
@@ -288,6 +330,9 @@ KF_EXPORT int kf_attn_fwd4(void* o, float* lse, const void* q, const void* k,
 // dumps what each lane receives for three addressing modes. Used once on
 // hardware to pin the lane->element mapping the v4 kernel assumes
 // (guide T10, m156/m162); kept for regression if the layout ever changes.
+// HW result (scripts/probe_tr16.py, MI355X): with addr_l = base + 8*l each
+// lane receives column (l&15) of the [4][16] row-major tile covered by its
+// 16-lane group's addresses; uniform addressing degenerates to lds[l&3].
 // ---------------------------------------------------------------------------
 __global__ void kf_tr16_probe_kernel(short* out, const short* in, int mode) {
   __shared__ short lds[1024];
