@@ -124,17 +124,22 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   const unsigned v_lane_off =
       (unsigned)(((g & 1) << 11) + ((g >> 1) << 8) + ((lane & 15) << 3));
 
-  // ---- S^T = mfma(K, Q) over one 32-kv sub-block ----
+  // ---- S^T = mfma(K, Q) over one 32-kv sub-block. All 8 K-fragment
+  // ds_reads issue back-to-back BEFORE the MFMA chain so the ~120-cycle
+  // LDS latency pipelines instead of serializing per MFMA (the same
+  // issue-all-then-compute discipline as the guide's GEMM ladder). ----
   auto qk_block = [&](int cur, int mt) {
+    kf_bf16x8v4 af[8];
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      af[kk] = *reinterpret_cast<const kf_bf16x8v4*>(
+          k_lds[cur] + kf_swz4(mt * 32 + l31, kk * 32 + hi * 16));
     kf_f32x16v4 st = kf_f32x16v4{0.f};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kk = 0; kk < 8; ++kk) {
-      kf_bf16x8v4 afrag = *reinterpret_cast<const kf_bf16x8v4*>(
-          k_lds[cur] + kf_swz4(mt * 32 + l31, kk * 32 + hi * 16));
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, qfrag[kk], st,
+    for (int kk = 0; kk < 8; ++kk)
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[kk], qfrag[kk], st,
                                                    0, 0, 0);
-    }
     __builtin_amdgcn_s_setprio(0);
     return st;
   };
