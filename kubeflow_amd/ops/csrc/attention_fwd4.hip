@@ -72,6 +72,10 @@ __device__ __forceinline__ float kf_exp2(float x) {
   asm volatile("ds_read_b64_tr_b16 %0, %1 offset:" OFFLIT               \
                : "=v"(dst) : "v"(addr))
 
+// ABL ablation bits (guide m164 methodology — ablate before optimizing):
+// 1 = skip softmax math (pb straight from scaled st), 2 = skip PV MFMA
+// (pb kept live), 4 = skip next-tile staging (compute on stale LDS).
+template <int ABL>
 __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     unsigned short* __restrict__ o, float* __restrict__ lse,
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
@@ -147,6 +151,25 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   // ---- defer-max online softmax (exp2 domain) + P repack (T12/T13) ----
   auto sm_block = [&](kf_f32x16v4& st, int kv_lo, bool need_mask,
                       kf_bf16x8v4* pb) {
+    if (ABL & 1) {  // NOSM: scale-only, straight to repack (keeps st live)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[r] *= scale2;
+      l_run += st[0];
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        const int base = step * 8;
+        unsigned int w0 = kf_cvt_pk_bf16_v4(st[base + 0], st[base + 1]);
+        unsigned int w1 = kf_cvt_pk_bf16_v4(st[base + 2], st[base + 3]);
+        unsigned int w2 = kf_cvt_pk_bf16_v4(st[base + 4], st[base + 5]);
+        unsigned int w3 = kf_cvt_pk_bf16_v4(st[base + 6], st[base + 7]);
+        auto s02 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
+        auto s13 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
+        unsigned int u[4] = {(unsigned)s02[0], (unsigned)s13[0],
+                             (unsigned)s02[1], (unsigned)s13[1]};
+        pb[step] = *reinterpret_cast<kf_bf16x8v4*>(u);
+      }
+      return;
+    }
     const int kv0 = kv_lo + hi * 4;
     float mx = -INFINITY;
     if (need_mask) {
@@ -203,6 +226,12 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
 
   // ---- O^T += V^T P^T with 1-deep tr_read prefetch ----
   auto pv_block = [&](const kf_bf16x8v4* pb, unsigned vbase) {
+    if (ABL & 2) {  // NOPV: keep pb live, skip tr_reads + MFMAs
+      asm volatile("" ::"v"(*(const unsigned*)&pb[0]),
+                   "v"(*(const unsigned*)&pb[1]), "v"(vbase));
+      oacc[0][0] += (float)pb[0][0];
+      return;
+    }
     kf_short4v4 t[2][4];
     KF_TR16(t[0][0], vbase, "0");
     KF_TR16(t[0][1], vbase, "128");
@@ -250,7 +279,7 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     const int cur = kt & 1;
     // ---- async-STAGE: issue next tile's global loads now; they land
     // under this tile's compute (T14) ----
-    const bool have_next = kt < last_kt;
+    const bool have_next = (ABL & 4) ? false : kt < last_kt;
     if (have_next) {
       const unsigned short* kg = kg0 + (int64_t)(kt + 1) * A4_KT * kts;
       const unsigned short* vg = vg0 + (int64_t)(kt + 1) * A4_KT * kts;
@@ -314,20 +343,52 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
         m_run * A4_LN2 + __logf(l_run);
 }
 
-KF_EXPORT int kf_attn_fwd4(void* o, float* lse, const void* q, const void* k,
-                           const void* v, int64_t B, int64_t S, int64_t Hq,
-                           int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
-                           float scale, int causal, void* stream) {
+template <int ABL>
+static int kf_attn_fwd4_launch(void* o, float* lse, const void* q,
+                               const void* k, const void* v, int64_t B,
+                               int64_t S, int64_t Hq, int64_t Hkv, int64_t D,
+                               int64_t qts, int64_t kts, float scale,
+                               int causal, void* stream) {
   if (D != A4_D || S % A4_QT || Hq % Hkv) return (int)hipErrorInvalidValue;
   if (qts == 0) qts = Hq * A4_D;
   if (kts == 0) kts = Hkv * A4_D;
   dim3 grid((unsigned)(S / A4_QT), (unsigned)Hq, (unsigned)B);
-  hipLaunchKernelGGL(kf_attn_fwd4_kernel, grid, dim3(A4_THREADS), 0,
+  hipLaunchKernelGGL(kf_attn_fwd4_kernel<ABL>, grid, dim3(A4_THREADS), 0,
                      (hipStream_t)stream, (unsigned short*)o, lse,
                      (const unsigned short*)q, (const unsigned short*)k,
                      (const unsigned short*)v, B, (int)S, (int)Hq, (int)Hkv,
                      qts, kts, scale, causal);
   return (int)hipGetLastError();
+}
+
+KF_EXPORT int kf_attn_fwd4(void* o, float* lse, const void* q, const void* k,
+                           const void* v, int64_t B, int64_t S, int64_t Hq,
+                           int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
+                           float scale, int causal, void* stream) {
+  return kf_attn_fwd4_launch<0>(o, lse, q, k, v, B, S, Hq, Hkv, D, qts, kts,
+                                scale, causal, stream);
+}
+
+// Ablation entry: mode bits as above (scratch results are NOT valid
+// attention outputs for mode != 0).
+KF_EXPORT int kf_attn_fwd4_abl(int mode, void* o, float* lse, const void* q,
+                               const void* k, const void* v, int64_t B,
+                               int64_t S, int64_t Hq, int64_t Hkv, int64_t D,
+                               int64_t qts, int64_t kts, float scale,
+                               int causal, void* stream) {
+  switch (mode) {
+    case 1: return kf_attn_fwd4_launch<1>(o, lse, q, k, v, B, S, Hq, Hkv, D,
+                                          qts, kts, scale, causal, stream);
+    case 2: return kf_attn_fwd4_launch<2>(o, lse, q, k, v, B, S, Hq, Hkv, D,
+                                          qts, kts, scale, causal, stream);
+    case 3: return kf_attn_fwd4_launch<3>(o, lse, q, k, v, B, S, Hq, Hkv, D,
+                                          qts, kts, scale, causal, stream);
+    case 4: return kf_attn_fwd4_launch<4>(o, lse, q, k, v, B, S, Hq, Hkv, D,
+                                          qts, kts, scale, causal, stream);
+    default:
+      return kf_attn_fwd4_launch<0>(o, lse, q, k, v, B, S, Hq, Hkv, D, qts,
+                                    kts, scale, causal, stream);
+  }
 }
 
 // ---------------------------------------------------------------------------
