@@ -419,20 +419,27 @@ __global__ __launch_bounds__(A8_THREADS, 2) void kf_attn_fwd8_kernel(
 }
 
 
-// v4 kernel (attention_fwd4.hip): async-STAGE double-buffer + tr_read V +
-// defer-max. KF_ATTN_V4=0 falls back to the v3 kernel for bisection.
+// Newer kernels: v4 (attention_fwd4.hip, 8-wave async-STAGE + tr_read V)
+// and v5 (attention_fwd5.hip, 4-wave x 64-q LDS-traffic-halved). Dispatch
+// order: KF_ATTN_IMPL=5 (default) -> 4 -> 3 for bisection.
 KF_EXPORT int kf_attn_fwd4(void* o, float* lse, const void* q, const void* k,
                            const void* v, int64_t B, int64_t S, int64_t Hq,
                            int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
                            float scale, int causal, void* stream);
+KF_EXPORT int kf_attn_fwd5(void* o, float* lse, const void* q, const void* k,
+                           const void* v, int64_t B, int64_t S, int64_t Hq,
+                           int64_t Hkv, int64_t D, int64_t qts, int64_t kts,
+                           float scale, int causal, void* stream);
 
-static bool kf_attn_v4_enabled() {
+static int kf_attn_impl() {
   static int cached = -1;
   if (cached < 0) {
-    const char* e = getenv("KF_ATTN_V4");
-    cached = (e == nullptr || e[0] != '0') ? 1 : 0;
+    const char* e = getenv("KF_ATTN_IMPL");
+    cached = (e && e[0] >= '3' && e[0] <= '5') ? e[0] - '0' : 5;
+    const char* v4 = getenv("KF_ATTN_V4");  // legacy bisection knob
+    if (v4 && v4[0] == '0' && cached > 3) cached = 3;
   }
-  return cached == 1;
+  return cached;
 }
 
 KF_EXPORT int kf_attn_fwd(void* o, float* lse, const void* q, const void* k,
@@ -442,7 +449,10 @@ KF_EXPORT int kf_attn_fwd(void* o, float* lse, const void* q, const void* k,
   if (D != AT_D || S % AT_QT || Hq % Hkv) return (int)hipErrorInvalidValue;
   if (qts == 0) qts = Hq * AT_D;
   if (kts == 0) kts = Hkv * AT_D;
-  if (S % A8_QT == 0 && kf_attn_v4_enabled())
+  if (S % A8_QT == 0 && kf_attn_impl() == 5)
+    return kf_attn_fwd5(o, lse, q, k, v, B, S, Hq, Hkv, D, qts, kts, scale,
+                        causal, stream);
+  if (S % A8_QT == 0 && kf_attn_impl() == 4)
     return kf_attn_fwd4(o, lse, q, k, v, B, S, Hq, Hkv, D, qts, kts, scale,
                         causal, stream);
   if (S % A8_QT == 0) {  // 8-wave swapped kernel for the training shapes
