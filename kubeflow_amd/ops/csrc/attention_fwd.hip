@@ -435,7 +435,7 @@ static int kf_attn_impl() {
   static int cached = -1;
   if (cached < 0) {
     const char* e = getenv("KF_ATTN_IMPL");
-    cached = (e && e[0] >= '3' && e[0] <= '5') ? e[0] - '0' : 5;
+    cached = (e && e[0] >= '3' && e[0] <= '5') ? e[0] - '0' : 4;
     const char* v4 = getenv("KF_ATTN_V4");  // legacy bisection knob
     if (v4 && v4[0] == '0' && cached > 3) cached = 3;
   }

@@ -7,7 +7,14 @@
 // ~18% no matter how well softmax/staging overlap.
 //
 // v5 halves the LDS traffic per MFMA: 4 waves x 64 q-rows per wave at
-// ONE wave per SIMD (1 block/CU, ~512-VGPR budget):
+// ONE wave per SIMD (1 block/CU, ~512-VGPR budget).
+//
+// MEASURED NEGATIVE RESULT (kept for the record): 253 TF vs v4's 445 on
+// B4 Hq32 Hkv8 S4096 causal. The register content (~390 VGPR incl. AGPR
+// aliasing; forcing 2 waves/SIMD spills 135) pins the kernel at 1
+// wave/SIMD, and without a second wave the exposed LDS/MFMA latencies and
+// softmax VALU cost more than the halved LDS traffic saves. v4 stays the
+// default; this file documents the structure-vs-occupancy tradeoff.
 //  * each K fragment (ds_read_b128) now feeds TWO QK^T MFMAs (two
 //    independent accumulator chains, one per 32-q half — the chains also
 //    interleave to fill the MFMA pipe);
