@@ -563,7 +563,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
       kf_bf16x8 ds0 = kf_exchange8(sacc, 0), ds1 = kf_exchange8(sacc, 8);
       const unsigned vbase =
           (unsigned)(size_t)(q_lds[cur]) + tr_off + (mt << 10);
-      KF_TR_ACC_LOOP(dkacc, vbase, ds0, ds1);
+      KF_TR_ACC_LOOP_B(dkacc, vbase, ds0, ds1);
     }
     if (have_next) stage_write(cur ^ 1);
     __syncthreads();
