@@ -330,6 +330,14 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
 
   const unsigned short* kvb_k =
       k + (b * S + kvrow_g) * kts + (int64_t)hkv * AB_D;
+  // persistent K B-fragments: loaded ONCE; without the explicit array the
+  // compiler rematerialized these 8 global loads inside every sub-block
+  // (regression caught by rocprof: dv8 2.97 ms vs round-1 2.16)
+  kf_bf16x8 kbf[8];
+#pragma unroll
+  for (int kk = 0; kk < 8; ++kk)
+    kbf[kk] =
+        *reinterpret_cast<const kf_bf16x8*>(kvb_k + kk * 16 + hi * 8);
 
   kf_f32x16 dvacc[4];
 #pragma unroll
@@ -394,9 +402,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
       for (int kk = 0; kk < 8; ++kk) {
         kf_bf16x8 qa = *reinterpret_cast<const kf_bf16x8*>(
             q_lds[cur] + kf_subrow8(mt * 32 + l31, kk * 32 + hi * 16));
-        kf_bf16x8 kb = *reinterpret_cast<const kf_bf16x8*>(
-            kvb_k + kk * 16 + hi * 8);
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kb, sacc, 0, 0, 0);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kbf[kk], sacc, 0,
+                                                       0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
 
@@ -462,6 +469,14 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
       k + (b * S + kvrow_g) * kts + (int64_t)hkv * AB_D;
   const unsigned short* kvb_v =
       v + (b * S + kvrow_g) * kts + (int64_t)hkv * AB_D;
+  kf_bf16x8 kbf[8], vbf[8];  // persistent B-fragments (see dv8 note)
+#pragma unroll
+  for (int kk = 0; kk < 8; ++kk) {
+    kbf[kk] =
+        *reinterpret_cast<const kf_bf16x8*>(kvb_k + kk * 16 + hi * 8);
+    vbf[kk] =
+        *reinterpret_cast<const kf_bf16x8*>(kvb_v + kk * 16 + hi * 8);
+  }
 
   kf_f32x16 dkacc[4];
 #pragma unroll
@@ -527,15 +542,12 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
       for (int kk = 0; kk < 8; ++kk) {
         kf_bf16x8 qa = *reinterpret_cast<const kf_bf16x8*>(
             q_lds[cur] + kf_subrow8(mt * 32 + l31, kk * 32 + hi * 16));
-        kf_bf16x8 kb = *reinterpret_cast<const kf_bf16x8*>(
-            kvb_k + kk * 16 + hi * 8);
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kb, sacc, 0, 0, 0);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kbf[kk], sacc, 0,
+                                                       0, 0);
         kf_bf16x8 da = *reinterpret_cast<const kf_bf16x8*>(
             do_lds[cur] + kf_subrow8(mt * 32 + l31, kk * 32 + hi * 16));
-        kf_bf16x8 vb = *reinterpret_cast<const kf_bf16x8*>(
-            kvb_v + kk * 16 + hi * 8);
-        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vb, dpacc, 0, 0,
-                                                        0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vbf[kk], dpacc,
+                                                        0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
 
