@@ -318,8 +318,13 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
   __shared__ unsigned char dot_lds[2][DKV8_QT * AB_D * 2]; // subtiled
   __shared__ float lse_s[2][DKV8_QT];                      // pre-mul log2e
 
-  const int kt = blockIdx.x, hkv = blockIdx.y;
-  const int64_t b = blockIdx.z;
+  // 1-D grid decoded kt-major: causal work decreases with kt, so the
+  // longest blocks (kt=0) dispatch first (LPT order — with only
+  // S/256 x Hkv x B blocks the schedule tail otherwise sets the wall)
+  const int kt = blockIdx.x / (Hkv * (int)B);
+  const int rest = blockIdx.x % (Hkv * (int)B);
+  const int hkv = rest % Hkv;
+  const int64_t b = rest / Hkv;
   const int g = Hq / Hkv;
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE;
@@ -455,8 +460,10 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
   __shared__ unsigned char do_lds[2][DKV8_QT * AB_D * 2];  // subtiled
   __shared__ float lse_s2[2][DKV8_QT], dlt_s2[2][DKV8_QT];
 
-  const int kt = blockIdx.x, hkv = blockIdx.y;
-  const int64_t b = blockIdx.z;
+  const int kt = blockIdx.x / (Hkv * (int)B);  // LPT decode (see dv8)
+  const int rest = blockIdx.x % (Hkv * (int)B);
+  const int hkv = rest % Hkv;
+  const int64_t b = rest / Hkv;
   const int g = Hq / Hkv;
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE;
@@ -614,7 +621,7 @@ KF_EXPORT int kf_attn_bwd8_dkv(void* dk, void* dv, const void* q,
                                int64_t B, int64_t S, int64_t Hq, int64_t Hkv,
                                int64_t qts, int64_t kts, int64_t dkts,
                                float scale, int causal, void* stream) {
-  dim3 gkv((unsigned)(S / DKV8_KT), (unsigned)Hkv, (unsigned)B);
+  dim3 gkv((unsigned)((S / DKV8_KT) * Hkv * B), 1, 1);
   hipLaunchKernelGGL(kf_attn_dv8_kernel, gkv, dim3(512), 0,
                      (hipStream_t)stream, (unsigned short*)dv,
                      (const unsigned short*)q, (const unsigned short*)k,
