@@ -165,8 +165,13 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
   __shared__ unsigned char k_lds[2][DQ8_KT * AB_D * 2];  // subtiled
   __shared__ unsigned char v_lds[2][DQ8_KT * AB_D * 2];  // subtiled
 
-  const int qt = blockIdx.x, hq = blockIdx.y;
-  const int64_t b = blockIdx.z;
+  // 1-D LPT grid: longest q-tiles (largest qt) dispatch first
+  const int nqt = gridDim.x / (Hq * (int)B);
+  const int qt = causal ? (nqt - 1 - blockIdx.x / (Hq * (int)B))
+                        : (int)(blockIdx.x / (Hq * (int)B));
+  const int rest = blockIdx.x % (Hq * (int)B);
+  const int hq = rest % Hq;
+  const int64_t b = rest / Hq;
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE;
@@ -605,7 +610,7 @@ KF_EXPORT int kf_attn_bwd8_dq(void* dq, const void* q, const void* k,
                               int64_t S, int64_t Hq, int64_t Hkv, int64_t qts,
                               int64_t kts, int64_t dqts, float scale,
                               int causal, void* stream) {
-  dim3 gq((unsigned)(S / DQ8_QT), (unsigned)Hq, (unsigned)B);
+  dim3 gq((unsigned)((S / DQ8_QT) * Hq * B), 1, 1);
   hipLaunchKernelGGL(kf_attn_dq8_kernel, gq, dim3(512), 0,
                      (hipStream_t)stream, (unsigned short*)dq,
                      (const unsigned short*)q, (const unsigned short*)k,

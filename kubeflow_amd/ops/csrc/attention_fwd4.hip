@@ -84,8 +84,14 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   __shared__ unsigned char k_lds[2][A4_KT * A4_D * 2];  // 16 KiB x2, swizzled
   __shared__ unsigned char v_lds[2][A4_KT * A4_D * 2];  // 16 KiB x2, subtiled
 
-  const int qt = blockIdx.x, hq = blockIdx.y;
-  const int64_t b = blockIdx.z;
+  // 1-D grid, qt-major DESCENDING: causal work grows with qt, so the
+  // longest q-tiles dispatch first (LPT order)
+  const int nqt = gridDim.x / (Hq * (int)B);
+  const int qt = causal ? (nqt - 1 - blockIdx.x / (Hq * (int)B))
+                        : (int)(blockIdx.x / (Hq * (int)B));
+  const int rest = blockIdx.x % (Hq * (int)B);
+  const int hq = rest % Hq;
+  const int64_t b = rest / Hq;
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE;
@@ -352,7 +358,7 @@ static int kf_attn_fwd4_launch(void* o, float* lse, const void* q,
   if (D != A4_D || S % A4_QT || Hq % Hkv) return (int)hipErrorInvalidValue;
   if (qts == 0) qts = Hq * A4_D;
   if (kts == 0) kts = Hkv * A4_D;
-  dim3 grid((unsigned)(S / A4_QT), (unsigned)Hq, (unsigned)B);
+  dim3 grid((unsigned)((S / A4_QT) * Hq * B), 1, 1);
   hipLaunchKernelGGL(kf_attn_fwd4_kernel<ABL>, grid, dim3(A4_THREADS), 0,
                      (hipStream_t)stream, (unsigned short*)o, lse,
                      (const unsigned short*)q, (const unsigned short*)k,
