@@ -42,11 +42,18 @@ def main():
     # tok/s, mb8 OOM (280 GB activations) — 6 is the per-GPU sweet spot.
     args = ap.parse_args()
 
+    if int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        os.environ.setdefault("KF_COMM_STATS", "1")  # comm/overlap stats
     rank, world, device = kdist.init_distributed()
     if world != args.gpus and rank == 0:
         print(f"# note: WORLD_SIZE={world} != --gpus {args.gpus}; using {world}",
               file=sys.stderr)
     n = world
+    print(f"# rank={rank}/{world} device={device} "
+          f"backend={'nccl' if device.type == 'cuda' and world > 1 else ('gloo' if world > 1 else 'none')} "
+          f"bucket_mb={os.environ.get('KF_DDP_BUCKET_MB', '64')} "
+          f"zero={os.environ.get('KF_ZERO') == '1'}", file=sys.stderr,
+          flush=True)
 
     torch.manual_seed(1234 + rank)
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
@@ -68,8 +75,12 @@ def main():
     if device.type == "cuda":
         torch.cuda.synchronize()
     t0 = time.time()
+    comm_ms = []
     for _ in range(args.steps):
         loss = trainer.step(tokens, targets)
+        c = getattr(trainer.ddp, "last_comm_ms", None)
+        if c is not None:
+            comm_ms.append(c)
     kdist.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
@@ -111,6 +122,12 @@ def main():
                 "loss": round(float(loss.item()), 4),
             },
         }
+        if comm_ms:
+            # GPU time the grad all-reduces occupied on the comm stream
+            # (they overlap backward; comparing against ms_per_step shows
+            # how much headroom the overlap has at this N / bucket size)
+            out["comm_ms_per_step"] = round(sum(comm_ms) / len(comm_ms), 2)
+            out["bucket_mb"] = float(os.environ.get("KF_DDP_BUCKET_MB", "64"))
         print(json.dumps(out), flush=True)
     if os.environ.get("KF_BENCH_MEM") and device.type == "cuda":
         peak = torch.cuda.max_memory_allocated(device) / (1 << 30)

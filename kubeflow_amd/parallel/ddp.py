@@ -21,13 +21,14 @@ from .flat import FlatParamSpace, _aligned
 
 
 class _Bucket:
-    __slots__ = ("start", "end", "param_ids", "pending", "work")
+    __slots__ = ("start", "end", "param_ids", "pending", "work", "ev")
 
     def __init__(self, start: int, end: int):
         self.start, self.end = start, end
         self.param_ids: List[int] = []
         self.pending = 0
         self.work = None
+        self.ev = None  # (start_event, end_event) when comm timing is on
 
 
 class BucketedDDP:
@@ -45,6 +46,13 @@ class BucketedDDP:
         self._avg_supported = (self.enabled and dist.is_initialized()
                                and dist.get_backend(process_group) == "nccl")
         self._sync = False
+        # comm/compute overlap stats for the scaling bench (KF_COMM_STATS):
+        # CUDA events bracket each bucket's all-reduce on the comm stream;
+        # last_comm_ms is their summed GPU time for the latest step.
+        self._timing = (os.environ.get("KF_COMM_STATS") == "1"
+                        and self.comm_stream is not None)
+        self.last_comm_ms: float | None = None
+        self.bucket_mb = bucket_mb
         self._build_buckets(int(bucket_mb * 1024 * 1024 / flat.grad.element_size()))
         if self.enabled:
             self._register_hooks()
@@ -90,8 +98,17 @@ class BucketedDDP:
         if self.comm_stream is not None:
             self.comm_stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self.comm_stream):
-                bucket.work = dist.all_reduce(view, op=op, group=self.pg,
-                                              async_op=True)
+                if self._timing:
+                    e0 = torch.cuda.Event(enable_timing=True)
+                    e1 = torch.cuda.Event(enable_timing=True)
+                    e0.record(self.comm_stream)
+                    bucket.work = dist.all_reduce(view, op=op, group=self.pg,
+                                                  async_op=True)
+                    e1.record(self.comm_stream)
+                    bucket.ev = (e0, e1)
+                else:
+                    bucket.work = dist.all_reduce(view, op=op, group=self.pg,
+                                                  async_op=True)
         else:
             bucket.work = dist.all_reduce(view, op=op, group=self.pg,
                                           async_op=True)
@@ -119,6 +136,14 @@ class BucketedDDP:
                 b.work.wait()
         if self.comm_stream is not None:
             torch.cuda.current_stream().wait_stream(self.comm_stream)
+        if self._timing:
+            torch.cuda.synchronize()
+            ms = 0.0
+            for b in self.buckets:
+                if b.ev is not None:
+                    ms += b.ev[0].elapsed_time(b.ev[1])
+                    b.ev = None
+            self.last_comm_ms = ms
         self._sync = False
         if not self._avg_supported:
             self.flat.grad.div_(self.world)
