@@ -38,13 +38,35 @@ class InferenceServiceReconciler(Reconciler):
     kind = "InferenceService"
 
     def __init__(self, store: ObjectStore, scheduler: GangScheduler,
-                 serving_dir: str):
+                 serving_dir: str, volumes_dir: Optional[str] = None):
         super().__init__(store)
         self.scheduler = scheduler
         self.serving_dir = serving_dir
+        self.volumes_dir = volumes_dir
         self.gangs: Dict[str, ProcessGang] = {}
         self.ports: Dict[str, int] = {}
         self.key_uid: Dict[tuple, str] = {}
+
+    def _resolve_storage_uri(self, uri: str, namespace: Optional[str]) -> str:
+        """KServe storageUri dialects in the single-node model: pvc://
+        resolves against the PVC-directory root (controllers/volume.py);
+        file:// and bare paths pass through; cloud schemes are rejected
+        like the tensorboard logspath dialect (no object stores here).
+        Raises ValueError (terminal InvalidSpec) on unusable URIs."""
+        if uri.startswith("pvc://"):
+            rest = uri[len("pvc://"):]
+            claim, _, sub = rest.partition("/")
+            if not claim or self.volumes_dir is None:
+                raise ValueError(f"unresolvable storageUri {uri!r}")
+            return os.path.join(self.volumes_dir, namespace or "default",
+                                claim, sub)
+        if uri.startswith("file://"):
+            return uri[len("file://"):]
+        if "://" in uri:
+            raise ValueError(
+                f"unsupported storageUri scheme {uri!r}: this platform "
+                "serves pvc:// and local paths (no gs://, s3://)")
+        return uri
 
     def reconcile(self, namespace: Optional[str], name: str) -> None:
         svc = self.store.get(self.kind, name, namespace)
@@ -137,6 +159,19 @@ class InferenceServiceReconciler(Reconciler):
             "max_slots": pred.get("maxSlots", 16),
             "world_size": 1,
         }
+        if pred.get("storageUri"):
+            try:
+                spec["ckpt_dir"] = self._resolve_storage_uri(
+                    pred["storageUri"], m.get("namespace"))
+                spec["storage_uri"] = pred["storageUri"]
+            except ValueError as e:
+                self.scheduler.release(uid)
+                set_condition(svc, "Failed", "True", "InvalidStorageUri",
+                              str(e))
+                self.store.update(svc, check_version=False)
+                self.store.record_event(svc, "InvalidStorageUri", str(e),
+                                        "Warning")
+                return
         poddefaults = self.store.list("PodDefault", m.get("namespace"))
         try:
             gang = launch_gang(

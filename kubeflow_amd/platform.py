@@ -55,7 +55,8 @@ class Platform:
         from kubeflow_amd.controllers.inference import InferenceServiceReconciler
         self.inference = InferenceServiceReconciler(
             self.store, self.scheduler,
-            os.path.join(self.root_dir, "serving"))
+            os.path.join(self.root_dir, "serving"),
+            volumes_dir=os.path.join(self.root_dir, "volumes"))
         self.manager.register(self.inference)
         self._controllers.append(self.inference)
         from kubeflow_amd.controllers.katib import (ExperimentReconciler,

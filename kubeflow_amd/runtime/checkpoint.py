@@ -147,3 +147,57 @@ def load(trainer, ckpt_dir: str, rank: int) -> int:
     if opt.get("cuda_rng") is not None and torch.cuda.is_available():
         torch.cuda.set_rng_state(opt["cuda_rng"])
     return trainer.step_num
+
+
+def load_model_weights(model, ckpt_path: str) -> int:
+    """Serving-side weight load (the KServe `storageUri` analog): restore a
+    built model's parameters from a training checkpoint WITHOUT a trainer
+    (no grad/optimizer buffers). Accepts the checkpoint root (reads its
+    `latest` marker) or a specific step-<K> directory. Returns the step.
+
+    The flat buffer layout matches FlatParamSpace (parallel/flat.py):
+    named_parameters() reversed, 256-element alignment — the same
+    build_model() on both sides guarantees identical ordering.
+    """
+    from kubeflow_amd.parallel.flat import _aligned
+
+    d = ckpt_path
+    if os.path.isdir(d) and os.path.exists(os.path.join(d, "latest")):
+        d = latest_dir(ckpt_path)
+    if d is None or not os.path.isdir(d):
+        raise FileNotFoundError(f"no checkpoint at {ckpt_path}")
+    meta = {}
+    meta_path = os.path.join(d, "meta.json")
+    if os.path.exists(meta_path):
+        with open(meta_path) as f:
+            meta = json.load(f)
+    if os.path.exists(os.path.join(d, "model-rank1.bin")) or \
+            os.path.exists(os.path.join(d, "model-rank1.pt")):
+        raise ValueError(
+            f"{d} holds a TP/PP-sharded checkpoint; serving loads require "
+            "a replicated (DDP/ZeRO) checkpoint or a merge pass")
+
+    params = [p for _, p in model.named_parameters() if p.requires_grad]
+    params.reverse()
+    total = sum(_aligned(p.numel()) for p in params)
+
+    if meta.get("format", 1) >= 2:
+        dtype = getattr(torch, meta.get("flat_dtype",
+                                        "bfloat16").replace("torch.", ""))
+        flat = torch.empty(total, dtype=dtype, device="cpu")
+        fastio.read_into(_model_path(d, 0, ".bin"), flat)
+    else:
+        blob = torch.load(_model_path(d, 0, ".pt"), map_location="cpu",
+                          weights_only=False)
+        flat = blob["flat_data"]
+        if flat.numel() != total:
+            raise ValueError(
+                f"checkpoint flat size {flat.numel()} != model {total}")
+    off = 0
+    with torch.no_grad():
+        for p in params:
+            n = p.numel()
+            p.data.copy_(flat[off:off + n].view(p.shape).to(
+                device=p.device, dtype=p.dtype))
+            off += _aligned(n)
+    return int(meta.get("step", 0))

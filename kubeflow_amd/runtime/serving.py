@@ -77,7 +77,8 @@ class Request:
 
 class InferenceEngine:
     def __init__(self, model_name: str, device=None, max_slots: int = 32,
-                 smax: int = 4096, max_batch: int = 32):
+                 smax: int = 4096, max_batch: int = 32,
+                 storage_uri: str | None = None):
         self.device = device or (torch.device("cuda", 0)
                                  if torch.cuda.is_available()
                                  else torch.device("cpu"))
@@ -85,6 +86,14 @@ class InferenceEngine:
                  else torch.float32)
         self.model: LlamaModel = build_model(model_name, device=self.device,
                                              dtype=dtype)
+        # KServe storageUri analog: load trained weights from a checkpoint
+        # directory (resolved to a filesystem path by the controller);
+        # without it the engine serves random-init weights (dev mode).
+        self.loaded_step: int | None = None
+        if storage_uri:
+            from kubeflow_amd.runtime import checkpoint as _ckpt
+            self.loaded_step = _ckpt.load_model_weights(self.model,
+                                                        storage_uri)
         self.model.eval()
         cfg = self.model.cfg
         smax = min(smax, cfg.max_seq_len)

@@ -34,6 +34,7 @@ def build_app(spec: dict) -> FastAPI:
         max_slots=int(spec.get("max_slots", 16)),
         smax=int(spec.get("max_seq_len", 2048)),
         max_batch=int(spec.get("max_batch", 16)),
+        storage_uri=spec.get("ckpt_dir") or None,
     ).start()
     app = FastAPI(title=f"kubeflow-amd inference: {name}")
     app.state.engine = engine
@@ -51,6 +52,8 @@ def build_app(spec: dict) -> FastAPI:
     def model_meta():
         return {"name": name, "ready": True,
                 "model": spec.get("model"),
+                "loaded_step": engine.loaded_step,
+                "storage_uri": spec.get("storage_uri"),
                 "stats": engine.stats}
 
     @app.post(f"/v1/models/{name}:predict")
