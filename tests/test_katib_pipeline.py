@@ -3,6 +3,7 @@
 Mirrors the reference's katib_studyjob_test (submit CR, poll conditions)
 and BASELINE config 5's DAG chaining.
 """
+import os
 import time
 
 import pytest
@@ -224,3 +225,48 @@ def test_space_unit_roundtrip_property():
         assert b["bs"] == a["bs"] and b["opt"] == a["opt"]
 
     check()
+
+
+def test_pipeline_train_then_deploy_trained_weights(tmp_path):
+    """BASELINE config 5's train->deploy seam with real weight flow: the
+    train task checkpoints to a PVC; the deploy task's InferenceService
+    uses storageUri=pvc://... and must serve the TRAINED weights (engine
+    reports the checkpoint step in its model metadata)."""
+    import json
+    import urllib.request
+
+    from kubeflow_amd.api.objects import get_condition
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        plat.store.create(new_object(
+            "PersistentVolumeClaim", "model-store", "default",
+            spec={"resources": {"requests": {"storage": "1Gi"}}}))
+        ck = os.path.join(str(tmp_path), "volumes", "default", "model-store",
+                          "llama")
+        run = new_object("PipelineRun", "pipe-deploy", "default", spec={
+            "tasks": [
+                {"name": "train", "dependencies": [],
+                 "template": {"model": "llama-tiny", "steps": 2,
+                              "gpus_per_replica": 0, "micro_batch": 2, "seq_len": 32,
+                              "checkpoint_dir": ck}},
+                {"name": "deploy", "dependencies": ["train"],
+                 "kind": "InferenceService",
+                 "template": {"model": "llama-tiny", "gpus": 0,
+                              "maxSlots": 2, "maxSeqLen": 128,
+                              "storageUri": "pvc://model-store/llama"}},
+            ],
+        }, api_version="pipelines.kubeflow.org/v1")
+        plat.store.create(run)
+        state, obj = _wait(plat.store, "PipelineRun", "pipe-deploy",
+                           "default", timeout=300)
+        assert state == "Succeeded", obj["status"]
+        # the deployed engine loaded the trained checkpoint
+        svc = plat.store.get("InferenceService", "pipe-deploy-deploy",
+                             "default")
+        ready = get_condition(svc, "Ready")
+        url = ready["message"]
+        with urllib.request.urlopen(
+                f"{url}/v1/models/pipe-deploy-deploy", timeout=10) as r:
+            meta = json.load(r)
+        assert meta["loaded_step"] == 2, meta
+        assert meta["storage_uri"] == "pvc://model-store/llama"
