@@ -35,7 +35,8 @@ import time
 from typing import Optional
 
 from fastapi import FastAPI, Request
-from fastapi.responses import JSONResponse, PlainTextResponse
+from fastapi.responses import (FileResponse, JSONResponse,
+                               PlainTextResponse)
 
 from kubeflow_amd.api import (ObjectStore, new_object, NotFoundError,
                               AlreadyExistsError, ConflictError)
@@ -457,7 +458,30 @@ def build_app(store: ObjectStore, scheduler=None,
                          rank, tail)
 
     # ------------------------------------------------ central dashboard UI
+    # ----------------------------------------------------------- dashboard
+    # SPA (kubeflow_amd/dashboard/static): namespace selector + routed
+    # resource views with exponential-backoff polling over this BFF —
+    # the central-dashboard + crud-web-app frontend layer rebuilt as
+    # hand-rolled ES modules (no build step). /classic keeps the
+    # server-rendered fallback table.
+    _static_dir = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "dashboard", "static")
+
     @app.get("/")
+    def dashboard_spa():
+        return FileResponse(os.path.join(_static_dir, "index.html"))
+
+    @app.get("/static/{fname}")
+    def dashboard_static(fname: str):
+        safe = os.path.basename(fname)
+        path = os.path.join(_static_dir, safe)
+        if not os.path.exists(path):
+            raise ApiError(404, f"no static asset {safe}")
+        media = {"js": "text/javascript", "css": "text/css",
+                 "html": "text/html"}.get(safe.rsplit(".", 1)[-1])
+        return FileResponse(path, media_type=media)
+
+    @app.get("/classic")
     def dashboard(request: Request):
         """Minimal server-rendered dashboard shell — the central-dashboard
         equivalent (namespaces, workloads, activities, GPU utilization)."""
