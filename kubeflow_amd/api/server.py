@@ -97,6 +97,10 @@ def build_app(store: ObjectStore, scheduler=None,
                                registry=registry)
         m_nb_deleted = Counter("notebook_delete_total", "deleted notebooks",
                                registry=registry)
+        g_nb_failed = Gauge("notebook_failed_total", "failed notebooks",
+                            registry=registry)
+        g_nb_culled = Gauge("notebook_culled_total", "culled notebooks",
+                            registry=registry)
     except ImportError:  # pragma: no cover
         registry = None
 
@@ -277,12 +281,29 @@ def build_app(store: ObjectStore, scheduler=None,
         return ok(user, message=f"{kind} {name} deleted")
 
     # ------------------------------------------------------------- kfam
+    def _owner_or_admin(me: str, ns: str) -> bool:
+        """isOwnerOrAdmin (access-management/kfam/api_default.go:116-118):
+        cluster admin, the namespace profile's owner, or a kfam admin."""
+        if is_cluster_admin(me):
+            return True
+        for p in store.list("Profile"):
+            if p["metadata"]["name"] == ns and _owner_name(p) == me:
+                return True
+        return bindings.role_for(me, ns) == "admin"
+
     @app.get("/kfam/v1/bindings")
     def kfam_list_bindings(request: Request, user: Optional[str] = None,
                            namespace: Optional[str] = None,
                            role: Optional[str] = None):
         me = user_of(request)
-        return ok(me, bindings=bindings.list(user, namespace, role))
+        out = bindings.list(user, namespace, role)
+        if not disable_auth and not is_cluster_admin(me):
+            # reads are permission-checked like writes: a non-admin sees
+            # only their own bindings plus namespaces they own/administer
+            out = [b for b in out
+                   if b.get("user", {}).get("name") == me
+                   or _owner_or_admin(me, b.get("referredNamespace", ""))]
+        return ok(me, bindings=out)
 
     @app.post("/kfam/v1/bindings")
     async def kfam_create_binding(request: Request):
@@ -290,9 +311,8 @@ def build_app(store: ObjectStore, scheduler=None,
         csrf_check(request)
         body = await request.json()
         ns = body["referredNamespace"]
-        if not (is_cluster_admin(me) or
-                bindings.role_for(me, ns) == "admin"):
-            raise ApiError(403, f"{me} is not admin of {ns}")
+        if not _owner_or_admin(me, ns):
+            raise ApiError(403, f"{me} is not owner or admin of {ns}")
         b = bindings.create(body["user"]["name"], ns,
                             body.get("roleRef", {}).get("name", "edit")
                             .replace("kubeflow-", ""))
@@ -304,9 +324,8 @@ def build_app(store: ObjectStore, scheduler=None,
         csrf_check(request)
         body = await request.json()
         ns = body["referredNamespace"]
-        if not (is_cluster_admin(me) or
-                bindings.role_for(me, ns) == "admin"):
-            raise ApiError(403, f"{me} is not admin of {ns}")
+        if not _owner_or_admin(me, ns):
+            raise ApiError(403, f"{me} is not owner or admin of {ns}")
         bindings.delete(body["user"]["name"], ns,
                         body.get("roleRef", {}).get("name", "edit")
                         .replace("kubeflow-", ""))
@@ -522,9 +541,22 @@ def build_app(store: ObjectStore, scheduler=None,
         if registry is None:
             return PlainTextResponse("")
         m_heartbeat.inc()
-        running = sum(1 for nb in store.list("Notebook")
-                      if nb.get("status", {}).get("readyReplicas"))
-        g_nb_running.set(running)
+        nbs = store.list("Notebook")
+        g_nb_running.set(sum(1 for nb in nbs
+                             if nb.get("status", {}).get("readyReplicas")))
+        # scraped on pull like the reference's custom Collector
+        # (notebook-controller/pkg/metrics/metrics.go:82-99)
+        g_nb_failed.set(sum(
+            1 for nb in nbs
+            if any(c.get("type") == "Failed" and c.get("status") == "True"
+                   for c in nb.get("status", {}).get("conditions", []))
+            or (nb.get("status", {}).get("containerState", {})
+                .get("waiting", {}).get("reason") in ("Error",
+                                                      "CrashLoopBackOff"))))
+        g_nb_culled.set(sum(
+            1 for nb in nbs
+            if nb["metadata"].get("annotations", {}).get(
+                "notebooks.kubeflow.org/culled") == "true"))
         return PlainTextResponse(generate_latest(registry).decode())
 
     return app

@@ -57,10 +57,71 @@ class NotebookReconciler(Reconciler):
     # ---------------------------------------------------------- versioning
     @staticmethod
     def normalize(nb: dict) -> dict:
-        """Accept v1alpha1/v1beta1/v1 and treat spec.template.spec uniformly
-        (conversion copies spec.template.spec verbatim in the reference)."""
-        if nb.get("apiVersion", "").endswith(("v1alpha1", "v1", "v1beta1")):
-            return nb
+        """Three-version conversion through the v1beta1 hub (the reference's
+        notebook_conversion.go:25-69: spec.template.spec and status copy
+        verbatim; conditions convert field-by-field). Accepts:
+          * v1alpha1 / v1 / v1beta1 pod-template shape
+            (spec.template.spec.containers[0] with image/resources/env), or
+          * the flat session shape ({image, cpu, memory, gpus, env}) that
+            the spawner API writes.
+        Returns the object rewritten to the v1beta1 hub carrying BOTH
+        shapes (flat keys for the session runtime, the template for
+        pod-shape consumers); the original apiVersion is preserved in the
+        `notebooks.kubeflow.org/original-api-version` annotation.
+        """
+        nb = dict(nb)
+        spec = dict(nb.get("spec") or {})
+        meta = nb.setdefault("metadata", {})
+        orig_version = nb.get("apiVersion", "kubeflow.org/v1beta1")
+        containers = ((spec.get("template") or {}).get("spec") or {}).get(
+            "containers") or []
+        if containers:
+            c0 = containers[0]
+            spec.setdefault("image", c0.get("image",
+                                            "kubeflow-amd/session:latest"))
+            limits = (c0.get("resources") or {}).get("limits") or {}
+            requests = (c0.get("resources") or {}).get("requests") or {}
+            if "gpus" not in spec:
+                spec["gpus"] = int(limits.get("amd.com/gpu", 0) or 0)
+            spec.setdefault("cpu", str(requests.get("cpu", limits.get(
+                "cpu", "2"))))
+            spec.setdefault("memory", str(requests.get("memory", limits.get(
+                "memory", "4Gi"))))
+            if "env" not in spec and c0.get("env"):
+                spec["env"] = {e["name"]: str(e.get("value", ""))
+                               for e in c0["env"]}
+        else:
+            # synthesize the pod-template shape from the flat session spec
+            limits = {}
+            if int(spec.get("gpus", 0) or 0) > 0:
+                limits["amd.com/gpu"] = int(spec["gpus"])
+            spec["template"] = {"spec": {"containers": [{
+                "name": meta.get("name", "notebook"),
+                "image": spec.get("image", "kubeflow-amd/session:latest"),
+                "resources": {
+                    "requests": {"cpu": str(spec.get("cpu", "2")),
+                                 "memory": str(spec.get("memory", "4Gi"))},
+                    "limits": limits,
+                },
+                "env": [{"name": k, "value": str(v)}
+                        for k, v in (spec.get("env") or {}).items()],
+            }]}}
+        # conditions convert field-by-field; unknown extra keys drop like
+        # the reference's typed conversion
+        status = dict(nb.get("status") or {})
+        conds = []
+        for c in status.get("conditions") or []:
+            conds.append({k: c[k] for k in
+                          ("type", "status", "reason", "message",
+                           "lastProbeTime", "lastTransitionTime",
+                           "lastUpdateTime") if k in c})
+        status["conditions"] = conds
+        if orig_version != "kubeflow.org/v1beta1":
+            meta.setdefault("annotations", {})[
+                "notebooks.kubeflow.org/original-api-version"] = orig_version
+        nb["apiVersion"] = "kubeflow.org/v1beta1"
+        nb["spec"] = spec
+        nb["status"] = status
         return nb
 
     def reconcile(self, namespace: Optional[str], name: str) -> None:
@@ -111,6 +172,10 @@ class NotebookReconciler(Reconciler):
                 if time.time() - last > self.idle_minutes * 60:
                     nb["metadata"]["annotations"][STOP_ANNOTATION] = \
                         time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+                    # distinguishes culler stops from user stops for the
+                    # culled-notebooks metric (metrics.go:22-99)
+                    nb["metadata"]["annotations"][
+                        "notebooks.kubeflow.org/culled"] = "true"
                     self.store.record_event(nb, "Culling",
                                             "idle beyond IDLE_TIME")
             except ValueError:
@@ -176,6 +241,10 @@ class NotebookReconciler(Reconciler):
 
     @staticmethod
     def _image(nb) -> str:
+        # normalize() guarantees both shapes; prefer the flat key
+        img = nb["spec"].get("image")
+        if img:
+            return img
         try:
             return nb["spec"]["template"]["spec"]["containers"][0]["image"]
         except (KeyError, IndexError):

@@ -217,3 +217,64 @@ def test_store_concurrent_writers():
         t.join(60)
     assert not errors
     assert store.get("Counter", "c", "ns")["spec"]["n"] == 200
+
+
+def test_kfam_binding_list_is_permission_checked(monkeypatch):
+    """Reads are owner-or-admin scoped like writes (api_default.go:116-118):
+    a non-admin sees only their own bindings."""
+    from kubeflow_amd.api.server import build_app
+    from kubeflow_amd.api.store import ObjectStore
+    from kubeflow_amd.kfam import BindingClient
+    from fastapi.testclient import TestClient
+
+    monkeypatch.setenv("KF_CLUSTER_ADMINS", "root@example.com")
+    monkeypatch.delenv("APP_DISABLE_AUTH", raising=False)
+    store = ObjectStore()
+    b = BindingClient(store)
+    b.create("alice@example.com", "team-a", "admin")
+    b.create("bob@example.com", "team-a", "edit")
+    b.create("carol@example.com", "team-c", "edit")
+    c = TestClient(build_app(store))
+
+    def names(resp):
+        return sorted({x["user"]["name"] for x in resp.json()["bindings"]})
+
+    # cluster admin sees everything
+    r = c.get("/kfam/v1/bindings",
+              headers={"kubeflow-userid": "root@example.com"})
+    assert names(r) == ["alice@example.com", "bob@example.com",
+                        "carol@example.com"]
+    # namespace admin sees their namespace's bindings, not others
+    r = c.get("/kfam/v1/bindings",
+              headers={"kubeflow-userid": "alice@example.com"})
+    assert names(r) == ["alice@example.com", "bob@example.com"]
+    # plain member sees only their own binding
+    r = c.get("/kfam/v1/bindings",
+              headers={"kubeflow-userid": "bob@example.com"})
+    assert names(r) == ["bob@example.com"]
+    # outsider sees nothing
+    r = c.get("/kfam/v1/bindings",
+              headers={"kubeflow-userid": "mallory@example.com"})
+    assert names(r) == []
+
+
+def test_notebook_failed_culled_metrics():
+    from kubeflow_amd.api.server import build_app
+    from kubeflow_amd.api.store import ObjectStore
+    from kubeflow_amd.api import new_object
+    from fastapi.testclient import TestClient
+
+    store = ObjectStore()
+    bad = new_object("Notebook", "nb-bad", "default",
+                     api_version="kubeflow.org/v1beta1")
+    bad["status"]["conditions"] = [{"type": "Failed", "status": "True"}]
+    store.create(bad)
+    culled = new_object("Notebook", "nb-culled", "default",
+                        api_version="kubeflow.org/v1beta1")
+    culled["metadata"]["annotations"][
+        "notebooks.kubeflow.org/culled"] = "true"
+    store.create(culled)
+    c = TestClient(build_app(store))
+    text = c.get("/metrics").text
+    assert "notebook_failed_total 1.0" in text
+    assert "notebook_culled_total 1.0" in text

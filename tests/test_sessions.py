@@ -182,3 +182,63 @@ def test_pvc_snapshot_and_restore(tmp_path):
             assert f.read() == "precious"
         evs = {e["reason"] for e in plat.store.events_for(obj)}
         assert "SnapshotRestored" in evs
+
+
+# ------------------------------------------------- notebook CRD versions
+
+def _v1_payload(name):
+    return {
+        "apiVersion": "kubeflow.org/v1", "kind": "Notebook",
+        "metadata": {"name": name, "namespace": "default"},
+        "spec": {"template": {"spec": {"containers": [{
+            "name": name, "image": "img:v1",
+            "resources": {"limits": {"amd.com/gpu": 2, "cpu": "4"},
+                          "requests": {"cpu": "4", "memory": "8Gi"}},
+            "env": [{"name": "A", "value": "1"}],
+        }]}}},
+        "status": {"conditions": [
+            {"type": "Running", "status": "True", "reason": "x",
+             "message": "m", "bogusField": "dropped"}]},
+    }
+
+
+def test_normalize_v1_pod_template():
+    from kubeflow_amd.controllers.notebook import NotebookReconciler
+    out = NotebookReconciler.normalize(_v1_payload("nb"))
+    assert out["apiVersion"] == "kubeflow.org/v1beta1"
+    assert out["metadata"]["annotations"][
+        "notebooks.kubeflow.org/original-api-version"] == "kubeflow.org/v1"
+    assert out["spec"]["image"] == "img:v1"
+    assert out["spec"]["gpus"] == 2
+    assert out["spec"]["memory"] == "8Gi"
+    assert out["spec"]["env"] == {"A": "1"}
+    # conditions convert field-by-field; unknown keys drop
+    c = out["status"]["conditions"][0]
+    assert c["type"] == "Running" and "bogusField" not in c
+
+
+def test_normalize_v1alpha1_pod_template():
+    from kubeflow_amd.controllers.notebook import NotebookReconciler
+    p = _v1_payload("nb2")
+    p["apiVersion"] = "kubeflow.org/v1alpha1"
+    out = NotebookReconciler.normalize(p)
+    assert out["apiVersion"] == "kubeflow.org/v1beta1"
+    assert out["spec"]["image"] == "img:v1"
+
+
+def test_normalize_flat_spawner_shape():
+    from kubeflow_amd.controllers.notebook import NotebookReconciler
+    out = NotebookReconciler.normalize({
+        "apiVersion": "kubeflow.org/v1beta1", "kind": "Notebook",
+        "metadata": {"name": "nb3", "namespace": "default"},
+        "spec": {"image": "img:flat", "gpus": 1, "cpu": "2",
+                 "memory": "4Gi", "env": {"B": "2"}},
+        "status": {},
+    })
+    c0 = out["spec"]["template"]["spec"]["containers"][0]
+    assert c0["image"] == "img:flat"
+    assert c0["resources"]["limits"]["amd.com/gpu"] == 1
+    assert c0["env"] == [{"name": "B", "value": "2"}]
+    # hub version untouched -> no original-version annotation
+    assert "notebooks.kubeflow.org/original-api-version" not in \
+        out["metadata"].get("annotations", {})
