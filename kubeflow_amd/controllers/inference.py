@@ -133,9 +133,18 @@ class InferenceServiceReconciler(Reconciler):
         import torch
         want_gpu = gpus > 0 and (torch.cuda.is_available()
                                  or os.environ.get("KF_FAKE_GPUS"))
+        ns = svc["metadata"].get("namespace")
+        if want_gpu:
+            from kubeflow_amd.scheduler.quota import QuotaExceeded, admit_gpus
+            try:
+                admit_gpus(self.store, self.scheduler, ns, gpus)
+            except QuotaExceeded as e:
+                self.store.record_event(svc, "QuotaExceeded", str(e),
+                                        "Warning")
+                raise RequeueAfter(2.0)
         try:
             if want_gpu:
-                alloc = self.scheduler.allocate(uid, gpus)
+                alloc = self.scheduler.allocate(uid, gpus, namespace=ns)
                 gpu_indices = alloc.gpu_indices
             else:
                 self.scheduler.allocate(uid, 0)

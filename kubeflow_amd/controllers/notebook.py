@@ -44,9 +44,11 @@ class _Session:
 class NotebookReconciler(Reconciler):
     kind = "Notebook"
 
-    def __init__(self, store: ObjectStore, sessions_dir: str):
+    def __init__(self, store: ObjectStore, sessions_dir: str,
+                 scheduler=None):
         super().__init__(store)
         self.sessions_dir = sessions_dir
+        self.scheduler = scheduler  # GPU sessions + quota accounting
         self.sessions: Dict[str, _Session] = {}
         self.key_uid: Dict[tuple, str] = {}
         self.enable_culling = os.environ.get(
@@ -194,10 +196,47 @@ class NotebookReconciler(Reconciler):
         except Exception:
             return False, None
 
+    @staticmethod
+    def _parse_mem(v) -> int:
+        v = str(v or "32Gi")
+        units = {"Ki": 1 << 10, "Mi": 1 << 20, "Gi": 1 << 30, "Ti": 1 << 40}
+        for suffix, mul in units.items():
+            if v.endswith(suffix):
+                return int(float(v[:-2]) * mul)
+        return int(float(v))
+
     def _start_session(self, nb):
         uid = nb["metadata"]["uid"]
         m = nb["metadata"]
         ns = m.get("namespace") or "default"
+        # GPU notebooks: shared (HBM-accounted) allocation through the gang
+        # scheduler + ResourceQuota admission — the reference's GPU spawn
+        # form path (jupyter form.py:262-287) backed by kube quota
+        import torch
+        gpus = int(nb["spec"].get("gpus", 0) or 0)
+        want_gpu = (gpus > 0 and self.scheduler is not None
+                    and (torch.cuda.is_available()
+                         or os.environ.get("KF_FAKE_GPUS")))
+        gpu_indices = []
+        if want_gpu:
+            from kubeflow_amd.scheduler import InsufficientResources
+            from kubeflow_amd.scheduler.quota import (QuotaExceeded,
+                                                      admit_gpus)
+            try:
+                admit_gpus(self.store, self.scheduler, ns, gpus)
+                alloc = self.scheduler.allocate(
+                    uid, gpus, exclusive=False,
+                    mem_per_gpu=self._parse_mem(nb["spec"].get("gpuMemory")),
+                    namespace=ns)
+                gpu_indices = alloc.gpu_indices
+            except (QuotaExceeded, InsufficientResources) as e:
+                reason = ("QuotaExceeded" if "Quota" in type(e).__name__
+                          else "InsufficientResources")
+                self.store.record_event(nb, reason, str(e), "Warning")
+                nb["status"]["containerState"] = {
+                    "waiting": {"reason": reason, "message": str(e)}}
+                self.store.update(nb, check_version=False)
+                raise RequeueAfter(3.0)
         port = free_port()
         workdir = os.path.join(self.sessions_dir, ns, m["name"])
         os.makedirs(workdir, exist_ok=True)
@@ -221,6 +260,8 @@ class NotebookReconciler(Reconciler):
                                     "Warning")
         env = dict(os.environ)
         env.update(explicit)
+        if gpu_indices:
+            env["HIP_VISIBLE_DEVICES"] = ",".join(map(str, gpu_indices))
         repo_root = os.path.dirname(os.path.dirname(os.path.dirname(
             os.path.abspath(__file__))))
         env["PYTHONPATH"] = repo_root + os.pathsep + env.get("PYTHONPATH", "")
@@ -251,6 +292,8 @@ class NotebookReconciler(Reconciler):
             return "kubeflow-amd/session:latest"
 
     def _stop_session(self, uid):
+        if self.scheduler is not None:
+            self.scheduler.release(uid)
         sess = self.sessions.pop(uid, None)
         if sess and sess.proc.poll() is None:
             sess.proc.terminate()

@@ -96,9 +96,25 @@ class TrainingJobReconciler(Reconciler):
         import torch
         want_gpu = gpus_per > 0 and (
             torch.cuda.is_available() or os.environ.get("KF_FAKE_GPUS"))
+        ns = job["metadata"].get("namespace")
+        if want_gpu:
+            from kubeflow_amd.scheduler.quota import QuotaExceeded, admit_gpus
+            try:
+                admit_gpus(self.store, self.scheduler, ns, n * gpus_per)
+            except QuotaExceeded as e:
+                # quota rejections requeue (a peer releasing frees budget),
+                # surfaced as Warning events like FailedCreate
+                self.store.record_event(job, "QuotaExceeded", str(e),
+                                        "Warning")
+                if not has_condition(job, "Created"):
+                    set_condition(job, "Created", "True", "JobCreated",
+                                  "blocked by ResourceQuota")
+                    self.store.update(job, check_version=False)
+                raise RequeueAfter(2.0)
         try:
             if want_gpu:
-                alloc = self.scheduler.allocate(uid, n * gpus_per)
+                alloc = self.scheduler.allocate(uid, n * gpus_per,
+                                                namespace=ns)
                 gpu_indices = alloc.gpu_indices
             else:
                 self.scheduler.allocate(uid, 0)

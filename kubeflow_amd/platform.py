@@ -72,7 +72,8 @@ class Platform:
         self._controllers.append(self.pipeline)
         from kubeflow_amd.controllers.notebook import NotebookReconciler
         self.notebook = NotebookReconciler(
-            self.store, os.path.join(self.root_dir, "notebooks"))
+            self.store, os.path.join(self.root_dir, "notebooks"),
+            scheduler=self.scheduler)
         from kubeflow_amd.controllers.tensorboard import TensorboardReconciler
         self.tensorboard = TensorboardReconciler(
             self.store, os.path.join(self.root_dir, "tensorboards"),

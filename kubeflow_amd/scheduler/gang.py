@@ -36,11 +36,14 @@ class GangScheduler:
         self._gpu_mem_used: Dict[int, int] = {g.index: 0 for g in self.inv.gpus}
         self._gpu_excl: Dict[int, Optional[str]] = {
             g.index: None for g in self.inv.gpus}
+        self._ns_of: Dict[str, str] = {}  # job_uid -> namespace (quota)
 
     def allocate(self, job_uid: str, n_gpus: int, mem_per_gpu: int = 0,
-                 exclusive: bool = True) -> Allocation:
+                 exclusive: bool = True,
+                 namespace: Optional[str] = None) -> Allocation:
         """All-or-nothing allocation of n_gpus. exclusive=True (training)
-        claims whole GPUs; exclusive=False co-schedules by HBM bytes."""
+        claims whole GPUs; exclusive=False co-schedules by HBM bytes.
+        `namespace` tags the allocation for ResourceQuota accounting."""
         with self._lock:
             if job_uid in self._alloc:
                 return self._alloc[job_uid]
@@ -69,6 +72,8 @@ class GangScheduler:
                 self._gpu_mem_used[idx] += mem_per_gpu
             alloc = Allocation(job_uid, chosen, mem_per_gpu)
             self._alloc[job_uid] = alloc
+            if namespace:
+                self._ns_of[job_uid] = namespace
             return alloc
 
     @staticmethod
@@ -82,8 +87,15 @@ class GangScheduler:
                 return window
         return free[:n]
 
+    def ns_gpu_usage(self, namespace: str) -> int:
+        """GPUs currently allocated to a namespace (quota accounting)."""
+        with self._lock:
+            return sum(len(a.gpu_indices) for uid, a in self._alloc.items()
+                       if self._ns_of.get(uid) == namespace)
+
     def release(self, job_uid: str):
         with self._lock:
+            self._ns_of.pop(job_uid, None)
             alloc = self._alloc.pop(job_uid, None)
             if alloc is None:
                 return

@@ -242,3 +242,38 @@ def test_normalize_flat_spawner_shape():
     # hub version untouched -> no original-version annotation
     assert "notebooks.kubeflow.org/original-api-version" not in \
         out["metadata"].get("annotations", {})
+
+
+def test_notebook_gpu_session_allocates_and_releases(tmp_path, monkeypatch):
+    """GPU notebooks flow through GangScheduler.allocate(exclusive=False)
+    with HBM accounting: the session env pins
+    HIP_VISIBLE_DEVICES and stopping the notebook releases the GPUs."""
+    import time as _t
+
+    from kubeflow_amd.api import new_object
+    from kubeflow_amd.platform import Platform
+
+    monkeypatch.setenv("KF_FAKE_GPUS", "2")
+    with Platform(root_dir=str(tmp_path)) as plat:
+        nb = new_object("Notebook", "gpu-nb", "default",
+                        spec={"image": "kubeflow-amd/session:latest",
+                              "gpus": 1, "gpuMemory": "16Gi"},
+                        api_version="kubeflow.org/v1beta1")
+        plat.store.create(nb)
+        deadline = _t.time() + 60
+        while _t.time() < deadline:
+            obj = plat.store.get("Notebook", "gpu-nb", "default")
+            if obj["status"].get("readyReplicas"):
+                break
+            _t.sleep(0.3)
+        assert obj["status"].get("readyReplicas") == 1, obj["status"]
+        assert plat.scheduler.ns_gpu_usage("default") == 1
+        # stop via the annotation -> GPUs released
+        obj["metadata"]["annotations"]["kubeflow-resource-stopped"] = "now"
+        plat.store.update(obj, check_version=False)
+        deadline = _t.time() + 30
+        while _t.time() < deadline:
+            if plat.scheduler.ns_gpu_usage("default") == 0:
+                break
+            _t.sleep(0.3)
+        assert plat.scheduler.ns_gpu_usage("default") == 0
