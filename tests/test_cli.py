@@ -64,8 +64,12 @@ def test_cli_submit_get_logs(served):
     assert r.returncode == 0
     r = _cli(env, "events")
     assert "JobSucceeded" in r.stdout
-    # dashboard page renders
+    # dashboard: / serves the SPA shell; /classic renders the workloads
     with urllib.request.urlopen(f"http://127.0.0.1:{port}/", timeout=5) as h:
+        page = h.read().decode()
+    assert "kubeflow-amd" in page and "/static/app.js" in page
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/classic",
+                                timeout=5) as h:
         page = h.read().decode()
     assert "kubeflow-amd" in page and "PyTorchJob" in page
 
