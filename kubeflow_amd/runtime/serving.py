@@ -69,6 +69,7 @@ class Request:
     slot: int = -1
     pos: int = 0
     generated: List[int] = field(default_factory=list)
+    token_times: List[float] = field(default_factory=list)  # per-token stamps
     done: threading.Event = field(default_factory=threading.Event)
     first_token_at: Optional[float] = None
     finished_at: Optional[float] = None
@@ -215,6 +216,7 @@ class InferenceEngine:
         req.pos = S
         req.generated.append(tok)
         req.first_token_at = time.time()
+        req.token_times.append(req.first_token_at)
         self.stats["prefill_tokens"] += S
 
     def _layer_prefill(self, layer, li, x, cos, sin, slot, S):
@@ -342,6 +344,7 @@ class InferenceEngine:
         for i, r in enumerate(acts):
             r.pos += 1
             r.generated.append(int(toks[i]))
+            r.token_times.append(now)
             self.stats["tokens_out"] += 1
             if (len(r.generated) >= r.max_new_tokens
                     or r.pos + 1 >= self.cache.smax):

@@ -21,16 +21,22 @@ def main():
     model = sys.argv[1] if len(sys.argv) > 1 else "llama3-8b"
     n_req = int(sys.argv[2]) if len(sys.argv) > 2 else 16
     max_new = int(sys.argv[3]) if len(sys.argv) > 3 else 64
-    plen = int(sys.argv[4]) if len(sys.argv) > 4 else 128
+    # prompt_len: an int, or "mix" = cycle 128/1024/4096 (the mixed-load
+    # p99 inter-token measurement of VERDICT round-1 item 9)
+    plen_arg = sys.argv[4] if len(sys.argv) > 4 else "128"
+    mixed = plen_arg == "mix"
+    lens = [128, 1024, 4096] if mixed else [int(plen_arg)]
+    plen = "mix" if mixed else int(plen_arg)
 
     eng = InferenceEngine(model, max_slots=32, smax=4096,
                           max_batch=32).start()
     # warm-up (captures the decode graphs for the buckets used)
-    eng.generate(list(range(1, plen + 1)), max_new_tokens=8, timeout=300)
+    eng.generate(list(range(1, lens[0] + 1)), max_new_tokens=8, timeout=300)
 
     results = []
     def run(i):
-        results.append(eng.generate(list(range(1 + i, plen + 1 + i)),
+        L = lens[i % len(lens)]
+        results.append(eng.generate(list(range(1 + i, L + 1 + i)),
                                     max_new_tokens=max_new, timeout=600))
 
     threads = [threading.Thread(target=run, args=(i,)) for i in range(n_req)]
@@ -46,6 +52,12 @@ def main():
                  for r in results if r.finished_at)
     ttft = sorted(1000 * (r.first_token_at - r.submitted)
                   for r in results if r.first_token_at)
+    # inter-token gaps across all streams: a long prefill stalling active
+    # decodes shows up here as p99 spikes
+    gaps = sorted(
+        1000 * (b - a)
+        for r in results
+        for a, b in zip(r.token_times, r.token_times[1:]))
     out = {
         "metric": "inferenceservice_tokens_per_s_out",
         "model": model,
@@ -56,6 +68,11 @@ def main():
         "latency_p50_ms": round(lat[len(lat) // 2], 1),
         "latency_p99_ms": round(lat[min(len(lat) - 1, int(len(lat) * 0.99))], 1),
         "ttft_p50_ms": round(ttft[len(ttft) // 2], 1),
+        "itl_p50_ms": round(gaps[len(gaps) // 2], 2) if gaps else None,
+        "itl_p99_ms": round(gaps[min(len(gaps) - 1,
+                                     int(len(gaps) * 0.99))], 2)
+                       if gaps else None,
+        "itl_max_ms": round(gaps[-1], 2) if gaps else None,
         "graph_replays": eng.stats.get("graph_replays", 0),
         "errors": sum(1 for r in results if r.error),
     }
