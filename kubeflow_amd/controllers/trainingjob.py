@@ -111,9 +111,21 @@ class TrainingJobReconciler(Reconciler):
                                   "blocked by ResourceQuota")
                     self.store.update(job, check_version=False)
                 raise RequeueAfter(2.0)
+        # shared GPU allocation (HBM-accounted co-scheduling) for small
+        # workloads like HPO trials: template {gpu_shared: true,
+        # gpu_memory: "24Gi"} packs multiple jobs per GPU
+        shared = bool(template.get("gpu_shared"))
+        mem = 0
+        if shared:
+            v = str(template.get("gpu_memory", "24Gi"))
+            units = {"Ki": 1 << 10, "Mi": 1 << 20, "Gi": 1 << 30}
+            mem = next((int(float(v[:-2]) * m) for sfx, m in units.items()
+                        if v.endswith(sfx)), 24 << 30)
         try:
             if want_gpu:
                 alloc = self.scheduler.allocate(uid, n * gpus_per,
+                                                exclusive=not shared,
+                                                mem_per_gpu=mem,
                                                 namespace=ns)
                 gpu_indices = alloc.gpu_indices
             else:
