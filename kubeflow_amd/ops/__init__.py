@@ -23,6 +23,7 @@ from . import _backend
 
 __all__ = [
     "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
+    "flash_attention_rect",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
     "swiglu", "fused_qkv_attention",
 ]
@@ -204,6 +205,64 @@ class _FlashAttnHip(torch.autograd.Function):
                             D, 0, 0, 0, 0, float(scale), int(causal),
                             _stream()), "attn_bwd")
         return dq, dk, dv, None, None
+
+
+def flash_attention_rect(q: torch.Tensor, k: torch.Tensor,
+                         v: torch.Tensor, q_offset: int,
+                         scale: Optional[float] = None) -> torch.Tensor:
+    """Rectangular-causal attention for chunked prefill (inference only,
+    no autograd): q [B, C, Hq, D] holds C query rows at global offset
+    q_offset; k/v [B, Skv, Hkv, D] hold the KV prefix INCLUDING the chunk
+    (Skv >= q_offset + C). Row i attends kv <= q_offset + i.
+
+    Native path: kf_attn_fwd4_rect (attention_fwd4.hip) — q is padded to a
+    256-row multiple and the kv buffers must have row capacity up to the
+    next 64 multiple of Skv (the serving KV-cache slabs do; this wrapper
+    re-pads otherwise)."""
+    B, C, Hq, D = q.shape
+    Skv = k.shape[1]
+    Hkv = k.shape[2]
+    if scale is None:
+        scale = D ** -0.5
+    assert Skv >= q_offset + C, (Skv, q_offset, C)
+    if _use_native(q):
+        lib = _backend.require()
+        pad = (256 - C % 256) % 256
+        if pad:
+            q = torch.cat([q, q.new_zeros(B, pad, Hq, D)], dim=1)
+        o = torch.empty_like(q)
+        lse = torch.empty(B, Hq, q.shape[1], dtype=torch.float32,
+                          device=q.device)
+        q = q.contiguous()
+        # rows must be contiguous [.., Hkv, D] and the buffer must extend
+        # to the next 64-row multiple of Skv (KV-cache slabs satisfy both)
+        need = (Skv + 63) // 64 * 64
+        def _rows_ok(t):
+            return (t.stride(1) == Hkv * D and t.stride(2) == D
+                    and t.stride(3) == 1
+                    and (B == 1 or t.stride(0) >= need * Hkv * D)
+                    and t.stride(0) // (Hkv * D) >= need)
+        if not (_rows_ok(k) and _rows_ok(v)):
+            kp = torch.zeros(B, need, Hkv, D, dtype=k.dtype, device=k.device)
+            vp = torch.zeros_like(kp)
+            kp[:, :Skv] = k[:, :Skv]
+            vp[:, :Skv] = v[:, :Skv]
+            k, v = kp, vp
+        _backend.check(
+            lib.kf_attn_fwd4_rect(_p(o), _fp(lse), _p(q), _p(k), _p(v),
+                                  B, q.shape[1], Skv, Hq, Hkv, D,
+                                  0, Hkv * D,
+                                  ctypes.c_float(float(scale)), q_offset,
+                                  _stream()), "attn_fwd4_rect")
+        return o[:, :C]
+    # reference: sdpa's tril(Sk - Sq) IS the rectangular-causal mask when
+    # the chunk sits at the END of the kv prefix
+    kv_end = q_offset + C
+    o = reference.sdpa(q.float().transpose(1, 2),
+                       k[:, :kv_end].float().transpose(1, 2),
+                       v[:, :kv_end].float().transpose(1, 2),
+                       causal=True, scale=scale).transpose(1, 2)
+    return o.to(q.dtype)
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,

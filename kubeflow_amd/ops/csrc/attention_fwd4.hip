@@ -80,7 +80,10 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     unsigned short* __restrict__ o, float* __restrict__ lse,
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, int64_t B, int S, int Hq, int Hkv,
-    int64_t qts, int64_t kts, float scale, int causal) {
+    int64_t qts, int64_t kts, float scale, int causal, int skv, int qoff) {
+  // rectangular-causal extension (chunked prefill): q rows are globally
+  // offset by qoff and attend kv in [0, min(qoff+qrow, skv-1)]; the plain
+  // causal square is the skv=S, qoff=0 special case.
   __shared__ unsigned char k_lds[2][A4_KT * A4_D * 2];  // 16 KiB x2, swizzled
   __shared__ unsigned char v_lds[2][A4_KT * A4_D * 2];  // 16 KiB x2, subtiled
 
@@ -118,9 +121,15 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   const int qrow_g = qt * A4_QT + w * 32 + l31;
   const int wave_qmin = qt * A4_QT + w * 32;
   const int wave_qmax = wave_qmin + 31;
+  // lane's highest visible kv row (pad q rows clamp to skv-1)
+  const int kv_limit = causal ? min(qrow_g + qoff, skv - 1) : skv - 1;
 
-  const int last_kt =
-      causal ? (qt * A4_QT + A4_QT - 1) / A4_KT : (S / A4_KT - 1);
+  const int kv_tiles = (skv + A4_KT - 1) / A4_KT;
+  int last_kt = kv_tiles - 1;
+  if (causal) {
+    const int c = (qoff + qt * A4_QT + A4_QT - 1) / A4_KT;
+    if (c < last_kt) last_kt = c;
+  }
 
   // staging: 512 threads x 2 vectors cover one 64x128 tile for K and V.
   const unsigned short* kg0 = k + (b * S) * kts + (int64_t)hkv * A4_D;
@@ -183,7 +192,7 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
       for (int r = 0; r < 16; ++r) {
         const int kv = kv0 + (r & 3) + 8 * (r >> 2);
         float sv = st[r] * scale2;
-        if (kv > qrow_g) sv = -INFINITY;
+        if (kv > kv_limit) sv = -INFINITY;
         st[r] = sv;
         mx = fmaxf(mx, sv);
       }
@@ -306,11 +315,15 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     if (do1) st1 = qk_block(cur, 1);
     kf_bf16x8v4 pb[2];
     if (do0) {
-      sm_block(st0, kv_lo0, causal && kv_lo0 + 31 > wave_qmin, pb);
+      sm_block(st0, kv_lo0,
+               (causal && kv_lo0 + 31 > wave_qmin + qoff) ||
+                   kv_lo0 + 31 >= skv, pb);
       pv_block(pb, vbase);
     }
     if (do1) {
-      sm_block(st1, kv_lo1, causal && kv_lo1 + 31 > wave_qmin, pb);
+      sm_block(st1, kv_lo1,
+               (causal && kv_lo1 + 31 > wave_qmin + qoff) ||
+                   kv_lo1 + 31 >= skv, pb);
       pv_block(pb, vbase + 1024);
     }
 
@@ -363,7 +376,28 @@ static int kf_attn_fwd4_launch(void* o, float* lse, const void* q,
                      (hipStream_t)stream, (unsigned short*)o, lse,
                      (const unsigned short*)q, (const unsigned short*)k,
                      (const unsigned short*)v, B, (int)S, (int)Hq, (int)Hkv,
-                     qts, kts, scale, causal);
+                     qts, kts, scale, causal, (int)S, 0);
+  return (int)hipGetLastError();
+}
+
+// Rectangular-causal forward for chunked prefill: Sq query rows (padded to
+// a 256 multiple) at global offset qoff attend skv cached kv rows. The
+// kv buffers must extend to ceil(skv/64)*64 rows (the KV-cache slab does).
+KF_EXPORT int kf_attn_fwd4_rect(void* o, float* lse, const void* q,
+                                const void* k, const void* v, int64_t B,
+                                int64_t Sq, int64_t Skv, int64_t Hq,
+                                int64_t Hkv, int64_t D, int64_t qts,
+                                int64_t kts, float scale, int64_t qoff,
+                                void* stream) {
+  if (D != A4_D || Sq % A4_QT || Hq % Hkv) return (int)hipErrorInvalidValue;
+  if (qts == 0) qts = Hq * A4_D;
+  if (kts == 0) kts = Hkv * A4_D;
+  dim3 grid((unsigned)((Sq / A4_QT) * Hq * B), 1, 1);
+  hipLaunchKernelGGL(kf_attn_fwd4_kernel<0>, grid, dim3(A4_THREADS), 0,
+                     (hipStream_t)stream, (unsigned short*)o, lse,
+                     (const unsigned short*)q, (const unsigned short*)k,
+                     (const unsigned short*)v, B, (int)Sq, (int)Hq,
+                     (int)Hkv, qts, kts, scale, 1, (int)Skv, (int)qoff);
   return (int)hipGetLastError();
 }
 

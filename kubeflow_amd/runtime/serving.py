@@ -103,6 +103,7 @@ class InferenceEngine:
         self.max_batch = max_batch
         self.pending: "queue.Queue[Request]" = queue.Queue()
         self.active: List[Request] = []
+        self._chunking = None  # in-progress chunked prefill state
         self._stop = False
         self._thread: Optional[threading.Thread] = None
         self.stats = {"requests": 0, "completed": 0, "tokens_out": 0,
@@ -153,17 +154,24 @@ class InferenceEngine:
         while not self._stop:
             self._admit()
             if not self.active:
-                time.sleep(0.002)  # queued-but-unadmittable or truly idle
+                if self._chunking is None:
+                    time.sleep(0.002)  # queued-but-unadmittable / idle
                 continue
             self._decode_step()
 
+    # prompts longer than this prefill in CHUNK-token slices interleaved
+    # with decode steps (rectangular-causal flash kernel
+    # kf_attn_fwd4_rect), bounding active streams' inter-token stalls to
+    # one chunk (~1/4 of a 4k prefill) instead of the whole prompt
+    PREFILL_CHUNK = 1024
+
     def _admit(self) -> int:
-        # Fairness: each admit runs a full inline prefill (~10s of ms for a
-        # 4k prompt), so while sequences are decoding admit at most ONE new
-        # request per iteration — a burst of long prompts then stalls the
-        # running decodes by one prefill, not the whole burst. With nothing
-        # active, drain the queue freely. (True chunked prefill needs a
-        # rectangular-causal flash kernel — round-2, docs/ROUND1.md item 3.)
+        # Fairness: one prefill unit (a full short prompt, or ONE chunk of
+        # a long one) per loop iteration while sequences decode; with
+        # nothing active, drain freely.
+        if self._chunking is not None:
+            self._advance_chunk()
+            return 0
         limit = 1 if self.active else self.max_batch
         n = 0
         while len(self.active) < self.max_batch and n < limit:
@@ -177,6 +185,8 @@ class InferenceEngine:
                 break
             self._start_request(req)
             n += 1
+            if self._chunking is not None:
+                break  # the long prompt continues next iteration
         return n
 
     def _start_request(self, req: Request):
@@ -186,6 +196,13 @@ class InferenceEngine:
             self.pending.put(req)
             return
         req.slot = slot
+        keep = max(1, self.cache.smax - req.max_new_tokens - 1)
+        prompt = req.prompt[-keep:]
+        if self.active and len(prompt) > self.PREFILL_CHUNK:
+            # chunked prefill: first chunk now, rest interleaved with decode
+            self._chunking = {"req": req, "prompt": prompt, "pos": 0}
+            self._advance_chunk()
+            return
         try:
             self._prefill(req)
             self.active.append(req)
@@ -196,6 +213,70 @@ class InferenceEngine:
             self.cache.free(slot)
             req.finished_at = time.time()
             req.done.set()
+
+    @torch.no_grad()
+    def _advance_chunk(self):
+        st = self._chunking
+        req, prompt, pos = st["req"], st["prompt"], st["pos"]
+        end = min(pos + self.PREFILL_CHUNK, len(prompt))
+        try:
+            x = self._prefill_chunk(req, prompt[pos:end], pos)
+            st["pos"] = end
+            if end == len(prompt):  # final chunk: sample the first token
+                x = self.model.final_norm(x[:, -1:])
+                logits = torch.nn.functional.linear(
+                    x, self.model.lm_head.weight)
+                tok = self._sample(logits[0, -1], req.temperature)
+                req.pos = len(prompt)
+                req.generated.append(tok)
+                req.first_token_at = time.time()
+                req.token_times.append(req.first_token_at)
+                self.stats["prefill_tokens"] += len(prompt)
+                self.active.append(req)
+                self._chunking = None
+        except Exception as e:  # pragma: no cover
+            import traceback
+            traceback.print_exc()
+            req.error = f"{type(e).__name__}: {e}"
+            self.cache.free(req.slot)
+            req.finished_at = time.time()
+            req.done.set()
+            self._chunking = None
+
+    def _prefill_chunk(self, req: Request, chunk, pos: int):
+        """One CHUNK-token slice through all layers: cache rows
+        [pos, pos+C) fill and the chunk attends the whole prefix via the
+        rectangular-causal kernel."""
+        cfg = self.model.cfg
+        C = len(chunk)
+        import torch.nn.functional as F
+        tokens = torch.tensor([chunk], dtype=torch.int64, device=self.device)
+        x = self.model.embed(tokens)
+        cos, sin = self.model.rope_cos, self.model.rope_sin
+        for li, layer in enumerate(self.model.layers):
+            qkv = F.linear(layer.attn_norm(x), layer.wqkv.weight)
+            q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
+                                 cfg.n_kv_heads * cfg.head_dim,
+                                 cfg.n_kv_heads * cfg.head_dim], dim=-1)
+            q = q.view(1, C, cfg.n_heads, cfg.head_dim)
+            k = k.view(1, C, cfg.n_kv_heads, cfg.head_dim)
+            v = v.view(1, C, cfg.n_kv_heads, cfg.head_dim)
+            q, k = ops.rope(q, k, cos, sin, pos)
+            self.cache.k[li][req.slot, pos:pos + C] = k[0]
+            self.cache.v[li][req.slot, pos:pos + C] = v[0]
+            kv_k = self.cache.k[li][req.slot:req.slot + 1, :pos + C]
+            kv_v = self.cache.v[li][req.slot:req.slot + 1, :pos + C]
+            o = ops.flash_attention_rect(q, kv_k, kv_v, q_offset=pos,
+                                         scale=cfg.head_dim ** -0.5)
+            o = layer.wo(o.reshape(1, C, cfg.n_heads * cfg.head_dim))
+            x = x + o
+            if layer.moe is not None:
+                x = x + layer.moe(layer.mlp_norm(x))
+            else:
+                g, u = F.linear(layer.mlp_norm(x), layer.w13.weight).split(
+                    [cfg.ffn_dim, cfg.ffn_dim], dim=-1)
+                x = x + layer.w2(F.silu(g) * u)
+        return x
 
     @torch.no_grad()
     def _prefill(self, req: Request):

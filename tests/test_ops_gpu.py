@@ -374,3 +374,41 @@ def test_serving_graph_matches_eager(gpu_device, monkeypatch):
     finally:
         eng_g.stop()
         eng_e.stop()
+
+
+@pytest.mark.gpu
+def test_attention_rect_gpu_parity():
+    """kf_attn_fwd4_rect (chunked prefill) vs the fp32 reference at a
+    non-aligned Skv with a q offset."""
+    torch.manual_seed(11)
+    dev = torch.device("cuda", 0)
+    B, Hq, Hkv, D = 1, 8, 2, 128
+    off, C = 640, 512           # chunk of 512 at offset 640
+    Skv = off + C               # 1152 (non-64-multiple + 64*18 = fine)
+    # kv buffer padded to the next 64-multiple capacity (cache-slab shape)
+    cap = (Skv + 63) // 64 * 64
+    q = torch.randn(B, C, Hq, D, device=dev, dtype=torch.bfloat16)
+    kbuf = torch.randn(B, cap, Hkv, D, device=dev, dtype=torch.bfloat16)
+    vbuf = torch.randn(B, cap, Hkv, D, device=dev, dtype=torch.bfloat16)
+    from kubeflow_amd import ops
+    o = ops.flash_attention_rect(q, kbuf[:, :Skv], vbuf[:, :Skv],
+                                 q_offset=off)
+    # reference on CPU in fp32 (rect mask via sdpa tril(Sk-Sq))
+    from kubeflow_amd.ops import reference as R
+    want = R.sdpa(q.float().cpu().transpose(1, 2),
+                  kbuf[:, :Skv].float().cpu().transpose(1, 2),
+                  vbuf[:, :Skv].float().cpu().transpose(1, 2),
+                  causal=True).transpose(1, 2)
+    err = ((o.cpu().float() - want).norm() / want.norm()).item()
+    assert err < 2e-2, err
+
+    # odd Skv (not a 64-multiple): padding rows must stay masked
+    Skv2 = off + C - 37
+    o2 = ops.flash_attention_rect(q[:, :256], kbuf[:, :Skv2],
+                                  vbuf[:, :Skv2], q_offset=Skv2 - 256)
+    want2 = R.sdpa(q[:, :256].float().cpu().transpose(1, 2),
+                   kbuf[:, :Skv2].float().cpu().transpose(1, 2),
+                   vbuf[:, :Skv2].float().cpu().transpose(1, 2),
+                   causal=True).transpose(1, 2)
+    err2 = ((o2.cpu().float() - want2).norm() / want2.norm()).item()
+    assert err2 < 2e-2, err2
