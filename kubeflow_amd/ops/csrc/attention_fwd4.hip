@@ -307,8 +307,9 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     // ---- compute: both QK^T clusters issue first so sub-block 0's
     // softmax VALU overlaps sub-block 1's MFMA tail ----
     const int kv_lo0 = kt * A4_KT, kv_lo1 = kt * A4_KT + 32;
-    const bool do0 = !causal || kv_lo0 <= wave_qmax;
-    const bool do1 = !causal || kv_lo1 <= wave_qmax;
+    const bool do0 = !causal || kv_lo0 <= wave_qmax + qoff;
+    const bool do1 = (!causal || kv_lo1 <= wave_qmax + qoff) &&
+                     kv_lo1 < skv;
     const unsigned vbase = (unsigned)(size_t)(v_lds[cur]) + v_lane_off;
     kf_f32x16v4 st0, st1;
     if (do0) st0 = qk_block(cur, 0);
