@@ -162,8 +162,11 @@ class InferenceEngine:
     # prompts longer than this prefill in CHUNK-token slices interleaved
     # with decode steps (rectangular-causal flash kernel
     # kf_attn_fwd4_rect), bounding active streams' inter-token stalls to
-    # one chunk (~1/4 of a 4k prefill) instead of the whole prompt
-    PREFILL_CHUNK = 1024
+    # one chunk instead of the whole prompt. Measured A/B at mixed
+    # 128/1k/4k load (profiles/r02_serve_mixed.md): inter-token p99
+    # 196 -> 68 ms for ~-25% aggregate tok/s — an SLO/throughput knob.
+    # KF_PREFILL_CHUNK overrides; 0 disables chunking.
+    PREFILL_CHUNK = int(os.environ.get("KF_PREFILL_CHUNK", "1024")) or (1 << 30)
 
     def _admit(self) -> int:
         # Fairness: one prefill unit (a full short prompt, or ONE chunk of
