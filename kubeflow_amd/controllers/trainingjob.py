@@ -43,13 +43,36 @@ class TrainingJobReconciler(Reconciler):
         self.poll_period = poll_period
         self.warm_pool = warm_pool
         self.gangs: Dict[str, ProcessGang] = {}
+        self.eval_gangs: Dict[str, ProcessGang] = {}
         self.restarts: Dict[str, int] = {}
         self.key_uid: Dict[tuple, str] = {}
 
     # ------------------------------------------------------------- helpers
+    # replica-type semantics (training-operator parity): Master/Chief is
+    # rank 0 and its template wins; Workers follow; Evaluator runs as a
+    # checkpoint-watching sidecar OUTSIDE the gang; PS has no analog in a
+    # torch runtime and is rejected loudly.
+    ROLE_ORDER = ("Master", "Chief", "Worker")
+
     def _replicas(self, job) -> int:
         specs = job["spec"].get(self.replica_field, {})
-        return sum(int(r.get("replicas", 1)) for r in specs.values())
+        return sum(int(r.get("replicas", 1)) for role, r in specs.items()
+                   if role != "Evaluator")
+
+    def _role_of_rank(self, job, rank: int) -> str:
+        specs = job["spec"].get(self.replica_field, {})
+        off = 0
+        for role in self.ROLE_ORDER:
+            if role in specs:
+                n = int(specs[role].get("replicas", 1))
+                if rank < off + n:
+                    return role
+                off += n
+        return "Worker"
+
+    def _evaluator_spec(self, job):
+        specs = job["spec"].get(self.replica_field, {})
+        return specs.get("Evaluator")
 
     def _template(self, job) -> dict:
         specs = job["spec"].get(self.replica_field, {})
@@ -90,6 +113,16 @@ class TrainingJobReconciler(Reconciler):
 
     def _start(self, job):
         uid = job["metadata"]["uid"]
+        specs = job["spec"].get(self.replica_field, {})
+        if "PS" in specs:
+            set_condition(job, "Failed", "True", "InvalidSpec",
+                          "PS replicas: parameter-server strategy has no "
+                          "analog in the torch runtime (use DDP Workers; "
+                          "tf.distribute PS-strategy is not supported)")
+            self.store.update(job, check_version=False)
+            self.store.record_event(job, "InvalidSpec",
+                                    "PS replicas unsupported", "Warning")
+            return
         n = self._replicas(job)
         template = self._template(job)
         gpus_per = int(template.get("gpus_per_replica", 1))
@@ -161,6 +194,23 @@ class TrainingJobReconciler(Reconciler):
             self.store.record_event(job, "InvalidSpec", str(e), "Warning")
             return
         self.gangs[uid] = gang
+        ev = self._evaluator_spec(job)
+        if ev is not None:
+            ev_spec = dict(template)
+            ev_spec.update(ev.get("template", {}))
+            ev_spec["role"] = "Evaluator"
+            ev_spec["world_size"] = 1
+            ev_spec.setdefault("checkpoint_dir",
+                               template.get("checkpoint_dir")
+                               or os.path.join(workdir, "checkpoints"))
+            try:
+                self.eval_gangs[uid] = launch_gang(
+                    uid + "-eval", os.path.join(workdir, "evaluator"),
+                    ev_spec, [], poddefaults=poddefaults,
+                    labels=job["metadata"].get("labels", {}))
+            except ValueError as e:
+                self.store.record_event(job, "EvaluatorFailed", str(e),
+                                        "Warning")
         set_condition(job, "Created", "True", "JobCreated", "gang launched")
         set_condition(job, "Running", "True", "JobRunning",
                       f"{n} replicas on GPUs {gpu_indices or 'cpu'}")
@@ -190,13 +240,27 @@ class TrainingJobReconciler(Reconciler):
         uid = job["metadata"]["uid"]
         state = gang.poll()
         ranks = self._read_rank_status(gang)
-        active = sum(1 for s in ranks.values() if s.get("state") == "running")
-        succeeded = sum(1 for s in ranks.values()
-                        if s.get("state") == "succeeded")
-        failed = sum(1 for s in ranks.values() if s.get("state") == "failed")
-        job["status"]["replicaStatuses"] = {
-            "Worker": {"active": active, "succeeded": succeeded,
-                       "failed": failed}}
+        by_role: Dict[str, Dict[str, int]] = {}
+        for rk, st in ranks.items():
+            role = self._role_of_rank(job, rk)
+            slot = by_role.setdefault(role, {"active": 0, "succeeded": 0,
+                                             "failed": 0})
+            key = {"running": "active", "succeeded": "succeeded",
+                   "failed": "failed"}.get(st.get("state"), "active")
+            slot[key] += 1
+        ev_gang = self.eval_gangs.get(uid)
+        if ev_gang is not None:
+            ev_state = ev_gang.poll()
+            ev_rank = self._read_rank_status(ev_gang).get(0, {})
+            job["status"]["replicaStatuses"] = job["status"].get(
+                "replicaStatuses", {})
+            by_role["Evaluator"] = {
+                "active": 1 if ev_state is None else 0,
+                "succeeded": 1 if ev_state == "Succeeded" else 0,
+                "failed": 1 if ev_state == "Failed" else 0}
+            if ev_rank.get("metrics", {}).get("eval_loss") is not None:
+                job["status"]["evalMetrics"] = ev_rank["metrics"]
+        job["status"]["replicaStatuses"] = by_role
         r0 = ranks.get(0, {})
         if r0.get("metrics"):
             job["status"]["trainingMetrics"] = r0["metrics"]
@@ -236,6 +300,9 @@ class TrainingJobReconciler(Reconciler):
         gang = self.gangs.pop(uid, None)
         if gang is not None:
             gang.terminate_and_wait()
+        ev = self.eval_gangs.pop(uid, None)
+        if ev is not None:
+            ev.terminate_and_wait()
         self.scheduler.release(uid)
 
     def on_deleted(self, namespace, name):

@@ -40,6 +40,69 @@ def write_status(rank_dir: str, state: str, step: int = 0, loss=None,
             f.write(json.dumps({"step": step, "metrics": metrics}) + "\n")
 
 
+
+def run_evaluator(spec: dict, workdir: str, rank_dir: str, stop: dict) -> int:
+    """TFJob Evaluator role: a sidecar process (NOT part of the training
+    gang) that watches the checkpoint directory, loads each new step's
+    weights, and reports eval loss over held-out synthetic batches — the
+    TF estimator evaluator loop mapped onto the checkpoint seam. Exits 0
+    when the trainers' final step appears (or on SIGTERM at job end)."""
+    import torch  # local: evaluator runs standalone
+    from kubeflow_amd.models import build_model
+    from kubeflow_amd.runtime import checkpoint as _ckpt
+
+    device = (torch.device("cuda", 0) if torch.cuda.is_available()
+              else torch.device("cpu"))
+    dtype = (torch.bfloat16 if device.type == "cuda"
+             and spec.get("dtype", "bf16") == "bf16" else torch.float32)
+    model = build_model(spec.get("model", "mnist-mlp"), device=device,
+                        dtype=dtype)
+    cfg = getattr(model, "cfg", None)
+    ckpt_dir = spec.get("checkpoint_dir") or os.path.join(workdir,
+                                                          "checkpoints")
+    final_step = int(spec.get("steps", 100))
+    eval_batches = int(spec.get("eval_batches", 4))
+    seen = -1
+    write_status(rank_dir, "running", 0)
+    deadline = time.time() + float(spec.get("eval_timeout", 600))
+    while not stop["flag"] and time.time() < deadline:
+        d = _ckpt.latest_dir(ckpt_dir)
+        step = seen
+        if d is not None:
+            try:
+                step = int(os.path.basename(d).split("-")[1])
+            except (IndexError, ValueError):
+                step = seen
+        if step > seen:
+            _ckpt.load_model_weights(model, d)
+            model.eval()
+            losses = []
+            with torch.no_grad():
+                for b in range(eval_batches):
+                    x, y = synthetic_batch(spec, cfg, device,
+                                           rank=1000 + b, step=step)
+                    out = model(x)
+                    if out.dim() == 3:  # lm
+                        loss = torch.nn.functional.cross_entropy(
+                            out.float().reshape(-1, out.shape[-1]),
+                            y.reshape(-1))
+                    else:
+                        loss = torch.nn.functional.cross_entropy(
+                            out.float(), y)
+                    losses.append(float(loss))
+            seen = step
+            write_status(rank_dir, "running", step,
+                         metrics={"eval_loss": sum(losses) / len(losses),
+                                  "eval_step": step})
+            if step >= final_step:
+                break
+        else:
+            time.sleep(0.2)
+    write_status(rank_dir, "succeeded", max(seen, 0),
+                 metrics={"eval_step": seen} if seen >= 0 else None)
+    return 0
+
+
 def synthetic_batch(spec: dict, cfg, device, rank: int, step: int):
     """Deterministic-per-(rank,step) synthetic data of the model's shape."""
     seed = (int(spec.get("seed", 0)) * 1000003 + rank * 9176 + step) % (2**31)
@@ -94,6 +157,9 @@ def main(argv=None):
 
     stop = {"flag": False}
     signal.signal(signal.SIGTERM, lambda *_: stop.update(flag=True))
+
+    if spec.get("role") == "Evaluator":
+        return run_evaluator(spec, workdir, rank_dir, stop)
 
     try:
         from kubeflow_amd.parallel.strategy import ParallelismSpec, Strategy

@@ -175,3 +175,59 @@ def test_poddefault_conflict_fails_job(tmp_path):
                if e["reason"] == "InvalidSpec"]
         assert len(evs) == 1
         assert not plat.pytorchjob.gangs
+
+
+def _wait_terminal(store, kind, name, ns, timeout=120):
+    deadline = time.time() + timeout
+    obj = store.get(kind, name, ns)
+    while time.time() < deadline:
+        obj = store.get(kind, name, ns)
+        if has_condition(obj, "Succeeded") or has_condition(obj, "Failed"):
+            return obj
+        time.sleep(0.3)
+    return obj
+
+
+def test_tfjob_chief_evaluator_roles(tmp_path):
+    """TFJob replica-type semantics: Chief is rank 0 (its template wins),
+    the Evaluator runs outside the gang watching checkpoints and reports
+    eval loss, and replicaStatuses is per role."""
+    with Platform(root_dir=str(tmp_path)) as plat:
+        job = new_object("TFJob", "tf-roles", "default", spec={
+            "tfReplicaSpecs": {
+                "Chief": {"replicas": 1, "restartPolicy": "Never",
+                          "template": {"model": "mnist-mlp", "steps": 6,
+                                       "gpus_per_replica": 0,
+                                       "save_every": 2, "status_every": 2}},
+                "Worker": {"replicas": 1},
+                "Evaluator": {"replicas": 1,
+                              "template": {"eval_batches": 2}},
+            }}, api_version="kubeflow.org/v1")
+        plat.store.create(job)
+        obj = _wait_terminal(plat.store, "TFJob", "tf-roles", "default",
+                             timeout=180)
+        assert has_condition(obj, "Succeeded"), obj["status"]
+        rs = obj["status"]["replicaStatuses"]
+        assert "Chief" in rs and "Worker" in rs, rs
+        assert rs["Chief"]["succeeded"] == 1
+        assert rs["Worker"]["succeeded"] == 1
+        # evaluator observed checkpoints and produced an eval loss
+        assert "Evaluator" in rs, rs
+        em = obj["status"].get("evalMetrics")
+        assert em and em.get("eval_loss") is not None, obj["status"]
+
+
+def test_tfjob_ps_rejected(tmp_path):
+    with Platform(root_dir=str(tmp_path)) as plat:
+        job = new_object("TFJob", "tf-ps", "default", spec={
+            "tfReplicaSpecs": {
+                "Worker": {"replicas": 1,
+                           "template": {"model": "mnist-mlp", "steps": 2}},
+                "PS": {"replicas": 2},
+            }}, api_version="kubeflow.org/v1")
+        plat.store.create(job)
+        obj = _wait_terminal(plat.store, "TFJob", "tf-ps", "default",
+                             timeout=60)
+        assert has_condition(obj, "Failed")
+        conds = {c["type"]: c for c in obj["status"]["conditions"]}
+        assert "parameter-server" in conds["Failed"]["message"]
