@@ -41,6 +41,36 @@ def write_status(rank_dir: str, state: str, step: int = 0, loss=None,
 
 
 
+def run_pytest(spec: dict, rank_dir: str) -> int:
+    """CI test task (the kind the ci/ workflow builders emit): runs a
+    pytest selection in a subprocess and reports pass/fail through the
+    normal rank status seam."""
+    import subprocess
+    import sys as _sys
+    args = list(spec.get("pytest_args") or [])
+    repo = os.path.dirname(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    write_status(rank_dir, "running", 0)
+    proc = subprocess.run(
+        [_sys.executable, "-m", "pytest", "-q", "-p", "no:cacheprovider",
+         "-m", "not gpu", *args],
+        cwd=repo, capture_output=True, text=True,
+        timeout=float(spec.get("timeout", 1200)))
+    tail = (proc.stdout or "")[-2000:]
+    with open(os.path.join(rank_dir, "pytest.log"), "w") as f:
+        f.write(proc.stdout or "")
+        f.write(proc.stderr or "")
+    if proc.returncode == 0:
+        write_status(rank_dir, "succeeded", 1,
+                     metrics={"pytest": tail.splitlines()[-1]
+                              if tail.splitlines() else ""})
+        return 0
+    write_status(rank_dir, "failed", 1,
+                 error=f"pytest rc={proc.returncode}: "
+                       f"{tail.splitlines()[-1] if tail.splitlines() else ''}")
+    return 1
+
+
 def run_evaluator(spec: dict, workdir: str, rank_dir: str, stop: dict) -> int:
     """TFJob Evaluator role: a sidecar process (NOT part of the training
     gang) that watches the checkpoint directory, loads each new step's
@@ -160,6 +190,8 @@ def main(argv=None):
 
     if spec.get("role") == "Evaluator":
         return run_evaluator(spec, workdir, rank_dir, stop)
+    if spec.get("task") == "pytest":
+        return run_pytest(spec, rank_dir)
 
     try:
         from kubeflow_amd.parallel.strategy import ParallelismSpec, Strategy
