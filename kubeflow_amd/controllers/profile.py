@@ -85,8 +85,37 @@ class ProfileReconciler(Reconciler):
                 q["metadata"]["ownerReferences"] = [owner_ref(prof)]
                 self.store.create(q)
 
+        # plugins (the reference's Apply loop, profile_controller.go:262-275
+        # — GCP/AWS IAM there; the local WorkloadIdentity analog + registry
+        # seam here, controllers/profile_plugins.py)
+        from kubeflow_amd.controllers.profile_plugins import (PluginError,
+                                                              apply_plugins)
+        try:
+            results = apply_plugins(self.store, prof, self.profiles_dir)
+            if results and prof["status"].get("plugins") != results:
+                prof["status"]["plugins"] = results
+                self.store.update(prof, check_version=False)
+        except PluginError as e:
+            set_condition(prof, "Ready", "False", "PluginFailed", str(e))
+            self.store.update(prof, check_version=False)
+            self.store.record_event(prof, "PluginFailed", str(e), "Warning")
+            return
+
         if not has_condition(prof, "Ready"):
             set_condition(prof, "Ready", "True", "ProfileReady",
                           f"namespace {name} provisioned")
             self.store.update(prof, check_version=False)
             self.store.record_event(prof, "ProfileReady", name)
+
+    def on_deleted(self, namespace, name):
+        # revoke plugin state (the finalizer teardown analog,
+        # profile_controller.go:277-312)
+        from kubeflow_amd.controllers.profile_plugins import revoke_plugins
+        try:
+            revoke_plugins(self.store,
+                           {"metadata": {"name": name}, "spec": {
+                               "plugins": [{"kind": k}
+                                           for k in ("WorkloadIdentity",)]}},
+                           self.profiles_dir)
+        except Exception:
+            pass

@@ -182,3 +182,61 @@ def test_store_log_compaction(tmp_path):
     # compact-on-load leaves exactly the live set
     with open(path) as f:
         assert sum(1 for _ in f) == 8
+
+
+def test_profile_plugin_seam(tmp_path):
+    """Plugin interface parity (profile_controller.go:78-84 + the cloud
+    IAM plugins' SA-annotation seam): the local WorkloadIdentity plugin
+    annotates default-editor and materializes a credential; unknown kinds
+    fail the profile loudly; deletion revokes."""
+    import json as _json
+    import os as _os
+    import time as _time
+
+    from kubeflow_amd.platform import Platform
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        prof = new_object("Profile", "team-wi", None,
+                          spec={"owner": {"kind": "User",
+                                          "name": "a@b.c"},
+                                "plugins": [{"kind": "WorkloadIdentity",
+                                             "spec": {"identity":
+                                                      "svc-team@local"}}]},
+                          api_version="kubeflow.org/v1")
+        plat.store.create(prof)
+        deadline = _time.time() + 30
+        while _time.time() < deadline:
+            obj = plat.store.get("Profile", "team-wi", None)
+            if has_condition(obj, "Ready"):
+                break
+            _time.sleep(0.2)
+        assert has_condition(obj, "Ready"), obj["status"]
+        sa = plat.store.get("ServiceAccount", "default-editor", "team-wi")
+        assert sa["metadata"]["annotations"][
+            "iam.kubeflow.org/local-identity"] == "svc-team@local"
+        cred = obj["status"]["plugins"]["WorkloadIdentity"]["credentialPath"]
+        assert _json.load(open(cred))["identity"] == "svc-team@local"
+
+        # unknown plugin kind -> Ready False with PluginFailed
+        bad = new_object("Profile", "team-bad", None,
+                         spec={"owner": {"kind": "User", "name": "a@b.c"},
+                               "plugins": [{"kind": "GcpOnlyThing"}]},
+                         api_version="kubeflow.org/v1")
+        plat.store.create(bad)
+        deadline = _time.time() + 30
+        while _time.time() < deadline:
+            objb = plat.store.get("Profile", "team-bad", None)
+            conds = {c["type"]: c for c in objb["status"]["conditions"]}
+            if "Ready" in conds and conds["Ready"]["status"] == "False":
+                break
+            _time.sleep(0.2)
+        conds = {c["type"]: c for c in objb["status"]["conditions"]}
+        assert conds["Ready"]["status"] == "False"
+        assert "GcpOnlyThing" in conds["Ready"]["message"]
+
+        # deletion revokes the credential
+        plat.store.delete("Profile", "team-wi", None)
+        deadline = _time.time() + 20
+        while _time.time() < deadline and _os.path.exists(cred):
+            _time.sleep(0.2)
+        assert not _os.path.exists(cred)
