@@ -235,8 +235,9 @@ def flash_attention_rect(q: torch.Tensor, k: torch.Tensor,
                           device=q.device)
         q = q.contiguous()
         # rows must be contiguous [.., Hkv, D] and the buffer must extend
-        # to the next 64-row multiple of Skv (KV-cache slabs satisfy both)
-        need = (Skv + 63) // 64 * 64
+        # to the next KV-TILE multiple of Skv (128 = A4_KT in
+        # attention_fwd4.hip; KV-cache slabs satisfy both)
+        need = (Skv + 127) // 128 * 128
         def _rows_ok(t):
             return (t.stride(1) == Hkv * D and t.stride(2) == D
                     and t.stride(3) == 1

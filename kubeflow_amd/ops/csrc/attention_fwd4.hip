@@ -37,7 +37,12 @@ typedef short kf_short4v4 __attribute__((ext_vector_type(4)));
 
 #define A4_D 128
 #define A4_QT 256      // q rows per block (8 waves x 32)
-#define A4_KT 64       // kv rows per LDS tile (double-buffered)
+// kv rows per LDS tile (double-buffered). 64 -> 64 KiB LDS; 128 -> 128 KiB
+// (still 1 block/CU at this register budget) with HALF the barriers.
+#ifndef A4_KT
+#define A4_KT 128
+#endif
+#define A4_SLAB (A4_KT / 4 * 64)   // elems per d16-slab in the V layout
 #define A4_THREADS 512
 #define A4_LOG2E 1.44269504f
 #define A4_LN2 0.69314718f
@@ -53,7 +58,7 @@ __device__ __forceinline__ int kf_swz4(int row, int byte_in_row) {
 // is 128 contiguous bytes; a 16-lane group's tr_read covers one subtile and
 // delivers column (lane&15): lane l receives V^T[d=base_d+(l&15)][kv=base+j].
 __device__ __forceinline__ int kf_vsub4(int kv, int d) {
-  return ((d >> 4) << 10) + ((kv >> 2) << 6) + ((kv & 3) << 4) + (d & 15);
+  return (d >> 4) * A4_SLAB + ((kv >> 2) << 6) + ((kv & 3) << 4) + (d & 15);
 }
 
 __device__ __forceinline__ unsigned int kf_cvt_pk_bf16_v4(float lo, float hi) {
@@ -134,14 +139,18 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   // staging: 512 threads x 2 vectors cover one 64x128 tile for K and V.
   const unsigned short* kg0 = k + (b * S) * kts + (int64_t)hkv * A4_D;
   const unsigned short* vg0 = v + (b * S) * kts + (int64_t)hkv * A4_D;
-  const int r0 = tid >> 4, c8 = tid & 15;          // j=0 row
-  const int r1 = (tid + A4_THREADS) >> 4;          // j=1 row
+  const int c8 = tid & 15;
+  int srow[A4_KT / 32];                            // staged rows per thread
+#pragma unroll
+  for (int j = 0; j < A4_KT / 32; ++j)
+    srow[j] = (tid + A4_THREADS * j) >> 4;
 
   // per-lane tr_read base byte address inside a V buffer: group g covers
   // slab (g&1), kv-subtile 2*(g>>1), column lane&15 (see kf_vsub4).
   const int g = lane >> 4;
   const unsigned v_lane_off =
-      (unsigned)(((g & 1) << 11) + ((g >> 1) << 8) + ((lane & 15) << 3));
+      (unsigned)(((g & 1) * A4_SLAB * 2) + ((g >> 1) << 8) +
+                 ((lane & 15) << 3));
 
   // ---- S^T = mfma(K, Q) over one 32-kv sub-block. All 8 K-fragment
   // ds_reads issue back-to-back BEFORE the MFMA chain so the ~120-cycle
@@ -255,7 +264,7 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
       if (dt < 3) {
-        const unsigned va = vbase + ((dt + 1) << 12);
+        const unsigned va = vbase + (unsigned)((dt + 1) * A4_SLAB * 4);
         KF_TR16(t[(dt + 1) & 1][0], va, "0");
         KF_TR16(t[(dt + 1) & 1][1], va, "128");
         KF_TR16(t[(dt + 1) & 1][2], va, "512");
@@ -279,15 +288,19 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
   };
 
   // ---- prologue: stage tile 0 into buffer 0 ----
-  kf_short8 kst0, kst1, vst0, vst1;
-  kst0 = *reinterpret_cast<const kf_short8*>(kg0 + r0 * kts + c8 * 8);
-  kst1 = *reinterpret_cast<const kf_short8*>(kg0 + r1 * kts + c8 * 8);
-  vst0 = *reinterpret_cast<const kf_short8*>(vg0 + r0 * kts + c8 * 8);
-  vst1 = *reinterpret_cast<const kf_short8*>(vg0 + r1 * kts + c8 * 8);
-  *reinterpret_cast<kf_short8*>(k_lds[0] + kf_swz4(r0, c8 * 16)) = kst0;
-  *reinterpret_cast<kf_short8*>(k_lds[0] + kf_swz4(r1, c8 * 16)) = kst1;
-  *reinterpret_cast<kf_short8*>(v_lds[0] + 2 * kf_vsub4(r0, c8 * 8)) = vst0;
-  *reinterpret_cast<kf_short8*>(v_lds[0] + 2 * kf_vsub4(r1, c8 * 8)) = vst1;
+  kf_short8 kst[A4_KT / 32], vst[A4_KT / 32];
+#pragma unroll
+  for (int j = 0; j < A4_KT / 32; ++j) {
+    kst[j] = *reinterpret_cast<const kf_short8*>(kg0 + srow[j] * kts + c8 * 8);
+    vst[j] = *reinterpret_cast<const kf_short8*>(vg0 + srow[j] * kts + c8 * 8);
+  }
+#pragma unroll
+  for (int j = 0; j < A4_KT / 32; ++j) {
+    *reinterpret_cast<kf_short8*>(k_lds[0] + kf_swz4(srow[j], c8 * 16)) =
+        kst[j];
+    *reinterpret_cast<kf_short8*>(v_lds[0] + 2 * kf_vsub4(srow[j], c8 * 8)) =
+        vst[j];
+  }
   __syncthreads();
 
   for (int kt = 0; kt <= last_kt; ++kt) {
@@ -298,34 +311,43 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     if (have_next) {
       const unsigned short* kg = kg0 + (int64_t)(kt + 1) * A4_KT * kts;
       const unsigned short* vg = vg0 + (int64_t)(kt + 1) * A4_KT * kts;
-      kst0 = *reinterpret_cast<const kf_short8*>(kg + r0 * kts + c8 * 8);
-      kst1 = *reinterpret_cast<const kf_short8*>(kg + r1 * kts + c8 * 8);
-      vst0 = *reinterpret_cast<const kf_short8*>(vg + r0 * kts + c8 * 8);
-      vst1 = *reinterpret_cast<const kf_short8*>(vg + r1 * kts + c8 * 8);
+#pragma unroll
+      for (int j = 0; j < A4_KT / 32; ++j) {
+        kst[j] =
+            *reinterpret_cast<const kf_short8*>(kg + srow[j] * kts + c8 * 8);
+        vst[j] =
+            *reinterpret_cast<const kf_short8*>(vg + srow[j] * kts + c8 * 8);
+      }
     }
 
-    // ---- compute: both QK^T clusters issue first so sub-block 0's
-    // softmax VALU overlaps sub-block 1's MFMA tail ----
-    const int kv_lo0 = kt * A4_KT, kv_lo1 = kt * A4_KT + 32;
-    const bool do0 = !causal || kv_lo0 <= wave_qmax + qoff;
-    const bool do1 = (!causal || kv_lo1 <= wave_qmax + qoff) &&
-                     kv_lo1 < skv;
+    // ---- compute: per 64-kv pair, both QK^T clusters issue first so
+    // sub-block A's softmax VALU overlaps sub-block B's MFMA tail ----
     const unsigned vbase = (unsigned)(size_t)(v_lds[cur]) + v_lane_off;
-    kf_f32x16v4 st0, st1;
-    if (do0) st0 = qk_block(cur, 0);
-    if (do1) st1 = qk_block(cur, 1);
-    kf_bf16x8v4 pb[2];
-    if (do0) {
-      sm_block(st0, kv_lo0,
-               (causal && kv_lo0 + 31 > wave_qmin + qoff) ||
-                   kv_lo0 + 31 >= skv, pb);
-      pv_block(pb, vbase);
-    }
-    if (do1) {
-      sm_block(st1, kv_lo1,
-               (causal && kv_lo1 + 31 > wave_qmin + qoff) ||
-                   kv_lo1 + 31 >= skv, pb);
-      pv_block(pb, vbase + 1024);
+#pragma unroll
+    for (int pair = 0; pair < A4_KT / 64; ++pair) {
+      const int kv_lo0 = kt * A4_KT + pair * 64;
+      const int kv_lo1 = kv_lo0 + 32;
+      const bool do0 = (!causal || kv_lo0 <= wave_qmax + qoff) &&
+                       kv_lo0 < skv;
+      const bool do1 = (!causal || kv_lo1 <= wave_qmax + qoff) &&
+                       kv_lo1 < skv;
+      if (!do0 && !do1) continue;
+      kf_f32x16v4 st0, st1;
+      if (do0) st0 = qk_block(cur, pair * 2);
+      if (do1) st1 = qk_block(cur, pair * 2 + 1);
+      kf_bf16x8v4 pb[2];
+      if (do0) {
+        sm_block(st0, kv_lo0,
+                 (causal && kv_lo0 + 31 > wave_qmin + qoff) ||
+                     kv_lo0 + 31 >= skv, pb);
+        pv_block(pb, vbase + (unsigned)(pair << 11));
+      }
+      if (do1) {
+        sm_block(st1, kv_lo1,
+                 (causal && kv_lo1 + 31 > wave_qmin + qoff) ||
+                     kv_lo1 + 31 >= skv, pb);
+        pv_block(pb, vbase + (unsigned)(pair << 11) + 1024);
+      }
     }
 
     // ---- write the staged tile into the other buffer; the compiler
@@ -333,12 +355,13 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
     // compute phase has covered the HBM latency ----
     if (have_next) {
       const int nxt = cur ^ 1;
-      *reinterpret_cast<kf_short8*>(k_lds[nxt] + kf_swz4(r0, c8 * 16)) = kst0;
-      *reinterpret_cast<kf_short8*>(k_lds[nxt] + kf_swz4(r1, c8 * 16)) = kst1;
-      *reinterpret_cast<kf_short8*>(v_lds[nxt] + 2 * kf_vsub4(r0, c8 * 8)) =
-          vst0;
-      *reinterpret_cast<kf_short8*>(v_lds[nxt] + 2 * kf_vsub4(r1, c8 * 8)) =
-          vst1;
+#pragma unroll
+      for (int j = 0; j < A4_KT / 32; ++j) {
+        *reinterpret_cast<kf_short8*>(k_lds[nxt] + kf_swz4(srow[j], c8 * 16)) =
+            kst[j];
+        *reinterpret_cast<kf_short8*>(v_lds[nxt] +
+                                      2 * kf_vsub4(srow[j], c8 * 8)) = vst[j];
+      }
     }
     __syncthreads();  // readers of [cur] done + writes to [nxt] visible
   }
