@@ -235,8 +235,8 @@ def flash_attention_rect(q: torch.Tensor, k: torch.Tensor,
                           device=q.device)
         q = q.contiguous()
         # rows must be contiguous [.., Hkv, D] and the buffer must extend
-        # to the next KV-TILE multiple of Skv (128 = A4_KT in
-        # attention_fwd4.hip; KV-cache slabs satisfy both)
+        # to the next KV-TILE multiple of Skv (A4_KT=64 in
+        # attention_fwd4.hip; 128 kept for headroom if the tile grows)
         need = (Skv + 127) // 128 * 128
         def _rows_ok(t):
             return (t.stride(1) == Hkv * D and t.stride(2) == D

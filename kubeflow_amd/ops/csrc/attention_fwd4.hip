@@ -39,8 +39,10 @@ typedef short kf_short4v4 __attribute__((ext_vector_type(4)));
 #define A4_QT 256      // q rows per block (8 waves x 32)
 // kv rows per LDS tile (double-buffered). 64 -> 64 KiB LDS; 128 -> 128 KiB
 // (still 1 block/CU at this register budget) with HALF the barriers.
+// (128 measured 656 TF vs 64's 674 on B4 Hq32 S4096 causal — the halved
+// barrier count does not pay for the deeper staging; keep 64)
 #ifndef A4_KT
-#define A4_KT 128
+#define A4_KT 64
 #endif
 #define A4_SLAB (A4_KT / 4 * 64)   // elems per d16-slab in the V layout
 #define A4_THREADS 512
