@@ -123,7 +123,21 @@ class InferenceEngine:
         self._pad_slot = self.cache.alloc() if self.use_graphs else None
 
     # ------------------------------------------------------------- public
-    def start(self):
+    def start(self, precapture: bool | None = None):
+        """Start the engine loop. By default (KF_SERVE_PRECAPTURE=1) every
+        power-of-two decode-batch bucket's hipGraph is captured up front:
+        lazy capture showed up as 100s-of-ms inter-token spikes the first
+        time each bucket appeared mid-traffic (profiles/r02_serve_mixed.md
+        itl_max)."""
+        if precapture is None:
+            precapture = os.environ.get("KF_SERVE_PRECAPTURE", "1") == "1"
+        if precapture and self.use_graphs:
+            b = 1
+            while True:
+                self._graph_for(b)
+                if b >= self.max_batch:
+                    break
+                b = min(b * 2, self.max_batch)
         self._thread = threading.Thread(target=self._loop, daemon=True,
                                         name="inference-engine")
         self._thread.start()
