@@ -68,3 +68,31 @@ def test_gpu_pytorchjob_and_inference(tmp_path):
             out = json.loads(resp.read())
         pred = out["predictions"][0]
         assert pred["error"] == "" and len(pred["tokens"]) == 6
+
+
+@pytest.mark.gpu
+def test_chunked_prefill_gpu_equality():
+    """On hardware, chunked prefill (rect kernel over the KV cache) must
+    generate the same greedy tokens as inline full prefill."""
+    import torch
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(5)
+    a = InferenceEngine("llama-tiny", max_slots=4, smax=1024, max_batch=4)
+    torch.manual_seed(5)
+    b = InferenceEngine("llama-tiny", max_slots=4, smax=1024, max_batch=4)
+    b.PREFILL_CHUNK = 256
+    a.start(precapture=False)
+    b.start(precapture=False)
+    try:
+        prompt = [(i * 13) % a.model.cfg.vocab_size for i in range(1, 700)]
+        # keep one stream active on b so the chunked path engages
+        a.generate([1, 2, 3], max_new_tokens=20, timeout=120)
+        b.generate([1, 2, 3], max_new_tokens=20, timeout=120)
+        ra = a.generate(prompt, max_new_tokens=16, timeout=120)
+        rb = b.generate(prompt, max_new_tokens=16, timeout=120)
+        assert not ra.error and not rb.error, (ra.error, rb.error)
+        assert ra.generated == rb.generated, (ra.generated, rb.generated)
+    finally:
+        a.stop()
+        b.stop()
