@@ -126,8 +126,16 @@ def build_app(store: ObjectStore, scheduler=None,
     def is_cluster_admin(user: str) -> bool:
         return user in cluster_admins or disable_auth
 
+    strict_authz = os.environ.get("KF_STRICT_AUTHZ", "0") == "1"
+
     def authz(user: str, namespace: Optional[str], verb: str):
-        """SubjectAccessReview analog (crud_backend/authz.py:25-132)."""
+        """SubjectAccessReview analog (crud_backend/authz.py:25-132).
+
+        A namespace with NO bindings at all is open by default (single-user
+        bootstrap — otherwise a fresh install locks everyone out); under
+        KF_STRICT_AUTHZ=1 the bootstrap hole closes: only profile owners
+        and cluster admins may touch binding-less namespaces, so deleting
+        the last binding no longer silently opens the namespace."""
         if disable_auth or namespace is None:
             return
         if is_cluster_admin(user):
@@ -136,7 +144,14 @@ def build_app(store: ObjectStore, scheduler=None,
         ns_bindings = [rb for rb in store.list("RoleBinding", namespace)
                        if "user" in rb["metadata"].get("annotations", {})]
         if not ns_bindings:
-            return  # unmanaged namespace: open (single-user bootstrap)
+            if not strict_authz:
+                return  # unmanaged namespace: open (single-user bootstrap)
+            for p in store.list("Profile"):
+                if p["metadata"]["name"] == namespace and                         _owner_name(p) == user:
+                    return
+            raise ApiError(
+                403, f"user {user} has no access to {namespace} "
+                "(strict authz: binding-less namespaces are owner-only)")
         if role is None:
             raise ApiError(403, f"user {user} has no access to {namespace}")
         if verb != "get" and role == "view":
