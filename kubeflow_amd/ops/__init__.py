@@ -170,30 +170,6 @@ def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
 
 # ------------------------------------------------------- Flash attention --
 
-def _attn_bwd_call(lib, dq_p, dk_p, dv_p, dout, q_p, k_p, v_p, o, lse,
-                   delta, B, S, Hq, Hkv, D, qts, kts, dqts, dkts, scale,
-                   causal, device):
-    """Dispatch the attention backward: the 8-wave head-split path (fp32
-    dK/dV per-head partials + reduce — kf_attn_bwd_ws) for training shapes,
-    the generic 4-wave path otherwise."""
-    if S % 256 == 0 and hasattr(lib, "kf_attn_bwd_ws"):
-        g = Hq // Hkv
-        ws = torch.empty(2, g, B, S, Hkv, D, dtype=torch.float32,
-                         device=device)
-        _backend.check(
-            lib.kf_attn_bwd_ws(dq_p, dk_p, dv_p, dout, q_p, k_p, v_p, o,
-                               _fp(lse), _fp(delta), _fp(ws[0]), _fp(ws[1]),
-                               B, S, Hq, Hkv, D, qts, kts, dqts, dkts,
-                               ctypes.c_float(float(scale)), int(causal),
-                               _stream()), "attn_bwd_ws")
-        return
-    _backend.check(
-        lib.kf_attn_bwd(dq_p, dk_p, dv_p, dout, q_p, k_p, v_p, o,
-                        _fp(lse), _fp(delta), B, S, Hq, Hkv, D, qts, kts,
-                        dqts, dkts, ctypes.c_float(float(scale)),
-                        int(causal), _stream()), "attn_bwd")
-
-
 class _FlashAttnHip(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
@@ -223,9 +199,11 @@ class _FlashAttnHip(torch.autograd.Function):
         dk = torch.empty_like(k)
         dv = torch.empty_like(v)
         delta = torch.empty(B, Hq, S, dtype=torch.float32, device=q.device)
-        _attn_bwd_call(lib, _p(dq), _p(dk), _p(dv), _p(dout), _p(q), _p(k),
-                       _p(v), _p(o), lse, delta, B, S, Hq, Hkv, D,
-                       0, 0, 0, 0, scale, causal, q.device)
+        _backend.check(
+            lib.kf_attn_bwd(_p(dq), _p(dk), _p(dv), _p(dout), _p(q), _p(k),
+                            _p(v), _p(o), _fp(lse), _fp(delta), B, S, Hq, Hkv,
+                            D, 0, 0, 0, 0, float(scale), int(causal),
+                            _stream()), "attn_bwd")
         return dq, dk, dv, None, None
 
 
@@ -405,11 +383,13 @@ class _FusedQkvAttentionHip(torch.autograd.Function):
         dout = dout.contiguous()
         dqkv = torch.empty_like(qkv)
         delta = torch.empty(B, Hq, S, dtype=torch.float32, device=qkv.device)
-        _attn_bwd_call(lib, _p(dqkv), _p_off(dqkv, Hq * D),
-                       _p_off(dqkv, (Hq + Hkv) * D), _p(dout), _p(qkv),
-                       _p_off(qkv, Hq * D), _p_off(qkv, (Hq + Hkv) * D),
-                       _p(o), lse, delta, B, S, Hq, Hkv, D,
-                       ts, ts, ts, ts, scale, causal, qkv.device)
+        _backend.check(
+            lib.kf_attn_bwd(_p(dqkv), _p_off(dqkv, Hq * D),
+                            _p_off(dqkv, (Hq + Hkv) * D), _p(dout), _p(qkv),
+                            _p_off(qkv, Hq * D), _p_off(qkv, (Hq + Hkv) * D),
+                            _p(o), _fp(lse), _fp(delta), B, S, Hq, Hkv, D,
+                            ts, ts, ts, ts, float(scale), int(causal),
+                            _stream()), "attn_bwd")
         # adjoint of the in-place rotation on the dq/dk regions
         _backend.check(
             lib.kf_rope(_p(dqkv), _p_off(dqkv, Hq * D), _fp(cos), _fp(sin),

@@ -85,11 +85,6 @@ def _configure(lib: ctypes.CDLL) -> None:
         lib.kf_attn_fwd4_rect.restype = I32
         lib.kf_attn_fwd4_rect.argtypes = [P, FP, P, P, P, I64, I64, I64,
                                           I64, I64, I64, I64, I64, F, I64, P]
-    if hasattr(lib, "kf_attn_bwd_ws"):
-        lib.kf_attn_bwd_ws.restype = I32
-        lib.kf_attn_bwd_ws.argtypes = [P, P, P, P, P, P, P, P, FP, FP, FP,
-                                       FP, I64, I64, I64, I64, I64, I64,
-                                       I64, I64, I64, F, I32, P]
     if hasattr(lib, "kf_attn_bwd"):
         lib.kf_attn_bwd.restype = I32
         lib.kf_attn_bwd.argtypes = [P, P, P, P, P, P, P, P, FP, FP, I64, I64,

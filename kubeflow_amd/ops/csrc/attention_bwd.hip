@@ -355,71 +355,11 @@ extern "C" int kf_attn_bwd8_dq(void*, const void*, const void*, const void*,
                                const void*, const float*, const float*,
                                int64_t, int64_t, int64_t, int64_t, int64_t,
                                int64_t, int64_t, float, int, void*);
-extern "C" int kf_attn_bwd8_dkv_part(float*, float*, const void*,
-                                     const void*, const void*, const void*,
-                                     const float*, const float*, int64_t,
-                                     int64_t, int64_t, int64_t, int64_t,
-                                     int64_t, int64_t, float, int, void*);
-
-// reduce the per-head fp32 dK/dV partials [g,B,S,Hkv,D] into the strided
-// bf16 outputs (deterministic fixed-order sum)
-__global__ __launch_bounds__(256) void kf_dkv_reduce_kernel(
-    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
-    const float* __restrict__ dkp, const float* __restrict__ dvp,
-    int64_t B, int S, int Hkv, int g, int64_t dkts) {
-  const int64_t n = (int64_t)B * S * Hkv * AB_D;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    float sk = 0.f, sv = 0.f;
-    for (int hg = 0; hg < g; ++hg) {
-      sk += dkp[(int64_t)hg * n + i];
-      sv += dvp[(int64_t)hg * n + i];
-    }
-    // i decodes as ((b*S + s)*Hkv + h)*D + d; outputs use row stride dkts
-    const int64_t row = i / ((int64_t)Hkv * AB_D);
-    const int64_t rem = i % ((int64_t)Hkv * AB_D);
-    const int64_t o = row * dkts + rem;
-    dk[o] = kf_f32_to_bf16(sk);
-    dv[o] = kf_f32_to_bf16(sv);
-  }
-}
-
-// Full backward with the 8-wave head-split dK/dV path: dk_ws/dv_ws are
-// caller-allocated fp32 workspaces of (Hq/Hkv)*B*S*Hkv*D elements.
-KF_EXPORT int kf_attn_bwd_ws(void* dq, void* dk, void* dv, const void* dout,
-                             const void* q, const void* k, const void* v,
-                             const void* o, const float* lse, float* delta,
-                             float* dk_ws, float* dv_ws,
-                             int64_t B, int64_t S, int64_t Hq, int64_t Hkv,
-                             int64_t D, int64_t qts, int64_t kts,
-                             int64_t dqts, int64_t dkts, float scale,
-                             int causal, void* stream) {
-  if (D != AB_D || S % 256 || Hq % Hkv) return (int)hipErrorInvalidValue;
-  if (qts == 0) qts = Hq * AB_D;
-  if (kts == 0) kts = Hkv * AB_D;
-  if (dqts == 0) dqts = Hq * AB_D;
-  if (dkts == 0) dkts = Hkv * AB_D;
-  hipLaunchKernelGGL(kf_attn_delta2_kernel,
-                     dim3(kf_grid_for(B * S * Hq, 4)), dim3(256), 0,
-                     (hipStream_t)stream, delta, (const unsigned short*)dout,
-                     (const unsigned short*)o, B, (int)S, (int)Hq);
-  int err = (int)hipGetLastError();
-  if (err) return err;
-  err = kf_attn_bwd8_dq(dq, q, k, v, dout, lse, delta, B, S, Hq, Hkv, qts,
-                        kts, dqts, scale, causal, stream);
-  if (err) return err;
-  err = kf_attn_bwd8_dkv_part(dk_ws, dv_ws, q, k, v, dout, lse, delta, B, S,
-                              Hq, Hkv, qts, kts, dkts, scale, causal,
-                              stream);
-  if (err) return err;
-  hipLaunchKernelGGL(kf_dkv_reduce_kernel,
-                     dim3(kf_grid_for(B * S * Hkv * AB_D, 256)), dim3(256),
-                     0, (hipStream_t)stream, (unsigned short*)dk,
-                     (unsigned short*)dv, dk_ws, dv_ws, B, (int)S, (int)Hkv,
-                     (int)(Hq / Hkv), dkts);
-  return (int)hipGetLastError();
-}
+extern "C" int kf_attn_bwd8_dkv(void*, void*, const void*, const void*,
+                                const void*, const void*, const float*,
+                                const float*, int64_t, int64_t, int64_t,
+                                int64_t, int64_t, int64_t, int64_t, float,
+                                int, void*);
 
 KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                           const void* q, const void* k, const void* v,
@@ -439,8 +379,13 @@ KF_EXPORT int kf_attn_bwd(void* dq, void* dk, void* dv, const void* dout,
                      (const unsigned short*)o, B, (int)S, (int)Hq);
   int err = (int)hipGetLastError();
   if (err) return err;
-  // (the 8-wave dK/dV path needs caller-provided fp32 workspaces — see
-  // kf_attn_bwd_ws; without them any S%64 shape takes the 4-wave kernels)
+  if (S % 256 == 0) {  // 8-wave swapped path for the training shapes
+    err = kf_attn_bwd8_dq(dq, q, k, v, dout, lse, delta, B, S, Hq, Hkv, qts,
+                          kts, dqts, scale, causal, stream);
+    if (err) return err;
+    return kf_attn_bwd8_dkv(dk, dv, q, k, v, dout, lse, delta, B, S, Hq, Hkv,
+                            qts, kts, dkts, scale, causal, stream);
+  }
   dim3 gq((unsigned)(S / 64), (unsigned)Hq, (unsigned)B);
   hipLaunchKernelGGL(kf_attn_dq_kernel, gq, dim3(256), 0, (hipStream_t)stream,
                      (unsigned short*)dq, (const unsigned short*)q,

@@ -314,29 +314,23 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dq8_kernel(
 #define DKV8_QT 64    // q rows per LDS tile
 
 __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
-    float* __restrict__ dvp, const unsigned short* __restrict__ q,
+    unsigned short* __restrict__ dv, const unsigned short* __restrict__ q,
     const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ dout, const float* __restrict__ lse,
     int64_t B, int S, int Hq, int Hkv, int64_t qts, int64_t kts,
     int64_t dkts, float scale, int causal) {
-  // HEAD-SPLIT (round 2): one block per (kv-block, GQA head) instead of a
-  // serial in-block head loop — 4x the blocks for LPT to balance, with
-  // fp32 per-head partials reduced by kf_dkv_reduce afterwards.
   __shared__ unsigned char q_lds[2][DKV8_QT * AB_D * 2];   // subtiled
   __shared__ unsigned char dot_lds[2][DKV8_QT * AB_D * 2]; // subtiled
   __shared__ float lse_s[2][DKV8_QT];                      // pre-mul log2e
 
   // 1-D grid decoded kt-major: causal work decreases with kt, so the
-  // longest blocks (kt=0) dispatch first (LPT order)
+  // longest blocks (kt=0) dispatch first (LPT order — with only
+  // S/256 x Hkv x B blocks the schedule tail otherwise sets the wall)
+  const int kt = blockIdx.x / (Hkv * (int)B);
+  const int rest = blockIdx.x % (Hkv * (int)B);
+  const int hkv = rest % Hkv;
+  const int64_t b = rest / Hkv;
   const int g = Hq / Hkv;
-  const int per_kt = g * Hkv * (int)B;
-  const int kt = blockIdx.x / per_kt;
-  const int rest = blockIdx.x % per_kt;
-  const int hg = rest / (Hkv * (int)B);
-  const int r2 = rest % (Hkv * (int)B);
-  const int hkv = r2 % Hkv;
-  const int64_t b = r2 / Hkv;
-  const int hq = hkv * g + hg;
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE;
   const int lane = tid & 63, l31 = lane & 31, hi = lane >> 5;
@@ -361,7 +355,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
 
   const int qt0 = causal ? (kt * DKV8_KT) / DKV8_QT : 0;
   const int nqt = S / DKV8_QT;
-  const int total = nqt - qt0;  // this head's q tiles
+  const int per_head = nqt - qt0;
+  const int total = g * per_head;
   const int sr0 = tid >> 4, sr1 = (tid + 512) >> 4, c8 = tid & 15;
   const unsigned tr_off = kf_tr_lane_off(lane);
 
@@ -370,7 +365,8 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
   kf_short8 sq0, sq1, sd0, sd1;
   float lse_ld = 0.f;
   auto stage_load = [&](int flat) {
-    const int qt = qt0 + flat;
+    const int hq = hkv * g + flat / per_head;
+    const int qt = qt0 + flat % per_head;
     const unsigned short* qg =
         q + (b * S + qt * DKV8_QT) * qts + (int64_t)hq * AB_D;
     const unsigned short* dog =
@@ -403,7 +399,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
     const int cur = flat & 1;
     const bool have_next = flat + 1 < total;
     if (have_next) stage_load(flat + 1);
-    const int qt = qt0 + flat;
+    const int qt = qt0 + flat % per_head;
 
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt) {
@@ -452,34 +448,28 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dv8_kernel(
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int kvr = kvbase + (r & 3) + 8 * (r >> 2);
-    const int64_t base =
-        ((((int64_t)hg * B + b) * S + kvr) * Hkv + hkv) * AB_D;
+    const int64_t base = (b * S + kvr) * dkts + (int64_t)hkv * AB_D;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      dvp[base + dt * 32 + l31] = dvacc[dt][r];
+      dv[base + dt * 32 + l31] = kf_f32_to_bf16(dvacc[dt][r]);
   }
 }
 
 __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
-    float* __restrict__ dkp, const unsigned short* __restrict__ q,
+    unsigned short* __restrict__ dk, const unsigned short* __restrict__ q,
     const unsigned short* __restrict__ k, const unsigned short* __restrict__ v,
     const unsigned short* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ delta, int64_t B, int S, int Hq, int Hkv,
     int64_t qts, int64_t kts, int64_t dkts, float scale, int causal) {
-  // head-split + fp32 partials like dv8 (see note there)
   __shared__ unsigned char q_lds[2][DKV8_QT * AB_D * 2];   // subtiled
   __shared__ unsigned char do_lds[2][DKV8_QT * AB_D * 2];  // subtiled
   __shared__ float lse_s2[2][DKV8_QT], dlt_s2[2][DKV8_QT];
 
-  const int g = Hq / Hkv;  // LPT head-split decode (see dv8)
-  const int per_kt = g * Hkv * (int)B;
-  const int kt = blockIdx.x / per_kt;
-  const int rest = blockIdx.x % per_kt;
-  const int hg = rest / (Hkv * (int)B);
-  const int r2 = rest % (Hkv * (int)B);
-  const int hkv = r2 % Hkv;
-  const int64_t b = r2 / Hkv;
-  const int hq = hkv * g + hg;
+  const int kt = blockIdx.x / (Hkv * (int)B);  // LPT decode (see dv8)
+  const int rest = blockIdx.x % (Hkv * (int)B);
+  const int hkv = rest % Hkv;
+  const int64_t b = rest / Hkv;
+  const int g = Hq / Hkv;
   const int tid = threadIdx.x;
   const int w = tid / KF_WAVE;
   const int lane = tid & 63, l31 = lane & 31, hi = lane >> 5;
@@ -506,14 +496,16 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
 
   const int qt0 = causal ? (kt * DKV8_KT) / DKV8_QT : 0;
   const int nqt = S / DKV8_QT;
-  const int total = nqt - qt0;
+  const int per_head = nqt - qt0;
+  const int total = g * per_head;
   const int sr0 = tid >> 4, sr1 = (tid + 512) >> 4, c8 = tid & 15;
   const unsigned tr_off = kf_tr_lane_off(lane);
 
   kf_short8 sq0, sq1, sd0, sd1;
   float lse_ld = 0.f, dlt_ld = 0.f;
   auto stage_load = [&](int flat) {
-    const int qt = qt0 + flat;
+    const int hq = hkv * g + flat / per_head;
+    const int qt = qt0 + flat % per_head;
     const unsigned short* qg =
         q + (b * S + qt * DKV8_QT) * qts + (int64_t)hq * AB_D;
     const unsigned short* dog =
@@ -550,7 +542,7 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
     const int cur = flat & 1;
     const bool have_next = flat + 1 < total;
     if (have_next) stage_load(flat + 1);
-    const int qt = qt0 + flat;
+    const int qt = qt0 + flat % per_head;
 
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt) {
@@ -605,11 +597,10 @@ __global__ __launch_bounds__(512, 2) void kf_attn_dk8_kernel(
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int kvr = kvbase + (r & 3) + 8 * (r >> 2);
-    const int64_t base =
-        ((((int64_t)hg * B + b) * S + kvr) * Hkv + hkv) * AB_D;
+    const int64_t base = (b * S + kvr) * dkts + (int64_t)hkv * AB_D;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      dkp[base + dt * 32 + l31] = dkacc[dt][r];
+      dk[base + dt * 32 + l31] = kf_f32_to_bf16(dkacc[dt][r]);
   }
 }
 
@@ -629,26 +620,22 @@ KF_EXPORT int kf_attn_bwd8_dq(void* dq, const void* q, const void* k,
   return (int)hipGetLastError();
 }
 
-// dK/dV partials, one block per (kv-block, head); the caller provides
-// fp32 workspaces of g*B*S*Hkv*D and reduces with kf_dkv_reduce.
-KF_EXPORT int kf_attn_bwd8_dkv_part(float* dkp, float* dvp, const void* q,
-                                    const void* k, const void* v,
-                                    const void* dout, const float* lse,
-                                    const float* delta, int64_t B, int64_t S,
-                                    int64_t Hq, int64_t Hkv, int64_t qts,
-                                    int64_t kts, int64_t dkts, float scale,
-                                    int causal, void* stream) {
-  const int64_t g = Hq / Hkv;
-  dim3 gkv((unsigned)((S / DKV8_KT) * g * Hkv * B), 1, 1);
+KF_EXPORT int kf_attn_bwd8_dkv(void* dk, void* dv, const void* q,
+                               const void* k, const void* v, const void* dout,
+                               const float* lse, const float* delta,
+                               int64_t B, int64_t S, int64_t Hq, int64_t Hkv,
+                               int64_t qts, int64_t kts, int64_t dkts,
+                               float scale, int causal, void* stream) {
+  dim3 gkv((unsigned)((S / DKV8_KT) * Hkv * B), 1, 1);
   hipLaunchKernelGGL(kf_attn_dv8_kernel, gkv, dim3(512), 0,
-                     (hipStream_t)stream, dvp,
+                     (hipStream_t)stream, (unsigned short*)dv,
                      (const unsigned short*)q, (const unsigned short*)k,
                      (const unsigned short*)dout, lse, B, (int)S, (int)Hq,
                      (int)Hkv, qts, kts, dkts, scale, causal);
   int err = (int)hipGetLastError();
   if (err) return err;
   hipLaunchKernelGGL(kf_attn_dk8_kernel, gkv, dim3(512), 0,
-                     (hipStream_t)stream, dkp,
+                     (hipStream_t)stream, (unsigned short*)dk,
                      (const unsigned short*)q, (const unsigned short*)k,
                      (const unsigned short*)v, (const unsigned short*)dout,
                      lse, delta, B, (int)S, (int)Hq, (int)Hkv, qts, kts,
