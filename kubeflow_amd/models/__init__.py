@@ -7,7 +7,8 @@ from __future__ import annotations
 from typing import Callable, Dict
 
 from .llama import (LlamaConfig, LlamaModel, llama3_8b, llama3_1b,
-                    llama3_70b, llama_moe_tiny, llama_tiny, llama_tiny_mha)
+                    llama3_70b, llama_moe_3b, llama_moe_tiny, llama_tiny,
+                    llama_tiny_mha)
 from .bert import (BertConfig, BertClassifier, bert_base, bert_base_hd128,
                    bert_tiny)
 from .mlp import MnistMLP
@@ -50,6 +51,7 @@ MODEL_CONFIGS: Dict[str, Callable] = {
     "llama-tiny": llama_tiny,
     "llama-tiny-mha": llama_tiny_mha,
     "llama-moe-tiny": llama_moe_tiny,
+    "llama-moe-3b": llama_moe_3b,
     "bert-base": bert_base,
     "bert-base-hd128": bert_base_hd128,
     "bert-tiny": bert_tiny,
@@ -71,6 +73,7 @@ MODEL_REGISTRY: Dict[str, Callable] = {
     "llama-tiny": _llama(llama_tiny),
     "llama-tiny-mha": _llama(llama_tiny_mha),
     "llama-moe-tiny": _llama(llama_moe_tiny),
+    "llama-moe-3b": _llama(llama_moe_3b),
     "bert-base": _bert(bert_base),
     "bert-base-hd128": _bert(bert_base_hd128),
     "bert-tiny": _bert(bert_tiny),

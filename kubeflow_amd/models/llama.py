@@ -89,6 +89,16 @@ def llama_moe_tiny(vocab: int = 512) -> LlamaConfig:
                        ffn_dim=512, max_seq_len=512, n_experts=4, top_k=2)
 
 
+def llama_moe_3b(vocab: int = 128256) -> LlamaConfig:
+    """Mid-size MoE (8 experts, top-2, ~3.4B params, ~0.9B active/token) —
+    a single-GPU-trainable Mixtral-style config for measuring the MoE
+    train path at real scale (the tiny config only covers numerics)."""
+    return LlamaConfig(name="llama-moe-3b", vocab_size=vocab,
+                       hidden_size=2048, n_layers=16, n_heads=16,
+                       n_kv_heads=4, ffn_dim=2048, max_seq_len=8192,
+                       n_experts=8, top_k=2)
+
+
 def llama_tiny_mha(vocab: int = 512) -> LlamaConfig:
     """Tiny MHA config (kv == q heads) whose head/ffn counts divide by 2 —
     used by the tensor-parallel tests and examples."""
