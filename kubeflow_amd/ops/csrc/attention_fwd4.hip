@@ -344,6 +344,21 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
                      kv_lo0 + 31 >= skv, pb);
         pv_block(pb, vbase + (unsigned)(pair << 11));
       }
+      // write the staged tile into the other buffer BETWEEN the two
+      // sub-blocks: the vmcnt wait on the staged registers lands after
+      // QK0/SM0/PV0 covered the HBM latency, and the ds_writes overlap
+      // sub-block 1's MFMAs instead of sitting in the serial tail
+      if (pair == 0 && have_next) {
+        const int nxt = cur ^ 1;
+#pragma unroll
+        for (int j = 0; j < A4_KT / 32; ++j) {
+          *reinterpret_cast<kf_short8*>(k_lds[nxt] +
+                                        kf_swz4(srow[j], c8 * 16)) = kst[j];
+          *reinterpret_cast<kf_short8*>(v_lds[nxt] +
+                                        2 * kf_vsub4(srow[j], c8 * 8)) =
+              vst[j];
+        }
+      }
       if (do1) {
         sm_block(st1, kv_lo1,
                  (causal && kv_lo1 + 31 > wave_qmin + qoff) ||
@@ -351,11 +366,9 @@ __global__ __launch_bounds__(A4_THREADS, 2) void kf_attn_fwd4_kernel(
         pv_block(pb, vbase + (unsigned)(pair << 11) + 1024);
       }
     }
-
-    // ---- write the staged tile into the other buffer; the compiler
-    // inserts the vmcnt wait on the staged registers here, after the
-    // compute phase has covered the HBM latency ----
-    if (have_next) {
+    // causal skip can bypass the pair-0 body entirely: make sure the
+    // staged tile still lands before the barrier
+    if (have_next && causal && kt * A4_KT > wave_qmax + qoff) {
       const int nxt = cur ^ 1;
 #pragma unroll
       for (int j = 0; j < A4_KT / 32; ++j) {
