@@ -76,6 +76,82 @@ def save(trainer, ckpt_dir: str, model_name: str, rank: int, world: int,
     return d
 
 
+class AsyncSave:
+    """Overlapped checkpointing (round-1 weak item 8): the training stall
+    shrinks to the device->host snapshot (PCIe-bound, ~1.8 s for the 8B
+    model's ~112 GB of flat state at 63 GB/s); the disk-bound libkfio
+    write + the commit run on a background thread while training
+    continues. One outstanding save at a time — a new save (or close)
+    waits for the previous write to become durable first, preserving the
+    all-ranks-durable-then-commit crash protocol for the completed step."""
+
+    def __init__(self):
+        self._thread = None
+        self._err = None
+
+    def wait(self):
+        """Block until the in-flight write (if any) is durable; re-raises
+        a background failure."""
+        if self._thread is not None:
+            self._thread.join()
+            self._thread = None
+        if self._err is not None:
+            err, self._err = self._err, None
+            raise err
+
+    def save(self, trainer, ckpt_dir: str, model_name: str, rank: int,
+             world: int, commit: bool = True):
+        import threading
+
+        self.wait()
+        step = trainer.step_num
+        d = os.path.join(ckpt_dir, f"step-{step}")
+        os.makedirs(d, exist_ok=True)
+        sharded = _is_sharded(trainer)
+        # snapshot to host NOW (training mutates these right after)
+        snap = {}
+        if sharded or rank == 0:
+            snap["model"] = trainer.flat.data.to("cpu", copy=True)
+        for tag, t in (("p32", trainer.p32), ("m", trainer.m),
+                       ("v", trainer.v)):
+            snap[tag] = t.to("cpu", copy=True)
+        rng = {"step": step, "rng": torch.get_rng_state(),
+               "cuda_rng": (torch.cuda.get_rng_state()
+                            if torch.cuda.is_available() else None)}
+        if trainer.flat.data.is_cuda:
+            torch.cuda.synchronize()
+
+        def _write():
+            try:
+                if "model" in snap:
+                    name = (f"model-rank{rank}.bin" if sharded
+                            else "model.bin")
+                    fastio.write_tensor(os.path.join(d, name), snap["model"])
+                for tag in ("p32", "m", "v"):
+                    fastio.write_tensor(
+                        os.path.join(d, f"optim-{tag}-rank{rank}.bin"),
+                        snap[tag])
+                torch.save(rng, os.path.join(d, f"optim-rank{rank}.pt"))
+                if rank == 0:
+                    with open(os.path.join(d, "meta.json"), "w") as f:
+                        json.dump({"step": step, "world_size": world,
+                                   "model": model_name, "sharded": sharded,
+                                   "format": 2,
+                                   "flat_dtype": str(
+                                       trainer.flat.data.dtype),
+                                   "param_names": trainer.flat.names,
+                                   "timestamp": time.time()}, f)
+                    if commit:
+                        commit_latest(ckpt_dir, step)
+            except BaseException as e:  # surfaced on the next wait()
+                self._err = e
+
+        self._thread = threading.Thread(target=_write, daemon=True,
+                                        name=f"ckpt-save-{step}")
+        self._thread.start()
+        return d
+
+
 def commit_latest(ckpt_dir: str, step: int):
     """Atomically point `latest` at step-<step> (call after ALL ranks'
     files are written)."""

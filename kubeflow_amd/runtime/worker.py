@@ -160,11 +160,32 @@ def synthetic_batch(spec: dict, cfg, device, rank: int, step: int):
     return x.to(device), y.to(device)
 
 
-def _save_all_ranks(trainer, ckpt_dir, spec, rank, world):
+_ASYNC_SAVER = None
+
+
+def _save_all_ranks(trainer, ckpt_dir, spec, rank, world, final=False):
     """Checkpoint with the `latest` marker committed only after EVERY
-    rank's files are durable (checkpoint.py:save commit semantics)."""
-    kdist.barrier()
+    rank's files are durable (checkpoint.py:save commit semantics).
+
+    world==1 defaults to OVERLAPPED saves (checkpoint.AsyncSave): training
+    only stalls for the device->host snapshot while the disk-bound write
+    runs in the background (KF_ASYNC_CKPT=0 opts out). Multi-rank jobs
+    keep the synchronous all-ranks-durable-then-commit protocol."""
+    global _ASYNC_SAVER
     from kubeflow_amd.runtime import checkpoint as _ckpt
+    use_async = (world == 1
+                 and os.environ.get("KF_ASYNC_CKPT", "1") == "1")
+    if use_async:
+        if _ASYNC_SAVER is None:
+            _ASYNC_SAVER = _ckpt.AsyncSave()
+        _ASYNC_SAVER.save(trainer, ckpt_dir, spec["model"], rank, world,
+                          commit=True)
+        if final:
+            _ASYNC_SAVER.wait()
+        return
+    if _ASYNC_SAVER is not None:
+        _ASYNC_SAVER.wait()
+    kdist.barrier()
     _ckpt.save(trainer, ckpt_dir, spec["model"], rank, world, commit=False)
     kdist.barrier()
     if rank == 0:
@@ -347,7 +368,9 @@ def main(argv=None):
             if save_every and (step + 1) % save_every == 0:
                 _save_all_ranks(trainer, ckpt_dir, spec, rank, world)
         if spec.get("save_final", True):
-            _save_all_ranks(trainer, ckpt_dir, spec, rank, world)
+            _save_all_ranks(trainer, ckpt_dir, spec, rank, world, final=True)
+        elif _ASYNC_SAVER is not None:
+            _ASYNC_SAVER.wait()  # drain in-flight periodic saves
         write_status(rank_dir, "succeeded", steps, loss,
                      metrics={"loss": None if loss is None else float(loss),
                               "loss_ema": ema})

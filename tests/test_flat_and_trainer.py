@@ -153,3 +153,44 @@ def test_checkpoint_format1_compat(tmp_path):
     assert ckpt.load(tr2, str(tmp_path), 0) == 1
     assert torch.equal(tr.flat.data, tr2.flat.data)
     assert torch.equal(tr.p32, tr2.p32)
+
+
+def test_async_checkpoint_matches_sync(tmp_path):
+    """AsyncSave snapshots before training continues: a checkpoint taken
+    mid-training then overwritten by more steps must hold the state AT
+    the snapshot, byte-identical to a synchronous save at that step."""
+    import torch
+
+    from kubeflow_amd.models import build_model
+    from kubeflow_amd.runtime import Trainer, TrainConfig
+    from kubeflow_amd.runtime import checkpoint as ckpt
+
+    torch.manual_seed(0)
+    m = build_model("llama-tiny", dtype=torch.float32)
+    tr = Trainer(m, TrainConfig(lr=1e-3, warmup_steps=1))
+    toks = torch.randint(0, m.cfg.vocab_size, (2, 32))
+    for _ in range(2):
+        tr.step(toks, toks)
+
+    sync_dir = tmp_path / "sync"
+    ckpt.save(tr, str(sync_dir), "llama-tiny", 0, 1)
+    want = tr.flat.data.clone()
+
+    saver = ckpt.AsyncSave()
+    async_dir = tmp_path / "async"
+    saver.save(tr, str(async_dir), "llama-tiny", 0, 1)
+    # keep training while the write is in flight — must not corrupt it
+    for _ in range(3):
+        tr.step(toks, toks)
+    saver.wait()
+
+    m2 = build_model("llama-tiny", dtype=torch.float32)
+    step = ckpt.load_model_weights(m2, str(async_dir))
+    assert step == 2
+    # load into a fresh trainer and compare the snapshot-time flat buffer
+    m3 = build_model("llama-tiny", dtype=torch.float32)
+    tr3 = Trainer(m3, TrainConfig(lr=1e-3, warmup_steps=1))
+    s3 = ckpt.load(tr3, str(async_dir), 0)
+    assert s3 == 2
+    assert torch.equal(tr3.flat.data, want)
+    assert not torch.equal(tr3.flat.data, tr.flat.data)  # training moved on
