@@ -288,10 +288,39 @@ class ObjectStore:
                 pass
 
     # -------------------------------------------------------------- events
+    # kube prunes Events by TTL (kube-apiserver --event-ttl, default 1h);
+    # without it a long-running platform accumulates them without bound
+    EVENT_TTL_S = float(os.environ.get("KF_EVENT_TTL_S", "3600"))
+    _EVENT_PRUNE_EVERY = 200
+
+    def _prune_events(self):
+        cutoff = time.time() - self.EVENT_TTL_S
+        stale = []
+        with self._lock:
+            for (kind, ns, name), obj in self._objs.items():
+                if kind != "Event":
+                    continue
+                ts = obj["metadata"].get("creationTimestamp", "")
+                try:
+                    import calendar
+                    t = calendar.timegm(time.strptime(ts,
+                                                      "%Y-%m-%dT%H:%M:%SZ"))
+                except (ValueError, TypeError):
+                    continue
+                if t < cutoff:
+                    stale.append((name, ns))
+        for name, ns in stale:
+            try:
+                self.delete("Event", name, ns)
+            except StoreError:
+                pass
+
     def record_event(self, involved: KfObject, reason: str, message: str,
                      etype: str = "Normal"):
         """k8s-Event-shaped record for the activities feed / status surfacing
         (jupyter status derivation reads these: apps/common/status.py:60-99)."""
+        if self._rv % self._EVENT_PRUNE_EVERY == 0:
+            self._prune_events()
         name = f"{involved['metadata']['name']}.{self._rv}.{int(time.time()*1000)}"
         ev = new_object("Event", name,
                         involved["metadata"].get("namespace") or "default",

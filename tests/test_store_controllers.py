@@ -240,3 +240,16 @@ def test_profile_plugin_seam(tmp_path):
         while _time.time() < deadline and _os.path.exists(cred):
             _time.sleep(0.2)
         assert not _os.path.exists(cred)
+
+
+def test_event_ttl_pruning(monkeypatch):
+    """Events expire after KF_EVENT_TTL_S like kube's --event-ttl."""
+    store = ObjectStore()
+    store.EVENT_TTL_S = 0.2
+    obj = store.create(new_object("PyTorchJob", "evt-job", "default"))
+    store.record_event(obj, "Old", "stale soon")
+    time.sleep(0.4)
+    store._prune_events()
+    store.record_event(obj, "New", "fresh")
+    reasons = [e["reason"] for e in store.events_for(obj)]
+    assert reasons == ["New"]
