@@ -74,17 +74,20 @@ class InferenceServiceReconciler(Reconciler):
         self.key_uid[(namespace, name)] = uid
         if has_condition(svc, "Failed"):  # invalid spec — terminal
             return
-        gang = self.gangs.get(uid)
+        gangs = self.gangs.get(uid)
 
-        if gang is None:
+        if gangs is None:
             self._start(svc)
             raise RequeueAfter(0.5)
 
-        state = gang.poll()
-        if state is not None:  # server process died -> restart (Always)
+        state = next((st for st in (g.poll() for g in gangs)
+                      if st is not None), None)
+        if state is not None:  # a server/proxy process died -> restart all
             self.store.record_event(svc, "PredictorCrashed",
                                     f"server exited ({state}); restarting",
                                     "Warning")
+            for g in gangs:
+                g.terminate_and_wait()
             self.gangs.pop(uid, None)
             self.scheduler.release(uid)
             set_condition(svc, "Ready", "False", "Restarting", "")
@@ -118,13 +121,14 @@ class InferenceServiceReconciler(Reconciler):
         uid = svc["metadata"]["uid"]
         pred = svc["spec"].get("predictor", {})
         gpus = int(pred.get("gpus", 1))
+        replicas = max(1, int(pred.get("replicas", 1)))
         if gpus > 1:
-            # BASELINE serving config is TP=1; multi-GPU serving (TP decode
-            # + broadcast-coordinated batching) is a declared v2 seam —
-            # fail loudly rather than strand an allocated-but-idle GPU.
+            # BASELINE serving config is TP=1; intra-model TP decode is a
+            # declared v2 seam — scale out with predictor.replicas instead
+            # (each replica owns a full copy on its own GPU, KServe-style).
             set_condition(svc, "Failed", "True", "InvalidSpec",
-                          f"predictor.gpus={gpus}: multi-GPU serving is not "
-                          "implemented in v1 (TP=1 per BASELINE config)")
+                          f"predictor.gpus={gpus}: TP>1 serving is not "
+                          "implemented (scale out with predictor.replicas)")
             self.store.update(svc, check_version=False)
             self.store.record_event(svc, "InvalidSpec",
                                     "multi-GPU serving not implemented",
@@ -137,14 +141,15 @@ class InferenceServiceReconciler(Reconciler):
         if want_gpu:
             from kubeflow_amd.scheduler.quota import QuotaExceeded, admit_gpus
             try:
-                admit_gpus(self.store, self.scheduler, ns, gpus)
+                admit_gpus(self.store, self.scheduler, ns, gpus * replicas)
             except QuotaExceeded as e:
                 self.store.record_event(svc, "QuotaExceeded", str(e),
                                         "Warning")
                 raise RequeueAfter(2.0)
         try:
             if want_gpu:
-                alloc = self.scheduler.allocate(uid, gpus, namespace=ns)
+                alloc = self.scheduler.allocate(uid, gpus * replicas,
+                                                namespace=ns)
                 gpu_indices = alloc.gpu_indices
             else:
                 self.scheduler.allocate(uid, 0)
@@ -154,15 +159,12 @@ class InferenceServiceReconciler(Reconciler):
                                     "Warning")
             raise RequeueAfter(2.0)
 
-        port = free_port()
-        self.ports[uid] = port
         m = svc["metadata"]
         workdir = os.path.join(self.serving_dir, m.get("namespace") or
                                "default", f"{m['name']}-{uid[:8]}")
-        spec = {
+        base_spec = {
             "name": m["name"],
             "model": pred.get("model", "llama-tiny"),
-            "port": port,
             "max_batch": pred.get("maxBatch", 16),
             "max_seq_len": pred.get("maxSeqLen", 2048),
             "max_slots": pred.get("maxSlots", 16),
@@ -170,9 +172,9 @@ class InferenceServiceReconciler(Reconciler):
         }
         if pred.get("storageUri"):
             try:
-                spec["ckpt_dir"] = self._resolve_storage_uri(
+                base_spec["ckpt_dir"] = self._resolve_storage_uri(
                     pred["storageUri"], m.get("namespace"))
-                spec["storage_uri"] = pred["storageUri"]
+                base_spec["storage_uri"] = pred["storageUri"]
             except ValueError as e:
                 self.scheduler.release(uid)
                 set_condition(svc, "Failed", "True", "InvalidStorageUri",
@@ -182,30 +184,58 @@ class InferenceServiceReconciler(Reconciler):
                                         "Warning")
                 return
         poddefaults = self.store.list("PodDefault", m.get("namespace"))
+        gangs = []
+        backend_ports = []
         try:
-            gang = launch_gang(
-                uid, workdir, spec, gpu_indices, poddefaults=poddefaults,
-                labels=m.get("labels", {}),
-                entry_module="kubeflow_amd.runtime.serving_server")
+            for i in range(replicas):
+                rport = free_port()
+                backend_ports.append(rport)
+                rspec = dict(base_spec)
+                rspec["port"] = rport
+                gangs.append(launch_gang(
+                    f"{uid}-r{i}",
+                    workdir if replicas == 1
+                    else os.path.join(workdir, f"replica-{i}"),
+                    rspec, gpu_indices[i * gpus:(i + 1) * gpus],
+                    poddefaults=poddefaults, labels=m.get("labels", {}),
+                    entry_module="kubeflow_amd.runtime.serving_server"))
+            if replicas > 1:
+                # the Service analog: round-robin proxy on the published
+                # port (KServe spreads replicas behind a k8s Service)
+                port = free_port()
+                gangs.append(launch_gang(
+                    f"{uid}-proxy", os.path.join(workdir, "proxy"),
+                    {"name": m["name"], "port": port,
+                     "backends": [f"http://127.0.0.1:{p}"
+                                  for p in backend_ports],
+                     "world_size": 1},
+                    [], poddefaults=[], labels=m.get("labels", {}),
+                    entry_module="kubeflow_amd.runtime.serving_proxy"))
+            else:
+                port = backend_ports[0]
         except ValueError as e:  # spec-level launch error — terminal
+            for g in gangs:
+                g.terminate_and_wait()
             self.scheduler.release(uid)
             set_condition(svc, "Failed", "True", "InvalidSpec", str(e))
             self.store.update(svc, check_version=False)
             self.store.record_event(svc, "InvalidSpec", str(e), "Warning")
             return
-        self.gangs[uid] = gang
+        self.ports[uid] = port
+        self.gangs[uid] = gangs
+        svc["status"]["replicas"] = replicas
         set_condition(svc, "Created", "True", "PredictorCreated",
-                      f"port {port}")
+                      f"port {port} ({replicas} replica(s))")
         self.store.update(svc, check_version=False)
         self.store.record_event(svc, "SuccessfulCreate",
-                                f"serving process on port {port}")
+                                f"{replicas} serving process(es), published "
+                                f"port {port}")
 
     def on_deleted(self, namespace, name):
         uid = self.key_uid.pop((namespace, name), None)
         if uid:
-            gang = self.gangs.pop(uid, None)
-            if gang is not None:
-                gang.terminate_and_wait()
+            for g in self.gangs.pop(uid, []) or []:
+                g.terminate_and_wait()
             self.scheduler.release(uid)
             self.ports.pop(uid, None)
 

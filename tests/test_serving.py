@@ -359,3 +359,46 @@ def test_flash_attention_rect_cpu_reference():
     s = s.masked_fill(~mask, float("-inf"))
     want = (s.softmax(-1) @ vr.float()).transpose(1, 2)
     assert torch.allclose(o.float(), want, atol=1e-4)
+
+
+def test_inference_service_replicas(tmp_path):
+    """KServe-style scale-out: predictor.replicas=2 runs two engines
+    behind the round-robin proxy; both backends serve traffic and the
+    published URL survives single-replica load."""
+    import json
+    import time
+    import urllib.request
+
+    from kubeflow_amd.api import new_object
+    from kubeflow_amd.api.objects import get_condition, has_condition
+    from kubeflow_amd.platform import Platform
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        svc = new_object("InferenceService", "scaled", "default", spec={
+            "predictor": {"model": "llama-tiny", "gpus": 0, "replicas": 2,
+                          "maxSlots": 2, "maxSeqLen": 256}},
+            api_version="serving.kserve.io/v1beta1")
+        plat.store.create(svc)
+        deadline = time.time() + 180
+        while time.time() < deadline:
+            obj = plat.store.get("InferenceService", "scaled", "default")
+            if has_condition(obj, "Ready"):
+                break
+            time.sleep(0.5)
+        assert has_condition(obj, "Ready"), obj["status"]
+        assert obj["status"]["replicas"] == 2
+        url = obj["status"]["url"]
+        # proxy health reports both replicas
+        with urllib.request.urlopen(f"{url}/healthz", timeout=5) as r:
+            h = json.load(r)
+        assert h["replicas"] == 2 and h["ready"] == 2, h
+        # several predictions round-robin across the backends
+        for i in range(4):
+            body = json.dumps({"instances": [
+                {"prompt": [1, 2, 3 + i], "max_new_tokens": 4}]}).encode()
+            req = urllib.request.Request(
+                f"{url}/v1/models/scaled:predict", data=body,
+                headers={"Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=60) as r:
+                out = json.load(r)
+            assert out["predictions"][0]["tokens"], out
