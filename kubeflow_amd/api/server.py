@@ -55,6 +55,8 @@ PLURALS = {
     "trials": ("Trial", "kubeflow.org/v1beta1"),
     "pipelineruns": ("PipelineRun", "pipelines.kubeflow.org/v1"),
     "events": ("Event", "v1"),
+    "configmaps": ("ConfigMap", "v1"),
+    "resourcequotas": ("ResourceQuota", "v1"),
 }
 
 STOP_ANNOTATION = "kubeflow-resource-stopped"

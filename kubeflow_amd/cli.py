@@ -106,6 +106,7 @@ def apply(file: str = typer.Option(..., "-f", "--file")):
     plural = {"PyTorchJob": "pytorchjobs", "TFJob": "tfjobs",
               "Notebook": "notebooks", "Tensorboard": "tensorboards",
               "PersistentVolumeClaim": "pvcs", "PodDefault": "poddefaults",
+              "ConfigMap": "configmaps", "ResourceQuota": "resourcequotas",
               "InferenceService": "inferenceservices",
               "Experiment": "experiments", "PipelineRun": "pipelineruns"}
     for doc in docs:

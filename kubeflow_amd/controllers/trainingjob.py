@@ -179,12 +179,16 @@ class TrainingJobReconciler(Reconciler):
         spec["world_size"] = n
         poddefaults = self.store.list("PodDefault",
                                       job["metadata"].get("namespace"))
+        configmaps = {c["metadata"]["name"]: c.get("data", {})
+                      for c in self.store.list(
+                          "ConfigMap", job["metadata"].get("namespace"))}
         numa = {g.index: g.numa_node for g in self.scheduler.inv.gpus}
         try:
             gang = launch_gang(uid, workdir, spec, gpu_indices,
                                poddefaults=poddefaults,
                                labels=job["metadata"].get("labels", {}),
-                               numa_nodes=numa, warm_pool=self.warm_pool)
+                               numa_nodes=numa, warm_pool=self.warm_pool,
+                               configmaps=configmaps)
         except ValueError as e:
             # spec-level launch error (e.g. PodDefault env conflict) —
             # terminal, not retryable: mark Failed instead of hot-looping
