@@ -25,7 +25,8 @@ def _llama(cfg_fn):
         import torch
         return LlamaModel(cfg_fn(), device=device,
                           dtype=dtype or torch.bfloat16, tp=kw.get("tp"),
-                          sp=kw.get("sp"), ep=kw.get("ep"))
+                          sp=kw.get("sp"), ep=kw.get("ep"),
+                          cp=kw.get("cp"))
     return build
 
 

@@ -168,14 +168,17 @@ class MoEMLP(nn.Module):
 
 
 class LlamaBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig, tp=None, sp=None, ep=None):
+    def __init__(self, cfg: LlamaConfig, tp=None, sp=None, ep=None,
+                 cp=None):
         super().__init__()
         h, d = cfg.hidden_size, cfg.head_dim
         self.cfg = cfg
         self.tp = tp if (tp is not None and tp.world > 1) else None
         self.sp = sp if (sp is not None and sp.world > 1) else None
-        if self.tp is not None and self.sp is not None:
-            raise ValueError("tp and sp (ulysses) are mutually exclusive")
+        self.cp = cp if (cp is not None and cp.world > 1) else None
+        if sum(x is not None for x in (self.tp, self.sp, self.cp)) > 1:
+            raise ValueError("tp / ulysses-sp / ring-cp are mutually "
+                             "exclusive per block")
         if ep is not None and cfg.n_experts <= 0:
             raise ValueError("expert parallelism requires an MoE config "
                              "(n_experts > 0)")
@@ -237,6 +240,20 @@ class LlamaBlock(nn.Module):
             o = ops.fused_qkv_attention(qkv, cos, sin, self.hq // n,
                                         self.hkv // n, cfg.head_dim)
             o = spmod.gather_heads_scatter_seq(o, self.sp)
+        elif self.cp is not None:
+            # ring/context parallel: this rank's contiguous chunk attends
+            # the full ring-rotated sequence (parallel/ring.py); rope uses
+            # the chunk's GLOBAL positions
+            from kubeflow_amd.parallel.ring import ring_attention
+            q, kk, vv = qkv.split([self.hq * cfg.head_dim,
+                                   self.hkv * cfg.head_dim,
+                                   self.hkv * cfg.head_dim], dim=-1)
+            q = q.view(B, S, self.hq, cfg.head_dim)
+            kk = kk.view(B, S, self.hkv, cfg.head_dim)
+            vv = vv.view(B, S, self.hkv, cfg.head_dim)
+            q, kk = ops.rope(q, kk, cos, sin, self.cp.rank * S)
+            o = ring_attention(q, kk, vv, self.cp, causal=True)
+            o = o.reshape(B, S, self.hq * cfg.head_dim)
         else:
             o = ops.fused_qkv_attention(qkv, cos, sin, self.hq,
                                         self.hkv, cfg.head_dim)
@@ -264,7 +281,7 @@ class LlamaBlock(nn.Module):
 
 class LlamaModel(nn.Module):
     def __init__(self, cfg: LlamaConfig, device=None, dtype=torch.bfloat16,
-                 tp=None, sp=None, ep=None):
+                 tp=None, sp=None, ep=None, cp=None):
         super().__init__()
         self.cfg = cfg
         self.tp = tp
@@ -274,7 +291,7 @@ class LlamaModel(nn.Module):
         with torch.device(device if device is not None else "cpu"):
             self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
             self.layers = nn.ModuleList(
-                [LlamaBlock(cfg, tp=tp, sp=sp, ep=ep)
+                [LlamaBlock(cfg, tp=tp, sp=sp, ep=ep, cp=cp)
                  for _ in range(cfg.n_layers)])
             self.final_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
             self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,

@@ -4,10 +4,9 @@ SURVEY.md §2.14: DDP is the only strategy any BASELINE config requires.
 Implemented: DDP (bucketed all-reduce, ddp.py), TP (Megatron-style
 head/ffn sharding, tp.py; composes with DP as a TP x DP mesh when
 degree < world_size), PP (1F1B/GPipe stages, pp.py), Ulysses SP (sp.py)
-and EP (expert parallel for MoE configs, ep.py) — llama family. Ring-SP/CP
-remains a declared-but-reserved enum so PyTorchJob specs stay
-forward-compatible, rejected with a clear error until a config demands it
-(it needs a ring-attention kernel). The seam is the per-rank environment
+EP (expert parallel for MoE configs, ep.py) and ring-SP/CP (rotating-KV
+context parallel, ring.py — "sp", aliases "ring"/"cp") — llama family.
+The seam is the per-rank environment
 the gang launcher already provides (RANK/WORLD_SIZE/LOCAL_RANK + this
 descriptor serialized into the worker spec as `parallelism`).
 
@@ -29,7 +28,8 @@ class Strategy(str, Enum):
                      # degree == world -> pure TP, degree < world -> TP x DP
     PP = "pp"        # implemented: 1F1B/GPipe stages (pp.py);
                      # degree == world -> pure PP, degree < world -> PP x DP
-    SP = "sp"        # reserved: sequence/context parallel (ring attention)
+    SP = "sp"        # implemented: ring attention context parallel
+                     # (ring.py); "ring"/"cp" accepted as aliases
     EP = "ep"        # implemented: expert parallel for MoE configs (ep.py);
                      # gather-compute-scatter, pure form in v1
     ULYSSES = "ulysses"  # implemented: attention head-scatter SP (sp.py);
@@ -37,7 +37,9 @@ class Strategy(str, Enum):
 
 
 IMPLEMENTED = {Strategy.DDP, Strategy.TP, Strategy.PP, Strategy.ULYSSES,
-               Strategy.EP}
+               Strategy.EP, Strategy.SP}
+
+_ALIASES = {"ring": "sp", "cp": "sp"}
 
 
 @dataclass
@@ -51,7 +53,8 @@ class ParallelismSpec:
         if isinstance(p, str):
             p = {"strategy": p}
         try:
-            strategy = Strategy(p.get("strategy", "ddp"))
+            raw = p.get("strategy", "ddp")
+            strategy = Strategy(_ALIASES.get(raw, raw))
         except ValueError:
             raise ValueError(
                 f"unknown parallelism strategy {p.get('strategy')!r}; "
@@ -61,8 +64,7 @@ class ParallelismSpec:
         return spec
 
     def validate(self):
-        if self.strategy not in IMPLEMENTED:
+        if self.strategy not in IMPLEMENTED:  # future additions fail loudly
             raise NotImplementedError(
                 f"parallelism strategy {self.strategy.value!r} is reserved "
-                "but not implemented (it needs a ring-attention kernel) — "
-                f"implemented: {sorted(s.value for s in IMPLEMENTED)}")
+                f"— implemented: {sorted(s.value for s in IMPLEMENTED)}")

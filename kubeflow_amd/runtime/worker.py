@@ -225,6 +225,7 @@ def main(argv=None):
         pp_ctx = None
         sp_ctx = None
         ep_ctx = None
+        cp_ctx = None
         dp_group = None
         data_rank = rank
         if pspec.strategy == Strategy.TP and world > 1:
@@ -247,6 +248,19 @@ def main(argv=None):
                     f"pure EP requires degree == world_size ({world}); "
                     f"got {pspec.degree}")
             ep_ctx = EpContext.from_group(None)
+        elif pspec.strategy == Strategy.SP and world > 1:
+            from kubeflow_amd.parallel.ring import RingContext
+            deg = pspec.degree if pspec.degree > 1 else world
+            if world % deg:
+                raise ValueError(f"world_size {world} not divisible by "
+                                 f"ring degree {deg}")
+            if deg == world:
+                cp_ctx = RingContext.from_group(None)
+                data_rank = 0
+            else:  # ring x DP mesh: contiguous ring groups, strided DP
+                cp_group, _dp_g, _cr, dp_rank = kdist.build_mesh(deg)
+                cp_ctx = RingContext.from_group(cp_group)
+                data_rank = dp_rank
         elif pspec.strategy == Strategy.ULYSSES and world > 1:
             from kubeflow_amd.parallel.sp import SpContext
             deg = pspec.degree if pspec.degree > 1 else world
@@ -287,7 +301,7 @@ def main(argv=None):
                                pp_ctx.world, device=device, dtype=dtype)
         else:
             model = build_model(spec["model"], device=device, dtype=dtype,
-                                tp=tp_ctx, sp=sp_ctx, ep=ep_ctx)
+                                tp=tp_ctx, sp=sp_ctx, ep=ep_ctx, cp=cp_ctx)
         if tp_ctx is not None:
             tp_ctx.sync_replicated(model)
         if ep_ctx is not None:
@@ -335,17 +349,18 @@ def main(argv=None):
             # model-parallel peers form one data replica: data_rank is 0
             # (pure) or the dp index (TP/PP x DP meshes)
             x, y = synthetic_batch(spec, cfg, device, data_rank, step)
-            if sp_ctx is not None:  # ulysses: take this rank's seq shard
-                if x.shape[1] % sp_ctx.world:
+            shard = sp_ctx or cp_ctx
+            if shard is not None:  # ulysses/ring: this rank's seq chunk
+                if x.shape[1] % shard.world:
                     raise ValueError(f"seq_len {x.shape[1]} not divisible "
-                                     f"by ulysses degree {sp_ctx.world}")
-                s = x.shape[1] // sp_ctx.world
-                x = x[:, sp_ctx.rank * s:(sp_ctx.rank + 1) * s].contiguous()
-                y = y[:, sp_ctx.rank * s:(sp_ctx.rank + 1) * s].contiguous()
+                                     f"by sp/ring degree {shard.world}")
+                s = x.shape[1] // shard.world
+                x = x[:, shard.rank * s:(shard.rank + 1) * s].contiguous()
+                y = y[:, shard.rank * s:(shard.rank + 1) * s].contiguous()
             loss = trainer.step(x, y)
             win_steps += 1
             win_items += x.numel()  # tokens (lm) or features processed
-            if sp_ctx is not None:  # status shows the global-mean loss
+            if (sp_ctx or cp_ctx) is not None:  # global-mean loss
                 import torch.distributed as tdist
                 lt = loss.to(torch.float32).clone()
                 tdist.all_reduce(lt)
