@@ -111,9 +111,11 @@ def test_parallelism_strategy_seam():
     ep = ParallelismSpec.from_spec({"parallelism": {"strategy": "ep",
                                                     "degree": 2}})
     assert ep.strategy == Strategy.EP
-    with pytest.raises(NotImplementedError):  # ring-SP still reserved
-        ParallelismSpec.from_spec({"parallelism": {"strategy": "sp",
-                                                   "degree": 2}})
+    # ring-SP graduated too (round 2); aliases resolve to Strategy.SP
+    for alias in ("sp", "ring", "cp"):
+        rg = ParallelismSpec.from_spec({"parallelism": {"strategy": alias,
+                                                        "degree": 2}})
+        assert rg.strategy == Strategy.SP and rg.degree == 2
     with pytest.raises(ValueError):
         ParallelismSpec.from_spec({"parallelism": {"strategy": "magic"}})
 
