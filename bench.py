@@ -66,7 +66,10 @@ def main():
 
     B, S = args.micro_batch, args.seq_len
     tokens = torch.randint(0, cfg.vocab_size, (B, S), device=device)
-    targets = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+    if hasattr(cfg, "n_classes"):  # bert-family classify head
+        targets = torch.randint(0, cfg.n_classes, (B,), device=device)
+    else:
+        targets = torch.randint(0, cfg.vocab_size, (B, S), device=device)
 
     t_start = time.time()
     for _ in range(args.warmup):
