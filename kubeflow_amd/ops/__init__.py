@@ -23,7 +23,7 @@ from . import _backend
 
 __all__ = [
     "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
-    "flash_attention_rect",
+    "flash_attention_rect", "skinny_linear",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
     "swiglu", "fused_qkv_attention",
 ]
@@ -166,6 +166,32 @@ def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
         return torch.cat(qs), torch.cat(ks)
     return (reference.rope_apply(q, cos, sin, pos_offset),
             reference.rope_apply(k, cos, sin, pos_offset))
+
+
+def skinny_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """Decode-batch linear y = x @ W^T for small leading dims (M <= 16):
+    the HIP weight-streaming kernel (skinny_gemm.hip) replaces hipBLASLt's
+    ~30-50%-of-BW GEMV path in the serving decode step. Falls back to
+    F.linear off-GPU or for larger M / unsupported shapes. Inference-only
+    (no autograd)."""
+    shape = x.shape
+    M = 1
+    for d in shape[:-1]:
+        M *= int(d)
+    K = shape[-1]
+    N = weight.shape[0]
+    if (not _use_native(x) or M > 16 or K % 32 or N % 16
+            or x.dtype != torch.bfloat16):
+        return torch.nn.functional.linear(x, weight)
+    lib = _backend.require()
+    x2 = x.reshape(M, K)
+    if not x2.is_contiguous():
+        x2 = x2.contiguous()
+    y = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    _backend.check(
+        lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), M, N, K, 0, 0, 0,
+                           _stream()), "skinny_gemm")
+    return y.view(*shape[:-1], N)
 
 
 # ------------------------------------------------------- Flash attention --

@@ -1,0 +1,80 @@
+// skinny_gemm.hip — decode-batch linear: C[M,N] = A[M,K] @ W[N,K]^T for
+// M <= 16 (the serving decode step's GEMV-shaped GEMMs).
+//
+// hipBLASLt runs these weight-streaming shapes at ~30-50% of HBM BW at
+// M=16 (profiles/r02: decode-step GEMMs dominate the 6.3 ms step whose
+// weight-read floor is ~2.5 ms). This kernel streams W exactly once at
+// full coalescing: block = 4 waves SHARING one 16-column N-tile with an
+// in-block K-split, each wave issuing mfma_f32_16x16x32_bf16 over its K
+// range (A operand rows = the M batch rows, zero-padded to 16), then a
+// 4-way LDS reduction. Lanes {l, l+16, l+32, l+48} read consecutive
+// 8-element chunks of the same W row, so each wave instruction covers
+// contiguous 64 B per row — full line utilization while W streams.
+//
+// Grid = N/16 blocks; K % 32 == 0; bf16 in/out, fp32 accumulate.
+
+#include "kf_common.h"
+
+typedef __bf16 kf_bf16x8s __attribute__((ext_vector_type(8)));
+typedef float kf_f32x4s __attribute__((ext_vector_type(4)));
+
+#define SK_NT 16      // N columns per block
+#define SK_WAVES 4    // K-split ways
+
+__global__ __launch_bounds__(SK_WAVES * 64, 4) void kf_skinny_gemm_kernel(
+    unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
+    const unsigned short* __restrict__ w, int M, int64_t N, int64_t K,
+    int64_t lda, int64_t ldw, int64_t ldc) {
+  __shared__ float red[SK_WAVES][SK_NT][SK_NT];  // per-wave C tiles
+
+  const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
+  const int tid = threadIdx.x;
+  const int wv = tid / KF_WAVE;
+  const int lane = tid & (KF_WAVE - 1);
+  const int l15 = lane & 15;
+  const int hi4 = lane >> 4;  // 0..3: k-subchunk within the 32-k step
+
+  // this wave's K range (each wave strides by SK_WAVES*32 for coalesced
+  // row-chunk progression shared with its sibling lanes)
+  kf_f32x4s acc = kf_f32x4s{0.f, 0.f, 0.f, 0.f};
+  const unsigned short* wrow = w + (n0 + l15) * ldw;
+  const bool arow_ok = l15 < M;
+  const unsigned short* arow = a + (arow_ok ? l15 : 0) * lda;
+  const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
+  for (int64_t k = (int64_t)wv * 32; k < K; k += SK_WAVES * 32) {
+    kf_bf16x8s af = arow_ok
+        ? *reinterpret_cast<const kf_bf16x8s*>(arow + k + hi4 * 8)
+        : zero8;
+    kf_bf16x8s wf =
+        *reinterpret_cast<const kf_bf16x8s*>(wrow + k + hi4 * 8);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, wf, acc, 0, 0, 0);
+  }
+  // C/D layout (guide §3, m89): col = lane&15, row = (lane>>4)*4 + j
+#pragma unroll
+  for (int j = 0; j < 4; ++j) red[wv][hi4 * 4 + j][l15] = acc[j];
+  __syncthreads();
+  // waves 0..3 reduce + write: thread (row, col) pairs
+  if (tid < SK_NT * SK_NT) {
+    const int row = tid / SK_NT, col = tid % SK_NT;
+    if (row < M && n0 + col < N) {
+      float s = red[0][row][col] + red[1][row][col] + red[2][row][col] +
+                red[3][row][col];
+      c[row * ldc + n0 + col] = kf_f32_to_bf16(s);
+    }
+  }
+}
+
+KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
+                             int64_t M, int64_t N, int64_t K, int64_t lda,
+                             int64_t ldw, int64_t ldc, void* stream) {
+  if (M < 1 || M > 16 || K % 32 || N % SK_NT) return (int)hipErrorInvalidValue;
+  if (lda == 0) lda = K;
+  if (ldw == 0) ldw = K;
+  if (ldc == 0) ldc = N;
+  dim3 grid((unsigned)(N / SK_NT), 1, 1);
+  hipLaunchKernelGGL(kf_skinny_gemm_kernel, grid, dim3(SK_WAVES * 64), 0,
+                     (hipStream_t)stream, (unsigned short*)c,
+                     (const unsigned short*)a, (const unsigned short*)w,
+                     (int)M, N, K, lda, ldw, ldc);
+  return (int)hipGetLastError();
+}

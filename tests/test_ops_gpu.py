@@ -412,3 +412,20 @@ def test_attention_rect_gpu_parity():
                    causal=True).transpose(1, 2)
     err2 = ((o2.cpu().float() - want2).norm() / want2.norm()).item()
     assert err2 < 2e-2, err2
+
+
+@pytest.mark.gpu
+def test_skinny_gemm_parity():
+    """Decode-batch linear kernel vs F.linear at the serving shapes."""
+    torch.manual_seed(2)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    for M, N, K in ((16, 6144, 4096), (1, 4096, 4096), (5, 28672, 4096),
+                    (16, 4096, 14336), (16, 128256, 4096)):
+        x = torch.randn(M, 1, K, device=dev, dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
+        got = ops.skinny_linear(x, w)
+        want = torch.nn.functional.linear(x, w)
+        err = ((got.float() - want.float()).norm() /
+               want.float().norm().clamp(min=1e-6)).item()
+        assert err < 2e-2, (M, N, K, err)
