@@ -81,6 +81,10 @@ def _configure(lib: ctypes.CDLL) -> None:
     lib.kf_ce_bwd.argtypes = [P, P, FP, PI64, FP, I64, I64, I64, P]
     lib.kf_attn_fwd.restype = I32
     lib.kf_attn_fwd.argtypes = [P, FP, P, P, P, I64, I64, I64, I64, I64, I64, I64, F, I32, P]
+    if hasattr(lib, "kf_skinny_gemm"):
+        lib.kf_skinny_gemm.restype = I32
+        lib.kf_skinny_gemm.argtypes = [P, P, P, I64, I64, I64, I64, I64,
+                                       I64, P]
     if hasattr(lib, "kf_attn_fwd4_rect"):
         lib.kf_attn_fwd4_rect.restype = I32
         lib.kf_attn_fwd4_rect.argtypes = [P, FP, P, P, P, I64, I64, I64,
