@@ -41,7 +41,26 @@ __global__ __launch_bounds__(SK_WAVES * 64, 4) void kf_skinny_gemm_kernel(
   const bool arow_ok = l15 < M;
   const unsigned short* arow = a + (arow_ok ? l15 : 0) * lda;
   const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
-  for (int64_t k = (int64_t)wv * 32; k < K; k += SK_WAVES * 32) {
+  // 8-deep unrolled stream: ~256 B of W per wave in flight so the HBM
+  // latency pipelines (a single outstanding load-pair left the wave
+  // latency-bound at ~45% of blaslt on the big shapes)
+  const int64_t step = (int64_t)SK_WAVES * 32;
+  int64_t k = (int64_t)wv * 32;
+  for (; k + 7 * step + 32 <= K; k += 8 * step) {
+    kf_bf16x8s afs[8], wfs[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int64_t ku = k + u * step + hi4 * 8;
+      afs[u] = arow_ok
+          ? *reinterpret_cast<const kf_bf16x8s*>(arow + ku) : zero8;
+      wfs[u] = *reinterpret_cast<const kf_bf16x8s*>(wrow + ku);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afs[u], wfs[u], acc,
+                                                    0, 0, 0);
+  }
+  for (; k < K; k += step) {
     kf_bf16x8s af = arow_ok
         ? *reinterpret_cast<const kf_bf16x8s*>(arow + k + hi4 * 8)
         : zero8;
