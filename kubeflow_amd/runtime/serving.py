@@ -348,10 +348,17 @@ class InferenceEngine:
         cfg = self.model.cfg
         import torch.nn.functional as F
         N = tokens.shape[0]
+        # decode GEMMs: KF_SKINNY routes the GEMV-shaped linears through
+        # the weight-streaming kernel (ops.skinny_linear) — "small" (the
+        # measured wins: qkv/wo), "all", or "off" (hipBLASLt everywhere)
+        mode = os.environ.get("KF_SKINNY", "small")
+        lin_small = (ops.skinny_linear if mode in ("small", "all")
+                     else F.linear)
+        lin_big = ops.skinny_linear if mode == "all" else F.linear
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
-            qkv = F.linear(layer.attn_norm(x), layer.wqkv.weight)
+            qkv = lin_small(layer.attn_norm(x), layer.wqkv.weight)
             q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
                                  cfg.n_kv_heads * cfg.head_dim,
                                  cfg.n_kv_heads * cfg.head_dim], dim=-1)
@@ -363,15 +370,16 @@ class InferenceEngine:
             self.cache.v[li][slots.long(), positions] = v[:, 0]
             o = ops.attention_decode(q[:, 0], self.cache.k[li],
                                      self.cache.v[li], slots, lens)
-            o = layer.wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim))
+            o = lin_small(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
+                          layer.wo.weight)
             x = x + o
             if layer.moe is not None:
                 x = x + layer.moe(layer.mlp_norm(x))
             else:
-                y = ops.swiglu(F.linear(layer.mlp_norm(x), layer.w13.weight))
-                x = x + layer.w2(y)
+                y = ops.swiglu(lin_big(layer.mlp_norm(x), layer.w13.weight))
+                x = x + lin_small(y, layer.w2.weight)
         x = self.model.final_norm(x)
-        logits = F.linear(x, self.model.lm_head.weight)  # [N,1,V]
+        logits = lin_big(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)
 
     def _graph_for(self, bucket: int):
