@@ -350,7 +350,10 @@ class InferenceEngine:
         N = tokens.shape[0]
         # decode GEMMs: KF_SKINNY routes the GEMV-shaped linears through
         # the weight-streaming kernel (ops.skinny_linear) — "small" (the
-        # measured wins: qkv/wo), "all", or "off" (hipBLASLt everywhere)
+        # shapes it wins at M<=16: qkv N6144/K4096, wo N4096/K4096 —
+        # profiles/r02_skinny_gemm.md), "all", or "off" (hipBLASLt
+        # everywhere). The K14336 / N28672 / lm_head shapes stay on
+        # hipBLASLt under "small": measured losses there.
         mode = os.environ.get("KF_SKINNY", "small")
         lin_small = (ops.skinny_linear if mode in ("small", "all")
                      else F.linear)
@@ -377,7 +380,7 @@ class InferenceEngine:
                 x = x + layer.moe(layer.mlp_norm(x))
             else:
                 y = ops.swiglu(lin_big(layer.mlp_norm(x), layer.w13.weight))
-                x = x + lin_small(y, layer.w2.weight)
+                x = x + lin_big(y, layer.w2.weight)
         x = self.model.final_norm(x)
         logits = lin_big(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)
