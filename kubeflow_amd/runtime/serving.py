@@ -355,13 +355,15 @@ class InferenceEngine:
         # everywhere). The K14336 / N28672 / lm_head shapes stay on
         # hipBLASLt under "small": measured losses there.
         mode = os.environ.get("KF_SKINNY", "small")
-        lin_small = (ops.skinny_linear if mode in ("small", "all")
-                     else F.linear)
+        lin_qkv = (ops.skinny_linear if mode in ("small", "all", "qkv")
+                   else F.linear)
+        lin_wo = (ops.skinny_linear if mode in ("small", "all", "wo")
+                  else F.linear)
         lin_big = ops.skinny_linear if mode == "all" else F.linear
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
-            qkv = lin_small(layer.attn_norm(x), layer.wqkv.weight)
+            qkv = lin_qkv(layer.attn_norm(x), layer.wqkv.weight)
             q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
                                  cfg.n_kv_heads * cfg.head_dim,
                                  cfg.n_kv_heads * cfg.head_dim], dim=-1)
@@ -373,8 +375,8 @@ class InferenceEngine:
             self.cache.v[li][slots.long(), positions] = v[:, 0]
             o = ops.attention_decode(q[:, 0], self.cache.k[li],
                                      self.cache.v[li], slots, lens)
-            o = lin_small(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
-                          layer.wo.weight)
+            o = lin_wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
+                       layer.wo.weight)
             x = x + o
             if layer.moe is not None:
                 x = x + layer.moe(layer.mlp_norm(x))
