@@ -23,7 +23,7 @@ from . import _backend
 
 __all__ = [
     "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
-    "flash_attention_rect", "skinny_linear",
+    "flash_attention_rect", "skinny_linear", "kv_store",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
     "swiglu", "fused_qkv_attention",
 ]
@@ -166,6 +166,32 @@ def rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
         return torch.cat(qs), torch.cat(ks)
     return (reference.rope_apply(q, cos, sin, pos_offset),
             reference.rope_apply(k, cos, sin, pos_offset))
+
+
+def kv_store(cache_k: torch.Tensor, cache_v: torch.Tensor,
+             k: torch.Tensor, v: torch.Tensor, slots: torch.Tensor,
+             positions: torch.Tensor) -> None:
+    """Scatter this decode step's k/v token rows into cache row
+    [slot[i], position[i]] for each sequence i — one launch instead of two
+    advanced-indexing kernels per layer (kv_store.hip). cache_k/v
+    [SLOTS,SMAX,Hkv,D]; k/v [N,1,Hkv,D] (or [N,Hkv,D]); slots int32 [N];
+    positions int64 [N]. Capture-safe; no autograd (serving only)."""
+    n = k.shape[0]
+    row = cache_k.shape[2] * cache_k.shape[3]
+    if _use_native(k) and k.dtype == torch.bfloat16 and row % 8 == 0:
+        lib = _backend.require()
+        kk, vv = k.reshape(n, row), v.reshape(n, row)
+        if not kk.is_contiguous():
+            kk = kk.contiguous()
+        if not vv.is_contiguous():
+            vv = vv.contiguous()
+        _backend.check(
+            lib.kf_kv_store(_p(cache_k), _p(cache_v), _p(kk), _p(vv),
+                            _i32p(slots), _p(positions), cache_k.shape[1],
+                            row, n, _stream()), "kv_store")
+        return
+    cache_k[slots.long(), positions] = k.reshape(n, *cache_k.shape[2:])
+    cache_v[slots.long(), positions] = v.reshape(n, *cache_v.shape[2:])
 
 
 def skinny_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:

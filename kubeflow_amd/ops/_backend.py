@@ -85,6 +85,9 @@ def _configure(lib: ctypes.CDLL) -> None:
         lib.kf_skinny_gemm.restype = I32
         lib.kf_skinny_gemm.argtypes = [P, P, P, I64, I64, I64, I64, I64,
                                        I64, P]
+    if hasattr(lib, "kf_kv_store"):
+        lib.kf_kv_store.restype = I32
+        lib.kf_kv_store.argtypes = [P, P, P, P, P, P, I64, I64, I64, P]
     if hasattr(lib, "kf_attn_fwd4_rect"):
         lib.kf_attn_fwd4_rect.restype = I32
         lib.kf_attn_fwd4_rect.argtypes = [P, FP, P, P, P, I64, I64, I64,

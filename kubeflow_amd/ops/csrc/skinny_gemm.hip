@@ -19,13 +19,17 @@ typedef __bf16 kf_bf16x8s __attribute__((ext_vector_type(8)));
 typedef float kf_f32x4s __attribute__((ext_vector_type(4)));
 
 #define SK_NT 16      // N columns per block
-#define SK_WAVES 4    // K-split ways
+// K-split ways (waves/block) is a template knob: small-N shapes (wo:
+// N=4096 -> 256 blocks == 1 block/CU == 1 wave/SIMD at 4 waves) are
+// occupancy-starved and want 8; profiled per shape in
+// profiles/r02_skinny_gemm.md.
 
-__global__ __launch_bounds__(SK_WAVES * 64, 4) void kf_skinny_gemm_kernel(
+template <int SKW>
+__global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned short* __restrict__ w, int M, int64_t N, int64_t K,
     int64_t lda, int64_t ldw, int64_t ldc) {
-  __shared__ float red[SK_WAVES][SK_NT][SK_NT];  // per-wave C tiles
+  __shared__ float red[SKW][SK_NT][SK_NT];  // per-wave C tiles
 
   const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
   const int tid = threadIdx.x;
@@ -44,7 +48,7 @@ __global__ __launch_bounds__(SK_WAVES * 64, 4) void kf_skinny_gemm_kernel(
   // 8-deep unrolled stream: ~256 B of W per wave in flight so the HBM
   // latency pipelines (a single outstanding load-pair left the wave
   // latency-bound at ~45% of blaslt on the big shapes)
-  const int64_t step = (int64_t)SK_WAVES * 32;
+  const int64_t step = (int64_t)SKW * 32;
   int64_t k = (int64_t)wv * 32;
   for (; k + 7 * step + 32 <= K; k += 8 * step) {
     kf_bf16x8s afs[8], wfs[8];
@@ -72,12 +76,13 @@ __global__ __launch_bounds__(SK_WAVES * 64, 4) void kf_skinny_gemm_kernel(
 #pragma unroll
   for (int j = 0; j < 4; ++j) red[wv][hi4 * 4 + j][l15] = acc[j];
   __syncthreads();
-  // waves 0..3 reduce + write: thread (row, col) pairs
+  // first waves reduce + write: thread (row, col) pairs
   if (tid < SK_NT * SK_NT) {
     const int row = tid / SK_NT, col = tid % SK_NT;
     if (row < M && n0 + col < N) {
-      float s = red[0][row][col] + red[1][row][col] + red[2][row][col] +
-                red[3][row][col];
+      float s = 0.f;
+#pragma unroll
+      for (int ww = 0; ww < SKW; ++ww) s += red[ww][row][col];
       c[row * ldc + n0 + col] = kf_f32_to_bf16(s);
     }
   }
@@ -91,9 +96,18 @@ KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
   if (ldw == 0) ldw = K;
   if (ldc == 0) ldc = N;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
-  hipLaunchKernelGGL(kf_skinny_gemm_kernel, grid, dim3(SK_WAVES * 64), 0,
-                     (hipStream_t)stream, (unsigned short*)c,
-                     (const unsigned short*)a, (const unsigned short*)w,
-                     (int)M, N, K, lda, ldw, ldc);
+  // 8 waves when the grid can't fill the chip with 4-wave blocks
+  // (<2 blocks/CU), 4 otherwise — measured matrix in
+  // profiles/r02_skinny_gemm.md.
+  if (N / SK_NT < 512)
+    hipLaunchKernelGGL(kf_skinny_gemm_kernel<8>, grid, dim3(8 * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)c,
+                       (const unsigned short*)a, (const unsigned short*)w,
+                       (int)M, N, K, lda, ldw, ldc);
+  else
+    hipLaunchKernelGGL(kf_skinny_gemm_kernel<4>, grid, dim3(4 * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)c,
+                       (const unsigned short*)a, (const unsigned short*)w,
+                       (int)M, N, K, lda, ldw, ldc);
   return (int)hipGetLastError();
 }

@@ -355,11 +355,13 @@ class InferenceEngine:
         # everywhere). The K14336 / N28672 / lm_head shapes stay on
         # hipBLASLt under "small": measured losses there.
         mode = os.environ.get("KF_SKINNY", "small")
-        lin_qkv = (ops.skinny_linear if mode in ("small", "all", "qkv")
-                   else F.linear)
-        lin_wo = (ops.skinny_linear if mode in ("small", "all", "wo")
-                  else F.linear)
-        lin_big = ops.skinny_linear if mode == "all" else F.linear
+        sel = {"small": {"qkv", "wo"}, "off": set(),
+               "all": {"qkv", "wo", "w13", "w2", "lm"}}.get(
+                   mode, set(mode.split(",")))
+        def _lin(name):
+            return ops.skinny_linear if name in sel else F.linear
+        lin_qkv, lin_wo = _lin("qkv"), _lin("wo")
+        lin_w13, lin_w2, lin_lm = _lin("w13"), _lin("w2"), _lin("lm")
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
@@ -371,8 +373,8 @@ class InferenceEngine:
             k = k.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
             v = v.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
             q, k = ops.rope(q, k, cos, sin, positions=positions)
-            self.cache.k[li][slots.long(), positions] = k[:, 0]
-            self.cache.v[li][slots.long(), positions] = v[:, 0]
+            ops.kv_store(self.cache.k[li], self.cache.v[li], k, v,
+                         slots, positions)
             o = ops.attention_decode(q[:, 0], self.cache.k[li],
                                      self.cache.v[li], slots, lens)
             o = lin_wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
@@ -381,10 +383,11 @@ class InferenceEngine:
             if layer.moe is not None:
                 x = x + layer.moe(layer.mlp_norm(x))
             else:
-                y = ops.swiglu(lin_big(layer.mlp_norm(x), layer.w13.weight))
-                x = x + lin_big(y, layer.w2.weight)
+                y = ops.swiglu(lin_w13(layer.mlp_norm(x),
+                                       layer.w13.weight))
+                x = x + lin_w2(y, layer.w2.weight)
         x = self.model.final_norm(x)
-        logits = lin_big(x, self.model.lm_head.weight)  # [N,1,V]
+        logits = lin_lm(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)
 
     def _graph_for(self, bucket: int):

@@ -429,3 +429,25 @@ def test_skinny_gemm_parity():
         err = ((got.float() - want.float()).norm() /
                want.float().norm().clamp(min=1e-6)).item()
         assert err < 2e-2, (M, N, K, err)
+
+
+@pytest.mark.gpu
+def test_kv_store_parity():
+    """Fused decode cache scatter vs advanced indexing."""
+    torch.manual_seed(3)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    SLOTS, SMAX, Hkv, D, N = 8, 64, 8, 128, 5
+    ck = torch.randn(SLOTS, SMAX, Hkv, D, device=dev, dtype=torch.bfloat16)
+    cv = torch.randn_like(ck)
+    ck2, cv2 = ck.clone(), cv.clone()
+    k = torch.randn(N, 1, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn_like(k)
+    slots = torch.tensor([7, 0, 3, 3, 5], dtype=torch.int32, device=dev)
+    positions = torch.tensor([0, 63, 10, 11, 32], dtype=torch.int64,
+                             device=dev)
+    ops.kv_store(ck, cv, k, v, slots, positions)
+    ck2[slots.long(), positions] = k[:, 0]
+    cv2[slots.long(), positions] = v[:, 0]
+    assert torch.equal(ck, ck2)
+    assert torch.equal(cv, cv2)
