@@ -123,8 +123,11 @@ class _RopeHip(torch.autograd.Function):
         lib = _backend.require()
         B, S, Hq, D = q.shape
         Hkv = k.shape[2]
-        q = q.contiguous().clone(memory_format=torch.contiguous_format)
-        k = k.contiguous().clone(memory_format=torch.contiguous_format)
+        # clone(contiguous_format) compacts strided inputs in ONE copy
+        # (the QKV-split views are always strided; .contiguous().clone()
+        # would copy twice)
+        q = q.clone(memory_format=torch.contiguous_format)
+        k = k.clone(memory_format=torch.contiguous_format)
         _backend.check(
             lib.kf_rope(_p(q), _p(k), _fp(cos), _fp(sin),
                         _ip_or_null(positions), B, S, Hq, Hkv, D,
@@ -141,8 +144,8 @@ class _RopeHip(torch.autograd.Function):
         cos, sin = ctx.saved_tensors[0], ctx.saved_tensors[1]
         positions = ctx.saved_tensors[2] if len(ctx.saved_tensors) > 2 else None
         B, S, Hq, Hkv, D, pos_offset = ctx.dims
-        dq = dq.contiguous().clone(memory_format=torch.contiguous_format)
-        dk = dk.contiguous().clone(memory_format=torch.contiguous_format)
+        dq = dq.clone(memory_format=torch.contiguous_format)
+        dk = dk.clone(memory_format=torch.contiguous_format)
         _backend.check(
             lib.kf_rope(_p(dq), _p(dk), _fp(cos), _fp(sin),
                         _ip_or_null(positions), B, S, Hq, Hkv, D,

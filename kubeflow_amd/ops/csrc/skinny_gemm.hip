@@ -88,6 +88,94 @@ __global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------
+// LDS-staged variant. The direct kernel's W loads put 16 rows x 64 B per
+// instruction on the wire — half of every 128 B line — capping it at
+// ~3 TB/s. Here the block's 8 waves stage W chunks cooperatively with
+// FULL-ROW 1 KB runs (wave wv loads rows {2wv, 2wv+1}, one b128 per lane
+// per row), then read their MFMA fragments from LDS. Padded row stride
+// (520 elems) keeps both LDS sides at worst 2-way bank-conflicted.
+// Double-buffered: one __syncthreads per 512-k chunk; the next chunk's
+// global loads issue before the barrier so HBM latency overlaps MFMA.
+
+#define SKL_W 8
+#define SKL_KC 512                      // staged K elems (1 KB per row)
+#define SKL_STRIDE (SKL_KC + 8)        // +8 elems: 2-way max conflicts
+
+__global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
+    unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
+    const unsigned short* __restrict__ w, int M, int64_t N, int64_t K,
+    int64_t lda, int64_t ldw, int64_t ldc) {
+  __shared__ unsigned short wbuf[2][SK_NT][SKL_STRIDE];
+  __shared__ float red[SKL_W][SK_NT][SK_NT];
+
+  const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
+  const int tid = threadIdx.x;
+  const int wv = tid / KF_WAVE;
+  const int lane = tid & (KF_WAVE - 1);
+  const int l15 = lane & 15;
+  const int hi4 = lane >> 4;
+
+  kf_f32x4s acc = kf_f32x4s{0.f, 0.f, 0.f, 0.f};
+  const bool arow_ok = l15 < M;
+  const unsigned short* arow = a + (arow_ok ? l15 : 0) * lda;
+  const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
+  // writer: wave wv stages rows {2wv, 2wv+1}, lane covers elems
+  // [lane*8, lane*8+8) of each 512-elem row slice
+  const unsigned short* wr0 = w + (n0 + 2 * wv) * ldw + lane * 8;
+  const unsigned short* wr1 = wr0 + ldw;
+
+  const int64_t nch = K / SKL_KC;
+  const int ke0 = wv * 64 + hi4 * 8;  // this wave's k slice (s=0; s=1 at +32)
+  kf_bf16x8s st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0);
+  kf_bf16x8s st1 = *reinterpret_cast<const kf_bf16x8s*>(wr1);
+  kf_bf16x8s af0 = arow_ok
+      ? *reinterpret_cast<const kf_bf16x8s*>(arow + ke0) : zero8;
+  kf_bf16x8s af1 = arow_ok
+      ? *reinterpret_cast<const kf_bf16x8s*>(arow + ke0 + 32) : zero8;
+  *reinterpret_cast<kf_bf16x8s*>(&wbuf[0][2 * wv][lane * 8]) = st0;
+  *reinterpret_cast<kf_bf16x8s*>(&wbuf[0][2 * wv + 1][lane * 8]) = st1;
+  for (int64_t ch = 0; ch < nch; ++ch) {
+    kf_bf16x8s a0 = af0, a1 = af1;
+    if (ch + 1 < nch) {
+      st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0 + (ch + 1) * SKL_KC);
+      st1 = *reinterpret_cast<const kf_bf16x8s*>(wr1 + (ch + 1) * SKL_KC);
+      if (arow_ok) {
+        af0 = *reinterpret_cast<const kf_bf16x8s*>(
+            arow + (ch + 1) * SKL_KC + ke0);
+        af1 = *reinterpret_cast<const kf_bf16x8s*>(
+            arow + (ch + 1) * SKL_KC + ke0 + 32);
+      }
+    }
+    // one barrier per chunk: makes buffer ch&1's writes visible AND
+    // guarantees last iteration's readers of buffer (ch+1)&1 are done
+    __syncthreads();
+    kf_bf16x8s wf0 =
+        *reinterpret_cast<const kf_bf16x8s*>(&wbuf[ch & 1][l15][ke0]);
+    kf_bf16x8s wf1 =
+        *reinterpret_cast<const kf_bf16x8s*>(&wbuf[ch & 1][l15][ke0 + 32]);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, wf0, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, wf1, acc, 0, 0, 0);
+    if (ch + 1 < nch) {
+      const int b = (int)((ch + 1) & 1);
+      *reinterpret_cast<kf_bf16x8s*>(&wbuf[b][2 * wv][lane * 8]) = st0;
+      *reinterpret_cast<kf_bf16x8s*>(&wbuf[b][2 * wv + 1][lane * 8]) = st1;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) red[wv][hi4 * 4 + j][l15] = acc[j];
+  __syncthreads();
+  if (tid < SK_NT * SK_NT) {
+    const int row = tid / SK_NT, col = tid % SK_NT;
+    if (row < M && n0 + col < N) {
+      float s = 0.f;
+#pragma unroll
+      for (int ww = 0; ww < SKL_W; ++ww) s += red[ww][row][col];
+      c[row * ldc + n0 + col] = kf_f32_to_bf16(s);
+    }
+  }
+}
+
 KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
                              int64_t M, int64_t N, int64_t K, int64_t lda,
                              int64_t ldw, int64_t ldc, void* stream) {
@@ -96,10 +184,14 @@ KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
   if (ldw == 0) ldw = K;
   if (ldc == 0) ldc = N;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
-  // 8 waves when the grid can't fill the chip with 4-wave blocks
-  // (<2 blocks/CU), 4 otherwise — measured matrix in
-  // profiles/r02_skinny_gemm.md.
-  if (N / SK_NT < 512)
+  if (K % SKL_KC == 0 && ldw % 8 == 0 && lda % 8 == 0)
+    hipLaunchKernelGGL(kf_skinny_lds_kernel, grid, dim3(SKL_W * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)c,
+                       (const unsigned short*)a, (const unsigned short*)w,
+                       (int)M, N, K, lda, ldw, ldc);
+  // direct-load fallback: 8 waves when the grid can't fill the chip with
+  // 4-wave blocks (<2 blocks/CU), 4 otherwise
+  else if (N / SK_NT < 512)
     hipLaunchKernelGGL(kf_skinny_gemm_kernel<8>, grid, dim3(8 * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
