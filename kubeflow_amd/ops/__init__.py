@@ -24,6 +24,7 @@ from . import _backend
 __all__ = [
     "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
     "flash_attention_rect", "skinny_linear", "kv_store",
+    "decode_rope_store",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
     "swiglu", "fused_qkv_attention",
 ]
@@ -197,12 +198,14 @@ def kv_store(cache_k: torch.Tensor, cache_v: torch.Tensor,
     cache_v[slots.long(), positions] = v.reshape(n, *cache_v.shape[2:])
 
 
-def skinny_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
-    """Decode-batch linear y = x @ W^T for small leading dims (M <= 16):
-    the HIP weight-streaming kernel (skinny_gemm.hip) replaces hipBLASLt's
-    ~30-50%-of-BW GEMV path in the serving decode step. Falls back to
-    F.linear off-GPU or for larger M / unsupported shapes. Inference-only
-    (no autograd)."""
+def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
+                  residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Decode-batch linear y = x @ W^T (+ residual) for small leading dims
+    (M <= 16): the LDS-staged weight-streaming kernel (skinny_gemm.hip)
+    replaces hipBLASLt's ~30-50%-of-BW GEMV path in the serving decode
+    step; `residual` fuses the following elementwise add into the
+    epilogue. Falls back to F.linear off-GPU or for larger M /
+    unsupported shapes. Inference-only (no autograd)."""
     shape = x.shape
     M = 1
     for d in shape[:-1]:
@@ -211,16 +214,60 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     N = weight.shape[0]
     if (not _use_native(x) or M > 16 or K % 32 or N % 16
             or x.dtype != torch.bfloat16):
-        return torch.nn.functional.linear(x, weight)
+        y = torch.nn.functional.linear(x, weight)
+        return y + residual if residual is not None else y
     lib = _backend.require()
     x2 = x.reshape(M, K)
     if not x2.is_contiguous():
         x2 = x2.contiguous()
+    r2 = None
+    if residual is not None:
+        r2 = residual.reshape(M, N)
+        if not r2.is_contiguous():
+            r2 = r2.contiguous()
     y = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _backend.check(
-        lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), M, N, K, 0, 0, 0,
-                           _stream()), "skinny_gemm")
+        lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), _p(r2), M, N, K,
+                           0, 0, 0, _stream()), "skinny_gemm")
     return y.view(*shape[:-1], N)
+
+
+def decode_rope_store(qkv: torch.Tensor, cache_k: torch.Tensor,
+                      cache_v: torch.Tensor, cos: torch.Tensor,
+                      sin: torch.Tensor, slots: torch.Tensor,
+                      positions: torch.Tensor, n_heads: int,
+                      n_kv_heads: int) -> torch.Tensor:
+    """Fused decode-step RoPE + cache scatter (decode_fused.hip): consume
+    the fused QKV projection [N, 1, (Hq+2*Hkv)*D] directly — rotate q and
+    k, write q to a fresh [N, Hq, D] tensor, scatter rotated k and copied
+    v into cache row [slot[i], position[i]]. Replaces split + rope clones
+    + kv_store (~4 launches and two D2D copies per layer). Serving only
+    (no autograd); capture-safe."""
+    D = cache_k.shape[3]
+    n = qkv.shape[0]
+    q = torch.empty(n, n_heads, D, dtype=qkv.dtype, device=qkv.device)
+    if (_use_native(qkv) and qkv.dtype == torch.bfloat16 and D == 128):
+        lib = _backend.require()
+        q3 = qkv.reshape(n, -1)
+        if not q3.is_contiguous():
+            q3 = q3.contiguous()
+        _backend.check(
+            lib.kf_decode_rope_store(_p(q), _p(cache_k), _p(cache_v),
+                                     _p(q3), _fp(cos), _fp(sin),
+                                     _i32p(slots), _p(positions), n,
+                                     n_heads, n_kv_heads, D,
+                                     cache_k.shape[1], _stream()),
+            "decode_rope_store")
+        return q
+    # reference path: split + rope + indexed store
+    qq, kk, vv = qkv.reshape(n, 1, -1).split(
+        [n_heads * D, n_kv_heads * D, n_kv_heads * D], dim=-1)
+    qq = qq.view(n, 1, n_heads, D)
+    kk = kk.view(n, 1, n_kv_heads, D)
+    vv = vv.view(n, 1, n_kv_heads, D)
+    qq, kk = rope(qq, kk, cos, sin, positions=positions)
+    kv_store(cache_k, cache_v, kk, vv, slots, positions)
+    return qq.reshape(n, n_heads, D)
 
 
 # ------------------------------------------------------- Flash attention --
@@ -539,10 +586,22 @@ def attention_decode(q: torch.Tensor, kcache: torch.Tensor,
         lib = _backend.require()
         q = q.contiguous()
         out = torch.empty_like(q)
+        # flash-decoding split: N*Hkv blocks alone underfill the 256-CU
+        # chip at serving batch sizes; split the sequence over grid.z and
+        # merge partials (one extra tiny kernel). Static per (N, Hkv) so
+        # hipGraph capture sees fixed shapes.
+        splits = min(8, max(1, 512 // max(1, N * Hkv)))
+        po = pm = None
+        if splits > 1:
+            po = torch.empty(N, Hq, splits, D, dtype=torch.float32,
+                             device=q.device)
+            pm = torch.empty(N, Hq, splits, 2, dtype=torch.float32,
+                             device=q.device)
         _backend.check(
-            lib.kf_attn_decode(_p(out), _p(q), _p(kcache), _p(vcache),
-                               _i32p(slots), _i32p(lens), N, kcache.shape[1],
-                               Hq, Hkv, D, float(scale), _stream()),
+            lib.kf_attn_decode(_p(out), _p(po), _p(pm), _p(q), _p(kcache),
+                               _p(vcache), _i32p(slots), _i32p(lens), N,
+                               kcache.shape[1], Hq, Hkv, D, splits,
+                               float(scale), _stream()),
             "attn_decode")
         return out
     # reference path: per-sequence sdpa over the cached prefix

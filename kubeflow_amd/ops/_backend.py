@@ -83,8 +83,8 @@ def _configure(lib: ctypes.CDLL) -> None:
     lib.kf_attn_fwd.argtypes = [P, FP, P, P, P, I64, I64, I64, I64, I64, I64, I64, F, I32, P]
     if hasattr(lib, "kf_skinny_gemm"):
         lib.kf_skinny_gemm.restype = I32
-        lib.kf_skinny_gemm.argtypes = [P, P, P, I64, I64, I64, I64, I64,
-                                       I64, P]
+        lib.kf_skinny_gemm.argtypes = [P, P, P, P, I64, I64, I64, I64,
+                                       I64, I64, P]
     if hasattr(lib, "kf_kv_store"):
         lib.kf_kv_store.restype = I32
         lib.kf_kv_store.argtypes = [P, P, P, P, P, P, I64, I64, I64, P]
@@ -108,8 +108,12 @@ def _configure(lib: ctypes.CDLL) -> None:
     lib.kf_swiglu_bwd.restype = I32
     lib.kf_swiglu_bwd.argtypes = [P, P, P, I64, I64, P]
     lib.kf_attn_decode.restype = I32
-    lib.kf_attn_decode.argtypes = [P, P, P, P, PI32, PI32, I64, I64, I64, I64,
-                                   I64, F, P]
+    lib.kf_attn_decode.argtypes = [P, P, P, P, P, P, PI32, PI32, I64, I64,
+                                   I64, I64, I64, I64, F, P]
+    if hasattr(lib, "kf_decode_rope_store"):
+        lib.kf_decode_rope_store.restype = I32
+        lib.kf_decode_rope_store.argtypes = [P, P, P, P, FP, FP, PI32, P,
+                                             I64, I64, I64, I64, I64, P]
 
 
 def check(err: int, name: str) -> None:

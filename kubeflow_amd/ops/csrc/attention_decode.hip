@@ -29,8 +29,15 @@ __device__ __forceinline__ int kf_swzd(int row, int byte_in_row) {
   return row * (AD_D * 2) + (byte_in_row ^ ((row & 7) << 4));
 }
 
+// Grid (N, Hkv, SPLITS): flash-decoding sequence split. N*Hkv blocks
+// alone (16x8 = 128 at the serving bucket) fill only half the chip and
+// serialize long contexts; each grid.z block handles a contiguous
+// [lo, hi) slice of the sequence and (for splits > 1) writes an
+// UNNORMALIZED partial (o, m, l) that kf_attn_decode_combine merges
+// with the online-softmax rules.
 __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
-    unsigned short* __restrict__ out, const unsigned short* __restrict__ q,
+    unsigned short* __restrict__ out, float* __restrict__ part_o,
+    float* __restrict__ part_ml, const unsigned short* __restrict__ q,
     const unsigned short* __restrict__ kcache,
     const unsigned short* __restrict__ vcache,
     const int* __restrict__ slots, const int* __restrict__ lens,
@@ -40,6 +47,7 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
   __shared__ float p_lds[4][AD_TILE];
 
   const int n = blockIdx.x, hkv = blockIdx.y;
+  const int z = blockIdx.z, splits = gridDim.z;
   const int g = Hq / Hkv;
   const int nw = blockDim.x / KF_WAVE;
   const int w = threadIdx.x / KF_WAVE;
@@ -48,6 +56,10 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
   const int64_t slot = slots[n];
   const int64_t cbase = (slot * smax * Hkv + hkv) * AD_D;
   const int64_t cstride = (int64_t)Hkv * AD_D;
+  // this split's row slice, tile-aligned so staging stays coalesced
+  const int chunk =
+      ((len + splits - 1) / splits + AD_TILE - 1) / AD_TILE * AD_TILE;
+  const int lo = z * chunk, hi = min(len, lo + chunk);
 
   for (int hbase = 0; hbase < g; hbase += nw) {
     const int hg = hbase + w;
@@ -66,8 +78,8 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
     float m_run = -INFINITY, l_run = 0.f;
     float o0 = 0.f, o1 = 0.f;  // lane's d-elements (d = 2*lane, 2*lane+1)
 
-    for (int t0 = 0; t0 < len; t0 += AD_TILE) {
-      const int rows = min(AD_TILE, len - t0);
+    for (int t0 = lo; t0 < hi; t0 += AD_TILE) {
+      const int rows = min(AD_TILE, hi - t0);
       __syncthreads();
       for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
         const int r = vi >> 4, c8 = vi & 15;
@@ -112,25 +124,73 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
       }
     }
     if (active) {
-      const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
-      unsigned short* orow = out + ((int64_t)n * Hq + hq) * AD_D;
-      orow[lane * 2] = kf_f32_to_bf16(o0 * inv_l);
-      orow[lane * 2 + 1] = kf_f32_to_bf16(o1 * inv_l);
+      if (splits == 1) {
+        const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+        unsigned short* orow = out + ((int64_t)n * Hq + hq) * AD_D;
+        orow[lane * 2] = kf_f32_to_bf16(o0 * inv_l);
+        orow[lane * 2 + 1] = kf_f32_to_bf16(o1 * inv_l);
+      } else {
+        const int64_t pb = ((int64_t)n * Hq + hq) * splits + z;
+        float* po = part_o + pb * AD_D;
+        po[lane * 2] = o0;
+        po[lane * 2 + 1] = o1;
+        if (lane == 0) {
+          part_ml[pb * 2] = m_run;
+          part_ml[pb * 2 + 1] = l_run;
+        }
+      }
     }
   }
 }
 
-KF_EXPORT int kf_attn_decode(void* out, const void* q, const void* kcache,
+// merge the split partials: one 64-lane wave per (n, hq) row
+__global__ __launch_bounds__(64) void kf_attn_decode_combine_kernel(
+    unsigned short* __restrict__ out, const float* __restrict__ part_o,
+    const float* __restrict__ part_ml, int splits) {
+  const int64_t row = blockIdx.x;  // n*Hq + hq
+  const int lane = threadIdx.x;
+  float M = -INFINITY;
+  for (int z = 0; z < splits; ++z)
+    M = fmaxf(M, part_ml[(row * splits + z) * 2]);
+  float L = 0.f, o0 = 0.f, o1 = 0.f;
+  for (int z = 0; z < splits; ++z) {
+    const float mz = part_ml[(row * splits + z) * 2];
+    if (mz == -INFINITY) continue;
+    const float wz = __expf(mz - M);
+    L += part_ml[(row * splits + z) * 2 + 1] * wz;
+    const float* po = part_o + (row * splits + z) * AD_D;
+    o0 += po[lane * 2] * wz;
+    o1 += po[lane * 2 + 1] * wz;
+  }
+  const float inv_l = L > 0.f ? 1.f / L : 0.f;
+  out[row * AD_D + lane * 2] = kf_f32_to_bf16(o0 * inv_l);
+  out[row * AD_D + lane * 2 + 1] = kf_f32_to_bf16(o1 * inv_l);
+}
+
+// splits > 1 requires part_o [N,Hq,splits,128] f32 and part_ml
+// [N,Hq,splits,2] f32 workspaces; pass splits = 0 to let the host pick
+// enough grid.z to fill the chip (2 blocks/CU).
+KF_EXPORT int kf_attn_decode(void* out, void* part_o, void* part_ml,
+                             const void* q, const void* kcache,
                              const void* vcache, const int* slots,
                              const int* lens, int64_t N, int64_t smax,
-                             int64_t Hq, int64_t Hkv, int64_t D, float scale,
-                             void* stream) {
+                             int64_t Hq, int64_t Hkv, int64_t D,
+                             int64_t splits, float scale, void* stream) {
   if (D != AD_D || Hq % Hkv) return (int)hipErrorInvalidValue;
-  dim3 grid((unsigned)N, (unsigned)Hkv);
+  if (splits < 1) return (int)hipErrorInvalidValue;
+  if (splits > 1 && (!part_o || !part_ml)) return (int)hipErrorInvalidValue;
+  dim3 grid((unsigned)N, (unsigned)Hkv, (unsigned)splits);
   hipLaunchKernelGGL(kf_attn_decode_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (unsigned short*)out,
+                     (float*)part_o, (float*)part_ml,
                      (const unsigned short*)q, (const unsigned short*)kcache,
                      (const unsigned short*)vcache, slots, lens, smax,
                      (int)Hq, (int)Hkv, scale);
+  if (splits > 1)
+    hipLaunchKernelGGL(kf_attn_decode_combine_kernel,
+                       dim3((unsigned)(N * Hq)), dim3(64), 0,
+                       (hipStream_t)stream, (unsigned short*)out,
+                       (const float*)part_o, (const float*)part_ml,
+                       (int)splits);
   return (int)hipGetLastError();
 }

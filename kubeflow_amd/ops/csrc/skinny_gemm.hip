@@ -27,7 +27,8 @@ typedef float kf_f32x4s __attribute__((ext_vector_type(4)));
 template <int SKW>
 __global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
-    const unsigned short* __restrict__ w, int M, int64_t N, int64_t K,
+    const unsigned short* __restrict__ w,
+    const unsigned short* __restrict__ res, int M, int64_t N, int64_t K,
     int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ float red[SKW][SK_NT][SK_NT];  // per-wave C tiles
 
@@ -83,6 +84,7 @@ __global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
       float s = 0.f;
 #pragma unroll
       for (int ww = 0; ww < SKW; ++ww) s += red[ww][row][col];
+      if (res) s += kf_bf16_to_f32(res[row * ldc + n0 + col]);
       c[row * ldc + n0 + col] = kf_f32_to_bf16(s);
     }
   }
@@ -104,7 +106,8 @@ __global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
 
 __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
-    const unsigned short* __restrict__ w, int M, int64_t N, int64_t K,
+    const unsigned short* __restrict__ w,
+    const unsigned short* __restrict__ res, int M, int64_t N, int64_t K,
     int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned short wbuf[2][SK_NT][SKL_STRIDE];
   __shared__ float red[SKL_W][SK_NT][SK_NT];
@@ -171,14 +174,18 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
       float s = 0.f;
 #pragma unroll
       for (int ww = 0; ww < SKL_W; ++ww) s += red[ww][row][col];
+      if (res) s += kf_bf16_to_f32(res[row * ldc + n0 + col]);
       c[row * ldc + n0 + col] = kf_f32_to_bf16(s);
     }
   }
 }
 
+// res (nullable, bf16, C layout): fused residual epilogue C = res + A@W^T
+// — removes the separate elementwise add after the wo / w2 projections.
 KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
-                             int64_t M, int64_t N, int64_t K, int64_t lda,
-                             int64_t ldw, int64_t ldc, void* stream) {
+                             const void* res, int64_t M, int64_t N,
+                             int64_t K, int64_t lda, int64_t ldw,
+                             int64_t ldc, void* stream) {
   if (M < 1 || M > 16 || K % 32 || N % SK_NT) return (int)hipErrorInvalidValue;
   if (lda == 0) lda = K;
   if (ldw == 0) ldw = K;
@@ -188,18 +195,21 @@ KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
     hipLaunchKernelGGL(kf_skinny_lds_kernel, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (int)M, N, K, lda, ldw, ldc);
+                       (const unsigned short*)res, (int)M, N, K, lda, ldw,
+                       ldc);
   // direct-load fallback: 8 waves when the grid can't fill the chip with
   // 4-wave blocks (<2 blocks/CU), 4 otherwise
   else if (N / SK_NT < 512)
     hipLaunchKernelGGL(kf_skinny_gemm_kernel<8>, grid, dim3(8 * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (int)M, N, K, lda, ldw, ldc);
+                       (const unsigned short*)res, (int)M, N, K, lda, ldw,
+                       ldc);
   else
     hipLaunchKernelGGL(kf_skinny_gemm_kernel<4>, grid, dim3(4 * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (int)M, N, K, lda, ldw, ldc);
+                       (const unsigned short*)res, (int)M, N, K, lda, ldw,
+                       ldc);
   return (int)hipGetLastError();
 }

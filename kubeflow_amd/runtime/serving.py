@@ -355,37 +355,36 @@ class InferenceEngine:
         # everywhere). The K14336 / N28672 / lm_head shapes stay on
         # hipBLASLt under "small": measured losses there.
         mode = os.environ.get("KF_SKINNY", "small")
-        sel = {"small": {"qkv", "wo"}, "off": set(),
+        sel = {"small": {"qkv", "wo", "w13", "w2"}, "off": set(),
                "all": {"qkv", "wo", "w13", "w2", "lm"}}.get(
                    mode, set(mode.split(",")))
         def _lin(name):
             return ops.skinny_linear if name in sel else F.linear
-        lin_qkv, lin_wo = _lin("qkv"), _lin("wo")
-        lin_w13, lin_w2, lin_lm = _lin("w13"), _lin("w2"), _lin("lm")
+        lin_qkv, lin_w13, lin_lm = _lin("qkv"), _lin("w13"), _lin("lm")
+        def _lin_res(name):
+            if name in sel:
+                return lambda t, w, r: ops.skinny_linear(t, w, residual=r)
+            return lambda t, w, r: F.linear(t, w) + r
+        lin_wo, lin_w2 = _lin_res("wo"), _lin_res("w2")
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
             qkv = lin_qkv(layer.attn_norm(x), layer.wqkv.weight)
-            q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
-                                 cfg.n_kv_heads * cfg.head_dim,
-                                 cfg.n_kv_heads * cfg.head_dim], dim=-1)
-            q = q.view(N, 1, cfg.n_heads, cfg.head_dim)
-            k = k.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
-            v = v.view(N, 1, cfg.n_kv_heads, cfg.head_dim)
-            q, k = ops.rope(q, k, cos, sin, positions=positions)
-            ops.kv_store(self.cache.k[li], self.cache.v[li], k, v,
-                         slots, positions)
-            o = ops.attention_decode(q[:, 0], self.cache.k[li],
+            # fused RoPE + cache scatter straight off the QKV projection
+            q = ops.decode_rope_store(qkv, self.cache.k[li],
+                                      self.cache.v[li], cos, sin, slots,
+                                      positions, cfg.n_heads,
+                                      cfg.n_kv_heads)
+            o = ops.attention_decode(q, self.cache.k[li],
                                      self.cache.v[li], slots, lens)
-            o = lin_wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
-                       layer.wo.weight)
-            x = x + o
+            x = lin_wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
+                       layer.wo.weight, x)  # residual fused in epilogue
             if layer.moe is not None:
                 x = x + layer.moe(layer.mlp_norm(x))
             else:
                 y = ops.swiglu(lin_w13(layer.mlp_norm(x),
                                        layer.w13.weight))
-                x = x + lin_w2(y, layer.w2.weight)
+                x = lin_w2(y, layer.w2.weight, x)
         x = self.model.final_norm(x)
         logits = lin_lm(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)

@@ -451,3 +451,48 @@ def test_kv_store_parity():
     cv2[slots.long(), positions] = v[:, 0]
     assert torch.equal(ck, ck2)
     assert torch.equal(cv, cv2)
+
+
+@pytest.mark.gpu
+def test_decode_rope_store_parity():
+    """Fused RoPE+scatter vs split/rope/kv_store reference composition."""
+    torch.manual_seed(4)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    SLOTS, SMAX, Hq, Hkv, D, N = 6, 64, 8, 2, 128, 5
+    ck = torch.zeros(SLOTS, SMAX, Hkv, D, device=dev, dtype=torch.bfloat16)
+    cv = torch.zeros_like(ck)
+    ck2, cv2 = ck.clone(), cv.clone()
+    qkv = torch.randn(N, 1, (Hq + 2 * Hkv) * D, device=dev,
+                      dtype=torch.bfloat16)
+    cos, sin = ops.rope_cos_sin(SMAX, D, device=dev)
+    slots = torch.tensor([5, 0, 2, 2, 4], dtype=torch.int32, device=dev)
+    positions = torch.tensor([0, 63, 10, 11, 32], dtype=torch.int64,
+                             device=dev)
+    q = ops.decode_rope_store(qkv, ck, cv, cos, sin, slots, positions,
+                              Hq, Hkv)
+    # reference composition (native rope kernel + kv_store)
+    qq, kk, vv = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    qq = qq.view(N, 1, Hq, D)
+    kk = kk.view(N, 1, Hkv, D)
+    vv = vv.view(N, 1, Hkv, D)
+    qr, kr = ops.rope(qq, kk, cos, sin, positions=positions)
+    ops.kv_store(ck2, cv2, kr, vv, slots, positions)
+    assert torch.equal(q, qr.reshape(N, Hq, D))
+    assert torch.equal(ck, ck2)
+    assert torch.equal(cv, cv2)
+
+
+@pytest.mark.gpu
+def test_skinny_residual_parity():
+    torch.manual_seed(5)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    M, N, K = 16, 4096, 4096
+    x = torch.randn(M, 1, K, device=dev, dtype=torch.bfloat16) * 0.5
+    w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
+    r = torch.randn(M, 1, N, device=dev, dtype=torch.bfloat16)
+    got = ops.skinny_linear(x, w, residual=r)
+    want = torch.nn.functional.linear(x.float(), w.float()) + r.float()
+    err = ((got.float() - want).norm() / want.norm()).item()
+    assert err < 2e-2, err
