@@ -32,7 +32,7 @@ def main():
     eng = InferenceEngine(model, max_slots=32, smax=4096, max_batch=32)
     for i in range(B):
         r = Request(rid=f"r{i}", prompt=list(range(1, plen + 1)),
-                    max_new_tokens=10 ** 9)
+                    max_new_tokens=steps * 4 + 64)
         eng.pending.put(r)
     eng._admit()
     while eng._chunking is not None:
