@@ -14,6 +14,7 @@ All ops are exposed as autograd-capable functions:
 from __future__ import annotations
 
 import ctypes
+import os
 from typing import Optional
 
 import torch
@@ -590,7 +591,8 @@ def attention_decode(q: torch.Tensor, kcache: torch.Tensor,
         # chip at serving batch sizes; split the sequence over grid.z and
         # merge partials (one extra tiny kernel). Static per (N, Hkv) so
         # hipGraph capture sees fixed shapes.
-        splits = min(8, max(1, 512 // max(1, N * Hkv)))
+        splits = int(os.environ.get("KF_DECODE_SPLITS", "0")) or \
+            min(8, max(1, 512 // max(1, N * Hkv)))
         po = pm = None
         if splits > 1:
             po = torch.empty(N, Hq, splits, D, dtype=torch.float32,

@@ -76,19 +76,47 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
         qreg[i][j] = kf_bf16_to_f32((unsigned short)qv[j]);
     }
     float m_run = -INFINITY, l_run = 0.f;
-    float o0 = 0.f, o1 = 0.f;  // lane's d-elements (d = 2*lane, 2*lane+1)
+    // PV accumulation is 4-row-group split for b128 V reads: lane l owns
+    // d-elems [(l&15)*8, +8) for rows r ≡ (l>>4) (mod 4); the 4 groups'
+    // partial o vectors reduce through LDS once per sequence slice.
+    const int rg = lane >> 4;          // this lane's row group
+    const int dcol = (lane & 15) * 8;  // this lane's 8-elem d-range
+    float ov[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
 
     for (int t0 = lo; t0 < hi; t0 += AD_TILE) {
       const int rows = min(AD_TILE, hi - t0);
       __syncthreads();
-      for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
-        const int r = vi >> 4, c8 = vi & 15;
-        kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
-            kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
-        *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kv8;
-        kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
-            vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
-        *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vv8;
+      if (rows == AD_TILE) {
+        // full tile: register-buffered unrolled staging (8 loads in
+        // flight per thread before any LDS write — the dynamic loop's
+        // load->write->load chain left staging HBM-latency-bound)
+        kf_short8 kb[4], vb[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int vi = (int)threadIdx.x + u * 256;
+          const int r = vi >> 4, c8 = vi & 15;
+          kb[u] = *reinterpret_cast<const kf_short8*>(
+              kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          vb[u] = *reinterpret_cast<const kf_short8*>(
+              vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int vi = (int)threadIdx.x + u * 256;
+          const int r = vi >> 4, c8 = vi & 15;
+          *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kb[u];
+          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vb[u];
+        }
+      } else {
+        for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
+          const int r = vi >> 4, c8 = vi & 15;
+          kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
+              kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kv8;
+          kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
+              vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vv8;
+        }
       }
       __syncthreads();
 
@@ -111,29 +139,50 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
       const float p = (s == -INFINITY) ? 0.f : __expf(s - m_new);
       m_run = m_new;
       l_run = l_run * alpha + kf_wave_sum(p);
-      o0 *= alpha;
-      o1 *= alpha;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) ov[e] *= alpha;
       p_lds[w][lane] = p;
       __builtin_amdgcn_s_waitcnt(0);  // wave-local p_lds write->read fence
-      for (int r = 0; r < rows; ++r) {
+      // 16 b128 V reads per tile per lane (was 64 b32): rows 4i + rg
+      for (int i = 0; i < AD_TILE / 4; ++i) {
+        const int r = 4 * i + rg;
+        if (r >= rows) break;
         const float pr = p_lds[w][r];
-        const unsigned int vv = *reinterpret_cast<const unsigned int*>(
-            v_lds + kf_swzd(r, lane * 4));
-        o0 += pr * kf_bf16_to_f32((unsigned short)(vv & 0xffff));
-        o1 += pr * kf_bf16_to_f32((unsigned short)(vv >> 16));
+        kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
+            v_lds + kf_swzd(r, dcol * 2));
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          ov[e] += pr * kf_bf16_to_f32((unsigned short)vv8[e]);
       }
     }
-    if (active) {
+    // fold the 4 row groups: lanes {l15, l15+16, l15+32, l15+48} hold
+    // partial o for the same d-range. Exchange via this wave's p_lds row
+    // (wave-local scratch; 3 rounds of 8 floats each).
+    float osum[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) osum[e] = ov[e];
+    for (int src = 1; src < 4; ++src) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        p_lds[w][lane] = ov[e];
+        __builtin_amdgcn_s_waitcnt(0);
+        osum[e] += p_lds[w][(lane + 16 * src) & 63];
+        __builtin_amdgcn_s_waitcnt(0);
+      }
+    }
+    if (active && rg == 0) {
+      // lane l15 writes d-range [dcol, dcol+8)
       if (splits == 1) {
         const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
-        unsigned short* orow = out + ((int64_t)n * Hq + hq) * AD_D;
-        orow[lane * 2] = kf_f32_to_bf16(o0 * inv_l);
-        orow[lane * 2 + 1] = kf_f32_to_bf16(o1 * inv_l);
+        unsigned short* orow = out + ((int64_t)n * Hq + hq) * AD_D + dcol;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          orow[e] = kf_f32_to_bf16(osum[e] * inv_l);
       } else {
         const int64_t pb = ((int64_t)n * Hq + hq) * splits + z;
-        float* po = part_o + pb * AD_D;
-        po[lane * 2] = o0;
-        po[lane * 2 + 1] = o1;
+        float* po = part_o + pb * AD_D + dcol;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) po[e] = osum[e];
         if (lane == 0) {
           part_ml[pb * 2] = m_run;
           part_ml[pb * 2 + 1] = l_run;
