@@ -42,8 +42,8 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
     const unsigned short* __restrict__ vcache,
     const int* __restrict__ slots, const int* __restrict__ lens,
     int64_t smax, int Hq, int Hkv, float scale) {
-  __shared__ unsigned char k_lds[2][AD_TILE * AD_D * 2];
-  __shared__ unsigned char v_lds[2][AD_TILE * AD_D * 2];
+  __shared__ unsigned char k_lds[AD_TILE * AD_D * 2];
+  __shared__ unsigned char v_lds[AD_TILE * AD_D * 2];
   __shared__ float p_lds[4][AD_TILE];
 
   const int n = blockIdx.x, hkv = blockIdx.y;
@@ -83,43 +83,41 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
     const int dcol = (lane & 15) * 8;  // this lane's 8-elem d-range
     float ov[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
 
-    // double-buffered tiles: each thread prefetches its 8 chunk loads
-    // for tile t+1 into registers BEFORE the barrier, consumes tile t
-    // from LDS, then retires the registers into the other buffer — one
-    // barrier per tile, staging HBM latency hidden behind QK/PV.
-    kf_short8 kb[4], vb[4];
-    auto stage_loads = [&](int t0) {
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int vi = (int)threadIdx.x + u * 256;
-        const int r = vi >> 4, c8 = vi & 15;
-        const bool ok = t0 + r < (int)smax;
-        const int64_t off =
-            cbase + (int64_t)(t0 + (ok ? r : 0)) * cstride + c8 * 8;
-        kb[u] = *reinterpret_cast<const kf_short8*>(kcache + off);
-        vb[u] = *reinterpret_cast<const kf_short8*>(vcache + off);
-      }
-    };
-    auto stage_store = [&](int buf) {
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int vi = (int)threadIdx.x + u * 256;
-        const int r = vi >> 4, c8 = vi & 15;
-        *reinterpret_cast<kf_short8*>(k_lds[buf] + kf_swzd(r, c8 * 16)) =
-            kb[u];
-        *reinterpret_cast<kf_short8*>(v_lds[buf] + kf_swzd(r, c8 * 16)) =
-            vb[u];
-      }
-    };
-    if (lo < hi) {
-      stage_loads(lo);
-      stage_store(0);
-    }
     for (int t0 = lo; t0 < hi; t0 += AD_TILE) {
       const int rows = min(AD_TILE, hi - t0);
-      const int buf = ((t0 - lo) / AD_TILE) & 1;
-      if (t0 + AD_TILE < hi) stage_loads(t0 + AD_TILE);
-      // one barrier per tile: buf's writes visible, buf^1's readers done
+      __syncthreads();
+      if (rows == AD_TILE) {
+        // full tile: register-buffered unrolled staging (8 loads in
+        // flight per thread before any LDS write — the dynamic loop's
+        // load->write->load chain left staging HBM-latency-bound)
+        kf_short8 kb[4], vb[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int vi = (int)threadIdx.x + u * 256;
+          const int r = vi >> 4, c8 = vi & 15;
+          kb[u] = *reinterpret_cast<const kf_short8*>(
+              kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          vb[u] = *reinterpret_cast<const kf_short8*>(
+              vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int vi = (int)threadIdx.x + u * 256;
+          const int r = vi >> 4, c8 = vi & 15;
+          *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kb[u];
+          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vb[u];
+        }
+      } else {
+        for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
+          const int r = vi >> 4, c8 = vi & 15;
+          kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
+              kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kv8;
+          kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
+              vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vv8;
+        }
+      }
       __syncthreads();
 
       float s = -INFINITY;
@@ -128,14 +126,13 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
 #pragma unroll
         for (int i = 0; i < AD_D / 8; ++i) {
           kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
-              k_lds[buf] + kf_swzd(lane, i * 16));
+              k_lds + kf_swzd(lane, i * 16));
 #pragma unroll
           for (int j = 0; j < 8; ++j)
             acc += qreg[i][j] * kf_bf16_to_f32((unsigned short)kv8[j]);
         }
         s = acc * scale;
       }
-      if (t0 + AD_TILE < hi) stage_store(buf ^ 1);
       const float tile_max = kf_wave_max(s);
       const float m_new = fmaxf(m_run, tile_max);
       const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
@@ -152,7 +149,7 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
         if (r >= rows) break;
         const float pr = p_lds[w][r];
         kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
-            v_lds[buf] + kf_swzd(r, dcol * 2));
+            v_lds + kf_swzd(r, dcol * 2));
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           ov[e] += pr * kf_bf16_to_f32((unsigned short)vv8[e]);
