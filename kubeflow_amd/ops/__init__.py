@@ -202,7 +202,7 @@ def kv_store(cache_k: torch.Tensor, cache_v: torch.Tensor,
 def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
                   residual: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Decode-batch linear y = x @ W^T (+ residual) for small leading dims
-    (M <= 16): the LDS-staged weight-streaming kernel (skinny_gemm.hip)
+    (M <= 32): the LDS-staged weight-streaming kernel (skinny_gemm.hip)
     replaces hipBLASLt's ~30-50%-of-BW GEMV path in the serving decode
     step; `residual` fuses the following elementwise add into the
     epilogue. Falls back to F.linear off-GPU or for larger M /
@@ -213,8 +213,8 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
         M *= int(d)
     K = shape[-1]
     N = weight.shape[0]
-    if (not _use_native(x) or M > 16 or K % 32 or N % 16
-            or x.dtype != torch.bfloat16):
+    if (not _use_native(x) or M > 32 or K % 32 or N % 16
+            or x.dtype != torch.bfloat16 or (M > 16 and K % 512)):
         y = torch.nn.functional.linear(x, weight)
         return y + residual if residual is not None else y
     lib = _backend.require()

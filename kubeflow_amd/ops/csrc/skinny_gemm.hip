@@ -104,13 +104,17 @@ __global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
 #define SKL_KC 512                      // staged K elems (1 KB per row)
 #define SKL_STRIDE (SKL_KC + 8)        // +8 elems: 2-way max conflicts
 
+// MT = M-tile (16 or 32 batch rows); W traffic is identical, the wider
+// tile just adds a second A fragment + accumulator (decode buckets > 16
+// otherwise fell back to hipBLASLt).
+template <int MT>
 __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned short* __restrict__ w,
     const unsigned short* __restrict__ res, int M, int64_t N, int64_t K,
     int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned short wbuf[2][SK_NT][SKL_STRIDE];
-  __shared__ float red[SKL_W][SK_NT][SK_NT];
+  __shared__ float red[SKL_W][MT][SK_NT];
 
   const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
   const int tid = threadIdx.x;
@@ -119,9 +123,16 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
   const int l15 = lane & 15;
   const int hi4 = lane >> 4;
 
-  kf_f32x4s acc = kf_f32x4s{0.f, 0.f, 0.f, 0.f};
-  const bool arow_ok = l15 < M;
-  const unsigned short* arow = a + (arow_ok ? l15 : 0) * lda;
+  const int NMT = MT / 16;  // A tiles (1 or 2)
+  kf_f32x4s acc[NMT];
+  bool arow_ok[NMT];
+  const unsigned short* arow[NMT];
+#pragma unroll
+  for (int t = 0; t < NMT; ++t) {
+    acc[t] = kf_f32x4s{0.f, 0.f, 0.f, 0.f};
+    arow_ok[t] = l15 + 16 * t < M;
+    arow[t] = a + (arow_ok[t] ? l15 + 16 * t : 0) * lda;
+  }
   const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
   // writer: wave wv stages rows {2wv, 2wv+1}, lane covers elems
   // [lane*8, lane*8+8) of each 512-elem row slice
@@ -132,23 +143,34 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
   const int ke0 = wv * 64 + hi4 * 8;  // this wave's k slice (s=0; s=1 at +32)
   kf_bf16x8s st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0);
   kf_bf16x8s st1 = *reinterpret_cast<const kf_bf16x8s*>(wr1);
-  kf_bf16x8s af0 = arow_ok
-      ? *reinterpret_cast<const kf_bf16x8s*>(arow + ke0) : zero8;
-  kf_bf16x8s af1 = arow_ok
-      ? *reinterpret_cast<const kf_bf16x8s*>(arow + ke0 + 32) : zero8;
+  kf_bf16x8s af0[NMT], af1[NMT];
+#pragma unroll
+  for (int t = 0; t < NMT; ++t) {
+    af0[t] = arow_ok[t]
+        ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0) : zero8;
+    af1[t] = arow_ok[t]
+        ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32) : zero8;
+  }
   *reinterpret_cast<kf_bf16x8s*>(&wbuf[0][2 * wv][lane * 8]) = st0;
   *reinterpret_cast<kf_bf16x8s*>(&wbuf[0][2 * wv + 1][lane * 8]) = st1;
   for (int64_t ch = 0; ch < nch; ++ch) {
-    kf_bf16x8s a0 = af0, a1 = af1;
+    kf_bf16x8s a0[NMT], a1[NMT];
+#pragma unroll
+    for (int t = 0; t < NMT; ++t) {
+      a0[t] = af0[t];
+      a1[t] = af1[t];
+    }
     if (ch + 1 < nch) {
       st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0 + (ch + 1) * SKL_KC);
       st1 = *reinterpret_cast<const kf_bf16x8s*>(wr1 + (ch + 1) * SKL_KC);
-      if (arow_ok) {
-        af0 = *reinterpret_cast<const kf_bf16x8s*>(
-            arow + (ch + 1) * SKL_KC + ke0);
-        af1 = *reinterpret_cast<const kf_bf16x8s*>(
-            arow + (ch + 1) * SKL_KC + ke0 + 32);
-      }
+#pragma unroll
+      for (int t = 0; t < NMT; ++t)
+        if (arow_ok[t]) {
+          af0[t] = *reinterpret_cast<const kf_bf16x8s*>(
+              arow[t] + (ch + 1) * SKL_KC + ke0);
+          af1[t] = *reinterpret_cast<const kf_bf16x8s*>(
+              arow[t] + (ch + 1) * SKL_KC + ke0 + 32);
+        }
     }
     // one barrier per chunk: makes buffer ch&1's writes visible AND
     // guarantees last iteration's readers of buffer (ch+1)&1 are done
@@ -157,8 +179,13 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
         *reinterpret_cast<const kf_bf16x8s*>(&wbuf[ch & 1][l15][ke0]);
     kf_bf16x8s wf1 =
         *reinterpret_cast<const kf_bf16x8s*>(&wbuf[ch & 1][l15][ke0 + 32]);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, wf0, acc, 0, 0, 0);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, wf1, acc, 0, 0, 0);
+#pragma unroll
+    for (int t = 0; t < NMT; ++t) {
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0[t], wf0, acc[t],
+                                                       0, 0, 0);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1[t], wf1, acc[t],
+                                                       0, 0, 0);
+    }
     if (ch + 1 < nch) {
       const int b = (int)((ch + 1) & 1);
       *reinterpret_cast<kf_bf16x8s*>(&wbuf[b][2 * wv][lane * 8]) = st0;
@@ -166,9 +193,12 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     }
   }
 #pragma unroll
-  for (int j = 0; j < 4; ++j) red[wv][hi4 * 4 + j][l15] = acc[j];
+  for (int t = 0; t < NMT; ++t)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      red[wv][16 * t + hi4 * 4 + j][l15] = acc[t][j];
   __syncthreads();
-  if (tid < SK_NT * SK_NT) {
+  if (tid < MT * SK_NT) {
     const int row = tid / SK_NT, col = tid % SK_NT;
     if (row < M && n0 + col < N) {
       float s = 0.f;
@@ -186,13 +216,21 @@ KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
                              const void* res, int64_t M, int64_t N,
                              int64_t K, int64_t lda, int64_t ldw,
                              int64_t ldc, void* stream) {
-  if (M < 1 || M > 16 || K % 32 || N % SK_NT) return (int)hipErrorInvalidValue;
+  if (M < 1 || M > 32 || K % 32 || N % SK_NT) return (int)hipErrorInvalidValue;
   if (lda == 0) lda = K;
   if (ldw == 0) ldw = K;
   if (ldc == 0) ldc = N;
+  const bool lds_ok = K % SKL_KC == 0 && ldw % 8 == 0 && lda % 8 == 0;
+  if (M > 16 && !lds_ok) return (int)hipErrorInvalidValue;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
-  if (K % SKL_KC == 0 && ldw % 8 == 0 && lda % 8 == 0)
-    hipLaunchKernelGGL(kf_skinny_lds_kernel, grid, dim3(SKL_W * 64), 0,
+  if (lds_ok && M > 16)
+    hipLaunchKernelGGL(kf_skinny_lds_kernel<32>, grid, dim3(SKL_W * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)c,
+                       (const unsigned short*)a, (const unsigned short*)w,
+                       (const unsigned short*)res, (int)M, N, K, lda, ldw,
+                       ldc);
+  else if (lds_ok)
+    hipLaunchKernelGGL(kf_skinny_lds_kernel<16>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
                        (const unsigned short*)res, (int)M, N, K, lda, ldw,

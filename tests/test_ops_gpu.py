@@ -421,7 +421,8 @@ def test_skinny_gemm_parity():
     dev = torch.device("cuda", 0)
     from kubeflow_amd import ops
     for M, N, K in ((16, 6144, 4096), (1, 4096, 4096), (5, 28672, 4096),
-                    (16, 4096, 14336), (16, 128256, 4096)):
+                    (16, 4096, 14336), (16, 128256, 4096),
+                    (32, 6144, 4096), (24, 4096, 14336)):
         x = torch.randn(M, 1, K, device=dev, dtype=torch.bfloat16) * 0.5
         w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
         got = ops.skinny_linear(x, w)
