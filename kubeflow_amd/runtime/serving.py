@@ -189,6 +189,26 @@ class InferenceEngine:
         if self._chunking is not None:
             self._advance_chunk()
             return 0
+        if not self.active:
+            # ramp-up: batch same-length queued prompts into ONE prefill
+            # forward — B=1 prefill is GEMM-efficiency bound (M = seq
+            # len), so grouping multiplies M at no latency cost while
+            # nothing is decoding. (While decodes run, prefills stay
+            # single so inter-token stalls stay bounded.)
+            group = self._take_prefill_group()
+            if group:
+                try:
+                    self._prefill_many(group)
+                    self.active.extend(group)
+                except Exception as e:  # pragma: no cover
+                    import traceback
+                    traceback.print_exc()
+                    for r in group:
+                        r.error = f"{type(e).__name__}: {e}"
+                        self.cache.free(r.slot)
+                        r.finished_at = time.time()
+                        r.done.set()
+                return len(group)
         limit = 1 if self.active else self.max_batch
         n = 0
         while len(self.active) < self.max_batch and n < limit:
@@ -205,6 +225,50 @@ class InferenceEngine:
             if self._chunking is not None:
                 break  # the long prompt continues next iteration
         return n
+
+    PREFILL_GROUP = int(os.environ.get("KF_PREFILL_GROUP", "4"))
+
+    def _take_prefill_group(self):
+        """Pop up to PREFILL_GROUP pending requests whose EFFECTIVE prompt
+        length matches the head request's; requeue mismatches (slight
+        reorder — a scheduler decision, not a protocol one). Returns []
+        when batching does not apply (it needs >=2 same-length prompts,
+        free slots, and an inline-size prompt)."""
+        if self.PREFILL_GROUP < 2:
+            return []
+
+        def eff_len(r):
+            return len(r.prompt[-max(1, self.cache.smax
+                                     - r.max_new_tokens - 1):])
+
+        taken, back = [], []
+        while len(taken) < self.PREFILL_GROUP:
+            try:
+                r = self.pending.get_nowait()
+            except queue.Empty:
+                break
+            if not taken:
+                taken.append(r)
+                continue
+            if eff_len(r) == eff_len(taken[0]):
+                taken.append(r)
+            else:
+                back.append(r)
+        for r in back:
+            self.pending.put(r)
+        if len(taken) < 2 or eff_len(taken[0]) > self.PREFILL_CHUNK:
+            for r in taken:
+                self.pending.put(r)
+            return []
+        group = []
+        for r in taken:
+            slot = self.cache.alloc()
+            if slot is None:
+                self.pending.put(r)
+                continue
+            r.slot = slot
+            group.append(r)
+        return group
 
     def _start_request(self, req: Request):
         slot = self.cache.alloc()
@@ -297,29 +361,35 @@ class InferenceEngine:
 
     @torch.no_grad()
     def _prefill(self, req: Request):
-        """Run the prompt through the model, filling this slot's cache and
-        producing the first generated token."""
-        cfg = self.model.cfg
-        keep = max(1, self.cache.smax - req.max_new_tokens - 1)
-        prompt = req.prompt[-keep:]
-        S = len(prompt)
-        tokens = torch.tensor([prompt], dtype=torch.int64, device=self.device)
+        self._prefill_many([req])
+
+    @torch.no_grad()
+    def _prefill_many(self, reqs):
+        """Run same-length prompts through the model as one batch, filling
+        each request's cache slot and producing its first token."""
+        keep = [max(1, self.cache.smax - r.max_new_tokens - 1) for r in reqs]
+        prompts = [r.prompt[-k:] for r, k in zip(reqs, keep)]
+        S = len(prompts[0])
+        tokens = torch.tensor(prompts, dtype=torch.int64, device=self.device)
         x = self.model.embed(tokens)
         cos, sin = self.model.rope_cos, self.model.rope_sin
+        slots = [r.slot for r in reqs]
         for li, layer in enumerate(self.model.layers):
-            x = self._layer_prefill(layer, li, x, cos, sin, req.slot, S)
+            x = self._layer_prefill(layer, li, x, cos, sin, slots, S)
         x = self.model.final_norm(x[:, -1:])
         logits = torch.nn.functional.linear(x, self.model.lm_head.weight)
-        tok = self._sample(logits[0, -1], req.temperature)
-        req.pos = S
-        req.generated.append(tok)
-        req.first_token_at = time.time()
-        req.token_times.append(req.first_token_at)
-        self.stats["prefill_tokens"] += S
+        now = time.time()
+        for i, r in enumerate(reqs):
+            tok = self._sample(logits[i, -1], r.temperature)
+            r.pos = S
+            r.generated.append(tok)
+            r.first_token_at = now
+            r.token_times.append(now)
+            self.stats["prefill_tokens"] += S
 
-    def _layer_prefill(self, layer, li, x, cos, sin, slot, S):
+    def _layer_prefill(self, layer, li, x, cos, sin, slots, S):
         cfg = self.model.cfg
-        B = 1
+        B = len(slots)
         import torch.nn.functional as F
         qkv = F.linear(layer.attn_norm(x), layer.wqkv.weight)
         q, k, v = qkv.split([cfg.n_heads * cfg.head_dim,
@@ -329,8 +399,9 @@ class InferenceEngine:
         k = k.view(B, S, cfg.n_kv_heads, cfg.head_dim)
         v = v.view(B, S, cfg.n_kv_heads, cfg.head_dim)
         q, k = ops.rope(q, k, cos, sin, 0)
-        self.cache.k[li][slot, :S] = k[0]
-        self.cache.v[li][slot, :S] = v[0]
+        for gi, slot in enumerate(slots):
+            self.cache.k[li][slot, :S] = k[gi]
+            self.cache.v[li][slot, :S] = v[gi]
         o = ops.flash_attention(q, k, v, causal=True)
         o = layer.wo(o.reshape(B, S, cfg.n_heads * cfg.head_dim))
         x = x + o

@@ -29,7 +29,8 @@ def main():
     plen = int(sys.argv[3]) if len(sys.argv) > 3 else 128
     steps = int(sys.argv[4]) if len(sys.argv) > 4 else 50
 
-    eng = InferenceEngine(model, max_slots=32, smax=4096, max_batch=32)
+    eng = InferenceEngine(model, max_slots=max(34, B + 2), smax=4096,
+                      max_batch=32)
     for i in range(B):
         r = Request(rid=f"r{i}", prompt=list(range(1, plen + 1)),
                     max_new_tokens=steps * 4 + 64)

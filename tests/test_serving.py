@@ -402,3 +402,61 @@ def test_inference_service_replicas(tmp_path):
             with urllib.request.urlopen(req, timeout=60) as r:
                 out = json.load(r)
             assert out["predictions"][0]["tokens"], out
+
+
+def test_batched_prefill_group_matches_serial():
+    """Same-length prompts batched at ramp-up decode the same greedy
+    tokens as serial prefill (KF_PREFILL_GROUP=1 disables batching)."""
+    import os
+
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    prompts = [[1 + i, 7, 9, 4, 2 + i] for i in range(3)]
+
+    def run(group):
+        os.environ["KF_PREFILL_GROUP"] = str(group)
+        try:
+            import torch
+            torch.manual_seed(7117)  # identical weights across runs
+            eng = InferenceEngine("llama-tiny", max_slots=8, smax=128,
+                                  max_batch=8)
+            eng.PREFILL_GROUP = group
+            eng.start(precapture=False)
+            try:
+                import threading
+                reqs = []
+                th = [threading.Thread(
+                    target=lambda p=p: reqs.append(
+                        eng.generate(p, max_new_tokens=6, timeout=60)))
+                    for p in prompts]
+                for t in th:
+                    t.start()
+                for t in th:
+                    t.join()
+                return {tuple(r.prompt): r.generated for r in reqs}
+            finally:
+                eng.stop()
+        finally:
+            os.environ.pop("KF_PREFILL_GROUP", None)
+
+    batched = run(4)
+    serial = run(1)
+    assert set(batched) == set(serial)
+    for k in serial:
+        assert len(batched[k]) == 6
+        assert batched[k] == serial[k], (k, batched[k], serial[k])
+
+
+def test_prefill_group_requeues_mismatched_lengths():
+    from kubeflow_amd.runtime.serving import InferenceEngine, Request
+
+    eng = InferenceEngine("llama-tiny", max_slots=8, smax=128, max_batch=8)
+    eng.PREFILL_GROUP = 4
+    eng.pending.put(Request(rid="a", prompt=[1, 2, 3]))
+    eng.pending.put(Request(rid="b", prompt=[4, 5]))
+    eng.pending.put(Request(rid="c", prompt=[6, 7, 8]))
+    group = eng._take_prefill_group()
+    assert sorted(r.rid for r in group) == ["a", "c"]
+    assert eng.pending.qsize() == 1  # "b" requeued
+    for r in group:
+        eng.cache.free(r.slot)
