@@ -35,10 +35,12 @@ def main():
         r = Request(rid=f"r{i}", prompt=list(range(1, plen + 1)),
                     max_new_tokens=steps * 4 + 64)
         eng.pending.put(r)
-    eng._admit()
-    while eng._chunking is not None:
-        eng._advance_chunk()
+    for _ in range(B * 8):
+        if len(eng.active) >= B:
+            break
         eng._admit()
+        while eng._chunking is not None:
+            eng._advance_chunk()
     assert len(eng.active) == B, len(eng.active)
 
     # warm + capture the bucket
