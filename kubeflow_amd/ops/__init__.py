@@ -25,7 +25,8 @@ from . import _backend
 __all__ = [
     "rms_norm", "layer_norm", "rope", "flash_attention", "attention_decode",
     "flash_attention_rect", "skinny_linear", "kv_store",
-    "decode_rope_store",
+    "decode_rope_store", "skinny_linear_q8", "quantize_fp8_rows",
+    "dequantize_fp8_rows",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
     "swiglu", "fused_qkv_attention",
 ]
@@ -230,6 +231,58 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
     _backend.check(
         lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), _p(r2), M, N, K,
                            0, 0, 0, _stream()), "skinny_gemm")
+    return y.view(*shape[:-1], N)
+
+
+def quantize_fp8_rows(w: torch.Tensor):
+    """Per-output-row OCP e4m3 weight quantization for the W8A16 decode
+    path: scale = absmax/448 per row, w8 = round(w/scale) in e4m3fn.
+    Returns (w8 uint8 [N,K], scale fp32 [N])."""
+    wf = w.float()
+    scale = wf.abs().amax(dim=1).clamp_min(1e-12) / 448.0
+    q = (wf / scale.unsqueeze(1)).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).contiguous(), scale.contiguous()
+
+
+def dequantize_fp8_rows(w8: torch.Tensor, scale: torch.Tensor,
+                        dtype=torch.bfloat16) -> torch.Tensor:
+    return (w8.view(torch.float8_e4m3fn).float()
+            * scale.unsqueeze(1)).to(dtype)
+
+
+def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
+                     scale: torch.Tensor,
+                     residual: Optional[torch.Tensor] = None
+                     ) -> torch.Tensor:
+    """Quantized decode linear y = x @ dequant(W8)^T (+ residual): fp8
+    weights halve the HBM traffic of the weight-BW-bound decode GEMMs
+    (skinny_gemm.hip kf_skinny_q8_kernel); activations stay bf16 and the
+    MFMA runs bf16 (W8A16). Serving-only, behind KF_SERVE_QUANT=fp8."""
+    shape = x.shape
+    M = 1
+    for d in shape[:-1]:
+        M *= int(d)
+    K = shape[-1]
+    N = w8.shape[0]
+    if (not _use_native(x) or M > 32 or K % 512 or N % 16
+            or x.dtype != torch.bfloat16):
+        y = torch.nn.functional.linear(
+            x, dequantize_fp8_rows(w8, scale, x.dtype))
+        return y + residual if residual is not None else y
+    lib = _backend.require()
+    x2 = x.reshape(M, K)
+    if not x2.is_contiguous():
+        x2 = x2.contiguous()
+    r2 = None
+    if residual is not None:
+        r2 = residual.reshape(M, N)
+        if not r2.is_contiguous():
+            r2 = r2.contiguous()
+    y = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    _backend.check(
+        lib.kf_skinny_gemm_q8(_p(y), _p(x2), _p(w8), _fp(scale), _p(r2),
+                              M, N, K, 0, 0, 0, _stream()),
+        "skinny_gemm_q8")
     return y.view(*shape[:-1], N)
 
 

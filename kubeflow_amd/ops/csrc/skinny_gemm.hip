@@ -212,6 +212,172 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
 
 // res (nullable, bf16, C layout): fused residual epilogue C = res + A@W^T
 // — removes the separate elementwise add after the wo / w2 projections.
+// ---------------------------------------------------------------------
+// W8A16 quantized variant: W stored as OCP e4m3 fp8 with one fp32 scale
+// per output row (absmax/448). The decode step is weight-BW-bound, so
+// halving W bytes ≈ halves GEMM time; fragments convert fp8 -> f32 ->
+// bf16 in-register (v_cvt_pk_f32_fp8 + v_cvt_pk_bf16_f32 — exact, since
+// e4m3's 3 mantissa bits embed in bf16's 8) and MFMA stays bf16, so the
+// activations keep full precision. Same LDS staging/shape rules as the
+// bf16 kernel: K % 512 == 0, N % 16 == 0, M <= 32.
+
+#define SKQ_STRIDE (SKL_KC + 16)  // fp8 row slice bytes + pad
+
+__device__ __forceinline__ kf_bf16x8s kf_fp8x8_to_bf16x8(
+    const unsigned char* p8) {
+  const unsigned int* pp = reinterpret_cast<const unsigned int*>(p8);
+  const unsigned int lo = pp[0], hi = pp[1];
+  typedef float kf_f32x2q __attribute__((ext_vector_type(2)));
+  kf_f32x2q f0 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
+  kf_f32x2q f1 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
+  kf_f32x2q f2 = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
+  kf_f32x2q f3 = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
+  unsigned int r0, r1, r2, r3;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r0) : "v"(f0[0]), "v"(f0[1]));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r1) : "v"(f1[0]), "v"(f1[1]));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r2) : "v"(f2[0]), "v"(f2[1]));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r3) : "v"(f3[0]), "v"(f3[1]));
+  union {
+    unsigned int u[4];
+    kf_bf16x8s v;
+  } out;
+  out.u[0] = r0;
+  out.u[1] = r1;
+  out.u[2] = r2;
+  out.u[3] = r3;
+  return out.v;
+}
+
+template <int MT>
+__global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
+    unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
+    const unsigned char* __restrict__ w8, const float* __restrict__ wscale,
+    const unsigned short* __restrict__ res, int M, int64_t N, int64_t K,
+    int64_t lda, int64_t ldw, int64_t ldc) {
+  __shared__ unsigned char wbuf8[2][SK_NT][SKQ_STRIDE];
+  __shared__ float red[SKL_W][MT][SK_NT];
+
+  const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
+  const int tid = threadIdx.x;
+  const int wv = tid / KF_WAVE;
+  const int lane = tid & (KF_WAVE - 1);
+  const int l15 = lane & 15;
+  const int hi4 = lane >> 4;
+
+  const int NMT = MT / 16;
+  kf_f32x4s acc[NMT];
+  bool arow_ok[NMT];
+  const unsigned short* arow[NMT];
+#pragma unroll
+  for (int t = 0; t < NMT; ++t) {
+    acc[t] = kf_f32x4s{0.f, 0.f, 0.f, 0.f};
+    arow_ok[t] = l15 + 16 * t < M;
+    arow[t] = a + (arow_ok[t] ? l15 + 16 * t : 0) * lda;
+  }
+  const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
+  const unsigned char* wr0 = w8 + (n0 + 2 * wv) * ldw + lane * 8;
+  const unsigned char* wr1 = wr0 + ldw;
+
+  const int64_t nch = K / SKL_KC;
+  const int ke0 = wv * 64 + hi4 * 8;
+  unsigned long long st0 = *reinterpret_cast<const unsigned long long*>(wr0);
+  unsigned long long st1 = *reinterpret_cast<const unsigned long long*>(wr1);
+  kf_bf16x8s af0[NMT], af1[NMT];
+#pragma unroll
+  for (int t = 0; t < NMT; ++t) {
+    af0[t] = arow_ok[t]
+        ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0) : zero8;
+    af1[t] = arow_ok[t]
+        ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32) : zero8;
+  }
+  *reinterpret_cast<unsigned long long*>(&wbuf8[0][2 * wv][lane * 8]) = st0;
+  *reinterpret_cast<unsigned long long*>(&wbuf8[0][2 * wv + 1][lane * 8]) =
+      st1;
+  for (int64_t ch = 0; ch < nch; ++ch) {
+    kf_bf16x8s a0[NMT], a1[NMT];
+#pragma unroll
+    for (int t = 0; t < NMT; ++t) {
+      a0[t] = af0[t];
+      a1[t] = af1[t];
+    }
+    if (ch + 1 < nch) {
+      st0 = *reinterpret_cast<const unsigned long long*>(
+          wr0 + (ch + 1) * SKL_KC);
+      st1 = *reinterpret_cast<const unsigned long long*>(
+          wr1 + (ch + 1) * SKL_KC);
+#pragma unroll
+      for (int t = 0; t < NMT; ++t)
+        if (arow_ok[t]) {
+          af0[t] = *reinterpret_cast<const kf_bf16x8s*>(
+              arow[t] + (ch + 1) * SKL_KC + ke0);
+          af1[t] = *reinterpret_cast<const kf_bf16x8s*>(
+              arow[t] + (ch + 1) * SKL_KC + ke0 + 32);
+        }
+    }
+    __syncthreads();
+    kf_bf16x8s wf0 = kf_fp8x8_to_bf16x8(&wbuf8[ch & 1][l15][ke0]);
+    kf_bf16x8s wf1 = kf_fp8x8_to_bf16x8(&wbuf8[ch & 1][l15][ke0 + 32]);
+#pragma unroll
+    for (int t = 0; t < NMT; ++t) {
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0[t], wf0, acc[t],
+                                                       0, 0, 0);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1[t], wf1, acc[t],
+                                                       0, 0, 0);
+    }
+    if (ch + 1 < nch) {
+      const int b = (int)((ch + 1) & 1);
+      *reinterpret_cast<unsigned long long*>(&wbuf8[b][2 * wv][lane * 8]) =
+          st0;
+      *reinterpret_cast<unsigned long long*>(
+          &wbuf8[b][2 * wv + 1][lane * 8]) = st1;
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < NMT; ++t)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      red[wv][16 * t + hi4 * 4 + j][l15] = acc[t][j];
+  __syncthreads();
+  if (tid < MT * SK_NT) {
+    const int row = tid / SK_NT, col = tid % SK_NT;
+    if (row < M && n0 + col < N) {
+      float s = 0.f;
+#pragma unroll
+      for (int ww = 0; ww < SKL_W; ++ww) s += red[ww][row][col];
+      s *= wscale[n0 + col];
+      if (res) s += kf_bf16_to_f32(res[row * ldc + n0 + col]);
+      c[row * ldc + n0 + col] = kf_f32_to_bf16(s);
+    }
+  }
+}
+
+KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
+                                const float* wscale, const void* res,
+                                int64_t M, int64_t N, int64_t K,
+                                int64_t lda, int64_t ldw, int64_t ldc,
+                                void* stream) {
+  if (M < 1 || M > 32 || K % SKL_KC || N % SK_NT)
+    return (int)hipErrorInvalidValue;
+  if (lda == 0) lda = K;
+  if (ldw == 0) ldw = K;
+  if (ldc == 0) ldc = N;
+  if (ldw % 8 || lda % 8) return (int)hipErrorInvalidValue;
+  dim3 grid((unsigned)(N / SK_NT), 1, 1);
+  if (M > 16)
+    hipLaunchKernelGGL(kf_skinny_q8_kernel<32>, grid, dim3(SKL_W * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)c,
+                       (const unsigned short*)a, (const unsigned char*)w8,
+                       wscale, (const unsigned short*)res, (int)M, N, K,
+                       lda, ldw, ldc);
+  else
+    hipLaunchKernelGGL(kf_skinny_q8_kernel<16>, grid, dim3(SKL_W * 64), 0,
+                       (hipStream_t)stream, (unsigned short*)c,
+                       (const unsigned short*)a, (const unsigned char*)w8,
+                       wscale, (const unsigned short*)res, (int)M, N, K,
+                       lda, ldw, ldc);
+  return (int)hipGetLastError();
+}
+
 KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
                              const void* res, int64_t M, int64_t N,
                              int64_t K, int64_t lda, int64_t ldw,

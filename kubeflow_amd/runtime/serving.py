@@ -97,6 +97,20 @@ class InferenceEngine:
                                                         storage_uri)
         self.model.eval()
         cfg = self.model.cfg
+        # KF_SERVE_QUANT=fp8: per-output-row OCP e4m3 weight-only quant
+        # for the DECODE linears (W8A16 — halves the weight traffic the
+        # decode step is bound by; prefill keeps the bf16 weights).
+        # Applied after storageUri load so trained weights quantize.
+        self.quant = os.environ.get("KF_SERVE_QUANT", "off") == "fp8"
+        self._qw = None
+        if self.quant and not getattr(cfg, "n_experts", 0):
+            self._qw = [
+                {n: ops.quantize_fp8_rows(getattr(layer, n).weight)
+                 for n in ("wqkv", "wo", "w13", "w2")}
+                for layer in self.model.layers
+            ]
+        elif self.quant:
+            self.quant = False  # MoE decode is eager/dense — not routed
         smax = min(smax, cfg.max_seq_len)
         self.cache = KVCache(cfg.n_layers, max_slots, smax,
                              cfg.n_kv_heads, cfg.head_dim, self.device, dtype)
@@ -429,18 +443,33 @@ class InferenceEngine:
         sel = {"small": {"qkv", "wo", "w13", "w2"}, "off": set(),
                "all": {"qkv", "wo", "w13", "w2", "lm"}}.get(
                    mode, set(mode.split(",")))
-        def _lin(name):
-            return ops.skinny_linear if name in sel else F.linear
-        lin_qkv, lin_w13, lin_lm = _lin("qkv"), _lin("w13"), _lin("lm")
-        def _lin_res(name):
+        qw = self._qw if self.quant else None
+
+        def _lin(name, wname):
+            if qw is not None and name in sel:
+                return lambda li, t, w: ops.skinny_linear_q8(
+                    t, *qw[li][wname])
             if name in sel:
-                return lambda t, w, r: ops.skinny_linear(t, w, residual=r)
-            return lambda t, w, r: F.linear(t, w) + r
-        lin_wo, lin_w2 = _lin_res("wo"), _lin_res("w2")
+                return lambda li, t, w: ops.skinny_linear(t, w)
+            return lambda li, t, w: F.linear(t, w)
+        lin_qkv = _lin("qkv", "wqkv")
+        lin_w13 = _lin("w13", "w13")
+        lin_lm = (ops.skinny_linear if "lm" in sel else F.linear)
+
+        def _lin_res(name, wname):
+            if qw is not None and name in sel:
+                return lambda li, t, w, r: ops.skinny_linear_q8(
+                    t, *qw[li][wname], residual=r)
+            if name in sel:
+                return lambda li, t, w, r: ops.skinny_linear(t, w,
+                                                             residual=r)
+            return lambda li, t, w, r: F.linear(t, w) + r
+        lin_wo = _lin_res("wo", "wo")
+        lin_w2 = _lin_res("w2", "w2")
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
-            qkv = lin_qkv(layer.attn_norm(x), layer.wqkv.weight)
+            qkv = lin_qkv(li, layer.attn_norm(x), layer.wqkv.weight)
             # fused RoPE + cache scatter straight off the QKV projection
             q = ops.decode_rope_store(qkv, self.cache.k[li],
                                       self.cache.v[li], cos, sin, slots,
@@ -448,14 +477,14 @@ class InferenceEngine:
                                       cfg.n_kv_heads)
             o = ops.attention_decode(q, self.cache.k[li],
                                      self.cache.v[li], slots, lens)
-            x = lin_wo(o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
+            x = lin_wo(li, o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
                        layer.wo.weight, x)  # residual fused in epilogue
             if layer.moe is not None:
                 x = x + layer.moe(layer.mlp_norm(x))
             else:
-                y = ops.swiglu(lin_w13(layer.mlp_norm(x),
+                y = ops.swiglu(lin_w13(li, layer.mlp_norm(x),
                                        layer.w13.weight))
-                x = lin_w2(y, layer.w2.weight, x)
+                x = lin_w2(li, y, layer.w2.weight, x)
         x = self.model.final_norm(x)
         logits = lin_lm(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)

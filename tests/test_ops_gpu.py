@@ -497,3 +497,22 @@ def test_skinny_residual_parity():
     want = torch.nn.functional.linear(x.float(), w.float()) + r.float()
     err = ((got.float() - want).norm() / want.norm()).item()
     assert err < 2e-2, err
+
+
+@pytest.mark.gpu
+def test_skinny_q8_parity():
+    """W8A16 kernel vs F.linear on the dequantized weights (same math up
+    to bf16 rounding), with and without residual, M 16 and 32."""
+    torch.manual_seed(6)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    for M, N, K in ((16, 6144, 4096), (32, 4096, 14336), (3, 4096, 4096)):
+        x = torch.randn(M, 1, K, device=dev, dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
+        w8, sc = ops.quantize_fp8_rows(w)
+        r = torch.randn(M, 1, N, device=dev, dtype=torch.bfloat16)
+        wd = ops.dequantize_fp8_rows(w8, sc, torch.float32)
+        got = ops.skinny_linear_q8(x, w8, sc, residual=r)
+        want = torch.nn.functional.linear(x.float(), wd) + r.float()
+        err = ((got.float() - want).norm() / want.norm()).item()
+        assert err < 2e-2, (M, N, K, err)

@@ -460,3 +460,24 @@ def test_prefill_group_requeues_mismatched_lengths():
     assert eng.pending.qsize() == 1  # "b" requeued
     for r in group:
         eng.cache.free(r.slot)
+
+
+def test_quantized_engine_generates(monkeypatch):
+    """KF_SERVE_QUANT=fp8 (W8A16 decode weights) still decodes sane
+    tokens on the CPU fallback path."""
+    import torch
+
+    monkeypatch.setenv("KF_SERVE_QUANT", "fp8")
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(11)
+    eng = InferenceEngine("llama-tiny", max_slots=4, smax=128, max_batch=4)
+    assert eng.quant and eng._qw is not None
+    eng.start(precapture=False)
+    try:
+        r = eng.generate([5, 3, 8, 1], max_new_tokens=5, timeout=60)
+        assert not r.error
+        assert len(r.generated) == 5
+        assert all(0 <= t < eng.model.cfg.vocab_size for t in r.generated)
+    finally:
+        eng.stop()
