@@ -516,3 +516,38 @@ def test_skinny_q8_parity():
         want = torch.nn.functional.linear(x.float(), wd) + r.float()
         err = ((got.float() - want).norm() / want.norm()).item()
         assert err < 2e-2, (M, N, K, err)
+
+
+@pytest.mark.gpu
+def test_quantized_decode_logits_close(monkeypatch):
+    """W8A16 decode logits stay close to the bf16 engine's (same weights,
+    cosine > 0.99) — the quant path changes precision, not semantics."""
+    import torch.nn.functional as TF
+
+    from kubeflow_amd.runtime.serving import InferenceEngine, Request
+
+    dev = torch.device("cuda", 0)
+    prompt = [3, 14, 15, 9, 2, 6, 1, 2]
+
+    def run(quant):
+        if quant:
+            monkeypatch.setenv("KF_SERVE_QUANT", "fp8")
+        else:
+            monkeypatch.delenv("KF_SERVE_QUANT", raising=False)
+        torch.manual_seed(0)
+        eng = InferenceEngine("llama-tiny", device=dev, max_slots=2,
+                              smax=256, max_batch=2)
+        req = Request(rid="t", prompt=list(prompt), max_new_tokens=4)
+        req.slot = eng.cache.alloc()
+        eng._prefill(req)
+        tokens = torch.tensor([[req.generated[-1]]], device=dev)
+        positions = torch.tensor([req.pos], device=dev)
+        slots = torch.tensor([req.slot], dtype=torch.int32, device=dev)
+        lens = torch.tensor([req.pos + 1], dtype=torch.int32, device=dev)
+        logits = eng._decode_forward(tokens, positions, slots, lens)
+        return logits.float().cpu()
+
+    lq = run(True)
+    lb = run(False)
+    cos = TF.cosine_similarity(lq.flatten(), lb.flatten(), dim=0).item()
+    assert cos > 0.99, cos
