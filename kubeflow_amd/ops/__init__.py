@@ -264,7 +264,7 @@ def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
         M *= int(d)
     K = shape[-1]
     N = w8.shape[0]
-    if (not _use_native(x) or M > 32 or K % 512 or N % 16
+    if (not _use_native(x) or M > 32 or K % 1024 or N % 16
             or x.dtype != torch.bfloat16):
         y = torch.nn.functional.linear(
             x, dequantize_fp8_rows(w8, scale, x.dtype))

@@ -221,7 +221,8 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
 // activations keep full precision. Same LDS staging/shape rules as the
 // bf16 kernel: K % 512 == 0, N % 16 == 0, M <= 32.
 
-#define SKQ_STRIDE (SKL_KC + 16)  // fp8 row slice bytes + pad
+#define SKQ_KC 1024               // fp8 K elems per staged chunk (1 KB/row)
+#define SKQ_STRIDE (SKQ_KC + 16)   // fp8 row slice bytes + pad
 
 __device__ __forceinline__ kf_bf16x8s kf_fp8x8_to_bf16x8(
     const unsigned char* p8) {
@@ -275,61 +276,58 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
     arow[t] = a + (arow_ok[t] ? l15 + 16 * t : 0) * lda;
   }
   const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
-  const unsigned char* wr0 = w8 + (n0 + 2 * wv) * ldw + lane * 8;
+  // 16 B per lane per row: a full 1 KB (1024 fp8) row slice per wave
+  // instruction — same load width as the bf16 kernel, twice the K
+  typedef unsigned int kf_u32x4q __attribute__((ext_vector_type(4)));
+  const unsigned char* wr0 = w8 + (n0 + 2 * wv) * ldw + lane * 16;
   const unsigned char* wr1 = wr0 + ldw;
 
-  const int64_t nch = K / SKL_KC;
-  const int ke0 = wv * 64 + hi4 * 8;
-  unsigned long long st0 = *reinterpret_cast<const unsigned long long*>(wr0);
-  unsigned long long st1 = *reinterpret_cast<const unsigned long long*>(wr1);
-  kf_bf16x8s af0[NMT], af1[NMT];
+  const int64_t nch = K / SKQ_KC;
+  const int ke0 = wv * 128 + hi4 * 8;  // k slices at +0,+32,+64,+96
+  kf_u32x4q st0 = *reinterpret_cast<const kf_u32x4q*>(wr0);
+  kf_u32x4q st1 = *reinterpret_cast<const kf_u32x4q*>(wr1);
+  kf_bf16x8s af[4][NMT];
 #pragma unroll
-  for (int t = 0; t < NMT; ++t) {
-    af0[t] = arow_ok[t]
-        ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0) : zero8;
-    af1[t] = arow_ok[t]
-        ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32) : zero8;
-  }
-  *reinterpret_cast<unsigned long long*>(&wbuf8[0][2 * wv][lane * 8]) = st0;
-  *reinterpret_cast<unsigned long long*>(&wbuf8[0][2 * wv + 1][lane * 8]) =
-      st1;
+  for (int si = 0; si < 4; ++si)
+#pragma unroll
+    for (int t = 0; t < NMT; ++t)
+      af[si][t] = arow_ok[t]
+          ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32 * si)
+          : zero8;
+  *reinterpret_cast<kf_u32x4q*>(&wbuf8[0][2 * wv][lane * 16]) = st0;
+  *reinterpret_cast<kf_u32x4q*>(&wbuf8[0][2 * wv + 1][lane * 16]) = st1;
   for (int64_t ch = 0; ch < nch; ++ch) {
-    kf_bf16x8s a0[NMT], a1[NMT];
+    kf_bf16x8s acur[4][NMT];
 #pragma unroll
-    for (int t = 0; t < NMT; ++t) {
-      a0[t] = af0[t];
-      a1[t] = af1[t];
-    }
+    for (int si = 0; si < 4; ++si)
+#pragma unroll
+      for (int t = 0; t < NMT; ++t) acur[si][t] = af[si][t];
     if (ch + 1 < nch) {
-      st0 = *reinterpret_cast<const unsigned long long*>(
-          wr0 + (ch + 1) * SKL_KC);
-      st1 = *reinterpret_cast<const unsigned long long*>(
-          wr1 + (ch + 1) * SKL_KC);
+      st0 = *reinterpret_cast<const kf_u32x4q*>(wr0 + (ch + 1) * SKQ_KC);
+      st1 = *reinterpret_cast<const kf_u32x4q*>(wr1 + (ch + 1) * SKQ_KC);
 #pragma unroll
-      for (int t = 0; t < NMT; ++t)
-        if (arow_ok[t]) {
-          af0[t] = *reinterpret_cast<const kf_bf16x8s*>(
-              arow[t] + (ch + 1) * SKL_KC + ke0);
-          af1[t] = *reinterpret_cast<const kf_bf16x8s*>(
-              arow[t] + (ch + 1) * SKL_KC + ke0 + 32);
-        }
+      for (int si = 0; si < 4; ++si)
+#pragma unroll
+        for (int t = 0; t < NMT; ++t)
+          if (arow_ok[t])
+            af[si][t] = *reinterpret_cast<const kf_bf16x8s*>(
+                arow[t] + (ch + 1) * SKQ_KC + ke0 + 32 * si);
     }
     __syncthreads();
-    kf_bf16x8s wf0 = kf_fp8x8_to_bf16x8(&wbuf8[ch & 1][l15][ke0]);
-    kf_bf16x8s wf1 = kf_fp8x8_to_bf16x8(&wbuf8[ch & 1][l15][ke0 + 32]);
 #pragma unroll
-    for (int t = 0; t < NMT; ++t) {
-      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0[t], wf0, acc[t],
-                                                       0, 0, 0);
-      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1[t], wf1, acc[t],
-                                                       0, 0, 0);
+    for (int si = 0; si < 4; ++si) {
+      kf_bf16x8s wf =
+          kf_fp8x8_to_bf16x8(&wbuf8[ch & 1][l15][ke0 + 32 * si]);
+#pragma unroll
+      for (int t = 0; t < NMT; ++t)
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(acur[si][t], wf,
+                                                         acc[t], 0, 0, 0);
     }
     if (ch + 1 < nch) {
       const int b = (int)((ch + 1) & 1);
-      *reinterpret_cast<unsigned long long*>(&wbuf8[b][2 * wv][lane * 8]) =
-          st0;
-      *reinterpret_cast<unsigned long long*>(
-          &wbuf8[b][2 * wv + 1][lane * 8]) = st1;
+      *reinterpret_cast<kf_u32x4q*>(&wbuf8[b][2 * wv][lane * 16]) = st0;
+      *reinterpret_cast<kf_u32x4q*>(&wbuf8[b][2 * wv + 1][lane * 16]) =
+          st1;
     }
   }
 #pragma unroll
@@ -356,7 +354,7 @@ KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
                                 int64_t M, int64_t N, int64_t K,
                                 int64_t lda, int64_t ldw, int64_t ldc,
                                 void* stream) {
-  if (M < 1 || M > 32 || K % SKL_KC || N % SK_NT)
+  if (M < 1 || M > 32 || K % SKQ_KC || N % SK_NT)
     return (int)hipErrorInvalidValue;
   if (lda == 0) lda = K;
   if (ldw == 0) ldw = K;
