@@ -170,6 +170,16 @@ class InferenceServiceReconciler(Reconciler):
             "max_slots": pred.get("maxSlots", 16),
             "world_size": 1,
         }
+        if pred.get("quantization"):
+            # W8A16 fp8 decode weights (ops/csrc/skinny_gemm.hip q8 path)
+            if pred["quantization"] != "fp8":
+                set_condition(svc, "Failed", "True", "InvalidQuantization",
+                              f"unsupported quantization "
+                              f"{pred['quantization']!r} (supported: fp8)")
+                self.scheduler.release(uid)
+                self.store.update(svc, check_version=False)
+                return
+            base_spec["quantization"] = pred["quantization"]
         if pred.get("storageUri"):
             try:
                 base_spec["ckpt_dir"] = self._resolve_storage_uri(

@@ -79,7 +79,8 @@ class Request:
 class InferenceEngine:
     def __init__(self, model_name: str, device=None, max_slots: int = 32,
                  smax: int = 4096, max_batch: int = 32,
-                 storage_uri: str | None = None):
+                 storage_uri: str | None = None,
+                 quant: str | None = None):
         self.device = device or (torch.device("cuda", 0)
                                  if torch.cuda.is_available()
                                  else torch.device("cpu"))
@@ -101,7 +102,12 @@ class InferenceEngine:
         # for the DECODE linears (W8A16 — halves the weight traffic the
         # decode step is bound by; prefill keeps the bf16 weights).
         # Applied after storageUri load so trained weights quantize.
-        self.quant = os.environ.get("KF_SERVE_QUANT", "off") == "fp8"
+        if quant is None:
+            quant = os.environ.get("KF_SERVE_QUANT", "off")
+        if quant not in ("off", "fp8"):
+            raise ValueError(f"unsupported quantization {quant!r} "
+                             "(supported: fp8)")
+        self.quant = quant == "fp8"
         self._qw = None
         if self.quant and not getattr(cfg, "n_experts", 0):
             self._qw = [

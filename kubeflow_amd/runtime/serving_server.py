@@ -35,6 +35,7 @@ def build_app(spec: dict) -> FastAPI:
         smax=int(spec.get("max_seq_len", 2048)),
         max_batch=int(spec.get("max_batch", 16)),
         storage_uri=spec.get("ckpt_dir") or None,
+        quant=spec.get("quantization") or None,
     ).start()
     app = FastAPI(title=f"kubeflow-amd inference: {name}")
     app.state.engine = engine

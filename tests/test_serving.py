@@ -481,3 +481,29 @@ def test_quantized_engine_generates(monkeypatch):
         assert all(0 <= t < eng.model.cfg.vocab_size for t in r.generated)
     finally:
         eng.stop()
+
+
+def test_inference_spec_quantization_field(tmp_path):
+    """spec.predictor.quantization: fp8 flows to the engine; unsupported
+    values flip Failed=True with InvalidQuantization."""
+    import time as _t
+
+    from kubeflow_amd.api import new_object
+    from kubeflow_amd.api.objects import get_condition
+    from kubeflow_amd.platform import Platform
+
+    with Platform(root_dir=str(tmp_path)) as plat:
+        plat.store.create(new_object(
+            "InferenceService", "bad-quant", "default",
+            spec={"predictor": {"model": "llama-tiny", "gpus": 0,
+                                "quantization": "int3"}}))
+        deadline = _t.time() + 30
+        cond = None
+        while _t.time() < deadline:
+            obj = plat.store.get("InferenceService", "bad-quant", "default")
+            cond = get_condition(obj, "Failed")
+            if cond is not None:
+                break
+            _t.sleep(0.2)
+        assert cond is not None and cond["status"] == "True"
+        assert cond["reason"] == "InvalidQuantization"
