@@ -166,6 +166,32 @@ class MoEMLP(nn.Module):
             out = epmod.reduce_scatter_sum(out, self.ep)
         return out.view(B, S, h)
 
+    def decode_dense(self, x):
+        """Capture-safe decode dispatch: every expert runs on every token
+        and the top-k gate weights (scattered to a dense [N, E] mask)
+        select the sum. Data-dependent routing (nonzero/index_add) would
+        replay the CAPTURED batch's routing under hipGraph; at decode
+        batches (N <= 32) running all experts is weight-BW-bound like the
+        dense MLP, so this is what makes MoE decode graphable. EP stays
+        on the routed path (serving is TP=1/EP=1)."""
+        if self.ep:
+            return self.forward(x)
+        cfg = self.cfg
+        N, S, h = x.shape
+        xf = x.reshape(N * S, h)
+        logits = torch.nn.functional.linear(xf, self.gate.weight)
+        topv, topi = logits.topk(cfg.top_k, dim=-1)
+        w = torch.softmax(topv.float(), dim=-1).to(x.dtype)
+        gatew = torch.zeros(N * S, cfg.n_experts, dtype=x.dtype,
+                            device=x.device)
+        gatew.scatter_(1, topi, w)
+        out = torch.zeros_like(xf)
+        for el in range(self.local_e):
+            y = ops.swiglu(ops.skinny_linear(xf, self.experts_w13[el]))
+            y = ops.skinny_linear(y, self.experts_w2[el])
+            out = out + gatew[:, self.e_lo + el].unsqueeze(1) * y
+        return out.view(N, S, h)
+
 
 class LlamaBlock(nn.Module):
     def __init__(self, cfg: LlamaConfig, tp=None, sp=None, ep=None,

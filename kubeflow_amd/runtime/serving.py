@@ -132,13 +132,10 @@ class InferenceEngine:
         # bucket (the ~10 kernels x 32 layers of launch overhead dominate
         # small-batch decode latency otherwise). One cache slot is reserved
         # as the padding target for bucket rows beyond the live batch.
-        # MoE routing (topk/nonzero/index_add) is data-dependent control
-        # flow — a captured graph would replay the CAPTURED batch's routing
-        # on every decode, silently corrupting outputs. Eager decode for
-        # MoE until a capture-safe dispatch (dense masked einsum) exists.
+        # MoE decode goes through MoEMLP.decode_dense (all experts on all
+        # tokens, dense gate mask) — fixed shapes, so it captures.
         self.use_graphs = (self.device.type == "cuda"
-                           and os.environ.get("KF_SERVE_GRAPH", "1") == "1"
-                           and getattr(cfg, "n_experts", 0) == 0)
+                           and os.environ.get("KF_SERVE_GRAPH", "1") == "1")
         self._graphs = {}
         self._pad_slot = self.cache.alloc() if self.use_graphs else None
 
@@ -486,7 +483,7 @@ class InferenceEngine:
             x = lin_wo(li, o.reshape(N, 1, cfg.n_heads * cfg.head_dim),
                        layer.wo.weight, x)  # residual fused in epilogue
             if layer.moe is not None:
-                x = x + layer.moe(layer.mlp_norm(x))
+                x = x + layer.moe.decode_dense(layer.mlp_norm(x))
             else:
                 y = ops.swiglu(lin_w13(li, layer.mlp_norm(x),
                                        layer.w13.weight))
