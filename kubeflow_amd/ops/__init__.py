@@ -698,7 +698,16 @@ class _CrossEntropyHip(torch.autograd.Function):
         logits, lse, targets, inv_valid = ctx.saved_tensors
         T, V = logits.shape
         scale = (grad_out.float() * inv_valid).contiguous()
-        dlogits = torch.empty_like(logits)
+        # dlogits is written INTO the saved logits buffer: the kernel is
+        # elementwise per position (read lv -> write ov at the same
+        # address), logits feed no other backward node (the lm_head GEMM
+        # backward uses x and W), and at mb6-7 x seq 4096 x 128256-vocab
+        # this buffer is ~7 GB — materializing a second one was the OOM
+        # that blocked micro-batch 7 (KF_CE_INPLACE=0 restores the copy).
+        if os.environ.get("KF_CE_INPLACE", "1") == "1":
+            dlogits = logits
+        else:
+            dlogits = torch.empty_like(logits)
         _backend.check(
             lib.kf_ce_bwd(_p(dlogits), _p(logits), _fp(lse), _ip(targets),
                           _fp(scale), T, V, ctx.ignore_index, _stream()),
