@@ -676,7 +676,7 @@ def attention_decode(q: torch.Tensor, kcache: torch.Tensor,
 
 class _CrossEntropyHip(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, logits, targets, ignore_index):
+    def forward(ctx, logits, targets, ignore_index, inplace_ok=False):
         lib = _backend.require()
         T, V = logits.shape
         logits = logits.contiguous()
@@ -690,6 +690,7 @@ class _CrossEntropyHip(torch.autograd.Function):
                                 torch.zeros((), device=logits.device))
         ctx.save_for_backward(logits, lse, targets, inv_valid)
         ctx.ignore_index = ignore_index
+        ctx.inplace_ok = inplace_ok
         return (loss_sum * inv_valid).squeeze(0)
 
     @staticmethod
@@ -698,13 +699,15 @@ class _CrossEntropyHip(torch.autograd.Function):
         logits, lse, targets, inv_valid = ctx.saved_tensors
         T, V = logits.shape
         scale = (grad_out.float() * inv_valid).contiguous()
-        # dlogits is written INTO the saved logits buffer: the kernel is
+        # dlogits is written INTO the saved logits buffer when the caller
+        # passed a NON-LEAF (the model path: logits = lm_head(x), whose
+        # backward uses x and W, never this output): the kernel is
         # elementwise per position (read lv -> write ov at the same
-        # address), logits feed no other backward node (the lm_head GEMM
-        # backward uses x and W), and at mb6-7 x seq 4096 x 128256-vocab
-        # this buffer is ~7 GB — materializing a second one was the OOM
-        # that blocked micro-batch 7 (KF_CE_INPLACE=0 restores the copy).
-        if os.environ.get("KF_CE_INPLACE", "1") == "1":
+        # address), and at mb6-7 x seq 4096 x 128256-vocab this buffer is
+        # ~7 GB — materializing a second one was the OOM that blocked
+        # micro-batch 7. Leaf inputs (user holds the tensor / wants
+        # .grad) always get a fresh buffer. KF_CE_INPLACE=0 disables.
+        if ctx.inplace_ok and os.environ.get("KF_CE_INPLACE", "1") == "1":
             dlogits = logits
         else:
             dlogits = torch.empty_like(logits)
@@ -712,14 +715,16 @@ class _CrossEntropyHip(torch.autograd.Function):
             lib.kf_ce_bwd(_p(dlogits), _p(logits), _fp(lse), _ip(targets),
                           _fp(scale), T, V, ctx.ignore_index, _stream()),
             "ce_bwd")
-        return dlogits, None, None
+        return dlogits, None, None, None
 
 
 def cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
                   ignore_index: int = -100):
     """Mean CE over non-ignored rows. logits [T,V] bf16, targets [T] int64."""
     if _use_native(logits):
-        return _CrossEntropyHip.apply(logits, targets, ignore_index)
+        inplace_ok = not logits.is_leaf and logits.is_contiguous()
+        return _CrossEntropyHip.apply(logits, targets, ignore_index,
+                                      inplace_ok)
     return reference.softmax_cross_entropy(logits, targets, ignore_index)
 
 
