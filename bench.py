@@ -37,9 +37,11 @@ def main():
     ap.add_argument("--seq-len", type=int,
                     default=int(os.environ.get("KF_BENCH_SEQ", "4096")))
     ap.add_argument("--micro-batch", type=int,
-                    default=int(os.environ.get("KF_BENCH_MB", "6")))
-    # mb sweep on MI355X (profiles/r01_mb_sweep.log): mb4 16.4k, mb6 16.8k
-    # tok/s, mb8 OOM (280 GB activations) — 6 is the per-GPU sweet spot.
+                    default=int(os.environ.get("KF_BENCH_MB", "7")))
+    # mb sweep on MI355X: mb4 16.4k, mb6 16.8k (r01_mb_sweep.log); round 2
+    # in-place CE backward freed the second ~7 GB dlogits buffer, so mb7
+    # fits (peak 277.6 GiB) and measures +1.2% over mb6 (20,789 vs 20,542
+    # same box); mb8 still OOMs on activations.
     args = ap.parse_args()
 
     if int(os.environ.get("WORLD_SIZE", "1")) > 1:
