@@ -35,6 +35,12 @@ __device__ __forceinline__ int kf_swzd(int row, int byte_in_row) {
 // [lo, hi) slice of the sequence and (for splits > 1) writes an
 // UNNORMALIZED partial (o, m, l) that kf_attn_decode_combine merges
 // with the online-softmax rules.
+// VDIRECT: read V straight from global in the PV phase (each row-group
+// instruction covers 256 contiguous bytes of one cache row, and the 4
+// waves' repeat reads hit the XCD's L2) instead of staging it — halves
+// the LDS per block, doubling resident blocks per CU. Measured per
+// context length; host picks.
+template <bool VDIRECT>
 __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
     unsigned short* __restrict__ out, float* __restrict__ part_o,
     float* __restrict__ part_ml, const unsigned short* __restrict__ q,
@@ -43,7 +49,7 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
     const int* __restrict__ slots, const int* __restrict__ lens,
     int64_t smax, int Hq, int Hkv, float scale) {
   __shared__ unsigned char k_lds[AD_TILE * AD_D * 2];
-  __shared__ unsigned char v_lds[AD_TILE * AD_D * 2];
+  __shared__ unsigned char v_lds[VDIRECT ? 1 : AD_TILE * AD_D * 2];
   __shared__ float p_lds[4][AD_TILE];
 
   const int n = blockIdx.x, hkv = blockIdx.y;
@@ -97,15 +103,18 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
           const int r = vi >> 4, c8 = vi & 15;
           kb[u] = *reinterpret_cast<const kf_short8*>(
               kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
-          vb[u] = *reinterpret_cast<const kf_short8*>(
-              vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          if (!VDIRECT)
+            vb[u] = *reinterpret_cast<const kf_short8*>(
+                vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
         }
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
           const int vi = (int)threadIdx.x + u * 256;
           const int r = vi >> 4, c8 = vi & 15;
           *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kb[u];
-          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vb[u];
+          if (!VDIRECT)
+            *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) =
+                vb[u];
         }
       } else {
         for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
@@ -113,9 +122,12 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
           kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
               kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
           *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kv8;
-          kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
-              vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
-          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) = vv8;
+          if (!VDIRECT) {
+            kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
+                vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+            *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) =
+                vv8;
+          }
         }
       }
       __syncthreads();
@@ -148,8 +160,13 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
         const int r = 4 * i + rg;
         if (r >= rows) break;
         const float pr = p_lds[w][r];
-        kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
-            v_lds + kf_swzd(r, dcol * 2));
+        kf_short8 vv8;
+        if (VDIRECT)
+          vv8 = *reinterpret_cast<const kf_short8*>(
+              vcache + cbase + (int64_t)(t0 + r) * cstride + dcol);
+        else
+          vv8 = *reinterpret_cast<const kf_short8*>(
+              v_lds + kf_swzd(r, dcol * 2));
 #pragma unroll
         for (int e = 0; e < 8; ++e)
           ov[e] += pr * kf_bf16_to_f32((unsigned short)vv8[e]);
@@ -229,12 +246,23 @@ KF_EXPORT int kf_attn_decode(void* out, void* part_o, void* part_ml,
   if (splits < 1) return (int)hipErrorInvalidValue;
   if (splits > 1 && (!part_o || !part_ml)) return (int)hipErrorInvalidValue;
   dim3 grid((unsigned)N, (unsigned)Hkv, (unsigned)splits);
-  hipLaunchKernelGGL(kf_attn_decode_kernel, grid, dim3(256), 0,
-                     (hipStream_t)stream, (unsigned short*)out,
-                     (float*)part_o, (float*)part_ml,
-                     (const unsigned short*)q, (const unsigned short*)kcache,
-                     (const unsigned short*)vcache, slots, lens, smax,
-                     (int)Hq, (int)Hkv, scale);
+  const char* vd = getenv("KF_DECODE_VDIRECT");
+  if (vd && vd[0] == '1')
+    hipLaunchKernelGGL(kf_attn_decode_kernel<true>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (unsigned short*)out,
+                       (float*)part_o, (float*)part_ml,
+                       (const unsigned short*)q,
+                       (const unsigned short*)kcache,
+                       (const unsigned short*)vcache, slots, lens, smax,
+                       (int)Hq, (int)Hkv, scale);
+  else
+    hipLaunchKernelGGL(kf_attn_decode_kernel<false>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (unsigned short*)out,
+                       (float*)part_o, (float*)part_ml,
+                       (const unsigned short*)q,
+                       (const unsigned short*)kcache,
+                       (const unsigned short*)vcache, slots, lens, smax,
+                       (int)Hq, (int)Hkv, scale);
   if (splits > 1)
     hipLaunchKernelGGL(kf_attn_decode_combine_kernel,
                        dim3((unsigned)(N * Hq)), dim3(64), 0,
