@@ -238,10 +238,13 @@ def quantize_fp8_rows(w: torch.Tensor):
     """Per-output-row OCP e4m3 weight quantization for the W8A16 decode
     path: scale = absmax/448 per row, w8 = round(w/scale) in e4m3fn.
     Returns (w8 uint8 [N,K], scale fp32 [N])."""
-    wf = w.float()
-    scale = wf.abs().amax(dim=1).clamp_min(1e-12) / 448.0
-    q = (wf / scale.unsqueeze(1)).to(torch.float8_e4m3fn)
-    return q.view(torch.uint8).contiguous(), scale.contiguous()
+    with torch.no_grad():  # w is usually a requires-grad Parameter — the
+        # fp32 temps would otherwise be RETAINED by autograd (141 GB of
+        # them for llama3-70b)
+        wf = w.detach().float()
+        scale = wf.abs().amax(dim=1).clamp_min(1e-12) / 448.0
+        q = (wf / scale.unsqueeze(1)).to(torch.float8_e4m3fn)
+        return q.view(torch.uint8).contiguous(), scale.contiguous()
 
 
 def dequantize_fp8_rows(w8: torch.Tensor, scale: torch.Tensor,
