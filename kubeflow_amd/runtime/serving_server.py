@@ -104,6 +104,18 @@ def build_app(spec: dict) -> FastAPI:
             f'kf_serving_completed_total{{model="{name}"}} {s["completed"]}',
             "# TYPE kf_serving_tokens_out_total counter",
             f'kf_serving_tokens_out_total{{model="{name}"}} {s["tokens_out"]}',
+            "# TYPE kf_serving_prefill_tokens_total counter",
+            f'kf_serving_prefill_tokens_total{{model="{name}"}} '
+            f'{s["prefill_tokens"]}',
+            "# TYPE kf_serving_graph_replays_total counter",
+            f'kf_serving_graph_replays_total{{model="{name}"}} '
+            f'{s["graph_replays"]}',
+            "# TYPE kf_serving_active_streams gauge",
+            f'kf_serving_active_streams{{model="{name}"}} '
+            f'{len(engine.active)}',
+            "# TYPE kf_serving_quantized gauge",
+            f'kf_serving_quantized{{model="{name}"}} '
+            f'{1 if engine.quant else 0}',
         ]
         return PlainTextResponse("\n".join(lines) + "\n")
 
