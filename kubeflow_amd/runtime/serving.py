@@ -71,6 +71,8 @@ class Request:
     generated: List[int] = field(default_factory=list)
     token_times: List[float] = field(default_factory=list)  # per-token stamps
     done: threading.Event = field(default_factory=threading.Event)
+    # streaming consumers: tokens are pushed as produced; None = finished
+    stream: "Optional[queue.Queue]" = None
     first_token_at: Optional[float] = None
     finished_at: Optional[float] = None
     error: str = ""
@@ -179,6 +181,23 @@ class InferenceEngine:
             req.error = req.error or "timeout"
         return req
 
+    def generate_stream(self, prompt: List[int], max_new_tokens: int = 32,
+                        temperature: float = 0.0, timeout: float = 120.0):
+        """Submit and yield tokens as they are produced (generator)."""
+        req = Request(rid=f"r{time.monotonic_ns()}", prompt=list(prompt),
+                      max_new_tokens=max_new_tokens,
+                      temperature=temperature, stream=queue.Queue())
+        self.submit(req)
+        while True:
+            try:
+                tok = req.stream.get(timeout=timeout)
+            except queue.Empty:
+                req.error = req.error or "timeout"
+                return
+            if tok is None:
+                return
+            yield tok
+
     # -------------------------------------------------------------- engine
     @torch.no_grad()
     def _loop(self):
@@ -224,6 +243,8 @@ class InferenceEngine:
                         r.error = f"{type(e).__name__}: {e}"
                         self.cache.free(r.slot)
                         r.finished_at = time.time()
+                        if r.stream is not None:
+                            r.stream.put(None)
                         r.done.set()
                 return len(group)
         limit = 1 if self.active else self.max_batch
@@ -310,6 +331,8 @@ class InferenceEngine:
             req.error = f"{type(e).__name__}: {e}"
             self.cache.free(slot)
             req.finished_at = time.time()
+            if req.stream is not None:
+                req.stream.put(None)
             req.done.set()
 
     @torch.no_grad()
@@ -329,6 +352,8 @@ class InferenceEngine:
                 req.generated.append(tok)
                 req.first_token_at = time.time()
                 req.token_times.append(req.first_token_at)
+                if req.stream is not None:
+                    req.stream.put(tok)
                 self.stats["prefill_tokens"] += len(prompt)
                 self.active.append(req)
                 self._chunking = None
@@ -338,6 +363,8 @@ class InferenceEngine:
             req.error = f"{type(e).__name__}: {e}"
             self.cache.free(req.slot)
             req.finished_at = time.time()
+            if req.stream is not None:
+                req.stream.put(None)
             req.done.set()
             self._chunking = None
 
@@ -402,6 +429,8 @@ class InferenceEngine:
             r.generated.append(tok)
             r.first_token_at = now
             r.token_times.append(now)
+            if r.stream is not None:
+                r.stream.put(tok)
             self.stats["prefill_tokens"] += S
 
     def _layer_prefill(self, layer, li, x, cos, sin, slots, S):
@@ -561,12 +590,16 @@ class InferenceEngine:
             r.pos += 1
             r.generated.append(int(toks[i]))
             r.token_times.append(now)
+            if r.stream is not None:
+                r.stream.put(int(toks[i]))
             self.stats["tokens_out"] += 1
             if (len(r.generated) >= r.max_new_tokens
                     or r.pos + 1 >= self.cache.smax):
                 r.finished_at = now
                 self.cache.free(r.slot)
                 self.stats["completed"] += 1
+                if r.stream is not None:
+                    r.stream.put(None)
                 r.done.set()
             else:
                 still.append(r)

@@ -94,6 +94,30 @@ def build_app(spec: dict) -> FastAPI:
                             temperature=float(body.get("temperature", 0.0)))
         return {"tokens": r.generated, "error": r.error}
 
+    @app.post(f"/v1/models/{name}:generate_stream")
+    async def generate_stream(req: HttpRequest):
+        """Server-sent events: one `data: {"token": t}` line per decoded
+        token, then `data: [DONE]` (the KServe/OpenAI streaming shape)."""
+        import anyio
+        from fastapi.responses import StreamingResponse
+
+        body = await req.json()
+        gen = engine.generate_stream(
+            body.get("prompt_tokens", [1]),
+            int(body.get("max_new_tokens", 16)),
+            temperature=float(body.get("temperature", 0.0)))
+
+        async def _events():
+            while True:
+                tok = await anyio.to_thread.run_sync(
+                    lambda: next(gen, None))
+                if tok is None:
+                    yield "data: [DONE]\n\n"
+                    return
+                yield f'data: {{"token": {tok}}}\n\n'
+
+        return StreamingResponse(_events(), media_type="text/event-stream")
+
     @app.get("/metrics")
     def metrics():
         s = engine.stats

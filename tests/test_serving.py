@@ -552,3 +552,27 @@ def test_moe_engine_decodes_with_graph_path():
         assert r.generated[0] == t1
     finally:
         eng.stop()
+
+
+def test_generate_stream_yields_tokens_then_done():
+    import torch
+
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(3)
+    eng = InferenceEngine("llama-tiny", max_slots=4, smax=128, max_batch=4)
+    eng.start(precapture=False)
+    try:
+        toks = list(eng.generate_stream([4, 7, 2], max_new_tokens=5,
+                                        timeout=60))
+        assert len(toks) == 5
+        assert all(isinstance(t, int) for t in toks)
+        # streamed tokens match a non-streamed run with the same state?
+        # (greedy + shared weights but cache state differs per request;
+        # just check a second stream also completes)
+        toks2 = list(eng.generate_stream([4, 7, 2], max_new_tokens=3,
+                                         timeout=60))
+        assert len(toks2) == 3
+        assert toks2 == toks[:3]  # greedy decode is deterministic
+    finally:
+        eng.stop()
