@@ -10,6 +10,7 @@ Endpoints (KServe data-plane shapes, which the reference's E2E suite probes
     POST /v1/models/<name>:predict    {"instances": [{"prompt_tokens": [...],
                                        "max_new_tokens": N}, ...]}
     POST /v2/generate                 single generate request
+    POST /v1/models/<name>:generate_stream   SSE token stream
     GET  /metrics                     Prometheus text format
 """
 from __future__ import annotations
