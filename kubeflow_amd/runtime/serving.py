@@ -57,6 +57,13 @@ class KVCache:
         with self._lock:
             self._free.append(slot)
 
+    def copy_rows(self, src: int, dst: int, n: int):
+        """Copy the first n cache rows of slot src into slot dst (all
+        layers) — the prefix-cache hit path. Device-side copies only."""
+        for l in range(len(self.k)):
+            self.k[l][dst, :n].copy_(self.k[l][src, :n])
+            self.v[l][dst, :n].copy_(self.v[l][src, :n])
+
 
 @dataclass
 class Request:
@@ -126,6 +133,12 @@ class InferenceEngine:
         self.pending: "queue.Queue[Request]" = queue.Queue()
         self.active: List[Request] = []
         self._chunking = None  # in-progress chunked prefill state
+        # prefix cache: finished sequences donate their slot (keyed by
+        # their full token history) so requests sharing a prompt prefix
+        # copy rows instead of recomputing them (KF_PREFIX_CACHE slots,
+        # 0 disables). LRU; evicted on demand when admission needs slots.
+        self.prefix_slots = int(os.environ.get("KF_PREFIX_CACHE", "4"))
+        self._prefix = {}   # tokens tuple -> slot (insertion = LRU order)
         self._stop = False
         self._thread: Optional[threading.Thread] = None
         self.stats = {"requests": 0, "completed": 0, "tokens_out": 0,
@@ -250,7 +263,7 @@ class InferenceEngine:
         limit = 1 if self.active else self.max_batch
         n = 0
         while len(self.active) < self.max_batch and n < limit:
-            if not self.cache.has_free():
+            if not self.cache.has_free() and not self._prefix:
                 break  # no KV slot: leave requests queued — decode frees
                 # slots as sequences finish (an unconditional loop here
                 # re-took requeued requests forever and starved decode)
@@ -300,7 +313,7 @@ class InferenceEngine:
             return []
         group = []
         for r in taken:
-            slot = self.cache.alloc()
+            slot = self._alloc_slot()
             if slot is None:
                 self.pending.put(r)
                 continue
@@ -308,8 +321,48 @@ class InferenceEngine:
             group.append(r)
         return group
 
-    def _start_request(self, req: Request):
+    def _alloc_slot(self):
+        """cache.alloc, evicting the oldest prefix-cache donation if the
+        pool is exhausted."""
         slot = self.cache.alloc()
+        if slot is None and self._prefix:
+            oldest = next(iter(self._prefix))
+            self.cache.free(self._prefix.pop(oldest))
+            slot = self.cache.alloc()
+        return slot
+
+    def _finish_slot(self, req: Request):
+        """Free a finished request's slot — or donate it to the prefix
+        cache when there is room and the history is worth caching."""
+        if (self.prefix_slots > 0 and len(req.prompt) >= 16
+                and len(self._prefix) < self.prefix_slots):
+            key = tuple(req.prompt[-max(1, self.cache.smax
+                                        - req.max_new_tokens - 1):]
+                        ) + tuple(req.generated)
+            if key not in self._prefix:
+                self._prefix[key] = req.slot
+                return
+        self.cache.free(req.slot)
+
+    def _prefix_hit(self, prompt):
+        """Longest cached history sharing a >=16-token prefix with
+        `prompt` (capped so >=1 prompt token remains to prefill)."""
+        best_len, best_key = 0, None
+        limit = len(prompt) - 1
+        for key in self._prefix:
+            n = 0
+            for a, b in zip(key, prompt):
+                if a != b or n >= limit:
+                    break
+                n += 1
+            if n > best_len:
+                best_len, best_key = n, key
+        if best_len >= 16:
+            return best_key, best_len
+        return None, 0
+
+    def _start_request(self, req: Request):
+        slot = self._alloc_slot()
         if slot is None:
             # no slot free: push back and decode on (slots free as seqs end)
             self.pending.put(req)
@@ -317,6 +370,19 @@ class InferenceEngine:
         req.slot = slot
         keep = max(1, self.cache.smax - req.max_new_tokens - 1)
         prompt = req.prompt[-keep:]
+        key, hit = self._prefix_hit(prompt)
+        if hit:
+            self.cache.copy_rows(self._prefix[key], slot, hit)
+            # re-insert as most recently used
+            self._prefix[key] = self._prefix.pop(key)
+            self.stats["prefix_hits"] = self.stats.get("prefix_hits", 0) + 1
+            self.stats["prefix_tokens_reused"] = (
+                self.stats.get("prefix_tokens_reused", 0) + hit)
+            # prefill only the tail via the rectangular-causal machinery
+            self._chunking = {"req": req, "prompt": prompt, "pos": hit,
+                              "start": hit}
+            self._advance_chunk()
+            return
         if self.active and len(prompt) > self.PREFILL_CHUNK:
             # chunked prefill: first chunk now, rest interleaved with decode
             self._chunking = {"req": req, "prompt": prompt, "pos": 0}
@@ -354,7 +420,8 @@ class InferenceEngine:
                 req.token_times.append(req.first_token_at)
                 if req.stream is not None:
                     req.stream.put(tok)
-                self.stats["prefill_tokens"] += len(prompt)
+                self.stats["prefill_tokens"] += (len(prompt)
+                                                  - st.get("start", 0))
                 self.active.append(req)
                 self._chunking = None
         except Exception as e:  # pragma: no cover
@@ -596,7 +663,7 @@ class InferenceEngine:
             if (len(r.generated) >= r.max_new_tokens
                     or r.pos + 1 >= self.cache.smax):
                 r.finished_at = now
-                self.cache.free(r.slot)
+                self._finish_slot(r)
                 self.stats["completed"] += 1
                 if r.stream is not None:
                     r.stream.put(None)

@@ -576,3 +576,52 @@ def test_generate_stream_yields_tokens_then_done():
         assert toks2 == toks[:3]  # greedy decode is deterministic
     finally:
         eng.stop()
+
+
+def test_prefix_cache_reuses_and_matches():
+    """A finished sequence's KV rows serve the next request with the same
+    prompt: prefill work drops by the hit length and greedy output is
+    identical."""
+    import torch
+
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(17)
+    eng = InferenceEngine("llama-tiny", max_slots=6, smax=128, max_batch=4)
+    eng.PREFILL_GROUP = 1  # isolate the prefix path
+    eng.start(precapture=False)
+    try:
+        prompt = list(range(2, 34))  # 32 tokens (>= the 16-token floor)
+        r1 = eng.generate(prompt, max_new_tokens=4, timeout=60)
+        assert not r1.error
+        pre1 = eng.stats["prefill_tokens"]
+        r2 = eng.generate(prompt, max_new_tokens=4, timeout=60)
+        assert not r2.error
+        assert r2.generated == r1.generated  # deterministic greedy
+        assert eng.stats.get("prefix_hits", 0) >= 1
+        reused = eng.stats.get("prefix_tokens_reused", 0)
+        assert reused >= 16
+        # second prefill only computed the tail
+        assert eng.stats["prefill_tokens"] - pre1 <= len(prompt) - reused + 4
+    finally:
+        eng.stop()
+
+
+def test_prefix_cache_evicts_under_pressure():
+    import torch
+
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(18)
+    eng = InferenceEngine("llama-tiny", max_slots=3, smax=128, max_batch=3)
+    eng.PREFILL_GROUP = 1
+    eng.start(precapture=False)
+    try:
+        # each finished request donates; with 3 slots total the next
+        # admissions must evict donations rather than starve
+        for i in range(5):
+            r = eng.generate(list(range(1 + i, 33 + i)), max_new_tokens=2,
+                             timeout=60)
+            assert not r.error, r.error
+    finally:
+        eng.stop()
