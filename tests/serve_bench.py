@@ -21,12 +21,15 @@ def main():
     model = sys.argv[1] if len(sys.argv) > 1 else "llama3-8b"
     n_req = int(sys.argv[2]) if len(sys.argv) > 2 else 16
     max_new = int(sys.argv[3]) if len(sys.argv) > 3 else 64
-    # prompt_len: an int, or "mix" = cycle 128/1024/4096 (the mixed-load
-    # p99 inter-token measurement of VERDICT round-1 item 9)
+    # prompt_len: an int, "mix" = cycle 128/1024/4096 (the mixed-load
+    # p99 inter-token measurement), or "same1024" = every request sends
+    # the SAME 1024-token prompt (prefix-cache hit path: TTFT after the
+    # first request is a row-copy + 1-token prefill)
     plen_arg = sys.argv[4] if len(sys.argv) > 4 else "128"
     mixed = plen_arg == "mix"
-    lens = [128, 1024, 4096] if mixed else [int(plen_arg)]
-    plen = "mix" if mixed else int(plen_arg)
+    same = plen_arg.startswith("same")
+    lens = [128, 1024, 4096] if mixed else         [int(plen_arg[4:])] if same else [int(plen_arg)]
+    plen = plen_arg if (mixed or same) else int(plen_arg)
 
     eng = InferenceEngine(model, max_slots=32, smax=4096,
                           max_batch=32).start()
@@ -36,7 +39,8 @@ def main():
     results = []
     def run(i):
         L = lens[i % len(lens)]
-        results.append(eng.generate(list(range(1 + i, L + 1 + i)),
+        off = 1 if same else 1 + i
+        results.append(eng.generate(list(range(off, L + off)),
                                     max_new_tokens=max_new, timeout=600))
 
     threads = [threading.Thread(target=run, args=(i,)) for i in range(n_req)]
@@ -47,6 +51,7 @@ def main():
         t.join()
     el = time.time() - t0
 
+    prefix_hits = eng.stats.get("prefix_hits", 0)
     toks = sum(len(r.generated) for r in results)
     lat = sorted(1000 * (r.finished_at - r.submitted)
                  for r in results if r.finished_at)
@@ -64,7 +69,7 @@ def main():
         "value": round(toks / el, 1),
         "n_requests": n_req,
         "max_new_tokens": max_new,
-        "prompt_len": plen,
+        "prompt_len": plen, "prefix_hits": prefix_hits,
         "latency_p50_ms": round(lat[len(lat) // 2], 1),
         "latency_p99_ms": round(lat[min(len(lat) - 1, int(len(lat) * 0.99))], 1),
         "ttft_p50_ms": round(ttft[len(ttft) // 2], 1),
