@@ -73,14 +73,18 @@ class BertLayer(nn.Module):
         self.fc2 = nn.Linear(cfg.ffn_dim, h)
         self.ln2 = KfLayerNorm(h, eps=cfg.norm_eps)
 
-    def forward(self, x):
+    def forward(self, x, kv_len=None):
         cfg = self.cfg
         B, S, _ = x.shape
         q, k, v = self.wqkv(x).split(cfg.n_heads * cfg.head_dim, dim=-1)
         q = q.view(B, S, cfg.n_heads, cfg.head_dim)
         k = k.view(B, S, cfg.n_heads, cfg.head_dim)
         v = v.view(B, S, cfg.n_heads, cfg.head_dim)
-        o = ops.flash_attention(q, k, v, causal=False)
+        if kv_len is not None and kv_len < S:
+            # padded serving batch: attend only the real kv_len tokens
+            o = ops.masked_attention(q, k, v, kv_len)
+        else:
+            o = ops.flash_attention(q, k, v, causal=False)
         o = self.wo(o.reshape(B, S, -1))
         x = self.ln1(x + o)
         return self.ln2(x + self.fc2(F.gelu(self.fc1(x))))
@@ -113,12 +117,12 @@ class BertClassifier(nn.Module):
                 mod.weight.fill_(1.0)
                 mod.bias.zero_()
 
-    def forward(self, tokens, targets=None):
+    def forward(self, tokens, targets=None, kv_len=None):
         B, S = tokens.shape
         pos = torch.arange(S, device=tokens.device)
         x = self.embed_ln(self.tok_embed(tokens) + self.pos_embed(pos))
         for layer in self.layers:
-            x = layer(x)
+            x = layer(x, kv_len=kv_len)
         logits = self.classifier(x[:, 0])  # [CLS]
         if targets is None:
             return logits

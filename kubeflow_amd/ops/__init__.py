@@ -28,6 +28,7 @@ __all__ = [
     "decode_rope_store", "skinny_linear_q8", "quantize_fp8_rows",
     "dequantize_fp8_rows",
     "cross_entropy", "fused_adamw", "rope_cos_sin", "native_available",
+    "masked_attention",
     "swiglu", "fused_qkv_attention",
 ]
 
@@ -422,6 +423,51 @@ def flash_attention_rect(q: torch.Tensor, k: torch.Tensor,
                        k[:, :kv_end].float().transpose(1, 2),
                        v[:, :kv_end].float().transpose(1, 2),
                        causal=True, scale=scale).transpose(1, 2)
+    return o.to(q.dtype)
+
+
+def masked_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     kv_len: int, scale: Optional[float] = None
+                     ) -> torch.Tensor:
+    """Bidirectional (non-causal) attention over only the first `kv_len`
+    kv rows — the padded-batch classifier case (BERT serving): rows
+    beyond kv_len are padding and must not be attended. Uses the
+    rectangular-causal kernel with q_offset = kv_len, which makes every
+    query's limit min(row + kv_len, kv_len-1) = kv_len-1 (inference
+    only, no autograd; D == 128 on the native path)."""
+    B, S, Hq, D = q.shape
+    if scale is None:
+        scale = D ** -0.5
+    if _use_native(q) and D == 128:
+        lib = _backend.require()
+        Hkv = k.shape[2]
+        pad = (256 - S % 256) % 256
+        if pad:
+            q = torch.cat([q, q.new_zeros(B, pad, Hq, D)], dim=1)
+        o = torch.empty_like(q)
+        lse = torch.empty(B, Hq, q.shape[1], dtype=torch.float32,
+                          device=q.device)
+        q = q.contiguous()
+        need = (kv_len + 127) // 128 * 128
+        if k.shape[1] < need:
+            kp = torch.zeros(B, need, Hkv, D, dtype=k.dtype,
+                             device=k.device)
+            vp = torch.zeros_like(kp)
+            kp[:, :kv_len] = k[:, :kv_len]
+            vp[:, :kv_len] = v[:, :kv_len]
+            k, v = kp, vp
+        k, v = k.contiguous(), v.contiguous()
+        _backend.check(
+            lib.kf_attn_fwd4_rect(_p(o), _fp(lse), _p(q), _p(k), _p(v),
+                                  B, q.shape[1], kv_len, Hq, Hkv, D,
+                                  0, Hkv * D,
+                                  ctypes.c_float(float(scale)), kv_len,
+                                  _stream()), "attn_fwd4_rect")
+        return o[:, :S]
+    o = reference.sdpa(q.float().transpose(1, 2),
+                       k[:, :kv_len].float().transpose(1, 2),
+                       v[:, :kv_len].float().transpose(1, 2),
+                       causal=False, scale=scale).transpose(1, 2)
     return o.to(q.dtype)
 
 

@@ -80,6 +80,7 @@ class Request:
     done: threading.Event = field(default_factory=threading.Event)
     # streaming consumers: tokens are pushed as produced; None = finished
     stream: "Optional[queue.Queue]" = None
+    scores: Optional[List[float]] = None  # classifier probabilities
     first_token_at: Optional[float] = None
     finished_at: Optional[float] = None
     error: str = ""
@@ -107,6 +108,28 @@ class InferenceEngine:
                                                         storage_uri)
         self.model.eval()
         cfg = self.model.cfg
+        # classifier models (BERT family): no KV cache / graphs — each
+        # request is one batched forward; same-length prompts batch, the
+        # rest pad to a 128 multiple with masked_attention(kv_len)
+        self.classify = hasattr(cfg, "n_classes")
+        if self.classify:
+            self.max_batch = max_batch
+            self.cache = None
+            self.quant = False
+            self._qw = None
+            self.use_graphs = False
+            self._graphs = {}
+            self._pad_slot = None
+            self._chunking = None
+            self.pending = queue.Queue()
+            self.active = []
+            self.prefix_slots = 0
+            self._prefix = {}
+            self._stop = False
+            self._thread = None
+            self.stats = {"requests": 0, "completed": 0, "tokens_out": 0,
+                          "prefill_tokens": 0, "graph_replays": 0}
+            return
         # KF_SERVE_QUANT=fp8: per-output-row OCP e4m3 weight-only quant
         # for the DECODE linears (W8A16 — halves the weight traffic the
         # decode step is bound by; prefill keeps the bf16 weights).
@@ -214,6 +237,11 @@ class InferenceEngine:
     # -------------------------------------------------------------- engine
     @torch.no_grad()
     def _loop(self):
+        if self.classify:
+            while not self._stop:
+                if not self._classify_batch():
+                    time.sleep(0.002)
+            return
         while not self._stop:
             self._admit()
             if not self.active:
@@ -230,6 +258,56 @@ class InferenceEngine:
     # 196 -> 68 ms for ~-25% aggregate tok/s — an SLO/throughput knob.
     # KF_PREFILL_CHUNK overrides; 0 disables chunking.
     PREFILL_CHUNK = int(os.environ.get("KF_PREFILL_CHUNK", "1024")) or (1 << 30)
+
+    @torch.no_grad()
+    def _classify_batch(self) -> int:
+        """Drain up to max_batch same-length classify requests and run
+        one padded forward (masked_attention over the real length)."""
+        taken, back = [], []
+        while len(taken) < self.max_batch:
+            try:
+                r = self.pending.get_nowait()
+            except queue.Empty:
+                break
+            if not taken or len(r.prompt) == len(taken[0].prompt):
+                taken.append(r)
+            else:
+                back.append(r)
+        for r in back:
+            self.pending.put(r)
+        if not taken:
+            return 0
+        try:
+            S = len(taken[0].prompt)
+            Sp = (S + 127) // 128 * 128
+            toks = torch.zeros(len(taken), Sp, dtype=torch.int64,
+                               device=self.device)
+            for i, r in enumerate(taken):
+                toks[i, :S] = torch.tensor(r.prompt, dtype=torch.int64)
+            logits = self.model(toks, kv_len=S if Sp > S else None)
+            probs = torch.softmax(logits.float(), dim=-1).cpu()
+            now = time.time()
+            for i, r in enumerate(taken):
+                r.scores = [round(float(p), 6) for p in probs[i]]
+                r.generated = [int(probs[i].argmax())]
+                r.first_token_at = r.finished_at = now
+                r.token_times.append(now)
+                self.stats["tokens_out"] += 1
+                self.stats["completed"] += 1
+                if r.stream is not None:
+                    r.stream.put(r.generated[0])
+                    r.stream.put(None)
+                r.done.set()
+        except Exception as e:  # pragma: no cover
+            import traceback
+            traceback.print_exc()
+            for r in taken:
+                r.error = f"{type(e).__name__}: {e}"
+                r.finished_at = time.time()
+                if r.stream is not None:
+                    r.stream.put(None)
+                r.done.set()
+        return len(taken)
 
     def _admit(self) -> int:
         # Fairness: one prefill unit (a full short prompt, or ONE chunk of

@@ -79,6 +79,8 @@ def build_app(spec: dict) -> FastAPI:
                 r.error = r.error or "timeout"
             preds.append({
                 "tokens": r.generated,
+                **({"class": r.generated[0] if r.generated else None,
+                    "scores": r.scores} if engine.classify else {}),
                 "error": r.error,
                 "ttft_ms": (None if r.first_token_at is None else
                             round(1000 * (r.first_token_at - r.submitted), 2)),

@@ -551,3 +551,51 @@ def test_quantized_decode_logits_close(monkeypatch):
     lb = run(False)
     cos = TF.cosine_similarity(lq.flatten(), lb.flatten(), dim=0).item()
     assert cos > 0.99, cos
+
+
+@pytest.mark.gpu
+def test_masked_attention_gpu_parity():
+    """masked_attention (rect kernel, q_offset=kv_len) vs fp32 sdpa over
+    only the real kv rows — the padded-classifier case, D=128."""
+    torch.manual_seed(7)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    B, S, Hq, Hkv, D = 2, 128, 6, 6, 128
+    kv_len = 37
+    q = torch.randn(B, S, Hq, D, device=dev, dtype=torch.bfloat16) * 0.5
+    k = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16) * 0.5
+    v = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16)
+    got = ops.masked_attention(q, k, v, kv_len)
+    ref = ops.reference.sdpa(
+        q.float().transpose(1, 2), k[:, :kv_len].float().transpose(1, 2),
+        v[:, :kv_len].float().transpose(1, 2), causal=False,
+    ).transpose(1, 2)
+    err = ((got.float() - ref).norm() / ref.norm()).item()
+    assert err < 2e-2, err
+
+
+@pytest.mark.gpu
+def test_classifier_engine_gpu_matches_forward():
+    """BERT-hd128 classify on GPU (native masked_attention path) ==
+    direct full forward at a 128-multiple length."""
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(10)
+    dev = torch.device("cuda", 0)
+    eng = InferenceEngine("bert-base-hd128", device=dev, max_batch=4)
+    eng.start()
+    try:
+        prompt = list(range(2, 53))  # odd length -> padded masked path
+        r = eng.generate(prompt, timeout=120)
+        assert not r.error, r.error
+        with torch.no_grad():
+            # reference: pad to 128 and use the same masked-attention
+            # entry (the unpadded direct forward is impossible on GPU —
+            # the flash kernel needs S % 128 == 0 — so ALSO check
+            # against a CPU fp32 clone of the same weights)
+            import copy
+            cpu = copy.deepcopy(eng.model).float().cpu()
+            want = int(cpu(torch.tensor([prompt]))[0].argmax())
+        assert r.generated[0] == want, (r.generated, want, r.scores)
+    finally:
+        eng.stop()

@@ -625,3 +625,44 @@ def test_prefix_cache_evicts_under_pressure():
             assert not r.error, r.error
     finally:
         eng.stop()
+
+
+def test_classifier_engine_serves_bert():
+    """InferenceService serves the BERT family: batched same-length
+    classify, mixed lengths handled across iterations, class == direct
+    model forward (masked_attention over the real length)."""
+    import threading
+
+    import torch
+
+    from kubeflow_amd.runtime.serving import InferenceEngine
+
+    torch.manual_seed(9)
+    eng = InferenceEngine("bert-base", max_batch=4)
+    assert eng.classify
+    eng.start()
+    try:
+        prompts = [list(range(3, 40)), list(range(3, 40)),
+                   list(range(2, 70)), list(range(1, 20))]
+        out = {}
+
+        def run(i):
+            out[i] = eng.generate(prompts[i], timeout=60)
+
+        th = [threading.Thread(target=run, args=(i,))
+              for i in range(len(prompts))]
+        for t in th:
+            t.start()
+        for t in th:
+            t.join()
+        for i, p in enumerate(prompts):
+            r = out[i]
+            assert not r.error, r.error
+            assert len(r.generated) == 1
+            assert r.scores is not None and len(r.scores) == 2
+            assert abs(sum(r.scores) - 1.0) < 1e-3
+            with torch.no_grad():
+                want = int(eng.model(torch.tensor([p]))[0].argmax())
+            assert r.generated[0] == want, (i, r.generated, want)
+    finally:
+        eng.stop()
