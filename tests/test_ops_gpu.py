@@ -589,13 +589,15 @@ def test_classifier_engine_gpu_matches_forward():
         r = eng.generate(prompt, timeout=120)
         assert not r.error, r.error
         with torch.no_grad():
-            # reference: pad to 128 and use the same masked-attention
-            # entry (the unpadded direct forward is impossible on GPU —
-            # the flash kernel needs S % 128 == 0 — so ALSO check
-            # against a CPU fp32 clone of the same weights)
+            # the unpadded direct GPU forward is impossible (the flash
+            # kernel needs S % 128 == 0), so check the class PROBABILITIES
+            # against a CPU fp32 clone of the same weights — argmax on
+            # random-init logits is too close to compare across dtypes
             import copy
             cpu = copy.deepcopy(eng.model).float().cpu()
-            want = int(cpu(torch.tensor([prompt]))[0].argmax())
-        assert r.generated[0] == want, (r.generated, want, r.scores)
+            want = torch.softmax(cpu(torch.tensor([prompt]))[0].float(),
+                                 dim=-1)
+        got = torch.tensor(r.scores)
+        assert torch.allclose(got, want, atol=0.05), (got, want)
     finally:
         eng.stop()
