@@ -448,18 +448,16 @@ def masked_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         lse = torch.empty(B, Hq, q.shape[1], dtype=torch.float32,
                           device=q.device)
         q = q.contiguous()
-        need = (kv_len + 127) // 128 * 128
-        if k.shape[1] < need:
-            kp = torch.zeros(B, need, Hkv, D, dtype=k.dtype,
-                             device=k.device)
-            vp = torch.zeros_like(kp)
-            kp[:, :kv_len] = k[:, :kv_len]
-            vp[:, :kv_len] = v[:, :kv_len]
-            k, v = kp, vp
-        k, v = k.contiguous(), v.contiguous()
+        # the kernel's kv batch stride is b * Sq_padded rows (training
+        # layout) — repack kv into that shape, zeroed past kv_len
+        Sp = q.shape[1]
+        kp = torch.zeros(B, Sp, Hkv, D, dtype=k.dtype, device=k.device)
+        vp = torch.zeros_like(kp)
+        kp[:, :kv_len] = k[:, :kv_len]
+        vp[:, :kv_len] = v[:, :kv_len]
         _backend.check(
-            lib.kf_attn_fwd4_rect(_p(o), _fp(lse), _p(q), _p(k), _p(v),
-                                  B, q.shape[1], kv_len, Hq, Hkv, D,
+            lib.kf_attn_fwd4_rect(_p(o), _fp(lse), _p(q), _p(kp), _p(vp),
+                                  B, Sp, kv_len, Hq, Hkv, D,
                                   0, Hkv * D,
                                   ctypes.c_float(float(scale)), kv_len,
                                   _stream()), "attn_fwd4_rect")
