@@ -398,13 +398,19 @@ def flash_attention_rect(q: torch.Tensor, k: torch.Tensor,
         # to the next KV-TILE multiple of Skv (A4_KT=64 in
         # attention_fwd4.hip; 128 kept for headroom if the tile grows)
         need = (Skv + 127) // 128 * 128
+        # the kernel's kv batch stride is b * Sq_padded rows, so the
+        # zero-copy fast path is B == 1 only (the serving slabs)
         def _rows_ok(t):
-            return (t.stride(1) == Hkv * D and t.stride(2) == D
+            return (B == 1 and t.stride(1) == Hkv * D and t.stride(2) == D
                     and t.stride(3) == 1
-                    and (B == 1 or t.stride(0) >= need * Hkv * D)
                     and t.stride(0) // (Hkv * D) >= need)
         if not (_rows_ok(k) and _rows_ok(v)):
-            kp = torch.zeros(B, need, Hkv, D, dtype=k.dtype, device=k.device)
+            rows = max(need, q.shape[1])
+            if B > 1 and rows != q.shape[1]:
+                raise ValueError(
+                    "batched rectangular attention requires Skv <= padded "
+                    f"Sq (kernel batch stride); got Skv={Skv} B={B}")
+            kp = torch.zeros(B, rows, Hkv, D, dtype=k.dtype, device=k.device)
             vp = torch.zeros_like(kp)
             kp[:, :Skv] = k[:, :Skv]
             vp[:, :Skv] = v[:, :Skv]
