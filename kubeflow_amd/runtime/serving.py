@@ -147,6 +147,9 @@ class InferenceEngine:
                  for n in ("wqkv", "wo", "w13", "w2")}
                 for layer in self.model.layers
             ]
+            # lm_head stays bf16 under "small" routing (sampling
+            # sensitivity); KF_SKINNY=all opts it into fp8 too
+            self._q_lm = ops.quantize_fp8_rows(self.model.lm_head.weight)
         elif self.quant:
             self.quant = False  # MoE decode is eager/dense — not routed
         smax = min(smax, cfg.max_seq_len)
@@ -631,7 +634,12 @@ class InferenceEngine:
             return lambda li, t, w: F.linear(t, w)
         lin_qkv = _lin("qkv", "wqkv")
         lin_w13 = _lin("w13", "w13")
-        lin_lm = (ops.skinny_linear if "lm" in sel else F.linear)
+        if "lm" in sel and qw is not None:
+            lin_lm = lambda t, w: ops.skinny_linear_q8(t, *self._q_lm)
+        elif "lm" in sel:
+            lin_lm = ops.skinny_linear
+        else:
+            lin_lm = F.linear
 
         def _lin_res(name, wname):
             if qw is not None and name in sel:
