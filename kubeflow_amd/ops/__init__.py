@@ -201,8 +201,18 @@ def kv_store(cache_k: torch.Tensor, cache_v: torch.Tensor,
     cache_v[slots.long(), positions] = v.reshape(n, *cache_v.shape[2:])
 
 
+def _rms_args(x, rms):
+    """rms = (RMSNorm module, fp32 gamma tensor) or None -> kernel args.
+    The fused path needs K %% 512 == 0 (LDS kernels only)."""
+    if rms is None:
+        return None, 0.0
+    module, g32 = rms
+    return g32, float(module.eps)
+
+
 def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
-                  residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+                  residual: Optional[torch.Tensor] = None,
+                  rms=None) -> torch.Tensor:
     """Decode-batch linear y = x @ W^T (+ residual) for small leading dims
     (M <= 32): the LDS-staged weight-streaming kernel (skinny_gemm.hip)
     replaces hipBLASLt's ~30-50%-of-BW GEMV path in the serving decode
@@ -216,7 +226,10 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
     K = shape[-1]
     N = weight.shape[0]
     if (not _use_native(x) or M > 32 or K % 32 or N % 16
-            or x.dtype != torch.bfloat16 or (M > 16 and K % 512)):
+            or x.dtype != torch.bfloat16 or (M > 16 and K % 512)
+            or (rms is not None and K % 512)):
+        if rms is not None:
+            x = rms[0](x)
         y = torch.nn.functional.linear(x, weight)
         return y + residual if residual is not None else y
     lib = _backend.require()
@@ -228,10 +241,12 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
         r2 = residual.reshape(M, N)
         if not r2.is_contiguous():
             r2 = r2.contiguous()
+    g32, eps = _rms_args(x, rms)
     y = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _backend.check(
-        lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), _p(r2), M, N, K,
-                           0, 0, 0, _stream()), "skinny_gemm")
+        lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), _p(r2), _fp(g32),
+                           eps, M, N, K, 0, 0, 0, _stream()),
+        "skinny_gemm")
     return y.view(*shape[:-1], N)
 
 
@@ -256,8 +271,8 @@ def dequantize_fp8_rows(w8: torch.Tensor, scale: torch.Tensor,
 
 def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
                      scale: torch.Tensor,
-                     residual: Optional[torch.Tensor] = None
-                     ) -> torch.Tensor:
+                     residual: Optional[torch.Tensor] = None,
+                     rms=None) -> torch.Tensor:
     """Quantized decode linear y = x @ dequant(W8)^T (+ residual): fp8
     weights halve the HBM traffic of the weight-BW-bound decode GEMMs
     (skinny_gemm.hip kf_skinny_q8_kernel); activations stay bf16 and the
@@ -270,6 +285,8 @@ def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
     N = w8.shape[0]
     if (not _use_native(x) or M > 32 or K % 1024 or N % 16
             or x.dtype != torch.bfloat16):
+        if rms is not None:
+            x = rms[0](x)
         y = torch.nn.functional.linear(
             x, dequantize_fp8_rows(w8, scale, x.dtype))
         return y + residual if residual is not None else y
@@ -282,10 +299,12 @@ def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
         r2 = residual.reshape(M, N)
         if not r2.is_contiguous():
             r2 = r2.contiguous()
+    g32, eps = _rms_args(x, rms)
     y = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _backend.check(
         lib.kf_skinny_gemm_q8(_p(y), _p(x2), _p(w8), _fp(scale), _p(r2),
-                              M, N, K, 0, 0, 0, _stream()),
+                              _fp(g32), eps, M, N, K, 0, 0, 0,
+                              _stream()),
         "skinny_gemm_q8")
     return y.view(*shape[:-1], N)
 

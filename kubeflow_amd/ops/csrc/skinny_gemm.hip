@@ -18,6 +18,43 @@
 typedef __bf16 kf_bf16x8s __attribute__((ext_vector_type(8)));
 typedef float kf_f32x4s __attribute__((ext_vector_type(4)));
 
+// Fused-RMSNorm support: when rmsg != null, A rows are normalized
+// in-flight — the block computes each A row's rstd once (it reads the
+// whole row over the chunk loop anyway; A is LLC-hot) and scales
+// fragments by rstd[row] * gamma[k]. Removes the two standalone rmsnorm
+// launches per decode layer (profiles/r02_decode_anatomy.md).
+typedef unsigned short kf_u16x8q __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ float kf_row_rstd(const unsigned short* row,
+                                             int64_t K, float eps) {
+  // all 64 lanes of the calling wave stride one row cooperatively
+  float acc = 0.f;
+  const int lane = threadIdx.x & (KF_WAVE - 1);
+  for (int64_t i = (int64_t)lane * 8; i < K; i += KF_WAVE * 8) {
+    const kf_u16x8q x = *reinterpret_cast<const kf_u16x8q*>(row + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float f = kf_bf16_to_f32(x[j]);
+      acc += f * f;
+    }
+  }
+  for (int off = 32; off; off >>= 1) acc += __shfl_xor(acc, off, KF_WAVE);
+  return rsqrtf(acc / (float)K + eps);
+}
+
+__device__ __forceinline__ kf_bf16x8s kf_rms_scale8(
+    kf_bf16x8s a, float rstd, const float* __restrict__ g) {
+  union {
+    kf_bf16x8s v;
+    unsigned short u[8];
+  } in, out;
+  in.v = a;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    out.u[j] = kf_f32_to_bf16(kf_bf16_to_f32(in.u[j]) * rstd * g[j]);
+  return out.v;
+}
+
 #define SK_NT 16      // N columns per block
 // K-split ways (waves/block) is a template knob: small-N shapes (wo:
 // N=4096 -> 256 blocks == 1 block/CU == 1 wave/SIMD at 4 waves) are
@@ -111,10 +148,12 @@ template <int MT>
 __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned short* __restrict__ w,
-    const unsigned short* __restrict__ res, int M, int64_t N, int64_t K,
-    int64_t lda, int64_t ldw, int64_t ldc) {
+    const unsigned short* __restrict__ res,
+    const float* __restrict__ rmsg, float rms_eps, int M, int64_t N,
+    int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned short wbuf[2][SK_NT][SKL_STRIDE];
   __shared__ float red[SKL_W][MT][SK_NT];
+  __shared__ float rstd_lds[MT];
 
   const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
   const int tid = threadIdx.x;
@@ -134,6 +173,17 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     arow[t] = a + (arow_ok[t] ? l15 + 16 * t : 0) * lda;
   }
   const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
+  float rstd[NMT];
+  if (rmsg) {
+    // wave wv computes rstd for rows {wv*MT/8 .. } (MT/8 rows per wave)
+    for (int r = wv * (MT / SKL_W); r < (wv + 1) * (MT / SKL_W); ++r) {
+      const float v = kf_row_rstd(a + (r < M ? r : 0) * lda, K, rms_eps);
+      if (lane == 0) rstd_lds[r] = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int t = 0; t < NMT; ++t) rstd[t] = rstd_lds[l15 + 16 * t];
+  }
   // writer: wave wv stages rows {2wv, 2wv+1}, lane covers elems
   // [lane*8, lane*8+8) of each 512-elem row slice
   const unsigned short* wr0 = w + (n0 + 2 * wv) * ldw + lane * 8;
@@ -179,6 +229,14 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
         *reinterpret_cast<const kf_bf16x8s*>(&wbuf[ch & 1][l15][ke0]);
     kf_bf16x8s wf1 =
         *reinterpret_cast<const kf_bf16x8s*>(&wbuf[ch & 1][l15][ke0 + 32]);
+    if (rmsg) {
+      const float* g0 = rmsg + ch * SKL_KC + ke0;
+#pragma unroll
+      for (int t = 0; t < NMT; ++t) {
+        a0[t] = kf_rms_scale8(a0[t], rstd[t], g0);
+        a1[t] = kf_rms_scale8(a1[t], rstd[t], g0 + 32);
+      }
+    }
 #pragma unroll
     for (int t = 0; t < NMT; ++t) {
       acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0[t], wf0, acc[t],
@@ -253,10 +311,12 @@ template <int MT>
 __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned char* __restrict__ w8, const float* __restrict__ wscale,
-    const unsigned short* __restrict__ res, int M, int64_t N, int64_t K,
-    int64_t lda, int64_t ldw, int64_t ldc) {
+    const unsigned short* __restrict__ res,
+    const float* __restrict__ rmsg, float rms_eps, int M, int64_t N,
+    int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned char wbuf8[2][SK_NT][SKQ_STRIDE];
   __shared__ float red[SKL_W][MT][SK_NT];
+  __shared__ float rstd_lds[MT];
 
   const int64_t n0 = (int64_t)blockIdx.x * SK_NT;
   const int tid = threadIdx.x;
@@ -276,6 +336,16 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
     arow[t] = a + (arow_ok[t] ? l15 + 16 * t : 0) * lda;
   }
   const kf_bf16x8s zero8 = kf_bf16x8s{0, 0, 0, 0, 0, 0, 0, 0};
+  float rstd[NMT];
+  if (rmsg) {
+    for (int r = wv * (MT / SKL_W); r < (wv + 1) * (MT / SKL_W); ++r) {
+      const float vv = kf_row_rstd(a + (r < M ? r : 0) * lda, K, rms_eps);
+      if (lane == 0) rstd_lds[r] = vv;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int t = 0; t < NMT; ++t) rstd[t] = rstd_lds[l15 + 16 * t];
+  }
   // 16 B per lane per row: a full 1 KB (1024 fp8) row slice per wave
   // instruction — same load width as the bf16 kernel, twice the K
   typedef unsigned int kf_u32x4q __attribute__((ext_vector_type(4)));
@@ -318,6 +388,12 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
     for (int si = 0; si < 4; ++si) {
       kf_bf16x8s wf =
           kf_fp8x8_to_bf16x8(&wbuf8[ch & 1][l15][ke0 + 32 * si]);
+      if (rmsg) {
+        const float* g = rmsg + ch * SKQ_KC + ke0 + 32 * si;
+#pragma unroll
+        for (int t = 0; t < NMT; ++t)
+          acur[si][t] = kf_rms_scale8(acur[si][t], rstd[t], g);
+      }
 #pragma unroll
       for (int t = 0; t < NMT; ++t)
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(acur[si][t], wf,
@@ -351,6 +427,7 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
 
 KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
                                 const float* wscale, const void* res,
+                                const float* rmsg, float rms_eps,
                                 int64_t M, int64_t N, int64_t K,
                                 int64_t lda, int64_t ldw, int64_t ldc,
                                 void* stream) {
@@ -365,19 +442,22 @@ KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
     hipLaunchKernelGGL(kf_skinny_q8_kernel<32>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned char*)w8,
-                       wscale, (const unsigned short*)res, (int)M, N, K,
-                       lda, ldw, ldc);
+                       wscale, (const unsigned short*)res, rmsg, rms_eps,
+                       (int)M, N, K, lda, ldw, ldc);
   else
     hipLaunchKernelGGL(kf_skinny_q8_kernel<16>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned char*)w8,
-                       wscale, (const unsigned short*)res, (int)M, N, K,
-                       lda, ldw, ldc);
+                       wscale, (const unsigned short*)res, rmsg, rms_eps,
+                       (int)M, N, K, lda, ldw, ldc);
   return (int)hipGetLastError();
 }
 
+// rmsg (nullable, fp32 [K]): fused input RMSNorm — C = res +
+// rmsnorm(A; rmsg, rms_eps) @ W^T (LDS-staged kernels only).
 KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
-                             const void* res, int64_t M, int64_t N,
+                             const void* res, const float* rmsg,
+                             float rms_eps, int64_t M, int64_t N,
                              int64_t K, int64_t lda, int64_t ldw,
                              int64_t ldc, void* stream) {
   if (M < 1 || M > 32 || K % 32 || N % SK_NT) return (int)hipErrorInvalidValue;
@@ -386,19 +466,20 @@ KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
   if (ldc == 0) ldc = N;
   const bool lds_ok = K % SKL_KC == 0 && ldw % 8 == 0 && lda % 8 == 0;
   if (M > 16 && !lds_ok) return (int)hipErrorInvalidValue;
+  if (rmsg && !lds_ok) return (int)hipErrorInvalidValue;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
   if (lds_ok && M > 16)
     hipLaunchKernelGGL(kf_skinny_lds_kernel<32>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (const unsigned short*)res, (int)M, N, K, lda, ldw,
-                       ldc);
+                       (const unsigned short*)res, rmsg, rms_eps, (int)M,
+                       N, K, lda, ldw, ldc);
   else if (lds_ok)
     hipLaunchKernelGGL(kf_skinny_lds_kernel<16>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (const unsigned short*)res, (int)M, N, K, lda, ldw,
-                       ldc);
+                       (const unsigned short*)res, rmsg, rms_eps, (int)M,
+                       N, K, lda, ldw, ldc);
   // direct-load fallback: 8 waves when the grid can't fill the chip with
   // 4-wave blocks (<2 blocks/CU), 4 otherwise
   else if (N / SK_NT < 512)

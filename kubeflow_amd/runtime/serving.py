@@ -152,6 +152,14 @@ class InferenceEngine:
             self._q_lm = ops.quantize_fp8_rows(self.model.lm_head.weight)
         elif self.quant:
             self.quant = False  # MoE decode is eager/dense — not routed
+        # fp32 norm gammas for the fused RMSNorm-in-GEMM decode path
+        # (the kernel wants fp32; converting per step would add launches)
+        self._g32 = [
+            (layer.attn_norm.weight.detach().float(),
+             None if layer.moe is not None
+             else layer.mlp_norm.weight.detach().float())
+            for layer in self.model.layers
+        ]
         smax = min(smax, cfg.max_seq_len)
         self.cache = KVCache(cfg.n_layers, max_slots, smax,
                              cfg.n_kv_heads, cfg.head_dim, self.device, dtype)
@@ -627,11 +635,13 @@ class InferenceEngine:
 
         def _lin(name, wname):
             if qw is not None and name in sel:
-                return lambda li, t, w: ops.skinny_linear_q8(
-                    t, *qw[li][wname])
+                return lambda li, t, w, rms=None: ops.skinny_linear_q8(
+                    t, *qw[li][wname], rms=rms)
             if name in sel:
-                return lambda li, t, w: ops.skinny_linear(t, w)
-            return lambda li, t, w: F.linear(t, w)
+                return lambda li, t, w, rms=None: ops.skinny_linear(
+                    t, w, rms=rms)
+            return (lambda li, t, w, rms=None:
+                    F.linear(rms[0](t) if rms else t, w))
         lin_qkv = _lin("qkv", "wqkv")
         lin_w13 = _lin("w13", "w13")
         if "lm" in sel and qw is not None:
@@ -654,7 +664,8 @@ class InferenceEngine:
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
-            qkv = lin_qkv(li, layer.attn_norm(x), layer.wqkv.weight)
+            qkv = lin_qkv(li, x, layer.wqkv.weight,
+                          rms=(layer.attn_norm, self._g32[li][0]))
             # fused RoPE + cache scatter straight off the QKV projection
             q = ops.decode_rope_store(qkv, self.cache.k[li],
                                       self.cache.v[li], cos, sin, slots,
@@ -667,8 +678,9 @@ class InferenceEngine:
             if layer.moe is not None:
                 x = x + layer.moe.decode_dense(layer.mlp_norm(x))
             else:
-                y = ops.swiglu(lin_w13(li, layer.mlp_norm(x),
-                                       layer.w13.weight))
+                y = ops.swiglu(lin_w13(li, x, layer.w13.weight,
+                                       rms=(layer.mlp_norm,
+                                            self._g32[li][1])))
                 x = lin_w2(li, y, layer.w2.weight, x)
         x = self.model.final_norm(x)
         logits = lin_lm(x, self.model.lm_head.weight)  # [N,1,V]

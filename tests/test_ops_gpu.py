@@ -601,3 +601,33 @@ def test_classifier_engine_gpu_matches_forward():
         assert torch.allclose(got, want, atol=0.05), (got, want)
     finally:
         eng.stop()
+
+
+@pytest.mark.gpu
+def test_skinny_fused_rms_parity():
+    """skinny_linear(rms=...) == F.linear(rms_norm(x)) for bf16/fp8."""
+    torch.manual_seed(11)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+    from kubeflow_amd.models.llama import RMSNorm
+
+    for M, N, K in ((16, 4096, 4096), (32, 4096, 14336)):
+        x = torch.randn(M, 1, K, device=dev, dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
+        norm = RMSNorm(K, 1e-5).to(device=dev, dtype=torch.bfloat16)
+        with torch.no_grad():
+            norm.weight.normal_(1.0, 0.1)
+        g32 = norm.weight.detach().float()
+        r = torch.randn(M, 1, N, device=dev, dtype=torch.bfloat16)
+        got = ops.skinny_linear(x, w, residual=r, rms=(norm, g32))
+        xn = ops.reference.rms_norm(x.float(), g32, 1e-5)
+        want = torch.nn.functional.linear(xn, w.float()) + r.float()
+        err = ((got.float() - want).norm() / want.norm()).item()
+        assert err < 3e-2, (M, N, K, err)
+        # fp8 path
+        w8, sc = ops.quantize_fp8_rows(w)
+        got8 = ops.skinny_linear_q8(x, w8, sc, residual=r, rms=(norm, g32))
+        wd = ops.dequantize_fp8_rows(w8, sc, torch.float32)
+        want8 = torch.nn.functional.linear(xn, wd) + r.float()
+        err8 = ((got8.float() - want8).norm() / want8.norm()).item()
+        assert err8 < 3e-2, (M, N, K, err8)
