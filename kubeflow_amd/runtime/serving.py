@@ -152,14 +152,6 @@ class InferenceEngine:
             self._q_lm = ops.quantize_fp8_rows(self.model.lm_head.weight)
         elif self.quant:
             self.quant = False  # MoE decode is eager/dense — not routed
-        # fp32 norm gammas for the fused RMSNorm-in-GEMM decode path
-        # (the kernel wants fp32; converting per step would add launches)
-        self._g32 = [
-            (layer.attn_norm.weight.detach().float(),
-             None if layer.moe is not None
-             else layer.mlp_norm.weight.detach().float())
-            for layer in self.model.layers
-        ]
         smax = min(smax, cfg.max_seq_len)
         self.cache = KVCache(cfg.n_layers, max_slots, smax,
                              cfg.n_kv_heads, cfg.head_dim, self.device, dtype)
@@ -664,8 +656,11 @@ class InferenceEngine:
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
-            qkv = lin_qkv(li, x, layer.wqkv.weight,
-                          rms=(layer.attn_norm, self._g32[li][0]))
+            # NOTE: fusing the RMSNorms into the GEMM prologue (rms=...)
+            # measured WORSE (replay 4.1 -> 5.8 ms): the per-block rstd
+            # pass stalls the weight stream behind 16 LLC row reads per
+            # block. Separate norm kernels stay (r02_decode_anatomy.md).
+            qkv = lin_qkv(li, layer.attn_norm(x), layer.wqkv.weight)
             # fused RoPE + cache scatter straight off the QKV projection
             q = ops.decode_rope_store(qkv, self.cache.k[li],
                                       self.cache.v[li], cos, sin, slots,
@@ -678,9 +673,8 @@ class InferenceEngine:
             if layer.moe is not None:
                 x = x + layer.moe.decode_dense(layer.mlp_norm(x))
             else:
-                y = ops.swiglu(lin_w13(li, x, layer.w13.weight,
-                                       rms=(layer.mlp_norm,
-                                            self._g32[li][1])))
+                y = ops.swiglu(lin_w13(li, layer.mlp_norm(x),
+                                       layer.w13.weight))
                 x = lin_w2(li, y, layer.w2.weight, x)
         x = self.model.final_norm(x)
         logits = lin_lm(x, self.model.lm_head.weight)  # [N,1,V]
