@@ -212,7 +212,7 @@ def _rms_args(x, rms):
 
 def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
                   residual: Optional[torch.Tensor] = None,
-                  rms=None) -> torch.Tensor:
+                  rms=None, fuse_swiglu: bool = False) -> torch.Tensor:
     """Decode-batch linear y = x @ W^T (+ residual) for small leading dims
     (M <= 32): the LDS-staged weight-streaming kernel (skinny_gemm.hip)
     replaces hipBLASLt's ~30-50%-of-BW GEMV path in the serving decode
@@ -223,17 +223,19 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
     M = 1
     for d in shape[:-1]:
         M *= int(d)
-    K = shape[-1]
+    K = shape[-1] // 2 if fuse_swiglu else shape[-1]
     N = weight.shape[0]
     if (not _use_native(x) or M > 32 or K % 32 or N % 16
             or x.dtype != torch.bfloat16 or (M > 16 and K % 512)
-            or (rms is not None and K % 512)):
+            or ((rms is not None or fuse_swiglu) and K % 512)):
         if rms is not None:
             x = rms[0](x)
+        if fuse_swiglu:
+            x = swiglu(x)
         y = torch.nn.functional.linear(x, weight)
         return y + residual if residual is not None else y
     lib = _backend.require()
-    x2 = x.reshape(M, K)
+    x2 = x.reshape(M, shape[-1])
     if not x2.is_contiguous():
         x2 = x2.contiguous()
     r2 = None
@@ -245,7 +247,8 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
     y = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _backend.check(
         lib.kf_skinny_gemm(_p(y), _p(x2), _p(weight), _p(r2), _fp(g32),
-                           eps, M, N, K, 0, 0, 0, _stream()),
+                           eps, int(fuse_swiglu), M, N, K,
+                           shape[-1], 0, 0, _stream()),
         "skinny_gemm")
     return y.view(*shape[:-1], N)
 
@@ -272,7 +275,7 @@ def dequantize_fp8_rows(w8: torch.Tensor, scale: torch.Tensor,
 def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
                      scale: torch.Tensor,
                      residual: Optional[torch.Tensor] = None,
-                     rms=None) -> torch.Tensor:
+                     rms=None, fuse_swiglu: bool = False) -> torch.Tensor:
     """Quantized decode linear y = x @ dequant(W8)^T (+ residual): fp8
     weights halve the HBM traffic of the weight-BW-bound decode GEMMs
     (skinny_gemm.hip kf_skinny_q8_kernel); activations stay bf16 and the
@@ -281,17 +284,19 @@ def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
     M = 1
     for d in shape[:-1]:
         M *= int(d)
-    K = shape[-1]
+    K = shape[-1] // 2 if fuse_swiglu else shape[-1]
     N = w8.shape[0]
     if (not _use_native(x) or M > 32 or K % 1024 or N % 16
             or x.dtype != torch.bfloat16):
         if rms is not None:
             x = rms[0](x)
+        if fuse_swiglu:
+            x = swiglu(x)
         y = torch.nn.functional.linear(
             x, dequantize_fp8_rows(w8, scale, x.dtype))
         return y + residual if residual is not None else y
     lib = _backend.require()
-    x2 = x.reshape(M, K)
+    x2 = x.reshape(M, shape[-1])
     if not x2.is_contiguous():
         x2 = x2.contiguous()
     r2 = None
@@ -303,8 +308,8 @@ def skinny_linear_q8(x: torch.Tensor, w8: torch.Tensor,
     y = torch.empty(M, N, dtype=x.dtype, device=x.device)
     _backend.check(
         lib.kf_skinny_gemm_q8(_p(y), _p(x2), _p(w8), _fp(scale), _p(r2),
-                              _fp(g32), eps, M, N, K, 0, 0, 0,
-                              _stream()),
+                              _fp(g32), eps, int(fuse_swiglu), M, N, K,
+                              shape[-1], 0, 0, _stream()),
         "skinny_gemm_q8")
     return y.view(*shape[:-1], N)
 

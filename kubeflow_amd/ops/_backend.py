@@ -84,11 +84,12 @@ def _configure(lib: ctypes.CDLL) -> None:
     if hasattr(lib, "kf_skinny_gemm"):
         lib.kf_skinny_gemm.restype = I32
         lib.kf_skinny_gemm.argtypes = [P, P, P, P, FP, F, I64, I64, I64,
-                                       I64, I64, I64, P]
+                                       I64, I64, I64, I64, P]
     if hasattr(lib, "kf_skinny_gemm_q8"):
         lib.kf_skinny_gemm_q8.restype = I32
         lib.kf_skinny_gemm_q8.argtypes = [P, P, P, FP, P, FP, F, I64,
-                                          I64, I64, I64, I64, I64, P]
+                                          I64, I64, I64, I64, I64, I64,
+                                          P]
     if hasattr(lib, "kf_kv_store"):
         lib.kf_kv_store.restype = I32
         lib.kf_kv_store.argtypes = [P, P, P, P, P, P, I64, I64, I64, P]

@@ -42,6 +42,29 @@ __device__ __forceinline__ float kf_row_rstd(const unsigned short* row,
   return rsqrtf(acc / (float)K + eps);
 }
 
+// Fused-SwiGLU support (the w2 projection): when swiglu != 0, A is the
+// RAW w13 output [M, 2K] (gate rows [0,K), up rows [K,2K)) and fragments
+// are silu(gate)*up computed in-flight — removes the standalone swiglu
+// kernel + the y round-trip from the decode layer. Pure elementwise on
+// the A side (no extra row pass, unlike the rms fusion which measured
+// negative).
+__device__ __forceinline__ kf_bf16x8s kf_swiglu8(kf_bf16x8s g,
+                                                 kf_bf16x8s u) {
+  union {
+    kf_bf16x8s v;
+    unsigned short us[8];
+  } gi, ui, o;
+  gi.v = g;
+  ui.v = u;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float gf = kf_bf16_to_f32(gi.us[j]);
+    const float uf = kf_bf16_to_f32(ui.us[j]);
+    o.us[j] = kf_f32_to_bf16(gf / (1.f + __expf(-gf)) * uf);
+  }
+  return o.v;
+}
+
 __device__ __forceinline__ kf_bf16x8s kf_rms_scale8(
     kf_bf16x8s a, float rstd, const float* __restrict__ g) {
   union {
@@ -149,8 +172,8 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned short* __restrict__ w,
     const unsigned short* __restrict__ res,
-    const float* __restrict__ rmsg, float rms_eps, int M, int64_t N,
-    int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
+    const float* __restrict__ rmsg, float rms_eps, int swiglu, int M,
+    int64_t N, int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned short wbuf[2][SK_NT][SKL_STRIDE];
   __shared__ float red[SKL_W][MT][SK_NT];
   __shared__ float rstd_lds[MT];
@@ -193,13 +216,21 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
   const int ke0 = wv * 64 + hi4 * 8;  // this wave's k slice (s=0; s=1 at +32)
   kf_bf16x8s st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0);
   kf_bf16x8s st1 = *reinterpret_cast<const kf_bf16x8s*>(wr1);
-  kf_bf16x8s af0[NMT], af1[NMT];
+  kf_bf16x8s af0[NMT], af1[NMT], uf0[NMT], uf1[NMT];
 #pragma unroll
   for (int t = 0; t < NMT; ++t) {
     af0[t] = arow_ok[t]
         ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0) : zero8;
     af1[t] = arow_ok[t]
         ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32) : zero8;
+    if (swiglu) {
+      uf0[t] = arow_ok[t]
+          ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + K + ke0)
+          : zero8;
+      uf1[t] = arow_ok[t]
+          ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + K + ke0 + 32)
+          : zero8;
+    }
   }
   *reinterpret_cast<kf_bf16x8s*>(&wbuf[0][2 * wv][lane * 8]) = st0;
   *reinterpret_cast<kf_bf16x8s*>(&wbuf[0][2 * wv + 1][lane * 8]) = st1;
@@ -207,8 +238,13 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     kf_bf16x8s a0[NMT], a1[NMT];
 #pragma unroll
     for (int t = 0; t < NMT; ++t) {
-      a0[t] = af0[t];
-      a1[t] = af1[t];
+      if (swiglu) {
+        a0[t] = kf_swiglu8(af0[t], uf0[t]);
+        a1[t] = kf_swiglu8(af1[t], uf1[t]);
+      } else {
+        a0[t] = af0[t];
+        a1[t] = af1[t];
+      }
     }
     if (ch + 1 < nch) {
       st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0 + (ch + 1) * SKL_KC);
@@ -220,6 +256,12 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
               arow[t] + (ch + 1) * SKL_KC + ke0);
           af1[t] = *reinterpret_cast<const kf_bf16x8s*>(
               arow[t] + (ch + 1) * SKL_KC + ke0 + 32);
+          if (swiglu) {
+            uf0[t] = *reinterpret_cast<const kf_bf16x8s*>(
+                arow[t] + K + (ch + 1) * SKL_KC + ke0);
+            uf1[t] = *reinterpret_cast<const kf_bf16x8s*>(
+                arow[t] + K + (ch + 1) * SKL_KC + ke0 + 32);
+          }
         }
     }
     // one barrier per chunk: makes buffer ch&1's writes visible AND
@@ -312,8 +354,8 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned char* __restrict__ w8, const float* __restrict__ wscale,
     const unsigned short* __restrict__ res,
-    const float* __restrict__ rmsg, float rms_eps, int M, int64_t N,
-    int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
+    const float* __restrict__ rmsg, float rms_eps, int swiglu, int M,
+    int64_t N, int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned char wbuf8[2][SK_NT][SKQ_STRIDE];
   __shared__ float red[SKL_W][MT][SK_NT];
   __shared__ float rstd_lds[MT];
@@ -356,14 +398,20 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
   const int ke0 = wv * 128 + hi4 * 8;  // k slices at +0,+32,+64,+96
   kf_u32x4q st0 = *reinterpret_cast<const kf_u32x4q*>(wr0);
   kf_u32x4q st1 = *reinterpret_cast<const kf_u32x4q*>(wr1);
-  kf_bf16x8s af[4][NMT];
+  kf_bf16x8s af[4][NMT], uf[4][NMT];
 #pragma unroll
   for (int si = 0; si < 4; ++si)
 #pragma unroll
-    for (int t = 0; t < NMT; ++t)
+    for (int t = 0; t < NMT; ++t) {
       af[si][t] = arow_ok[t]
           ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32 * si)
           : zero8;
+      if (swiglu)
+        uf[si][t] = arow_ok[t]
+            ? *reinterpret_cast<const kf_bf16x8s*>(
+                  arow[t] + K + ke0 + 32 * si)
+            : zero8;
+    }
   *reinterpret_cast<kf_u32x4q*>(&wbuf8[0][2 * wv][lane * 16]) = st0;
   *reinterpret_cast<kf_u32x4q*>(&wbuf8[0][2 * wv + 1][lane * 16]) = st1;
   for (int64_t ch = 0; ch < nch; ++ch) {
@@ -371,7 +419,9 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
 #pragma unroll
     for (int si = 0; si < 4; ++si)
 #pragma unroll
-      for (int t = 0; t < NMT; ++t) acur[si][t] = af[si][t];
+      for (int t = 0; t < NMT; ++t)
+        acur[si][t] =
+            swiglu ? kf_swiglu8(af[si][t], uf[si][t]) : af[si][t];
     if (ch + 1 < nch) {
       st0 = *reinterpret_cast<const kf_u32x4q*>(wr0 + (ch + 1) * SKQ_KC);
       st1 = *reinterpret_cast<const kf_u32x4q*>(wr1 + (ch + 1) * SKQ_KC);
@@ -379,9 +429,13 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
       for (int si = 0; si < 4; ++si)
 #pragma unroll
         for (int t = 0; t < NMT; ++t)
-          if (arow_ok[t])
+          if (arow_ok[t]) {
             af[si][t] = *reinterpret_cast<const kf_bf16x8s*>(
                 arow[t] + (ch + 1) * SKQ_KC + ke0 + 32 * si);
+            if (swiglu)
+              uf[si][t] = *reinterpret_cast<const kf_bf16x8s*>(
+                  arow[t] + K + (ch + 1) * SKQ_KC + ke0 + 32 * si);
+          }
     }
     __syncthreads();
 #pragma unroll
@@ -428,9 +482,9 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
 KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
                                 const float* wscale, const void* res,
                                 const float* rmsg, float rms_eps,
-                                int64_t M, int64_t N, int64_t K,
-                                int64_t lda, int64_t ldw, int64_t ldc,
-                                void* stream) {
+                                int64_t swiglu, int64_t M, int64_t N,
+                                int64_t K, int64_t lda, int64_t ldw,
+                                int64_t ldc, void* stream) {
   if (M < 1 || M > 32 || K % SKQ_KC || N % SK_NT)
     return (int)hipErrorInvalidValue;
   if (lda == 0) lda = K;
@@ -443,43 +497,45 @@ KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned char*)w8,
                        wscale, (const unsigned short*)res, rmsg, rms_eps,
-                       (int)M, N, K, lda, ldw, ldc);
+                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
   else
     hipLaunchKernelGGL(kf_skinny_q8_kernel<16>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned char*)w8,
                        wscale, (const unsigned short*)res, rmsg, rms_eps,
-                       (int)M, N, K, lda, ldw, ldc);
+                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
   return (int)hipGetLastError();
 }
 
 // rmsg (nullable, fp32 [K]): fused input RMSNorm — C = res +
 // rmsnorm(A; rmsg, rms_eps) @ W^T (LDS-staged kernels only).
+// swiglu != 0: A is the raw [M, 2K] w13 output; fragments are
+// silu(A[:, :K]) * A[:, K:] computed in-flight (LDS kernels only).
 KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
                              const void* res, const float* rmsg,
-                             float rms_eps, int64_t M, int64_t N,
-                             int64_t K, int64_t lda, int64_t ldw,
-                             int64_t ldc, void* stream) {
+                             float rms_eps, int64_t swiglu, int64_t M,
+                             int64_t N, int64_t K, int64_t lda,
+                             int64_t ldw, int64_t ldc, void* stream) {
   if (M < 1 || M > 32 || K % 32 || N % SK_NT) return (int)hipErrorInvalidValue;
   if (lda == 0) lda = K;
   if (ldw == 0) ldw = K;
   if (ldc == 0) ldc = N;
   const bool lds_ok = K % SKL_KC == 0 && ldw % 8 == 0 && lda % 8 == 0;
   if (M > 16 && !lds_ok) return (int)hipErrorInvalidValue;
-  if (rmsg && !lds_ok) return (int)hipErrorInvalidValue;
+  if ((rmsg || swiglu) && !lds_ok) return (int)hipErrorInvalidValue;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
   if (lds_ok && M > 16)
     hipLaunchKernelGGL(kf_skinny_lds_kernel<32>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (const unsigned short*)res, rmsg, rms_eps, (int)M,
-                       N, K, lda, ldw, ldc);
+                       (const unsigned short*)res, rmsg, rms_eps,
+                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
   else if (lds_ok)
     hipLaunchKernelGGL(kf_skinny_lds_kernel<16>, grid, dim3(SKL_W * 64), 0,
                        (hipStream_t)stream, (unsigned short*)c,
                        (const unsigned short*)a, (const unsigned short*)w,
-                       (const unsigned short*)res, rmsg, rms_eps, (int)M,
-                       N, K, lda, ldw, ldc);
+                       (const unsigned short*)res, rmsg, rms_eps,
+                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
   // direct-load fallback: 8 waves when the grid can't fill the chip with
   // 4-wave blocks (<2 blocks/CU), 4 otherwise
   else if (N / SK_NT < 512)

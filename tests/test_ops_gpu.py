@@ -631,3 +631,32 @@ def test_skinny_fused_rms_parity():
         want8 = torch.nn.functional.linear(xn, wd) + r.float()
         err8 = ((got8.float() - want8).norm() / want8.norm()).item()
         assert err8 < 3e-2, (M, N, K, err8)
+
+
+@pytest.mark.gpu
+def test_skinny_fused_swiglu_parity():
+    """skinny_linear(fuse_swiglu=True) on the raw [M, 2K] w13 output ==
+    F.linear(swiglu(h)) for bf16 and fp8 weights."""
+    torch.manual_seed(12)
+    dev = torch.device("cuda", 0)
+    from kubeflow_amd import ops
+
+    for M, FFN, H in ((16, 14336, 4096), (32, 1024, 4096)):
+        h = torch.randn(M, 1, 2 * FFN, device=dev,
+                        dtype=torch.bfloat16) * 0.5
+        w2 = torch.randn(H, FFN, device=dev, dtype=torch.bfloat16) * 0.02
+        r = torch.randn(M, 1, H, device=dev, dtype=torch.bfloat16)
+        got = ops.skinny_linear(h, w2, residual=r, fuse_swiglu=True)
+        hf = h.float()
+        g, u = hf.split(FFN, dim=-1)
+        y = torch.nn.functional.silu(g) * u
+        want = torch.nn.functional.linear(y, w2.float()) + r.float()
+        err = ((got.float() - want).norm() / want.norm()).item()
+        assert err < 3e-2, (M, FFN, err)
+        w8, sc = ops.quantize_fp8_rows(w2)
+        got8 = ops.skinny_linear_q8(h, w8, sc, residual=r,
+                                    fuse_swiglu=True)
+        wd = ops.dequantize_fp8_rows(w8, sc, torch.float32)
+        want8 = torch.nn.functional.linear(y, wd) + r.float()
+        err8 = ((got8.float() - want8).norm() / want8.norm()).item()
+        assert err8 < 3e-2, (M, FFN, err8)
