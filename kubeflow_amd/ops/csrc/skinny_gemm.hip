@@ -167,12 +167,12 @@ __global__ __launch_bounds__(SKW * 64, 2) void kf_skinny_gemm_kernel(
 // MT = M-tile (16 or 32 batch rows); W traffic is identical, the wider
 // tile just adds a second A fragment + accumulator (decode buckets > 16
 // otherwise fell back to hipBLASLt).
-template <int MT>
+template <int MT, bool SWIGLU>
 __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned short* __restrict__ w,
     const unsigned short* __restrict__ res,
-    const float* __restrict__ rmsg, float rms_eps, int swiglu, int M,
+    const float* __restrict__ rmsg, float rms_eps, int M,
     int64_t N, int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned short wbuf[2][SK_NT][SKL_STRIDE];
   __shared__ float red[SKL_W][MT][SK_NT];
@@ -216,14 +216,15 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
   const int ke0 = wv * 64 + hi4 * 8;  // this wave's k slice (s=0; s=1 at +32)
   kf_bf16x8s st0 = *reinterpret_cast<const kf_bf16x8s*>(wr0);
   kf_bf16x8s st1 = *reinterpret_cast<const kf_bf16x8s*>(wr1);
-  kf_bf16x8s af0[NMT], af1[NMT], uf0[NMT], uf1[NMT];
+  kf_bf16x8s af0[NMT], af1[NMT];
+  kf_bf16x8s uf0[SWIGLU ? NMT : 1], uf1[SWIGLU ? NMT : 1];
 #pragma unroll
   for (int t = 0; t < NMT; ++t) {
     af0[t] = arow_ok[t]
         ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0) : zero8;
     af1[t] = arow_ok[t]
         ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32) : zero8;
-    if (swiglu) {
+    if constexpr (SWIGLU) {
       uf0[t] = arow_ok[t]
           ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + K + ke0)
           : zero8;
@@ -238,7 +239,7 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
     kf_bf16x8s a0[NMT], a1[NMT];
 #pragma unroll
     for (int t = 0; t < NMT; ++t) {
-      if (swiglu) {
+      if constexpr (SWIGLU) {
         a0[t] = kf_swiglu8(af0[t], uf0[t]);
         a1[t] = kf_swiglu8(af1[t], uf1[t]);
       } else {
@@ -256,7 +257,7 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_lds_kernel(
               arow[t] + (ch + 1) * SKL_KC + ke0);
           af1[t] = *reinterpret_cast<const kf_bf16x8s*>(
               arow[t] + (ch + 1) * SKL_KC + ke0 + 32);
-          if (swiglu) {
+          if constexpr (SWIGLU) {
             uf0[t] = *reinterpret_cast<const kf_bf16x8s*>(
                 arow[t] + K + (ch + 1) * SKL_KC + ke0);
             uf1[t] = *reinterpret_cast<const kf_bf16x8s*>(
@@ -349,12 +350,12 @@ __device__ __forceinline__ kf_bf16x8s kf_fp8x8_to_bf16x8(
   return out.v;
 }
 
-template <int MT>
+template <int MT, bool SWIGLU>
 __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
     unsigned short* __restrict__ c, const unsigned short* __restrict__ a,
     const unsigned char* __restrict__ w8, const float* __restrict__ wscale,
     const unsigned short* __restrict__ res,
-    const float* __restrict__ rmsg, float rms_eps, int swiglu, int M,
+    const float* __restrict__ rmsg, float rms_eps, int M,
     int64_t N, int64_t K, int64_t lda, int64_t ldw, int64_t ldc) {
   __shared__ unsigned char wbuf8[2][SK_NT][SKQ_STRIDE];
   __shared__ float red[SKL_W][MT][SK_NT];
@@ -398,7 +399,8 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
   const int ke0 = wv * 128 + hi4 * 8;  // k slices at +0,+32,+64,+96
   kf_u32x4q st0 = *reinterpret_cast<const kf_u32x4q*>(wr0);
   kf_u32x4q st1 = *reinterpret_cast<const kf_u32x4q*>(wr1);
-  kf_bf16x8s af[4][NMT], uf[4][NMT];
+  kf_bf16x8s af[4][NMT];
+  kf_bf16x8s uf[4][SWIGLU ? NMT : 1];
 #pragma unroll
   for (int si = 0; si < 4; ++si)
 #pragma unroll
@@ -406,7 +408,7 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
       af[si][t] = arow_ok[t]
           ? *reinterpret_cast<const kf_bf16x8s*>(arow[t] + ke0 + 32 * si)
           : zero8;
-      if (swiglu)
+      if constexpr (SWIGLU)
         uf[si][t] = arow_ok[t]
             ? *reinterpret_cast<const kf_bf16x8s*>(
                   arow[t] + K + ke0 + 32 * si)
@@ -419,9 +421,12 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
 #pragma unroll
     for (int si = 0; si < 4; ++si)
 #pragma unroll
-      for (int t = 0; t < NMT; ++t)
-        acur[si][t] =
-            swiglu ? kf_swiglu8(af[si][t], uf[si][t]) : af[si][t];
+      for (int t = 0; t < NMT; ++t) {
+        if constexpr (SWIGLU)
+          acur[si][t] = kf_swiglu8(af[si][t], uf[si][t]);
+        else
+          acur[si][t] = af[si][t];
+      }
     if (ch + 1 < nch) {
       st0 = *reinterpret_cast<const kf_u32x4q*>(wr0 + (ch + 1) * SKQ_KC);
       st1 = *reinterpret_cast<const kf_u32x4q*>(wr1 + (ch + 1) * SKQ_KC);
@@ -432,7 +437,7 @@ __global__ __launch_bounds__(SKL_W * 64, 2) void kf_skinny_q8_kernel(
           if (arow_ok[t]) {
             af[si][t] = *reinterpret_cast<const kf_bf16x8s*>(
                 arow[t] + (ch + 1) * SKQ_KC + ke0 + 32 * si);
-            if (swiglu)
+            if constexpr (SWIGLU)
               uf[si][t] = *reinterpret_cast<const kf_bf16x8s*>(
                   arow[t] + K + (ch + 1) * SKQ_KC + ke0 + 32 * si);
           }
@@ -492,18 +497,20 @@ KF_EXPORT int kf_skinny_gemm_q8(void* c, const void* a, const void* w8,
   if (ldc == 0) ldc = N;
   if (ldw % 8 || lda % 8) return (int)hipErrorInvalidValue;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
-  if (M > 16)
-    hipLaunchKernelGGL(kf_skinny_q8_kernel<32>, grid, dim3(SKL_W * 64), 0,
-                       (hipStream_t)stream, (unsigned short*)c,
-                       (const unsigned short*)a, (const unsigned char*)w8,
-                       wscale, (const unsigned short*)res, rmsg, rms_eps,
-                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
-  else
-    hipLaunchKernelGGL(kf_skinny_q8_kernel<16>, grid, dim3(SKL_W * 64), 0,
-                       (hipStream_t)stream, (unsigned short*)c,
-                       (const unsigned short*)a, (const unsigned char*)w8,
-                       wscale, (const unsigned short*)res, rmsg, rms_eps,
-                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
+#define KF_SKQ_LAUNCH(MT, SW)                                            \
+  hipLaunchKernelGGL((kf_skinny_q8_kernel<MT, SW>), grid,                \
+                     dim3(SKL_W * 64), 0, (hipStream_t)stream,           \
+                     (unsigned short*)c, (const unsigned short*)a,       \
+                     (const unsigned char*)w8, wscale,                   \
+                     (const unsigned short*)res, rmsg, rms_eps, (int)M,  \
+                     N, K, lda, ldw, ldc)
+  if (M > 16) {
+    if (swiglu) KF_SKQ_LAUNCH(32, true);
+    else KF_SKQ_LAUNCH(32, false);
+  } else {
+    if (swiglu) KF_SKQ_LAUNCH(16, true);
+    else KF_SKQ_LAUNCH(16, false);
+  }
   return (int)hipGetLastError();
 }
 
@@ -524,18 +531,20 @@ KF_EXPORT int kf_skinny_gemm(void* c, const void* a, const void* w,
   if (M > 16 && !lds_ok) return (int)hipErrorInvalidValue;
   if ((rmsg || swiglu) && !lds_ok) return (int)hipErrorInvalidValue;
   dim3 grid((unsigned)(N / SK_NT), 1, 1);
-  if (lds_ok && M > 16)
-    hipLaunchKernelGGL(kf_skinny_lds_kernel<32>, grid, dim3(SKL_W * 64), 0,
-                       (hipStream_t)stream, (unsigned short*)c,
-                       (const unsigned short*)a, (const unsigned short*)w,
-                       (const unsigned short*)res, rmsg, rms_eps,
-                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
-  else if (lds_ok)
-    hipLaunchKernelGGL(kf_skinny_lds_kernel<16>, grid, dim3(SKL_W * 64), 0,
-                       (hipStream_t)stream, (unsigned short*)c,
-                       (const unsigned short*)a, (const unsigned short*)w,
-                       (const unsigned short*)res, rmsg, rms_eps,
-                       (int)swiglu, (int)M, N, K, lda, ldw, ldc);
+#define KF_SKL_LAUNCH(MT, SW)                                            \
+  hipLaunchKernelGGL((kf_skinny_lds_kernel<MT, SW>), grid,               \
+                     dim3(SKL_W * 64), 0, (hipStream_t)stream,           \
+                     (unsigned short*)c, (const unsigned short*)a,       \
+                     (const unsigned short*)w,                           \
+                     (const unsigned short*)res, rmsg, rms_eps, (int)M,  \
+                     N, K, lda, ldw, ldc)
+  if (lds_ok && M > 16) {
+    if (swiglu) KF_SKL_LAUNCH(32, true);
+    else KF_SKL_LAUNCH(32, false);
+  } else if (lds_ok) {
+    if (swiglu) KF_SKL_LAUNCH(16, true);
+    else KF_SKL_LAUNCH(16, false);
+  }
   // direct-load fallback: 8 waves when the grid can't fill the chip with
   // 4-wave blocks (<2 blocks/CU), 4 otherwise
   else if (N / SK_NT < 512)
