@@ -653,7 +653,7 @@ class InferenceEngine:
                     t, w, residual=r, fuse_swiglu=fuse_swiglu)
             return lambda li, t, w, r: F.linear(t, w) + r
         lin_wo = _lin_res("wo", "wo")
-        lin_w2 = _lin_res("w2", "w2", fuse_swiglu=True)
+        lin_w2 = _lin_res("w2", "w2")
         x = self.model.embed(tokens)  # [N,1,H]
         cos, sin = self.model.rope_cos, self.model.rope_sin
         for li, layer in enumerate(self.model.layers):
@@ -674,13 +674,14 @@ class InferenceEngine:
             if layer.moe is not None:
                 x = x + layer.moe.decode_dense(layer.mlp_norm(x))
             else:
-                h13 = lin_w13(li, layer.mlp_norm(x), layer.w13.weight)
-                # swiglu fused into the w2 fragment loader when w2 runs
-                # on the skinny kernel; explicit kernel otherwise
-                if "w2" in sel:
-                    x = lin_w2(li, h13, layer.w2.weight, x)
-                else:
-                    x = lin_w2(li, ops.swiglu(h13), layer.w2.weight, x)
+                # NOTE: fusing swiglu into the w2 fragment loader
+                # (fuse_swiglu=True) measured WORSE (replay 4.3 -> 5.0,
+                # quant 3.5 -> 4.6): the silu exp chain lands on the
+                # MFMA dependency path and breaks the load pipeline.
+                # The standalone swiglu kernel stays.
+                y = ops.swiglu(lin_w13(li, layer.mlp_norm(x),
+                                       layer.w13.weight))
+                x = lin_w2(li, y, layer.w2.weight, x)
         x = self.model.final_norm(x)
         logits = lin_lm(x, self.model.lm_head.weight)  # [N,1,V]
         return logits[:, -1]  # [N,V] (sampling happens outside the graph)
