@@ -666,3 +666,41 @@ def test_classifier_engine_serves_bert():
             assert r.generated[0] == want, (i, r.generated, want)
     finally:
         eng.stop()
+
+
+def test_serving_server_stream_and_metrics_endpoints():
+    """The FastAPI surface directly: SSE :generate_stream emits one data
+    line per token then [DONE]; /metrics carries the serving counters."""
+    import torch
+    from starlette.testclient import TestClient
+
+    from kubeflow_amd.runtime import serving_server
+
+    torch.manual_seed(23)
+    app = serving_server.build_app({"name": "tiny", "model": "llama-tiny",
+                                    "max_slots": 4, "max_seq_len": 128,
+                                    "max_batch": 4})
+    try:
+        with TestClient(app) as c:
+            r = c.post("/v1/models/tiny:generate_stream",
+                       json={"prompt_tokens": [4, 9, 2],
+                             "max_new_tokens": 5})
+            assert r.status_code == 200
+            lines = [l for l in r.text.splitlines() if l.startswith("data:")]
+            assert lines[-1] == "data: [DONE]"
+            assert len(lines) == 6  # 5 tokens + DONE
+            m = c.get("/metrics").text
+            for name in ("kf_serving_requests_total",
+                         "kf_serving_tokens_out_total",
+                         "kf_serving_prefill_tokens_total",
+                         "kf_serving_graph_replays_total",
+                         "kf_serving_active_streams",
+                         "kf_serving_quantized"):
+                assert name in m, name
+            p = c.post("/v1/models/tiny:predict",
+                       json={"instances": [{"prompt_tokens": [1, 2, 3],
+                                            "max_new_tokens": 3}]})
+            assert p.status_code == 200
+            assert len(p.json()["predictions"][0]["tokens"]) == 3
+    finally:
+        app.state.engine.stop()
