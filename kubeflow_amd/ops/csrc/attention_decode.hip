@@ -89,39 +89,47 @@ __global__ __launch_bounds__(256) void kf_attn_decode_kernel(
     const int dcol = (lane & 15) * 8;  // this lane's 8-elem d-range
     float ov[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
 
-    // register-buffered staging with NEXT-tile load prefetch: the tile
-    // t+1 global loads issue right after tile t's LDS writes, so HBM
-    // latency overlaps the QK/PV phases — single LDS buffer (a 2-buffer
-    // variant halved occupancy and measured worse). Per-row guard
-    // clamps the prefetched addresses inside the slot's smax rows.
-    kf_short8 kb[4], vb[4];
-    auto tile_loads = [&](int t0) {
-#pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int vi = (int)threadIdx.x + u * 256;
-        const int r = vi >> 4, c8 = vi & 15;
-        const int rr = t0 + r < (int)smax ? t0 + r : 0;
-        kb[u] = *reinterpret_cast<const kf_short8*>(
-            kcache + cbase + (int64_t)rr * cstride + c8 * 8);
-        if (!VDIRECT)
-          vb[u] = *reinterpret_cast<const kf_short8*>(
-              vcache + cbase + (int64_t)rr * cstride + c8 * 8);
-      }
-    };
-    if (lo < hi) tile_loads(lo);
     for (int t0 = lo; t0 < hi; t0 += AD_TILE) {
       const int rows = min(AD_TILE, hi - t0);
-      __syncthreads();  // previous tile's consumers done: buffer free
+      __syncthreads();
+      if (rows == AD_TILE) {
+        // full tile: register-buffered unrolled staging (8 loads in
+        // flight per thread before any LDS write — the dynamic loop's
+        // load->write->load chain left staging HBM-latency-bound)
+        kf_short8 kb[4], vb[4];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const int vi = (int)threadIdx.x + u * 256;
-        const int r = vi >> 4, c8 = vi & 15;
-        *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kb[u];
-        if (!VDIRECT)
-          *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) =
-              vb[u];
+        for (int u = 0; u < 4; ++u) {
+          const int vi = (int)threadIdx.x + u * 256;
+          const int r = vi >> 4, c8 = vi & 15;
+          kb[u] = *reinterpret_cast<const kf_short8*>(
+              kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          if (!VDIRECT)
+            vb[u] = *reinterpret_cast<const kf_short8*>(
+                vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int vi = (int)threadIdx.x + u * 256;
+          const int r = vi >> 4, c8 = vi & 15;
+          *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kb[u];
+          if (!VDIRECT)
+            *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) =
+                vb[u];
+        }
+      } else {
+        for (int vi = threadIdx.x; vi < rows * 16; vi += blockDim.x) {
+          const int r = vi >> 4, c8 = vi & 15;
+          kf_short8 kv8 = *reinterpret_cast<const kf_short8*>(
+              kcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+          *reinterpret_cast<kf_short8*>(k_lds + kf_swzd(r, c8 * 16)) = kv8;
+          if (!VDIRECT) {
+            kf_short8 vv8 = *reinterpret_cast<const kf_short8*>(
+                vcache + cbase + (int64_t)(t0 + r) * cstride + c8 * 8);
+            *reinterpret_cast<kf_short8*>(v_lds + kf_swzd(r, c8 * 16)) =
+                vv8;
+          }
+        }
       }
-      if (t0 + AD_TILE < hi) tile_loads(t0 + AD_TILE);
       __syncthreads();
 
       float s = -INFINITY;
